@@ -1,0 +1,130 @@
+"""Op dispatch for the MI355X engine.
+
+On a GPU box the hand-written gfx950 HIP kernels (_kernels extension) are
+REQUIRED — if a CUDA/HIP device is visible and the extension is missing,
+every op raises instead of silently falling back to eager PyTorch (the
+round-end harness records which .so files the GPU processes load). On CPU
+the fp32 torch references (torch_ref) run so the whole engine and gateway
+are testable without a GPU.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import torch_ref
+
+try:
+    from . import _kernels  # compiled by setup.py build_ext --inplace
+
+    HAS_KERNELS = True
+except ImportError:  # pragma: no cover - absent only before first build
+    _kernels = None
+    HAS_KERNELS = False
+
+
+def _require_kernels() -> None:
+    if not HAS_KERNELS:
+        raise RuntimeError(
+            "inference_gateway_amd.ops._kernels is not built but a GPU tensor "
+            "was passed. Build it in-tree with: "
+            "PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace"
+        )
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        _require_kernels()
+        out = torch.empty_like(x)
+        _kernels.rmsnorm(out, x, weight, eps)
+        return out
+    return torch_ref.rmsnorm_ref(x, weight, eps)
+
+
+def fused_add_rmsnorm(x, residual, weight, eps: float):
+    """In-place: residual += x; x = rmsnorm(residual) * w. Returns (x, residual)."""
+    if x.is_cuda:
+        _require_kernels()
+        _kernels.fused_add_rmsnorm(x, residual, weight, eps)
+        return x, residual
+    nx, nres = torch_ref.fused_add_rmsnorm_ref(x, residual, weight, eps)
+    x.copy_(nx)
+    residual.copy_(nres)
+    return x, residual
+
+
+def rotary_embedding(positions, q, k, head_dim: int, cos_sin_cache):
+    """In-place NeoX RoPE on q [T, NQ*HD] and k [T, NKV*HD]."""
+    if q.is_cuda:
+        _require_kernels()
+        _kernels.rotary_embedding(positions, q, k, head_dim, cos_sin_cache)
+        return q, k
+    nq, nk = torch_ref.rope_ref(positions, q, k, head_dim, cos_sin_cache)
+    q.copy_(nq)
+    k.copy_(nk)
+    return q, k
+
+
+def append_kv_cache(k, v, kcache, vcache, slot_mapping):
+    if k.is_cuda:
+        _require_kernels()
+        _kernels.append_kv_cache(k, v, kcache, vcache, slot_mapping)
+        return
+    torch_ref.append_kv_cache_ref(k, v, kcache, vcache, slot_mapping)
+
+
+def paged_attn_decode(q, kcache, vcache, block_table, ctx_lens, scale: float):
+    if q.is_cuda:
+        _require_kernels()
+        out = torch.empty_like(q)
+        _kernels.paged_attn_decode(
+            out, q, kcache, vcache, block_table, ctx_lens, scale
+        )
+        return out
+    return torch_ref.paged_attn_decode_ref(
+        q, kcache, vcache, block_table, ctx_lens, scale
+    )
+
+
+def flash_prefill(q, k, v, cu_seqlens, seq_lens, scale: float):
+    """Varlen causal prefill attention. seq_lens: python list of ints."""
+    if q.is_cuda:
+        _require_kernels()
+        out = torch.empty_like(q)
+        tile_seq, tile_off = torch_ref.make_prefill_tiles(
+            seq_lens, tile_m=64, device=q.device
+        )
+        _kernels.flash_prefill(
+            out, q, k, v, cu_seqlens, tile_seq, tile_off, scale
+        )
+        return out
+    return torch_ref.flash_prefill_ref(q, k, v, cu_seqlens, scale)
+
+
+def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
+    if gate_up.is_cuda:
+        _require_kernels()
+        inter = gate_up.shape[-1] // 2
+        out = torch.empty(
+            (*gate_up.shape[:-1], inter),
+            dtype=gate_up.dtype,
+            device=gate_up.device,
+        )
+        _kernels.silu_mul(out, gate_up)
+        return out
+    return torch_ref.silu_mul_ref(gate_up)
+
+
+def sample(logits, temperature, top_k, top_p, uniforms):
+    """logits [B, V] fp32 -> sampled token ids [B] int32."""
+    if logits.is_cuda:
+        _require_kernels()
+        out = torch.empty(
+            logits.shape[0], dtype=torch.int32, device=logits.device
+        )
+        _kernels.sample(out, logits, temperature, top_k, top_p, uniforms)
+        return out
+    return torch_ref.sample_ref(logits, temperature, top_k, top_p, uniforms)
+
+
+build_cos_sin_cache = torch_ref.build_cos_sin_cache
+make_prefill_tiles = torch_ref.make_prefill_tiles

@@ -1,0 +1,48 @@
+// Python bindings for the MI355X (gfx950) HIP kernel library.
+//
+// Every op here is a hand-written CDNA4 kernel (see the .hip files); the
+// Python side (inference_gateway_amd/ops/__init__.py) requires this
+// extension whenever a GPU is present — there is no silent eager fallback
+// on the GPU path.
+
+#include <torch/extension.h>
+
+namespace igw {
+void rmsnorm(torch::Tensor out, torch::Tensor in, torch::Tensor weight,
+             double eps);
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor weight, double eps);
+void rotary_embedding(torch::Tensor positions, torch::Tensor q,
+                      torch::Tensor k, int64_t head_dim,
+                      torch::Tensor cos_sin_cache);
+void append_kv_cache(torch::Tensor k, torch::Tensor v, torch::Tensor kcache,
+                     torch::Tensor vcache, torch::Tensor slot_mapping);
+void paged_attn_decode(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor kcache, torch::Tensor vcache,
+                       torch::Tensor block_table, torch::Tensor ctx_lens,
+                       double scale);
+void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                   torch::Tensor v, torch::Tensor cu_seqlens,
+                   torch::Tensor tile_seq, torch::Tensor tile_off,
+                   double scale);
+void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
+            torch::Tensor top_k, torch::Tensor top_p, torch::Tensor uniforms);
+void silu_mul(torch::Tensor out, torch::Tensor gate_up);
+}  // namespace igw
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &igw::rmsnorm, "RMSNorm (bf16, gfx950)");
+  m.def("fused_add_rmsnorm", &igw::fused_add_rmsnorm,
+        "residual += x; x = rmsnorm(residual) (bf16, gfx950)");
+  m.def("rotary_embedding", &igw::rotary_embedding,
+        "NeoX-style RoPE in-place on q/k (bf16, gfx950)");
+  m.def("append_kv_cache", &igw::append_kv_cache,
+        "scatter step K/V into the paged cache (bf16, gfx950)");
+  m.def("paged_attn_decode", &igw::paged_attn_decode,
+        "GQA paged-attention decode (bf16, gfx950)");
+  m.def("flash_prefill", &igw::flash_prefill,
+        "varlen causal MFMA flash-attention prefill (bf16, gfx950)");
+  m.def("sample", &igw::sample,
+        "fused temperature/top-k/top-p sampling (fp32 logits, gfx950)");
+  m.def("silu_mul", &igw::silu_mul, "fused SwiGLU activation (bf16, gfx950)");
+}

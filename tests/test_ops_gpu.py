@@ -1,0 +1,151 @@
+"""GPU numerics tests: every hand-written gfx950 HIP kernel vs the plain
+PyTorch fp32 reference (SURVEY.md §4 'GPU-kernel test tier').
+
+All tensors use asymmetric random data (guide §5.4 rule 16: symmetric inputs
+miss operand/output transposes).
+"""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from inference_gateway_amd import ops
+from inference_gateway_amd.ops import torch_ref
+
+
+def _assert_close_bf16(got, ref, atol=2e-2, rtol=2e-2):
+    torch.testing.assert_close(
+        got.float().cpu(), ref.float().cpu(), atol=atol, rtol=rtol
+    )
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    torch.manual_seed(1234)
+
+
+def test_kernels_extension_loaded():
+    # On a GPU box the native extension must be present — no eager fallback.
+    assert ops.HAS_KERNELS, "gfx950 _kernels extension not built/loaded"
+
+
+@pytest.mark.parametrize("hidden", [4096, 8192, 6144])
+def test_rmsnorm(hidden):
+    x = torch.randn(33, hidden, dtype=torch.bfloat16, device="cuda") * 3
+    w = torch.randn(hidden, dtype=torch.bfloat16, device="cuda")
+    out = ops.rmsnorm(x, w, 1e-5)
+    ref = torch_ref.rmsnorm_ref(x.cpu(), w.cpu(), 1e-5)
+    _assert_close_bf16(out, ref)
+
+
+def test_fused_add_rmsnorm():
+    hidden = 4096
+    x = torch.randn(17, hidden, dtype=torch.bfloat16, device="cuda")
+    res = torch.randn(17, hidden, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(hidden, dtype=torch.bfloat16, device="cuda")
+    xr, rr = torch_ref.fused_add_rmsnorm_ref(x.cpu(), res.cpu(), w.cpu(), 1e-5)
+    ops.fused_add_rmsnorm(x, res, w, 1e-5)
+    _assert_close_bf16(x, xr)
+    _assert_close_bf16(res, rr)
+
+
+def test_rope():
+    T, nq, nkv, hd = 41, 8, 2, 128
+    q = torch.randn(T, nq * hd, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, nkv * hd, dtype=torch.bfloat16, device="cuda")
+    pos = torch.randint(0, 2048, (T,), dtype=torch.int32, device="cuda")
+    cs = torch_ref.build_cos_sin_cache(hd, 4096, device="cuda")
+    qr, kr = torch_ref.rope_ref(pos.cpu(), q.cpu(), k.cpu(), hd, cs.cpu())
+    ops.rotary_embedding(pos, q, k, hd, cs)
+    _assert_close_bf16(q, qr)
+    _assert_close_bf16(k, kr)
+
+
+def test_append_kv_cache():
+    T, nkv, bs, hd, nb = 37, 4, 16, 128, 64
+    k = torch.randn(T, nkv, hd, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, nkv, hd, dtype=torch.bfloat16, device="cuda")
+    kc = torch.zeros(nb, nkv, bs, hd, dtype=torch.bfloat16, device="cuda")
+    vc = torch.zeros_like(kc)
+    slots = torch.randperm(nb * bs, device="cuda")[:T].to(torch.int32)
+    kc_ref, vc_ref = kc.cpu().clone(), vc.cpu().clone()
+    torch_ref.append_kv_cache_ref(k.cpu(), v.cpu(), kc_ref, vc_ref, slots.cpu())
+    ops.append_kv_cache(k, v, kc, vc, slots)
+    _assert_close_bf16(kc, kc_ref, atol=0, rtol=0)
+    _assert_close_bf16(vc, vc_ref, atol=0, rtol=0)
+
+
+@pytest.mark.parametrize("G", [1, 4, 8])
+def test_paged_attn_decode(G):
+    torch.manual_seed(7 + G)
+    B, nkv, bs, hd = 5, 8, 16, 128
+    nq = nkv * G
+    ctx = torch.tensor([1, 17, 16, 300, 1024], dtype=torch.int32, device="cuda")
+    max_blocks = 1024 // bs
+    nb = B * max_blocks + 8
+    q = torch.randn(B, nq, hd, dtype=torch.bfloat16, device="cuda")
+    kc = torch.randn(nb, nkv, bs, hd, dtype=torch.bfloat16, device="cuda")
+    vc = torch.randn(nb, nkv, bs, hd, dtype=torch.bfloat16, device="cuda")
+    perm = torch.randperm(nb, device="cuda")[: B * max_blocks].to(torch.int32)
+    bt = perm.view(B, max_blocks)
+    scale = hd**-0.5
+    out = ops.paged_attn_decode(q, kc, vc, bt, ctx, scale)
+    ref = torch_ref.paged_attn_decode_ref(
+        q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(), ctx.cpu(), scale
+    )
+    _assert_close_bf16(out, ref)
+
+
+@pytest.mark.parametrize(
+    "seq_lens", [[64], [1, 37, 64, 200], [128, 1024], [63, 65, 129]]
+)
+def test_flash_prefill(seq_lens):
+    torch.manual_seed(len(seq_lens))
+    nq, nkv, hd = 8, 2, 128
+    T = sum(seq_lens)
+    q = torch.randn(T, nq, hd, dtype=torch.bfloat16, device="cuda")
+    k = torch.randn(T, nkv, hd, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(T, nkv, hd, dtype=torch.bfloat16, device="cuda")
+    cu = torch.tensor(
+        [0] + list(torch.cumsum(torch.tensor(seq_lens), 0)),
+        dtype=torch.int32,
+        device="cuda",
+    )
+    scale = hd**-0.5
+    out = ops.flash_prefill(q, k, v, cu, seq_lens, scale)
+    ref = torch_ref.flash_prefill_ref(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), scale)
+    _assert_close_bf16(out, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_silu_mul():
+    gu = torch.randn(65, 2 * 14336, dtype=torch.bfloat16, device="cuda")
+    out = ops.silu_mul(gu)
+    ref = torch_ref.silu_mul_ref(gu.cpu())
+    _assert_close_bf16(out, ref)
+
+
+def test_sample_greedy():
+    B, V = 16, 128256
+    logits = torch.randn(B, V, dtype=torch.float32, device="cuda")
+    temp = torch.zeros(B, device="cuda")
+    tk = torch.zeros(B, dtype=torch.int32, device="cuda")
+    tp = torch.ones(B, device="cuda")
+    u = torch.rand(B, device="cuda")
+    out = ops.sample(logits, temp, tk, tp, u)
+    assert torch.equal(out.cpu(), logits.argmax(-1).to(torch.int32).cpu())
+
+
+@pytest.mark.parametrize("k,p", [(0, 1.0), (50, 1.0), (0, 0.9), (40, 0.95)])
+def test_sample_topk_topp(k, p):
+    torch.manual_seed(99)
+    B, V = 8, 32000
+    logits = torch.randn(B, V, dtype=torch.float32, device="cuda") * 4
+    temp = torch.full((B,), 0.8, device="cuda")
+    tk = torch.full((B,), k, dtype=torch.int32, device="cuda")
+    tp = torch.full((B,), p, device="cuda")
+    u = torch.rand(B, device="cuda").clamp(0.01, 0.99)
+    out = ops.sample(logits, temp, tk, tp, u)
+    ref = torch_ref.sample_ref(
+        logits.cpu(), temp.cpu(), tk.cpu(), tp.cpu(), u.cpu()
+    )
+    assert torch.equal(out.cpu(), ref)
