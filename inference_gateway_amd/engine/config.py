@@ -1,0 +1,146 @@
+"""Engine configuration: model architectures, cache sizing, scheduler knobs.
+
+The reference gateway has no engine; these configs define the
+net-new local provider models named by BASELINE.json: Llama-3 8B / 70B and
+Mixtral 8x7B, plus tiny variants for CPU tests and smoke().
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+
+@dataclass
+class ModelConfig:
+    name: str
+    hidden_size: int
+    num_layers: int
+    num_heads: int
+    num_kv_heads: int
+    head_dim: int
+    intermediate_size: int
+    vocab_size: int
+    rope_base: float = 500000.0
+    max_position: int = 8192
+    norm_eps: float = 1e-5
+    tie_embeddings: bool = False
+    # MoE (Mixtral): 0 experts => dense MLP
+    num_experts: int = 0
+    num_experts_per_tok: int = 2
+
+    @property
+    def is_moe(self) -> bool:
+        return self.num_experts > 0
+
+    def kv_bytes_per_token(self) -> int:
+        # bf16 K + V per layer
+        return 2 * self.num_kv_heads * self.head_dim * 2 * self.num_layers
+
+
+# Architectures from the public model cards (random-init weights — no network).
+MODEL_PRESETS = {
+    "llama-3-8b": ModelConfig(
+        name="llama-3-8b",
+        hidden_size=4096,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=14336,
+        vocab_size=128256,
+        rope_base=500000.0,
+    ),
+    "llama-3-70b": ModelConfig(
+        name="llama-3-70b",
+        hidden_size=8192,
+        num_layers=80,
+        num_heads=64,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=28672,
+        vocab_size=128256,
+        rope_base=500000.0,
+    ),
+    "mixtral-8x7b": ModelConfig(
+        name="mixtral-8x7b",
+        hidden_size=4096,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=14336,
+        vocab_size=32000,
+        rope_base=1000000.0,
+        num_experts=8,
+        num_experts_per_tok=2,
+    ),
+    # tiny models for CPU tests / smoke
+    "tiny-llama": ModelConfig(
+        name="tiny-llama",
+        hidden_size=512,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        intermediate_size=1024,
+        vocab_size=512,
+        rope_base=10000.0,
+        max_position=2048,
+    ),
+    "tiny-moe": ModelConfig(
+        name="tiny-moe",
+        hidden_size=512,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        intermediate_size=512,
+        vocab_size=512,
+        rope_base=10000.0,
+        max_position=2048,
+        num_experts=4,
+        num_experts_per_tok=2,
+    ),
+}
+
+
+@dataclass
+class SamplingParams:
+    max_new_tokens: int = 128
+    temperature: float = 1.0
+    top_k: int = 0          # <=0 disables
+    top_p: float = 1.0      # >=1 disables
+    stop_token_ids: tuple = ()
+    ignore_eos: bool = False
+
+
+@dataclass
+class EngineConfig:
+    model: ModelConfig
+    block_size: int = 16
+    num_gpu_blocks: Optional[int] = None   # None => size from free HBM
+    gpu_memory_utilization: float = 0.90
+    max_num_seqs: int = 256
+    max_batched_tokens: int = 8192
+    max_model_len: int = 8192
+    use_hip_graphs: bool = True
+    enforce_eager: bool = False
+    tensor_parallel_size: int = 1
+    seed: int = 0
+
+    @staticmethod
+    def tiny_llama(**kw) -> "EngineConfig":
+        defaults = dict(
+            model=MODEL_PRESETS["tiny-llama"],
+            num_gpu_blocks=64,
+            max_num_seqs=8,
+            max_batched_tokens=2048,
+            max_model_len=512,
+            use_hip_graphs=False,
+        )
+        defaults.update(kw)
+        return EngineConfig(**defaults)
+
+    @staticmethod
+    def for_model(name: str, **kw) -> "EngineConfig":
+        return EngineConfig(model=MODEL_PRESETS[name], **kw)

@@ -1,0 +1,216 @@
+"""LLMEngine: the synchronous continuous-batching inference core.
+
+Net-new scope (SURVEY.md §7 stages 3-5): this is the 'local provider' the
+gateway calls in-process instead of the reference's self-proxy HTTP hop
+(SURVEY.md §3.2 — the rebuild short-circuits it by design). One engine per
+GPU process; TP ranks run this same loop SPMD with RCCL collectives inside
+the model and identical sampling RNG, so no cross-rank token broadcast is
+needed on the hot path.
+"""
+from __future__ import annotations
+
+import itertools
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from .config import EngineConfig, SamplingParams
+from .kvcache import BlockManager
+from .models.batch import ForwardBatch
+from .models.llama import LlamaForCausalLM
+from .sampler import Sampler
+from .scheduler import ScheduledBatch, Scheduler, SeqState, Sequence
+
+
+@dataclass
+class StepOutput:
+    seq_id: int
+    token_id: int
+    finished: bool
+    finish_reason: Optional[str] = None
+
+
+@dataclass
+class RequestOutput:
+    seq_id: int
+    token_ids: List[int] = field(default_factory=list)
+    finished: bool = False
+    finish_reason: Optional[str] = None
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig, device: str = "cuda:0"):
+        self.cfg = cfg
+        self.device = torch.device(
+            device if torch.cuda.is_available() else "cpu")
+        self.dtype = torch.bfloat16
+        if cfg.model.is_moe:
+            from .models.mixtral import MixtralForCausalLM
+
+            self.model = MixtralForCausalLM(cfg.model, self.device, self.dtype)
+        else:
+            self.model = LlamaForCausalLM(cfg.model, self.device, self.dtype)
+        self.model.eval()
+
+        num_blocks = cfg.num_gpu_blocks or self._auto_num_blocks()
+        self.block_manager = BlockManager(num_blocks, cfg.block_size)
+        self.scheduler = Scheduler(
+            self.block_manager, cfg.max_num_seqs, cfg.max_batched_tokens,
+            cfg.max_model_len)
+        self.sampler = Sampler(self.device, cfg.seed)
+        self.kv_caches = self._allocate_kv(num_blocks)
+        self._seq_counter = itertools.count()
+        self._seqs: Dict[int, Sequence] = {}
+        self._graphs = None  # hipGraph runner, installed by capture_graphs()
+        self.max_blocks_per_seq = (
+            cfg.max_model_len + cfg.block_size - 1) // cfg.block_size
+
+    # ---------------- setup ----------------
+    def _auto_num_blocks(self) -> int:
+        from ..parallel.comm import tp_world_size
+
+        m = self.cfg.model
+        bytes_per_block = (
+            m.kv_bytes_per_token() * self.cfg.block_size // tp_world_size())
+        if self.device.type == "cuda":
+            free, _total = torch.cuda.mem_get_info(self.device)
+            budget = int(free * self.cfg.gpu_memory_utilization) - (2 << 30)
+        else:
+            budget = 1 << 30
+        return max(64, budget // bytes_per_block)
+
+    def _allocate_kv(self, num_blocks: int):
+        from ..parallel.comm import tp_world_size
+
+        m = self.cfg.model
+        nkv = max(1, m.num_kv_heads // tp_world_size())
+        shape = (num_blocks, nkv, self.cfg.block_size, m.head_dim)
+        caches = []
+        for _ in range(m.num_layers):
+            k = torch.zeros(shape, dtype=self.dtype, device=self.device)
+            v = torch.zeros(shape, dtype=self.dtype, device=self.device)
+            caches.append((k, v))
+        return caches
+
+    # ---------------- request API ----------------
+    def add_request(self, prompt_token_ids: List[int],
+                    sampling: Optional[SamplingParams] = None,
+                    seq_id: Optional[int] = None) -> int:
+        if seq_id is None:
+            seq_id = next(self._seq_counter)
+        seq = Sequence(
+            seq_id=seq_id,
+            prompt_token_ids=list(prompt_token_ids),
+            sampling=sampling or SamplingParams(),
+        )
+        self._seqs[seq_id] = seq
+        self.scheduler.add(seq)
+        return seq_id
+
+    def abort_request(self, seq_id: int) -> None:
+        seq = self._seqs.get(seq_id)
+        if seq is not None and seq.state != SeqState.FINISHED:
+            self.scheduler.abort(seq)
+
+    def is_finished(self) -> bool:
+        return not self.scheduler.has_work
+
+    def get_output(self, seq_id: int) -> RequestOutput:
+        seq = self._seqs[seq_id]
+        return RequestOutput(
+            seq_id=seq_id,
+            token_ids=list(seq.output_token_ids),
+            finished=seq.state == SeqState.FINISHED,
+            finish_reason=seq.finish_reason,
+        )
+
+    # ---------------- step ----------------
+    @torch.inference_mode()
+    def step(self) -> List[StepOutput]:
+        batch = self.scheduler.schedule()
+        if batch is None:
+            return []
+        fb = self._build_forward_batch(batch)
+        hidden = self.model(fb, self.kv_caches)
+        hidden_last = hidden[fb.last_token_indices]
+        logits = self.model.compute_logits(hidden_last)
+        tokens = self.sampler.sample(logits, batch.seqs)
+        return self._append_tokens(batch, tokens)
+
+    def _build_forward_batch(self, batch: ScheduledBatch) -> ForwardBatch:
+        dev = self.device
+        bs = self.cfg.block_size
+        if batch.is_prefill:
+            token_ids, positions, slots, cu, seq_lens, last = [], [], [], [0], [], []
+            for seq in batch.seqs:
+                L = seq.prompt_len
+                token_ids.extend(seq.prompt_token_ids)
+                positions.extend(range(L))
+                for i in range(L):
+                    blk = seq.block_ids[i // bs]
+                    slots.append(blk * bs + i % bs)
+                cu.append(cu[-1] + L)
+                seq_lens.append(L)
+                last.append(cu[-1] - 1)
+            return ForwardBatch(
+                token_ids=torch.tensor(token_ids, dtype=torch.int64, device=dev),
+                positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+                slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+                is_prefill=True,
+                cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+                seq_lens=seq_lens,
+                last_token_indices=torch.tensor(last, dtype=torch.int64,
+                                                device=dev),
+            )
+        # decode
+        B = len(batch.seqs)
+        token_ids = [s.output_token_ids[-1] if s.output_token_ids
+                     else s.prompt_token_ids[-1] for s in batch.seqs]
+        positions = [s.num_tokens - 1 for s in batch.seqs]
+        slots = []
+        maxb = max(len(s.block_ids) for s in batch.seqs)
+        bt = torch.zeros((B, maxb), dtype=torch.int32)
+        ctx = torch.tensor([s.num_tokens for s in batch.seqs],
+                           dtype=torch.int32)
+        for i, s in enumerate(batch.seqs):
+            pos = s.num_tokens - 1
+            blk = s.block_ids[pos // bs]
+            slots.append(blk * bs + pos % bs)
+            bt[i, : len(s.block_ids)] = torch.tensor(s.block_ids,
+                                                     dtype=torch.int32)
+        return ForwardBatch(
+            token_ids=torch.tensor(token_ids, dtype=torch.int64, device=dev),
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+            is_prefill=False,
+            block_table=bt.to(dev),
+            ctx_lens=ctx.to(dev),
+            last_token_indices=torch.arange(B, dtype=torch.int64, device=dev),
+        )
+
+    def _append_tokens(self, batch: ScheduledBatch,
+                       tokens: List[int]) -> List[StepOutput]:
+        import time
+
+        outputs = []
+        now = time.monotonic()
+        for seq, tok in zip(batch.seqs, tokens):
+            # decode positions were counted before this new token: the token
+            # generated now becomes part of the sequence
+            seq.output_token_ids.append(tok)
+            if seq.first_token_time is None:
+                seq.first_token_time = now
+            reason = None
+            sp = seq.sampling
+            if not sp.ignore_eos and tok in sp.stop_token_ids:
+                reason = "stop"
+            elif len(seq.output_token_ids) >= sp.max_new_tokens:
+                reason = "length"
+            elif seq.num_tokens >= self.cfg.max_model_len:
+                reason = "length"
+            if reason:
+                self.scheduler.finish(seq, reason)
+            outputs.append(
+                StepOutput(seq.seq_id, tok, reason is not None, reason))
+        return outputs

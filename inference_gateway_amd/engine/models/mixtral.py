@@ -1,0 +1,151 @@
+"""Mixtral-8x7B MoE decoder for the MI355X engine (BASELINE config 5).
+
+Same attention stack as Llama (hand-written gfx950 kernels); the MLP is a
+top-2 router + 8 SwiGLU experts. Expert parallelism strategy on the 8-GPU
+xGMI node: experts are TP-sharded like dense MLPs (each rank holds a slice
+of every expert), so dispatch stays local and the existing row-parallel
+all-reduce covers the combine — the right choice for the fully-connected
+xGMI topology at Mixtral's scale (8 experts / 8 GPUs would leave EP
+all-to-all bound by single-link bandwidth with no GEMM-shape benefit at
+decode batch sizes). Token routing/gather is done with index_select /
+index_add (hipBLASLt batched GEMMs per expert); a fused grouped-GEMM HIP
+kernel is the planned next optimization.
+"""
+from __future__ import annotations
+
+from typing import List, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ... import ops
+from ...parallel.comm import tp_rank, tp_world_size
+from ...parallel.layers import (ColumnParallelLinear, ReplicatedLinear,
+                                RowParallelLinear, _init_weight)
+from ..config import ModelConfig
+from .batch import ForwardBatch
+from .llama import LlamaAttention, _seed
+
+
+class MixtralMoE(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer: int, device, dtype):
+        super().__init__()
+        tp, rank = tp_world_size(), tp_rank()
+        h, inter = cfg.hidden_size, cfg.intermediate_size
+        self.num_experts = cfg.num_experts
+        self.top_k = cfg.num_experts_per_tok
+        self.router = ReplicatedLinear(h, cfg.num_experts, device, dtype,
+                                       _seed(f"l{layer}.router"))
+        assert inter % tp == 0
+        shard = inter // tp
+        # per-expert fused [gate|up] and down weights, stacked:
+        # gate_up: [E, 2*shard, h], down: [E, h, shard]
+        gus, downs = [], []
+        for e in range(cfg.num_experts):
+            g = ColumnParallelLinear(h, inter, rank, tp, device, dtype,
+                                     _seed(f"l{layer}.e{e}.gate"))
+            u = ColumnParallelLinear(h, inter, rank, tp, device, dtype,
+                                     _seed(f"l{layer}.e{e}.up"))
+            d = RowParallelLinear(inter, h, rank, tp, device, dtype,
+                                  _seed(f"l{layer}.e{e}.down"))
+            gus.append(torch.cat([g.weight.data, u.weight.data], 0))
+            downs.append(d.weight.data)
+        self.gate_up = nn.Parameter(torch.stack(gus).contiguous(),
+                                    requires_grad=False)
+        self.down = nn.Parameter(torch.stack(downs).contiguous(),
+                                 requires_grad=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ...parallel.comm import tp_all_reduce
+
+        T, h = x.shape
+        router_logits = self.router(x).float()            # [T, E]
+        weights, topk_idx = torch.topk(router_logits, self.top_k, dim=-1)
+        weights = torch.softmax(weights, dim=-1).to(x.dtype)  # [T, K]
+        out = torch.zeros_like(x)
+        flat_expert = topk_idx.reshape(-1)                # [T*K]
+        flat_token = (
+            torch.arange(T, device=x.device)
+            .unsqueeze(1)
+            .expand(T, self.top_k)
+            .reshape(-1)
+        )
+        # sort tokens by expert so each expert runs one contiguous GEMM
+        order = torch.argsort(flat_expert, stable=True)
+        sorted_expert = flat_expert[order]
+        sorted_token = flat_token[order]
+        counts = torch.bincount(sorted_expert, minlength=self.num_experts)
+        counts_list = counts.tolist()
+        start = 0
+        for e, cnt in enumerate(counts_list):
+            if cnt == 0:
+                continue
+            toks = sorted_token[start:start + cnt]
+            xe = x.index_select(0, toks)
+            gu = F.linear(xe, self.gate_up[e])
+            ye = F.linear(ops.silu_mul(gu), self.down[e])
+            w = weights.reshape(-1)[order[start:start + cnt]].unsqueeze(1)
+            out.index_add_(0, toks, ye * w)
+            start += cnt
+        return tp_all_reduce(out)
+
+
+class MixtralLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer: int, device, dtype):
+        super().__init__()
+        self.input_norm_w = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=dtype, device=device),
+            requires_grad=False)
+        self.post_norm_w = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=dtype, device=device),
+            requires_grad=False)
+        self.attn = LlamaAttention(cfg, layer, device, dtype)
+        self.moe = MixtralMoE(cfg, layer, device, dtype)
+        self.eps = cfg.norm_eps
+
+    def forward(self, x, residual, fb, kv_cache, cos_sin):
+        if residual is None:
+            residual = x.clone()
+            x = ops.rmsnorm(x, self.input_norm_w, self.eps)
+        else:
+            x, residual = ops.fused_add_rmsnorm(
+                x, residual, self.input_norm_w, self.eps)
+        x = self.attn(x, fb, kv_cache, cos_sin)
+        x, residual = ops.fused_add_rmsnorm(x, residual, self.post_norm_w,
+                                            self.eps)
+        x = self.moe(x)
+        return x, residual
+
+
+class MixtralForCausalLM(nn.Module):
+    def __init__(self, cfg: ModelConfig, device, dtype=torch.bfloat16):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = nn.Parameter(
+            _init_weight((cfg.vocab_size, cfg.hidden_size), device, dtype,
+                         _seed("embed")),
+            requires_grad=False)
+        self.layers = nn.ModuleList(
+            MixtralLayer(cfg, i, device, dtype) for i in range(cfg.num_layers))
+        self.final_norm_w = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=dtype, device=device),
+            requires_grad=False)
+        self.lm_head = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size,
+                                        device, dtype, _seed("lm_head"))
+        self.cos_sin = ops.build_cos_sin_cache(
+            cfg.head_dim, cfg.max_position, cfg.rope_base, device=device)
+
+    def forward(self, fb: ForwardBatch,
+                kv_caches: List[Tuple[torch.Tensor, torch.Tensor]]
+                ) -> torch.Tensor:
+        x = F.embedding(fb.token_ids, self.embed)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer(x, residual, fb, kv_caches[i], self.cos_sin)
+        x, _ = ops.fused_add_rmsnorm(x, residual, self.final_norm_w,
+                                     self.cfg.norm_eps)
+        return x
+
+    def compute_logits(self, hidden_last: torch.Tensor) -> torch.Tensor:
+        return self.lm_head(hidden_last).float()

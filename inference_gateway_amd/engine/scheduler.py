@@ -1,0 +1,137 @@
+"""Continuous-batching scheduler.
+
+Net-new engine scope (SURVEY.md §7 stage 3c): admission queue + running set
+with prefill-priority interleave and LIFO preemption when the paged cache
+runs dry. Sized for 288 GB HBM per GPU — on a real MI355X the block pool is
+large enough that preemption is the exception, not the steady state.
+"""
+from __future__ import annotations
+
+import enum
+import time
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from .config import SamplingParams
+from .kvcache import BlockManager
+
+
+class SeqState(enum.Enum):
+    WAITING = "waiting"
+    RUNNING = "running"
+    FINISHED = "finished"
+
+
+@dataclass
+class Sequence:
+    seq_id: int
+    prompt_token_ids: List[int]
+    sampling: SamplingParams
+    arrival_time: float = field(default_factory=time.monotonic)
+    output_token_ids: List[int] = field(default_factory=list)
+    block_ids: List[int] = field(default_factory=list)
+    state: SeqState = SeqState.WAITING
+    finish_reason: Optional[str] = None
+    first_token_time: Optional[float] = None
+
+    @property
+    def num_tokens(self) -> int:
+        return len(self.prompt_token_ids) + len(self.output_token_ids)
+
+    @property
+    def prompt_len(self) -> int:
+        return len(self.prompt_token_ids)
+
+
+@dataclass
+class ScheduledBatch:
+    is_prefill: bool
+    seqs: List[Sequence]
+
+
+class Scheduler:
+    def __init__(self, block_manager: BlockManager, max_num_seqs: int,
+                 max_batched_tokens: int, max_model_len: int):
+        self.bm = block_manager
+        self.max_num_seqs = max_num_seqs
+        self.max_batched_tokens = max_batched_tokens
+        self.max_model_len = max_model_len
+        self.waiting: deque[Sequence] = deque()
+        self.running: List[Sequence] = []
+
+    def add(self, seq: Sequence) -> None:
+        if seq.prompt_len > self.max_model_len:
+            raise ValueError(
+                f"prompt length {seq.prompt_len} exceeds max_model_len "
+                f"{self.max_model_len}"
+            )
+        self.waiting.append(seq)
+
+    @property
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def schedule(self) -> Optional[ScheduledBatch]:
+        # Prefill-priority: admit as many waiting prompts as fit the token
+        # budget and the block pool.
+        prefills: List[Sequence] = []
+        tokens = 0
+        while self.waiting and len(self.running) + len(prefills) < self.max_num_seqs:
+            seq = self.waiting[0]
+            if tokens + seq.prompt_len > self.max_batched_tokens and prefills:
+                break
+            if not self.bm.can_allocate(seq.prompt_len):
+                break
+            if tokens + seq.prompt_len > self.max_batched_tokens:
+                break  # single over-budget prompt: wait for its own batch
+            self.waiting.popleft()
+            seq.block_ids = self.bm.allocate(seq.prompt_len)
+            seq.state = SeqState.RUNNING
+            prefills.append(seq)
+            tokens += seq.prompt_len
+        if prefills:
+            self.running.extend(prefills)
+            return ScheduledBatch(is_prefill=True, seqs=prefills)
+
+        if not self.running:
+            return None
+
+        # Decode: every running sequence generates one token; make room first.
+        # LIFO preemption — newest sequence goes back to waiting, whole-prompt
+        # recompute (matches the 288GB-HBM design point where this is rare).
+        while True:
+            try:
+                for seq in self.running:
+                    # this step writes KV at position num_tokens-1
+                    self.bm.ensure_capacity(seq.block_ids, seq.num_tokens)
+                break
+            except RuntimeError:
+                victim = self.running.pop()  # newest arrival
+                self.bm.free(victim.block_ids)
+                # recompute from scratch later, keep generated tokens as part
+                # of the new prompt so the request resumes transparently
+                victim.prompt_token_ids = (
+                    victim.prompt_token_ids + victim.output_token_ids
+                )
+                victim.output_token_ids = []
+                victim.state = SeqState.WAITING
+                self.waiting.appendleft(victim)
+                if not self.running:
+                    return None
+        return ScheduledBatch(is_prefill=False, seqs=list(self.running))
+
+    def finish(self, seq: Sequence, reason: str) -> None:
+        seq.state = SeqState.FINISHED
+        seq.finish_reason = reason
+        self.bm.free(seq.block_ids)
+        if seq in self.running:
+            self.running.remove(seq)
+
+    def abort(self, seq: Sequence) -> None:
+        if seq.state == SeqState.WAITING:
+            try:
+                self.waiting.remove(seq)
+            except ValueError:
+                pass
+        self.finish(seq, "abort")
