@@ -1,0 +1,81 @@
+"""Async front-end over LLMEngine for the gateway.
+
+The engine loop runs on a dedicated thread (GPU work must not block the
+asyncio event loop serving SSE clients); token outputs are delivered to
+per-request asyncio queues via call_soon_threadsafe. This replaces the
+reference's streaming goroutine + channel relay (providers/core/
+provider.go:262-298) with an in-process path.
+"""
+from __future__ import annotations
+
+import asyncio
+import threading
+from typing import AsyncIterator, Dict, Optional
+
+from .config import EngineConfig, SamplingParams
+from .engine import LLMEngine, StepOutput
+
+
+class AsyncEngine:
+    _SENTINEL = object()
+
+    def __init__(self, cfg: EngineConfig, device: str = "cuda:0"):
+        self.engine = LLMEngine(cfg, device=device)
+        self._lock = threading.Lock()
+        self._wake = threading.Event()
+        self._stop = False
+        self._queues: Dict[int, asyncio.Queue] = {}
+        self._loops: Dict[int, asyncio.AbstractEventLoop] = {}
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name="igw-engine")
+        self._thread.start()
+
+    # ---- engine thread --------------------------------------------------
+    def _run(self) -> None:
+        while not self._stop:
+            with self._lock:
+                has_work = self.engine.scheduler.has_work
+            if not has_work:
+                self._wake.wait(timeout=0.05)
+                self._wake.clear()
+                continue
+            with self._lock:
+                outputs = self.engine.step()
+            for out in outputs:
+                self._dispatch(out)
+
+    def _dispatch(self, out: StepOutput) -> None:
+        q = self._queues.get(out.seq_id)
+        loop = self._loops.get(out.seq_id)
+        if q is None or loop is None:
+            return
+        loop.call_soon_threadsafe(q.put_nowait, out)
+        if out.finished:
+            loop.call_soon_threadsafe(q.put_nowait, self._SENTINEL)
+
+    # ---- async API ------------------------------------------------------
+    async def generate(self, prompt_token_ids, sampling: SamplingParams
+                       ) -> AsyncIterator[StepOutput]:
+        loop = asyncio.get_running_loop()
+        q: asyncio.Queue = asyncio.Queue()
+        with self._lock:
+            rid = self.engine.add_request(prompt_token_ids, sampling)
+            self._queues[rid] = q
+            self._loops[rid] = loop
+        self._wake.set()
+        try:
+            while True:
+                item = await q.get()
+                if item is self._SENTINEL:
+                    break
+                yield item
+        finally:
+            with self._lock:
+                self._queues.pop(rid, None)
+                self._loops.pop(rid, None)
+                self.engine.abort_request(rid)
+
+    def shutdown(self) -> None:
+        self._stop = True
+        self._wake.set()
+        self._thread.join(timeout=5)

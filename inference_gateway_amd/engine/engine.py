@@ -62,9 +62,16 @@ class LLMEngine:
         self.kv_caches = self._allocate_kv(num_blocks)
         self._seq_counter = itertools.count()
         self._seqs: Dict[int, Sequence] = {}
-        self._graphs = None  # hipGraph runner, installed by capture_graphs()
         self.max_blocks_per_seq = (
             cfg.max_model_len + cfg.block_size - 1) // cfg.block_size
+        self.graph_runner = None
+        if (self.device.type == "cuda" and cfg.use_hip_graphs
+                and not cfg.enforce_eager):
+            from .graph_runner import DecodeGraphRunner
+
+            self.graph_runner = DecodeGraphRunner(
+                self.model, self.kv_caches, cfg, self.device,
+                self.max_blocks_per_seq)
 
     # ---------------- setup ----------------
     def _auto_num_blocks(self) -> int:
@@ -131,9 +138,19 @@ class LLMEngine:
         batch = self.scheduler.schedule()
         if batch is None:
             return []
-        fb = self._build_forward_batch(batch)
-        hidden = self.model(fb, self.kv_caches)
-        hidden_last = hidden[fb.last_token_indices]
+        if self.graph_runner is not None:
+            # keep the per-slot device block tables current
+            for seq in batch.seqs:
+                if seq.synced_blocks < len(seq.block_ids):
+                    self.graph_runner.write_block_row(
+                        seq.slot, seq.block_ids, start=seq.synced_blocks)
+                    seq.synced_blocks = len(seq.block_ids)
+        if not batch.is_prefill and self.graph_runner is not None:
+            hidden_last = self.graph_runner.execute(batch.seqs)
+        else:
+            fb = self._build_forward_batch(batch)
+            hidden = self.model(fb, self.kv_caches)
+            hidden_last = hidden[fb.last_token_indices]
         logits = self.model.compute_logits(hidden_last)
         tokens = self.sampler.sample(logits, batch.seqs)
         return self._append_tokens(batch, tokens)

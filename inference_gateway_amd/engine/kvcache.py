@@ -15,7 +15,9 @@ class BlockManager:
     def __init__(self, num_blocks: int, block_size: int):
         self.num_blocks = num_blocks
         self.block_size = block_size
-        self._free: List[int] = list(range(num_blocks - 1, -1, -1))
+        # block 0 is reserved as the scratch target for hipGraph padding
+        # rows (graph_runner.py) and is never handed out
+        self._free: List[int] = list(range(num_blocks - 1, 0, -1))
 
     @property
     def num_free(self) -> int:

@@ -34,6 +34,8 @@ class Sequence:
     state: SeqState = SeqState.WAITING
     finish_reason: Optional[str] = None
     first_token_time: Optional[float] = None
+    slot: Optional[int] = None        # persistent batch slot (graph runner)
+    synced_blocks: int = 0            # device block-table rows written so far
 
     @property
     def num_tokens(self) -> int:
@@ -59,6 +61,7 @@ class Scheduler:
         self.max_model_len = max_model_len
         self.waiting: deque[Sequence] = deque()
         self.running: List[Sequence] = []
+        self._free_slots: List[int] = list(range(max_num_seqs - 1, -1, -1))
 
     def add(self, seq: Sequence) -> None:
         if seq.prompt_len > self.max_model_len:
@@ -87,6 +90,8 @@ class Scheduler:
                 break  # single over-budget prompt: wait for its own batch
             self.waiting.popleft()
             seq.block_ids = self.bm.allocate(seq.prompt_len)
+            seq.slot = self._free_slots.pop()
+            seq.synced_blocks = 0
             seq.state = SeqState.RUNNING
             prefills.append(seq)
             tokens += seq.prompt_len
@@ -109,6 +114,9 @@ class Scheduler:
             except RuntimeError:
                 victim = self.running.pop()  # newest arrival
                 self.bm.free(victim.block_ids)
+                if victim.slot is not None:
+                    self._free_slots.append(victim.slot)
+                    victim.slot = None
                 # recompute from scratch later, keep generated tokens as part
                 # of the new prompt so the request resumes transparently
                 victim.prompt_token_ids = (
@@ -125,6 +133,9 @@ class Scheduler:
         seq.state = SeqState.FINISHED
         seq.finish_reason = reason
         self.bm.free(seq.block_ids)
+        if seq.slot is not None:
+            self._free_slots.append(seq.slot)
+            seq.slot = None
         if seq in self.running:
             self.running.remove(seq)
 

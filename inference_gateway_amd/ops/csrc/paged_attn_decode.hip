@@ -64,14 +64,18 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
 
   const int tok = lane >> 2;        // 16 tokens per block tile
   const int seg = (lane & 3) * 32;  // 32-dim segment for the K dot
-  const int dv = lane * 2;          // 2 output dims per lane in the V pass
+  // V pass: 16 lanes x 8 contiguous dims (one b128 read per lane), 4 token
+  // subgroups; per instruction the wave reads 4 complete 256 B V rows.
+  const int vlane = lane & 15;      // dim group: dims [vlane*8, vlane*8+8)
+  const int vtok = lane >> 4;       // token subgroup: t = vtok + 4*t4
 
-  float m[G], l[G], acc[G][2];
+  float m[G], l[G], acc[G][8];
 #pragma unroll
   for (int j = 0; j < G; ++j) {
     m[j] = -INFINITY;
     l[j] = 0.0f;
-    acc[j][0] = acc[j][1] = 0.0f;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[j][e] = 0.0f;
   }
 
   for (int b = wave; b < nblocks; b += NWAVES) {
@@ -108,8 +112,8 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
       if (pmax > m[j]) {
         const float alpha = __expf(m[j] - pmax);
         l[j] *= alpha;
-        acc[j][0] *= alpha;
-        acc[j][1] *= alpha;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc[j][e] *= alpha;
         m[j] = pmax;
       }
       p[j] = (s == -INFINITY) ? 0.0f : __expf(s - m[j]);
@@ -119,19 +123,34 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
     // intra-wave LDS write->read ordering is handled by the compiler's
     // lgkmcnt tracking (same wave, same LDS object)
 
-    // ---- V pass: 2 dims per lane, loop over the 16 tokens ----
+    // ---- V pass: b128 per lane; 4 full V rows per wave instruction ----
     const int tlimit = min(BS, ctx - b * BS);  // tokens valid in this block
-#pragma unroll 4
-    for (int t = 0; t < tlimit; ++t) {
-      const uint16_t* vp = vbase + t * HD + dv;
-      const float v0 = bf16_to_f32(vp[0]);
-      const float v1 = bf16_to_f32(vp[1]);
+#pragma unroll
+    for (int t4 = 0; t4 < 4; ++t4) {
+      const int t = vtok + 4 * t4;
+      if (t >= tlimit) continue;  // stale-block data may be non-finite
+      ushort8 v8 = *reinterpret_cast<const ushort8*>(
+          vbase + t * HD + vlane * 8);
+      float vf[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) vf[e] = bf16_to_f32(v8.v[e]);
 #pragma unroll
       for (int j = 0; j < G; ++j) {
         const float pj = p_lds[wave][j][t];
-        acc[j][0] = fmaf(pj, v0, acc[j][0]);
-        acc[j][1] = fmaf(pj, v1, acc[j][1]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc[j][e] = fmaf(pj, vf[e], acc[j][e]);
       }
+    }
+  }
+
+  // fold the 4 token-subgroup partial sums (lanes l, l+16, l+32, l+48 hold
+  // the same dims)
+#pragma unroll
+  for (int j = 0; j < G; ++j) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      acc[j][e] += __shfl_xor(acc[j][e], 16, WAVE_SIZE);
+      acc[j][e] += __shfl_xor(acc[j][e], 32, WAVE_SIZE);
     }
   }
 
@@ -142,8 +161,11 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
       m_lds[wave][j] = m[j];
       l_lds[wave][j] = l[j];
     }
-    acc_lds[wave][j][dv] = acc[j][0];
-    acc_lds[wave][j][dv + 1] = acc[j][1];
+    if (vtok == 0) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        acc_lds[wave][j][vlane * 8 + e] = acc[j][e];
+    }
   }
   __syncthreads();
 

@@ -3,20 +3,23 @@
 // Net-new engine scope (SURVEY.md §7 stage 3b). One workgroup per sequence,
 // operating on the fp32 logits row (vocab ~128k):
 //
-//   1. block max M and expsum Z (softmax denominator) in one pass;
+//   1. block max M in one pass;
 //   2. greedy path (temperature == 0): argmax with lowest-index tie break;
-//   3. top-k threshold: 4-level radix select over the monotone uint32 key
-//      image of the fp32 logit (256-bin LDS histogram per level);
-//   4. top-p (nucleus) threshold: same radix walk but with exp-weighted
-//      histograms, stopping when the kept probability mass reaches p * Z;
-//   5. final inverse-CDF sample over the kept set in index order, with the
-//      caller-supplied uniform u (host draws it from torch's RNG so runs are
-//      reproducible under torch.manual_seed).
+//   3. joint 4-level radix walk over the monotone uint32 key image of the
+//      fp32 logit: each level is ONE pass that builds a count histogram
+//      (for the top-k threshold) and an exp-weighted histogram (for the
+//      top-p mass threshold) simultaneously, into WAVE-PRIVATE LDS bins
+//      (logit exponents cluster in a handful of bins, so shared-bin atomics
+//      would serialize across the whole block); Z falls out of level 3's
+//      weight histogram for free;
+//   4. kept-mass pass, then an inverse-CDF walk over the kept set in index
+//      order with the caller-supplied uniform u (host draws it from torch's
+//      RNG so runs are reproducible under torch.manual_seed).
 //
-// The Python reference implementation (ops/torch_ref.py::sample_ref) follows
-// exactly these semantics (value-threshold top-k keeps ties; nucleus set is
-// the smallest value-threshold set with mass >= p; index-order inverse CDF),
-// so GPU vs CPU tests compare token-for-token with fixed u.
+// Thresholds are VALUE thresholds: top-k keeps ties of the k-th value; the
+// nucleus set is the smallest value-threshold set with mass >= p*Z. The
+// Python reference (ops/torch_ref.py::sample_ref) implements exactly these
+// semantics, so GPU vs CPU tests compare token-for-token with fixed u.
 
 #include <ATen/cuda/CUDAContext.h>
 #include <torch/extension.h>
@@ -43,15 +46,18 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
     const float* __restrict__ top_p,      // [B] (>=1 => disabled)
     const float* __restrict__ uniforms,   // [B]
     int V) {
-  __shared__ float redf[BLOCK / WAVE_SIZE];
-  __shared__ float fbins[256];
-  __shared__ uint32_t ubins[256];
+  constexpr int NW = BLOCK / WAVE_SIZE;
+  __shared__ float redf[NW];
+  __shared__ uint32_t ubins[NW][256];
+  __shared__ float fbins[NW][256];
   __shared__ uint32_t s_u32[2];
-  __shared__ float s_f32[2];
+  __shared__ float s_f32[4];
   __shared__ float seg_sums[BLOCK + 1];
 
   const int seq = blockIdx.x;
   const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
   const float* row = logits + (int64_t)seq * V;
   const float temp = temperature[seq];
   const bool greedy = (temp <= 0.0f);
@@ -60,22 +66,18 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
   // ---- pass 1: max ----
   float mx = -INFINITY;
   for (int i = tid; i < V; i += BLOCK) mx = fmaxf(mx, row[i] * inv_t);
-  {
-    // block reduce max via LDS
-    int lane = tid & 63, wave = tid >> 6;
-    mx = wave_reduce_max(mx);
-    if (lane == 0) redf[wave] = mx;
-    __syncthreads();
-    float r = (tid < BLOCK / WAVE_SIZE) ? redf[tid] : -INFINITY;
-    for (int off = 1; off < BLOCK / WAVE_SIZE; off <<= 1)
-      r = fmaxf(r, __shfl_xor(r, off, WAVE_SIZE));
-    if (tid == 0) s_f32[0] = r;
-    __syncthreads();
+  mx = wave_reduce_max(mx);
+  if (lane == 0) redf[wave] = mx;
+  __syncthreads();
+  if (tid == 0) {
+    float r = redf[0];
+    for (int w = 1; w < NW; ++w) r = fmaxf(r, redf[w]);
+    s_f32[0] = r;
   }
+  __syncthreads();
   const float M = s_f32[0];
 
   if (greedy) {
-    // argmax, lowest index wins ties
     if (tid == 0) s_u32[0] = 0xffffffffu;
     __syncthreads();
     for (int i = tid; i < V; i += BLOCK) {
@@ -86,98 +88,98 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
     return;
   }
 
-  // ---- pass 2: expsum Z ----
-  float z = 0.0f;
-  for (int i = tid; i < V; i += BLOCK) z += __expf(row[i] * inv_t - M);
-  z = block_reduce_sum<BLOCK>(z, redf);
-  const float Z = z;
-
   const int k = top_k[seq];
   const float p = top_p[seq];
+  const bool do_k = (k > 0 && k < V);
+  const bool do_p = (p < 1.0f);
 
-  // ---- top-k radix select (over uint32 keys, MSB byte first) ----
-  uint32_t tk_key = 0;  // keep everything by default
-  if (k > 0 && k < V) {
-    uint32_t prefix = 0;
-    uint32_t remaining = (uint32_t)k;
-    for (int level = 3; level >= 0; --level) {
-      for (int b = tid; b < 256; b += BLOCK) ubins[b] = 0;
-      __syncthreads();
-      for (int i = tid; i < V; i += BLOCK) {
-        const uint32_t key = f32_key(row[i] * inv_t);
-        const bool match =
-            (level == 3) || ((key >> ((level + 1) * 8)) == prefix);
-        if (match) atomicAdd(&ubins[(key >> (level * 8)) & 0xff], 1u);
+  // ---- joint radix walk: 4 levels, one pass each ----
+  uint32_t tk_prefix = 0, tp_prefix = 0;
+  uint32_t tk_remaining = do_k ? (uint32_t)k : 0;
+  float tp_target = 0.0f;  // set after level 3 (needs Z)
+  float Z = 0.0f;
+
+  for (int level = 3; level >= 0; --level) {
+    for (int b = tid; b < 256; b += BLOCK) {
+#pragma unroll
+      for (int w = 0; w < NW; ++w) {
+        ubins[w][b] = 0;
+        fbins[w][b] = 0.0f;
       }
-      __syncthreads();
-      if (tid == 0) {
+    }
+    __syncthreads();
+    const int shift = level * 8;
+    for (int i = tid; i < V; i += BLOCK) {
+      const float x = row[i] * inv_t;
+      const uint32_t key = f32_key(x);
+      const uint32_t hi = (level == 3) ? 0 : (key >> (shift + 8));
+      const uint32_t b = (key >> shift) & 0xff;
+      if (do_k && (level == 3 || hi == tk_prefix)) {
+        atomicAdd(&ubins[wave][b], 1u);
+      }
+      if (level == 3 || hi == tp_prefix) {
+        // level 3 weight histogram doubles as Z (do_p or not)
+        atomicAdd(&fbins[wave][b], __expf(x - M));
+      }
+    }
+    __syncthreads();
+    if (tid == 0) {
+      if (level == 3) {
+        float z = 0.0f;
+        for (int b = 0; b < 256; ++b)
+          for (int w = 0; w < NW; ++w) z += fbins[w][b];
+        Z = z;
+        s_f32[1] = z;
+        tp_target = p * z;
+      }
+      if (do_k) {
         uint32_t cum = 0;
         int bstar = 0;
         for (int b = 255; b >= 0; --b) {
-          if (cum + ubins[b] >= remaining) {
+          uint32_t c = 0;
+          for (int w = 0; w < NW; ++w) c += ubins[w][b];
+          if (cum + c >= tk_remaining) {
             bstar = b;
             break;
           }
-          cum += ubins[b];
+          cum += c;
         }
-        s_u32[0] = (prefix << 8) | (uint32_t)bstar;
-        s_u32[1] = remaining - cum;
+        tk_prefix = (tk_prefix << 8) | (uint32_t)bstar;
+        tk_remaining -= cum;
+        s_u32[0] = tk_prefix;
       }
-      __syncthreads();
-      prefix = s_u32[0];
-      remaining = s_u32[1];
-    }
-    tk_key = prefix;
-  }
-
-  // ---- top-p radix select (exp-weighted histogram) ----
-  uint32_t tp_key = 0;
-  if (p < 1.0f) {
-    uint32_t prefix = 0;
-    float target = p * Z;
-    for (int level = 3; level >= 0; --level) {
-      for (int b = tid; b < 256; b += BLOCK) fbins[b] = 0.0f;
-      __syncthreads();
-      for (int i = tid; i < V; i += BLOCK) {
-        const float x = row[i] * inv_t;
-        const uint32_t key = f32_key(x);
-        const bool match =
-            (level == 3) || ((key >> ((level + 1) * 8)) == prefix);
-        if (match) atomicAdd(&fbins[(key >> (level * 8)) & 0xff], __expf(x - M));
-      }
-      __syncthreads();
-      if (tid == 0) {
+      if (do_p) {
         float cum = 0.0f;
         int bstar = 0;
         for (int b = 255; b >= 0; --b) {
-          if (cum + fbins[b] >= target) {
+          float c = 0.0f;
+          for (int w = 0; w < NW; ++w) c += fbins[w][b];
+          if (cum + c >= tp_target) {
             bstar = b;
             break;
           }
-          cum += fbins[b];
+          cum += c;
         }
-        s_u32[0] = (prefix << 8) | (uint32_t)bstar;
-        s_f32[1] = target - cum;
+        tp_prefix = (tp_prefix << 8) | (uint32_t)bstar;
+        tp_target -= cum;
+        s_f32[2] = tp_target;
+        s_u32[1] = tp_prefix;
       }
-      __syncthreads();
-      prefix = s_u32[0];
-      target = s_f32[1];
     }
-    tp_key = prefix;
+    __syncthreads();
+    if (do_k) tk_prefix = s_u32[0];
+    if (do_p) {
+      tp_prefix = s_u32[1];
+      tp_target = s_f32[2];
+    }
+    // each thread needs tk_remaining only on thread 0; prefixes broadcast
+    if (!do_k && !do_p && level == 3) break;  // pure-temperature sampling
   }
 
-  const uint32_t tau = max(tk_key, tp_key);
+  const uint32_t tau =
+      max(do_k ? tk_prefix : 0u, do_p ? tp_prefix : 0u);
 
-  // ---- kept-mass Zf ----
-  float zf = 0.0f;
-  for (int i = tid; i < V; i += BLOCK) {
-    const float x = row[i] * inv_t;
-    if (f32_key(x) >= tau) zf += __expf(x - M);
-  }
-  zf = block_reduce_sum<BLOCK>(zf, redf);
-
-  // ---- inverse-CDF sample in index order ----
-  const float target = uniforms[seq] * zf;
+  // ---- kept mass + per-thread segment sums (one fused pass) ----
   const int seg = (V + BLOCK - 1) / BLOCK;
   const int s0 = tid * seg;
   const int s1 = min(V, s0 + seg);
@@ -188,7 +190,6 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
   }
   seg_sums[tid] = segsum;
   __syncthreads();
-  // exclusive scan over BLOCK segment sums (single thread is fine: 256 adds)
   if (tid == 0) {
     float run = 0.0f;
     for (int t = 0; t < BLOCK; ++t) {
@@ -196,14 +197,17 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
       seg_sums[t] = run;
       run += s;
     }
-    seg_sums[BLOCK] = run;
+    seg_sums[BLOCK] = run;  // = Zf
     s_u32[0] = 0xffffffffu;
   }
   __syncthreads();
-  // the thread whose segment contains the target walks it
+
+  // ---- inverse-CDF walk in index order ----
+  const float zf = seg_sums[BLOCK];
+  const float target = uniforms[seq] * zf;
   const float base = seg_sums[tid];
   const float next = seg_sums[tid + 1];
-  if (target >= base && (target < next || (tid == BLOCK - 1))) {
+  if (target >= base && (target < next || tid == BLOCK - 1)) {
     float run = base;
     int chosen = -1;
     for (int i = s0; i < s1; ++i) {
@@ -220,12 +224,10 @@ __global__ __launch_bounds__(BLOCK) void sample_kernel(
   if (tid == 0) {
     uint32_t c = s_u32[0];
     if (c == 0xffffffffu) {
-      // numeric corner (u ~ 1.0 and rounding): fall back to last kept token
-      // -> handled by scanning from the top thread; choose argmax instead.
+      // numeric corner (u ~ 1.0): fall back to argmax
       c = 0;
       float best = -INFINITY;
       for (int i = 0; i < V; ++i) {
-        // cheap host-style fallback; practically unreachable
         const float x = row[i] * inv_t;
         if (x > best) {
           best = x;

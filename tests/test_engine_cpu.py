@@ -10,15 +10,16 @@ from inference_gateway_amd.engine.kvcache import BlockManager
 
 
 def test_block_manager_alloc_free():
-    bm = BlockManager(8, 16)
+    bm = BlockManager(8, 16)  # block 0 is reserved scratch => 7 usable
     a = bm.allocate(33)  # 3 blocks
-    assert len(a) == 3 and bm.num_free == 5
+    assert len(a) == 3 and bm.num_free == 4
+    assert 0 not in a
     bm.ensure_capacity(a, 48)
     assert len(a) == 3
     bm.ensure_capacity(a, 49)
-    assert len(a) == 4 and bm.num_free == 4
+    assert len(a) == 4 and bm.num_free == 3
     bm.free(a)
-    assert bm.num_free == 8 and a == []
+    assert bm.num_free == 7 and a == []
 
 
 def _engine(**kw):
