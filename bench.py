@@ -29,6 +29,12 @@ sys.path.insert(0, str(Path(__file__).resolve().parent))
 
 import torch  # noqa: E402
 
+# GEMM backend experiment knob: IGW_BLAS=hipblaslt routes torch GEMMs
+# through hipBLASLt instead of rocBLAS/Tensile (decode GEMMs are skinny-M;
+# see profiles/r01).
+if os.environ.get("IGW_BLAS", "") == "hipblaslt" and torch.cuda.is_available():
+    torch.backends.cuda.preferred_blas_library("cublaslt")
+
 
 def main() -> None:
     ap = argparse.ArgumentParser()

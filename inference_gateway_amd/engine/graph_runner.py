@@ -115,20 +115,19 @@ class DecodeGraphRunner:
 
         st = self.staging
         pad_slot = seqs[0].slot
-        for i, s in enumerate(seqs):
-            pos = s.num_tokens - 1
-            st[0, i] = (s.output_token_ids[-1] if s.output_token_ids
-                        else s.prompt_token_ids[-1])
-            st[1, i] = pos
-            st[2, i] = s.block_ids[pos // bs] * bs + pos % bs
-            st[3, i] = s.num_tokens
-            self.slot_idx_staging[i] = s.slot
-        for i in range(B, bucket):
-            st[0, i] = 0
-            st[1, i] = 0
-            st[2, i] = 0          # scratch block 0
-            st[3, i] = 1
-            self.slot_idx_staging[i] = pad_slot
+        npad = bucket - B
+        tokens = [(s.output_token_ids[-1] if s.output_token_ids
+                   else s.prompt_token_ids[-1]) for s in seqs] + [0] * npad
+        positions = [s.num_tokens - 1 for s in seqs] + [0] * npad
+        # padding rows write their dummy KV into reserved scratch block 0
+        slot_map = [s.block_ids[(s.num_tokens - 1) // bs] * bs
+                    + (s.num_tokens - 1) % bs for s in seqs] + [0] * npad
+        ctx = [s.num_tokens for s in seqs] + [1] * npad
+        slots = [s.slot for s in seqs] + [pad_slot] * npad
+        st[:, :bucket] = torch.tensor(
+            [tokens, positions, slot_map, ctx], dtype=torch.int64)
+        self.slot_idx_staging[:bucket] = torch.tensor(slots,
+                                                      dtype=torch.int64)
 
         dev_st = st[:, :bucket].to(self.device, non_blocking=True)
         self.in_tokens[:bucket].copy_(dev_st[0])

@@ -3,18 +3,23 @@
 // Net-new engine scope (SURVEY.md §7 stage 3b): single-query attention over a
 // paged KV cache, GQA-aware. One workgroup per (sequence, kv_head); the GQA
 // group's G query heads ride along so each K/V byte is read exactly once per
-// step — decode attention on this chip is HBM-bound, so the kernel is built
-// around minimal, fully-coalesced traffic:
+// step. Decode attention is HBM-bound; the kernel is built around coalesced
+// traffic and LATENCY HIDING (profiles/r01: the naive per-head loop was
+// latency-bound at 2.5 TB/s, 8x the per-iteration roofline):
 //
 //   * cache layout [num_blocks, n_kv_heads, block_size=16, head_dim] gives a
 //     contiguous 16x128 bf16 tile (4 KiB) per (block, head);
-//   * K pass: the wave reads a whole tile with 64 lanes x 64 B (4 lanes per
-//     token x 32-dim segment), dot-products against Q (pre-scaled, staged in
-//     LDS as fp32), 4-lane shfl reduce -> per-token scores;
-//   * online softmax per wave (m, l, acc in registers; fp32);
-//   * V pass: lanes switch roles to 2 output dims each; per token the wave
-//     reads one contiguous 256 B row of V; probabilities broadcast via LDS;
-//   * waves stripe over cache blocks and merge via LDS at the end
+//   * K pass: 64 lanes x 64 B (4 lanes per token x 32-dim segment); all G
+//     head dot-products are computed back-to-back so their shfl reduction
+//     chains overlap (ILP across heads instead of serial per-head chains);
+//   * token-axis reductions use 4 shfl_xor steps (offsets 4,8,16,32 — the
+//     token index lives in lane bits 2..5), not a full 6-step wave reduce;
+//   * V rows for the CURRENT block and K rows for the NEXT block are issued
+//     before the softmax shfl chain, so HBM latency hides under it;
+//   * P values reach the V-pass lanes via ds_bpermute (no LDS round trip);
+//   * V pass: 16 lanes x 8 contiguous dims (b128 reads), 4 token subgroups,
+//     partials folded with 2 shfl_xor at the end;
+//   * waves stripe over cache blocks and merge via LDS once at the end
 //     (flash-decoding style rescale).
 //
 // Wavefront-64 throughout; no CUDA-isms, no Triton.
@@ -25,6 +30,23 @@
 #include "common.h"
 
 namespace igw {
+
+// xor-reduce over the token axis (lane bits 2..5); values stay replicated
+DEVINLINE float token_reduce_max(float x) {
+  x = fmaxf(x, __shfl_xor(x, 4, WAVE_SIZE));
+  x = fmaxf(x, __shfl_xor(x, 8, WAVE_SIZE));
+  x = fmaxf(x, __shfl_xor(x, 16, WAVE_SIZE));
+  x = fmaxf(x, __shfl_xor(x, 32, WAVE_SIZE));
+  return x;
+}
+
+DEVINLINE float token_reduce_sum(float x) {
+  x += __shfl_xor(x, 4, WAVE_SIZE);
+  x += __shfl_xor(x, 8, WAVE_SIZE);
+  x += __shfl_xor(x, 16, WAVE_SIZE);
+  x += __shfl_xor(x, 32, WAVE_SIZE);
+  return x;
+}
 
 template <int HD, int G, int NWAVES>
 __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
@@ -46,28 +68,35 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
   const int lane = tid & (WAVE_SIZE - 1);
   const int wave = tid >> 6;
 
-  // LDS: Q (fp32, pre-scaled), per-wave P broadcast, merge scratch.
-  __shared__ float q_lds[G][HD];
-  __shared__ float p_lds[NWAVES][G][BS];
+  __shared__ uint16_t q_lds[G][HD];
   __shared__ float m_lds[NWAVES][G];
   __shared__ float l_lds[NWAVES][G];
   __shared__ float acc_lds[NWAVES][G][HD];
 
-  // stage Q*scale into LDS
-  for (int i = tid; i < G * HD; i += NWAVES * WAVE_SIZE) {
-    const int j = i / HD;
-    const int d = i % HD;
-    q_lds[j][d] =
-        bf16_to_f32(q[((int64_t)seq * nq + kvh * G + j) * HD + d]) * scale;
+  // stage Q into LDS (raw bf16; scale is applied to the scores so Q keeps
+  // full precision and the per-lane register image stays PACKED — an
+  // unpacked fp32 Q image is 128 VGPRs for G=4 and collapses occupancy)
+  for (int i = tid * 8; i < G * HD; i += NWAVES * WAVE_SIZE * 8) {
+    *reinterpret_cast<ushort8*>(&q_lds[0][0] + i) =
+        *reinterpret_cast<const ushort8*>(
+            q + ((int64_t)seq * nq + kvh * G) * HD + i);
   }
   __syncthreads();
 
-  const int tok = lane >> 2;        // 16 tokens per block tile
+  const int tok = lane >> 2;        // K pass: 16 tokens per tile
   const int seg = (lane & 3) * 32;  // 32-dim segment for the K dot
-  // V pass: 16 lanes x 8 contiguous dims (one b128 read per lane), 4 token
-  // subgroups; per instruction the wave reads 4 complete 256 B V rows.
-  const int vlane = lane & 15;      // dim group: dims [vlane*8, vlane*8+8)
-  const int vtok = lane >> 4;       // token subgroup: t = vtok + 4*t4
+  const int vlane = lane & 15;      // V pass: dims [vlane*8, vlane*8+8)
+  const int vtok = lane >> 4;       // V pass: tokens t = vtok + 4*t4
+
+  // this lane's Q segment, packed bf16 (16 dwords per head)
+  uint32_t qpk[G][16];
+#pragma unroll
+  for (int j = 0; j < G; ++j)
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      *reinterpret_cast<uint4*>(&qpk[j][c * 4]) =
+          *reinterpret_cast<const uint4*>(&q_lds[j][seg + c * 8]);
+    }
 
   float m[G], l[G], acc[G][8];
 #pragma unroll
@@ -78,73 +107,118 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
     for (int e = 0; e < 8; ++e) acc[j][e] = 0.0f;
   }
 
-  for (int b = wave; b < nblocks; b += NWAVES) {
-    const int64_t pb = block_table[(int64_t)seq * max_blocks + b];
-    const uint16_t* kbase =
-        kcache + (((int64_t)pb * nkv + kvh) * BS) * HD;
-    const uint16_t* vbase =
-        vcache + (((int64_t)pb * nkv + kvh) * BS) * HD;
+  const int64_t head_off = (int64_t)kvh * BS * HD;
+  const int32_t* bt_row = block_table + (int64_t)seq * max_blocks;
 
-    // ---- K pass: 64 B of K per lane (token tok, dims [seg, seg+32)) ----
-    float kf[32];
-    {
-      const uint16_t* kp = kbase + tok * HD + seg;
+  // raw (unconverted) K tile registers for the current block
+  ushort8 kraw[4];
+
+  auto load_k = [&](int b, ushort8* dst) {
+    const uint16_t* kp = kcache + ((int64_t)bt_row[b] * nkv) * BS * HD +
+                         head_off + tok * HD + seg;
 #pragma unroll
-      for (int c = 0; c < 4; ++c) {
-        ushort8 kv8 = *reinterpret_cast<const ushort8*>(kp + c * 8);
+    for (int c = 0; c < 4; ++c)
+      dst[c] = *reinterpret_cast<const ushort8*>(kp + c * 8);
+  };
+
+  const int b0 = wave;
+  if (b0 < nblocks) load_k(b0, kraw);
+
+  for (int b = b0; b < nblocks; b += NWAVES) {
+    // ---- scores for all G heads; K and Q stay PACKED in registers and are
+    //      unpacked pairwise in-loop (dims outer, heads inner) so live
+    //      register state stays under the 3-waves/SIMD budget ----
+    float s[G];
 #pragma unroll
-        for (int e = 0; e < 8; ++e) kf[c * 8 + e] = bf16_to_f32(kv8.v[e]);
+    for (int j = 0; j < G; ++j) s[j] = 0.0f;
+    const uint32_t* kp32 = reinterpret_cast<const uint32_t*>(&kraw[0]);
+#pragma unroll
+    for (int c = 0; c < 16; ++c) {
+      const uint32_t kp = kp32[c];
+      const float k0 = bf16_to_f32((uint16_t)(kp & 0xffff));
+      const float k1 = bf16_to_f32((uint16_t)(kp >> 16));
+#pragma unroll
+      for (int j = 0; j < G; ++j) {
+        uint32_t qp = qpk[j][c];
+        // opaque redefinition: stops the compiler hoisting the unpack of
+        // the loop-invariant packed Q out of the block loop (which would
+        // cost 128 fp32 VGPRs and collapse occupancy)
+        asm("" : "+v"(qp));
+        s[j] = fmaf(bf16_to_f32((uint16_t)(qp & 0xffff)), k0, s[j]);
+        s[j] = fmaf(bf16_to_f32((uint16_t)(qp >> 16)), k1, s[j]);
       }
     }
-    const bool valid = (b * BS + tok) < ctx;
 
-    float p[G];
+    // ---- issue V loads (current block) and K loads (next block) now so
+    //      their latency hides under the shfl chains below ----
+    const int tlimit = min(BS, ctx - b * BS);
+    ushort8 vraw[4];
+    const uint16_t* vp = vcache + ((int64_t)bt_row[b] * nkv) * BS * HD +
+                         head_off + vlane * 8;
+    if (tlimit == BS) {
+#pragma unroll
+      for (int t4 = 0; t4 < 4; ++t4)
+        vraw[t4] = *reinterpret_cast<const ushort8*>(
+            vp + (vtok + 4 * t4) * HD);
+    } else {
+#pragma unroll
+      for (int t4 = 0; t4 < 4; ++t4)
+        if (vtok + 4 * t4 < tlimit)
+          vraw[t4] = *reinterpret_cast<const ushort8*>(
+              vp + (vtok + 4 * t4) * HD);
+    }
+
+    // ---- finish scores: pairwise dot reduce + mask ----
+    const bool valid = (b * BS + tok) < ctx;
 #pragma unroll
     for (int j = 0; j < G; ++j) {
-      float s = 0.0f;
-#pragma unroll
-      for (int d = 0; d < 32; ++d) s += q_lds[j][seg + d] * kf[d];
-      s = group_reduce_sum<4>(s);  // full dot at all 4 lanes of the token
-      if (!valid) s = -INFINITY;
+      float v = s[j] + __shfl_xor(s[j], 1, WAVE_SIZE);
+      v += __shfl_xor(v, 2, WAVE_SIZE);
+      s[j] = valid ? v * scale : -INFINITY;
+    }
 
-      // online softmax (per wave)
-      const float pmax = wave_reduce_max(s);
-      if (pmax > m[j]) {
-        const float alpha = __expf(m[j] - pmax);
+    // ---- online softmax (all-head ILP) ----
+    float pmax[G], p[G];
+#pragma unroll
+    for (int j = 0; j < G; ++j) pmax[j] = token_reduce_max(s[j]);
+#pragma unroll
+    for (int j = 0; j < G; ++j) {
+      if (pmax[j] > m[j]) {
+        const float alpha = __expf(m[j] - pmax[j]);
         l[j] *= alpha;
 #pragma unroll
         for (int e = 0; e < 8; ++e) acc[j][e] *= alpha;
-        m[j] = pmax;
+        m[j] = pmax[j];
       }
-      p[j] = (s == -INFINITY) ? 0.0f : __expf(s - m[j]);
-      l[j] += wave_reduce_sum((lane & 3) == 0 ? p[j] : 0.0f);
-      if ((lane & 3) == 0) p_lds[wave][j][tok] = p[j];
+      p[j] = (s[j] == -INFINITY) ? 0.0f : __expf(s[j] - m[j]);
     }
-    // intra-wave LDS write->read ordering is handled by the compiler's
-    // lgkmcnt tracking (same wave, same LDS object)
+#pragma unroll
+    for (int j = 0; j < G; ++j)
+      l[j] += token_reduce_sum((lane & 3) == 0 ? p[j] : 0.0f);
 
-    // ---- V pass: b128 per lane; 4 full V rows per wave instruction ----
-    const int tlimit = min(BS, ctx - b * BS);  // tokens valid in this block
+    // ---- V accumulate: p reaches the V lanes via bpermute ----
 #pragma unroll
     for (int t4 = 0; t4 < 4; ++t4) {
       const int t = vtok + 4 * t4;
-      if (t >= tlimit) continue;  // stale-block data may be non-finite
-      ushort8 v8 = *reinterpret_cast<const ushort8*>(
-          vbase + t * HD + vlane * 8);
+      float pj[G];
+#pragma unroll
+      for (int j = 0; j < G; ++j)
+        pj[j] = __shfl(p[j], 4 * t, WAVE_SIZE);
+      if (t >= tlimit) continue;
       float vf[8];
 #pragma unroll
-      for (int e = 0; e < 8; ++e) vf[e] = bf16_to_f32(v8.v[e]);
+      for (int e = 0; e < 8; ++e) vf[e] = bf16_to_f32(vraw[t4].v[e]);
 #pragma unroll
-      for (int j = 0; j < G; ++j) {
-        const float pj = p_lds[wave][j][t];
+      for (int j = 0; j < G; ++j)
 #pragma unroll
-        for (int e = 0; e < 8; ++e) acc[j][e] = fmaf(pj, vf[e], acc[j][e]);
-      }
+        for (int e = 0; e < 8; ++e)
+          acc[j][e] = fmaf(pj[j], vf[e], acc[j][e]);
     }
+
+    if (b + NWAVES < nblocks) load_k(b + NWAVES, kraw);
   }
 
-  // fold the 4 token-subgroup partial sums (lanes l, l+16, l+32, l+48 hold
-  // the same dims)
+  // fold the 4 token-subgroup partials (lanes l, l+16, l+32, l+48 share dims)
 #pragma unroll
   for (int j = 0; j < G; ++j) {
 #pragma unroll
@@ -178,7 +252,8 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
     float L = 0.0f, o = 0.0f;
 #pragma unroll
     for (int w = 0; w < NWAVES; ++w) {
-      const float f = (m_lds[w][j] == -INFINITY) ? 0.0f : __expf(m_lds[w][j] - M);
+      const float f =
+          (m_lds[w][j] == -INFINITY) ? 0.0f : __expf(m_lds[w][j] - M);
       L = fmaf(f, l_lds[w][j], L);
       o = fmaf(f, acc_lds[w][j][d], o);
     }
