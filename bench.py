@@ -41,7 +41,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--batch", type=int, default=256,
+    ap.add_argument("--batch", type=int, default=512,
                     help="concurrent sequences per GPU")
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--model", type=str, default="llama-3-8b")
