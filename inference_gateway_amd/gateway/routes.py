@@ -1,0 +1,316 @@
+"""Gateway route handlers.
+
+Re-implements the reference's 12-handler router (api/routes.go:44-57) over
+the direct-call provider layer — same surface and semantics, no self-proxy
+hop (SURVEY.md §3.2 note: the hop doubled middleware cost by design; the
+rebuild short-circuits it in-process).
+
+Handlers: health, list_models, chat_completions (stream + non-stream),
+messages (Anthropic passthrough), responses (OpenAI passthrough), images
+(generations/edits/variations passthrough), proxy, mcp tools, metrics
+ingestion (otel/ingest.py).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+from typing import Optional
+
+from ..providers import routing
+from ..providers.constants import (ANTHROPIC_MESSAGES_ENDPOINT,
+                                   LOCAL_PROVIDER_ID,
+                                   OPENAI_IMAGES_EDITS_ENDPOINT,
+                                   OPENAI_IMAGES_ENDPOINT,
+                                   OPENAI_IMAGES_VARIATIONS_ENDPOINT,
+                                   OPENAI_RESPONSES_ENDPOINT)
+from ..providers.core import HTTPError
+from .http import Request, Response
+
+
+class Router:
+    def __init__(self, cfg, registry, logger, selector=None, mcp=None,
+                 telemetry=None):
+        self.cfg = cfg
+        self.registry = registry
+        self.logger = logger
+        self.selector = selector     # routing.Selector or None
+        self.mcp = mcp               # mcp layer (client + agent) or None
+        self.telemetry = telemetry
+
+    # ---------------- /health ----------------
+    async def health(self, req: Request) -> Response:
+        return Response.json({"status": "healthy"})
+
+    # ---------------- GET /v1/models ----------------
+    async def list_models(self, req: Request) -> Response:
+        provider_q = req.query.get("provider")
+        if provider_q:
+            try:
+                p = self.registry.build(provider_q)
+            except HTTPError as e:
+                return Response.error(e.status, e.message)
+            try:
+                result = await p.list_models()
+            except HTTPError as e:
+                return Response.error(502, f"upstream error: {e.message}")
+            except Exception as e:
+                return Response.error(502, f"upstream error: {e}")
+            models = result["data"]
+        else:
+            # fan-out over all configured providers with a shared timeout
+            # (reference routes.go:503-539); failures are skipped
+            async def fetch(pid):
+                try:
+                    p = self.registry.build(pid)
+                    return (await p.list_models())["data"]
+                except Exception:
+                    return []
+
+            results = await asyncio.gather(
+                *(fetch(pid) for pid in self.registry.configured_ids()),
+                return_exceptions=False)
+            models = [m for r in results for m in r]
+        models = routing.filter_models(
+            models, self.cfg.general.allowed_models,
+            self.cfg.general.disallowed_models)
+        return Response.json(
+            {"object": "list", "data": models})
+
+    # ---------------- POST /v1/chat/completions ----------------
+    def _resolve(self, req_model: str, provider_q: Optional[str]):
+        """alias pools -> prefix routing -> ?provider= (reference
+        routes.go:669-693). Returns (provider_id, model) or raises."""
+        if self.selector is not None:
+            dep = self.selector.select(req_model)
+            if dep is not None:
+                return dep.provider, dep.model
+        pid, model = routing.determine_provider_and_model(req_model)
+        if pid is None:
+            if not provider_q:
+                raise HTTPError(
+                    400,
+                    "unable to determine provider for model; use "
+                    "'provider/model' or the ?provider= query parameter")
+            pid = provider_q.lower()
+        return pid, model
+
+    async def chat_completions(self, req: Request) -> Response:
+        body = req.state.get("parsed_body")
+        if body is None:
+            try:
+                body = req.json()
+            except Exception:
+                return Response.error(400, "invalid JSON body")
+        if not isinstance(body, dict) or not body.get("model"):
+            return Response.error(400, "model is required")
+        try:
+            pid, model = self._resolve(body["model"],
+                                       req.query.get("provider"))
+        except HTTPError as e:
+            return Response.error(e.status, e.message)
+        if routing.model_denied(body["model"],
+                                self.cfg.general.allowed_models,
+                                self.cfg.general.disallowed_models):
+            return Response.error(403, f"model not allowed: {body['model']}")
+        try:
+            provider = self.registry.build(pid)
+        except HTTPError as e:
+            return Response.error(e.status, e.message)
+        body = dict(body)
+        body["model"] = model
+        # vision gate (reference routes.go:710-740): when vision is
+        # disabled, strip image parts rather than erroring
+        if not self.cfg.general.enable_vision:
+            self._strip_images(body)
+
+        stream = bool(body.get("stream"))
+        req.state["provider_id"] = pid
+        req.state["model"] = model
+
+        if self.mcp is not None and self.mcp.applies(req, body):
+            return await self.mcp.run_chat(provider, body, stream, req)
+
+        if stream:
+            try:
+                gen = self._relay_stream(provider, body)
+                return Response.sse(gen)
+            except HTTPError as e:
+                return Response.error(e.status, e.message)
+        try:
+            result = await provider.chat_completions(body)
+        except HTTPError as e:
+            status = 502 if e.status >= 500 else e.status
+            return Response.error(status, e.message)
+        except asyncio.TimeoutError:
+            return Response.error(504, "upstream timeout")
+        except Exception as e:
+            return Response.error(502, f"upstream error: {e}")
+        return Response.json(result)
+
+    @staticmethod
+    def _strip_images(body: dict) -> None:
+        for msg in body.get("messages") or []:
+            c = msg.get("content")
+            if isinstance(c, list):
+                text = [p for p in c if not (isinstance(p, dict)
+                                             and p.get("type") == "image_url")]
+                msg["content"] = text
+
+    async def _relay_stream(self, provider, body):
+        """Relay upstream SSE lines; guarantees exactly one [DONE]."""
+        done_sent = False
+        try:
+            async for line in provider.stream_chat_completions(body):
+                if not line:
+                    continue
+                if line.startswith("data:"):
+                    if line[5:].strip() == "[DONE]":
+                        done_sent = True
+                    yield line
+        except HTTPError as e:
+            yield ("data: " + json.dumps(
+                {"error": {"message": e.message, "code": e.status}}))
+        except Exception as e:
+            yield ("data: " + json.dumps({"error": {"message": str(e)}}))
+        finally:
+            if not done_sent:
+                yield "data: [DONE]"
+
+    # ---------------- byte passthrough handlers ----------------
+    async def messages(self, req: Request) -> Response:
+        """Anthropic-native /v1/messages (reference routes.go:842-1005):
+        byte-for-byte passthrough; only `model` is sniffed/rewritten."""
+        return await self._passthrough(
+            req, required_provider="anthropic",
+            endpoint=ANTHROPIC_MESSAGES_ENDPOINT,
+            error_shape="anthropic")
+
+    async def responses(self, req: Request) -> Response:
+        """OpenAI /v1/responses passthrough (routes.go:1018-1179)."""
+        return await self._passthrough(
+            req, required_provider="openai",
+            endpoint=OPENAI_RESPONSES_ENDPOINT, error_shape="openai")
+
+    async def _passthrough(self, req: Request, required_provider: str,
+                           endpoint: str, error_shape: str) -> Response:
+        def err(status, msg):
+            if error_shape == "anthropic":
+                return Response.json(
+                    {"type": "error",
+                     "error": {"type": "invalid_request_error",
+                               "message": msg}}, status)
+            return Response.json({"error": {"message": msg}}, status)
+
+        try:
+            body = req.json()
+        except Exception:
+            return err(400, "invalid JSON body")
+        if not isinstance(body, dict) or not body.get("model"):
+            return err(400, "model is required")
+        pid, model = routing.determine_provider_and_model(body["model"])
+        if pid is None:
+            pid = req.query.get("provider", required_provider).lower()
+        if pid != required_provider:
+            return err(400,
+                       f"this endpoint only supports the {required_provider} "
+                       f"provider")
+        try:
+            provider = self.registry.build(pid)
+        except HTTPError as e:
+            return err(e.status, e.message)
+        raw = req.body
+        if model != body["model"]:
+            body = dict(body)
+            body["model"] = model
+            raw = json.dumps(body).encode()
+        stream = bool(body.get("stream"))
+        headers = {"content-type": "application/json"}
+        if stream:
+            async def gen():
+                async with provider.stream_raw("POST", endpoint, raw,
+                                               headers) as r:
+                    if r.status_code >= 400:
+                        payload = (await r.aread()).decode()
+                        yield "data: " + payload
+                        return
+                    async for line in r.aiter_lines():
+                        if line:
+                            yield line
+            return Response.sse(gen())
+        try:
+            r = await provider.forward_raw("POST", endpoint, raw, headers)
+        except Exception as e:
+            return err(502, f"upstream error: {e}")
+        return Response(r.status_code, r.content,
+                        {"content-type": r.headers.get("content-type",
+                                                       "application/json")})
+
+    async def images(self, req: Request) -> Response:
+        if not self.cfg.general.enable_images:
+            return Response.error(404, "images API is disabled")
+        endpoint = {
+            "/v1/images/generations": OPENAI_IMAGES_ENDPOINT,
+            "/v1/images/edits": OPENAI_IMAGES_EDITS_ENDPOINT,
+            "/v1/images/variations": OPENAI_IMAGES_VARIATIONS_ENDPOINT,
+        }[req.path]
+        try:
+            provider = self.registry.build("openai")
+        except HTTPError as e:
+            return Response.error(e.status, e.message)
+        ct = req.headers.get("content-type", "application/json")
+        try:
+            r = await provider.forward_raw("POST", endpoint, req.body,
+                                           {"content-type": ct})
+        except Exception as e:
+            return Response.error(502, f"upstream error: {e}")
+        return Response(r.status_code, r.content,
+                        {"content-type": r.headers.get("content-type",
+                                                       "application/json")})
+
+    # ---------------- /proxy/{provider}/... ----------------
+    async def proxy(self, req: Request) -> Response:
+        """Generic reverse proxy (reference ProxyHandler routes.go:102-330):
+        caller's Authorization is stripped; provider credential injected."""
+        pid = req.path_params.get("provider", "")
+        rest = req.path_params.get("rest", "")
+        try:
+            provider = self.registry.build(pid)
+        except HTTPError as e:
+            return Response.error(e.status, e.message)
+        if pid == LOCAL_PROVIDER_ID:
+            return Response.error(400, "local provider has no proxy path")
+        endpoint = "/" + rest if rest else "/"
+        headers = {}
+        ct = req.headers.get("content-type")
+        if ct:
+            headers["content-type"] = ct
+        accept = req.headers.get("accept", "")
+        want_stream = "text/event-stream" in accept
+        try:
+            if want_stream:
+                async def gen():
+                    async with provider.stream_raw(req.method, endpoint,
+                                                   req.body, headers) as r:
+                        async for line in r.aiter_lines():
+                            if line:
+                                yield line
+                return Response.sse(gen())
+            r = await provider.forward_raw(req.method, endpoint, req.body,
+                                           headers)
+        except Exception as e:
+            return Response.error(502, f"upstream error: {e}")
+        return Response(r.status_code, r.content,
+                        {"content-type": r.headers.get("content-type",
+                                                       "application/json")})
+
+    # ---------------- GET /v1/mcp/tools ----------------
+    async def list_tools(self, req: Request) -> Response:
+        if self.mcp is None or not self.cfg.mcp.expose:
+            return Response.error(
+                403, "MCP tools endpoint is not exposed")
+        tools = self.mcp.client.chat_tools()
+        return Response.json({"object": "list",
+                              "data": [t["function"] | {
+                                  "server": self.mcp.client.tool_server(
+                                      t["function"]["name"]) or ""}
+                                  for t in tools]})

@@ -1,0 +1,153 @@
+"""The `local` provider: the in-process MI355X inference engine exposed
+through the same provider interface as the 15 remote providers.
+
+This is the architectural core of the rebuild: where the
+reference proxies every request to a remote HTTP API through a self-proxy
+hop, the local provider calls the engine directly — zero HTTP hops, tokens
+stream from the hipGraph decode loop straight into the SSE writer.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from typing import AsyncIterator, Optional
+
+from ..engine.async_engine import AsyncEngine
+from ..engine.config import EngineConfig, MODEL_PRESETS, SamplingParams
+from ..utils.tokenizer import apply_chat_template, get_tokenizer
+from .constants import LOCAL_PROVIDER_ID
+
+
+class LocalProvider:
+    def __init__(self, local_cfg, device: str = "cuda:0",
+                 engine: Optional[AsyncEngine] = None):
+        self.id = LOCAL_PROVIDER_ID
+        self.model_name = local_cfg.served_model_name or local_cfg.model
+        preset = MODEL_PRESETS[local_cfg.model]
+        ecfg = EngineConfig(
+            model=preset,
+            max_num_seqs=local_cfg.max_num_seqs,
+            max_model_len=local_cfg.max_model_len,
+            gpu_memory_utilization=local_cfg.gpu_memory_utilization,
+            enforce_eager=local_cfg.enforce_eager,
+            num_gpu_blocks=getattr(local_cfg, "num_gpu_blocks", None),
+        )
+        self.engine = engine or AsyncEngine(ecfg, device=device)
+        self.tokenizer = get_tokenizer(
+            preset.vocab_size, getattr(local_cfg, "tokenizer_path", None))
+
+    # ---- provider interface --------------------------------------------
+    async def list_models(self) -> dict:
+        return {
+            "provider": self.id,
+            "object": "list",
+            "data": [{
+                "id": f"{self.id}/{self.model_name}",
+                "object": "model",
+                "created": int(time.time()),
+                "owned_by": self.id,
+                "served_by": self.id,
+            }],
+        }
+
+    def _sampling_from_request(self, req: dict) -> SamplingParams:
+        max_tokens = req.get("max_completion_tokens") or req.get(
+            "max_tokens") or 256
+        stop_ids = ()
+        if self.tokenizer.eos_token_id is not None:
+            stop_ids = (self.tokenizer.eos_token_id,)
+        return SamplingParams(
+            max_new_tokens=int(max_tokens),
+            temperature=float(req.get("temperature", 1.0)),
+            top_p=float(req.get("top_p", 1.0)),
+            top_k=int(req.get("top_k", 0)),
+            stop_token_ids=stop_ids,
+        )
+
+    def _encode_request(self, req: dict):
+        prompt = apply_chat_template(req.get("messages") or [])
+        return self.tokenizer.encode(prompt)
+
+    async def chat_completions(self, req: dict,
+                               extra_headers=None) -> dict:
+        prompt_ids = self._encode_request(req)
+        sampling = self._sampling_from_request(req)
+        token_ids = []
+        finish_reason = "stop"
+        async for out in self.engine.generate(prompt_ids, sampling):
+            token_ids.append(out.token_id)
+            if out.finished:
+                finish_reason = ("stop" if out.finish_reason == "stop"
+                                 else "length")
+        text = self.tokenizer.decode(token_ids)
+        now = int(time.time())
+        return {
+            "id": f"chatcmpl-{uuid.uuid4().hex[:24]}",
+            "object": "chat.completion",
+            "created": now,
+            "model": f"{self.id}/{self.model_name}",
+            "choices": [{
+                "index": 0,
+                "message": {"role": "assistant", "content": text},
+                "finish_reason": finish_reason,
+            }],
+            "usage": {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": len(token_ids),
+                "total_tokens": len(prompt_ids) + len(token_ids),
+            },
+        }
+
+    async def stream_chat_completions(self, req: dict,
+                                      extra_headers=None
+                                      ) -> AsyncIterator[str]:
+        """Yields SSE lines ('data: {...}') matching OpenAI stream chunks;
+        per-token flush (reference bar: per-line SSE flush)."""
+        prompt_ids = self._encode_request(req)
+        sampling = self._sampling_from_request(req)
+        cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        created = int(time.time())
+        model = f"{self.id}/{self.model_name}"
+
+        def chunk(delta: dict, finish: Optional[str] = None,
+                  usage: Optional[dict] = None) -> str:
+            d = {
+                "id": cid,
+                "object": "chat.completion.chunk",
+                "created": created,
+                "model": model,
+                "choices": [{
+                    "index": 0,
+                    "delta": delta,
+                    "finish_reason": finish,
+                }],
+            }
+            if usage is not None:
+                d["usage"] = usage
+            return "data: " + json.dumps(d, separators=(",", ":"))
+
+        yield chunk({"role": "assistant", "content": ""})
+        ntok = 0
+        finish_reason = "stop"
+        pending: list = []
+        async for out in self.engine.generate(prompt_ids, sampling):
+            ntok += 1
+            pending.append(out.token_id)
+            text = self.tokenizer.decode(pending)
+            # hold back partial UTF-8 sequences (byte tokenizer)
+            if text.endswith("�"):
+                continue
+            pending = []
+            if out.finished:
+                finish_reason = ("stop" if out.finish_reason == "stop"
+                                 else "length")
+            yield chunk({"content": text})
+        usage = {
+            "prompt_tokens": len(prompt_ids),
+            "completion_tokens": ntok,
+            "total_tokens": len(prompt_ids) + ntok,
+        }
+        yield chunk({}, finish=finish_reason, usage=usage)
+        yield "data: [DONE]"

@@ -1,3 +1,5 @@
+import asyncio
+import inspect
 import sys
 from pathlib import Path
 
@@ -10,6 +12,22 @@ def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires an MI355X (or any ROCm GPU) device"
     )
+    config.addinivalue_line(
+        "markers", "asyncio: coroutine test (run via asyncio.run)"
+    )
+
+
+def pytest_pyfunc_call(pyfuncitem):
+    """Run `async def` tests with asyncio.run (no pytest-asyncio in the
+    offline wheelhouse)."""
+    fn = pyfuncitem.obj
+    if inspect.iscoroutinefunction(fn):
+        kwargs = {
+            name: pyfuncitem.funcargs[name]
+            for name in pyfuncitem._fixtureinfo.argnames
+        }
+        asyncio.run(fn(**kwargs))
+        return True
 
 
 def pytest_collection_modifyitems(config, items):
