@@ -1,0 +1,226 @@
+"""OIDC auth middleware + telemetry/metrics tests."""
+import base64
+import hashlib
+import hmac
+import json
+import time
+
+import httpx
+import pytest
+
+from inference_gateway_amd import config as config_mod
+from inference_gateway_amd.gateway.app import Gateway
+from inference_gateway_amd.gateway.middlewares.auth import (OIDCAuthenticator,
+                                                            verify_rs256)
+from inference_gateway_amd.logger import noop_logger
+from inference_gateway_amd.otel.metrics import Telemetry
+
+from test_gateway import gw_client, make_upstream
+
+ISSUER = "http://keycloak:8080/realms/test"
+
+
+def b64url(data: bytes) -> str:
+    return base64.urlsafe_b64encode(data).rstrip(b"=").decode()
+
+
+def make_hs256(claims: dict, secret: str) -> str:
+    header = b64url(json.dumps({"alg": "HS256", "typ": "JWT"}).encode())
+    payload = b64url(json.dumps(claims).encode())
+    signing = f"{header}.{payload}".encode()
+    sig = hmac.new(secret.encode(), signing, hashlib.sha256).digest()
+    return f"{header}.{payload}.{b64url(sig)}"
+
+
+# test-only RSA key generated deterministically (512-bit primes)
+def _is_prime(n: int) -> bool:
+    if n % 2 == 0:
+        return False
+    d, r = n - 1, 0
+    while d % 2 == 0:
+        d //= 2
+        r += 1
+    for a in (2, 3, 5, 7, 11, 13, 17, 19, 23, 29, 31, 37):
+        x = pow(a, d, n)
+        if x in (1, n - 1):
+            continue
+        for _ in range(r - 1):
+            x = x * x % n
+            if x == n - 1:
+                break
+        else:
+            return False
+    return True
+
+
+def _gen_prime(seed: int, bits: int = 512) -> int:
+    import random
+
+    rng = random.Random(seed)
+    while True:
+        c = rng.getrandbits(bits) | (1 << (bits - 1)) | 1
+        if _is_prime(c):
+            return c
+
+
+P = _gen_prime(1)
+Q = _gen_prime(2)
+N = P * Q
+E = 65537
+D = pow(E, -1, (P - 1) * (Q - 1))
+K = (N.bit_length() + 7) // 8
+_SHA256_PREFIX = bytes.fromhex("3031300d060960864801650304020105000420")
+
+
+def make_rs256(claims: dict) -> str:
+    header = b64url(json.dumps({"alg": "RS256", "kid": "k1"}).encode())
+    payload = b64url(json.dumps(claims).encode())
+    signing = f"{header}.{payload}".encode()
+    digest = hashlib.sha256(signing).digest()
+    em = (b"\x00\x01" + b"\xff" * (K - len(digest) - len(_SHA256_PREFIX) - 3)
+          + b"\x00" + _SHA256_PREFIX + digest)
+    sig = pow(int.from_bytes(em, "big"), D, N).to_bytes(K, "big")
+    return f"{header}.{payload}.{b64url(sig)}"
+
+
+def auth_handler(request: httpx.Request) -> httpx.Response:
+    if request.url.path.endswith("openid-configuration"):
+        return httpx.Response(200, json={"jwks_uri": ISSUER + "/jwks"})
+    if request.url.path.endswith("/jwks"):
+        n_bytes = N.to_bytes(K, "big")
+        return httpx.Response(200, json={"keys": [{
+            "kty": "RSA", "kid": "k1", "alg": "RS256",
+            "n": b64url(n_bytes),
+            "e": b64url(E.to_bytes(3, "big")),
+        }]})
+    return httpx.Response(404)
+
+
+def test_verify_rs256_roundtrip():
+    token = make_rs256({"sub": "u"})
+    h, p, s = token.split(".")
+    sig = base64.urlsafe_b64decode(s + "==")
+    assert verify_rs256(N, E, f"{h}.{p}".encode(), sig)
+    assert not verify_rs256(N, E, f"{h}.{p}x".encode(), sig)
+
+
+def make_auth_gateway(secret="shh"):
+    upstream, calls = make_upstream()
+
+    def handler(request):
+        if request.url.host == "keycloak:8080".split(":")[0] or \
+                "keycloak" in str(request.url.host):
+            return auth_handler(request)
+        return upstream(request)
+
+    env = {"GROQ_API_KEY": "k", "AUTH_ENABLED": "true",
+           "AUTH_OIDC_ISSUER": ISSUER,
+           "AUTH_OIDC_CLIENT_SECRET": secret}
+    cfg = config_mod.load(env)
+    client = httpx.AsyncClient(transport=httpx.MockTransport(handler))
+    return Gateway(cfg=cfg, logger=noop_logger(), http_client=client), calls
+
+
+@pytest.mark.asyncio
+async def test_auth_middleware_flow():
+    gw, _ = make_auth_gateway()
+    await gw.authenticator.initialize()
+    body = {"model": "groq/m", "messages": [{"role": "user", "content": "x"}]}
+    async with gw_client(gw) as c:
+        r = await c.post("/v1/chat/completions", json=body)
+        assert r.status_code == 401
+        r = await c.post("/v1/chat/completions", json=body,
+                         headers={"Authorization": "Bearer garbage"})
+        assert r.status_code == 401
+        good = make_hs256({"sub": "u1", "iss": ISSUER,
+                           "exp": time.time() + 60}, "shh")
+        r = await c.post("/v1/chat/completions", json=body,
+                         headers={"Authorization": f"Bearer {good}"})
+        assert r.status_code == 200
+        rs = make_rs256({"sub": "u2", "iss": ISSUER,
+                         "exp": time.time() + 60})
+        r = await c.post("/v1/chat/completions", json=body,
+                         headers={"Authorization": f"Bearer {rs}"})
+        assert r.status_code == 200
+        expired = make_hs256({"sub": "u1", "iss": ISSUER,
+                              "exp": time.time() - 10}, "shh")
+        r = await c.post("/v1/chat/completions", json=body,
+                         headers={"Authorization": f"Bearer {expired}"})
+        assert r.status_code == 401
+        # health stays open
+        r = await c.get("/health")
+        assert r.status_code == 200
+
+
+def make_telemetry_gateway():
+    handler, calls = make_upstream()
+    env = {"GROQ_API_KEY": "k", "TELEMETRY_ENABLED": "true",
+           "TELEMETRY_METRICS_PUSH_ENABLED": "true"}
+    cfg = config_mod.load(env)
+    client = httpx.AsyncClient(transport=httpx.MockTransport(handler))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client)
+    return gw, calls
+
+
+@pytest.mark.asyncio
+async def test_telemetry_records_usage_nonstream():
+    gw, _ = make_telemetry_gateway()
+    async with gw_client(gw) as c:
+        r = await c.post("/v1/chat/completions",
+                         json={"model": "groq/m",
+                               "messages": [{"role": "user",
+                                             "content": "x"}]})
+        assert r.status_code == 200
+    text = gw.telemetry.exposition().decode()
+    assert 'gen_ai_client_token_usage_count{model="m",provider="groq"' in text
+    assert "gen_ai_server_request_duration_seconds_count" in text
+
+
+@pytest.mark.asyncio
+async def test_telemetry_records_usage_stream_from_tail_chunks():
+    gw, _ = make_telemetry_gateway()
+    async with gw_client(gw) as c:
+        async with c.stream("POST", "/v1/chat/completions",
+                            json={"model": "groq/m", "stream": True,
+                                  "messages": [{"role": "user",
+                                                "content": "x"}]}) as r:
+            async for _ in r.aiter_lines():
+                pass
+    text = gw.telemetry.exposition().decode()
+    assert 'token_type="output"' in text
+    assert "gen_ai_server_time_to_first_token_seconds_count" in text
+
+
+@pytest.mark.asyncio
+async def test_otlp_json_ingestion():
+    gw, _ = make_telemetry_gateway()
+    payload = {
+        "resourceMetrics": [{"scopeMetrics": [{"metrics": [{
+            "name": "gen_ai.client.operation.duration",
+            "histogram": {
+                "aggregationTemporality": 2,
+                "dataPoints": [{
+                    "attributes": [
+                        {"key": "provider",
+                         "value": {"stringValue": "groq"}},
+                        {"key": "model", "value": {"stringValue": "m"}},
+                        {"key": "evil_attr", "value": {"stringValue": "x"}},
+                    ],
+                    "bucketCounts": ["1", "2", "0"],
+                    "explicitBounds": [0.1, 1.0],
+                }],
+            },
+        }]}]}],
+    }
+    async with gw_client(gw) as c:
+        r = await c.post("/v1/metrics", json=payload)
+        assert r.status_code == 200
+        assert r.json()["accepted"] == 1
+        # cumulative temporality is rejected
+        payload["resourceMetrics"][0]["scopeMetrics"][0]["metrics"][0][
+            "histogram"]["aggregationTemporality"] = 1
+        r = await c.post("/v1/metrics", json=payload)
+        assert r.json()["accepted"] == 0 and r.json()["rejected"] >= 1
+    text = gw.telemetry.exposition().decode()
+    assert "gen_ai_client_operation_duration_seconds_count" in text
+    assert "evil_attr" not in text
