@@ -142,5 +142,18 @@ class EngineConfig:
         return EngineConfig(**defaults)
 
     @staticmethod
+    def tiny_moe(**kw) -> "EngineConfig":
+        defaults = dict(
+            model=MODEL_PRESETS["tiny-moe"],
+            num_gpu_blocks=64,
+            max_num_seqs=8,
+            max_batched_tokens=2048,
+            max_model_len=512,
+            use_hip_graphs=False,
+        )
+        defaults.update(kw)
+        return EngineConfig(**defaults)
+
+    @staticmethod
     def for_model(name: str, **kw) -> "EngineConfig":
         return EngineConfig(model=MODEL_PRESETS[name], **kw)

@@ -65,6 +65,7 @@ class LLMEngine:
         self.max_blocks_per_seq = (
             cfg.max_model_len + cfg.block_size - 1) // cfg.block_size
         self.graph_runner = None
+        self._gr_version = -1
         if (self.device.type == "cuda" and cfg.use_hip_graphs
                 and not cfg.enforce_eager):
             from .graph_runner import DecodeGraphRunner
@@ -138,22 +139,31 @@ class LLMEngine:
         batch = self.scheduler.schedule()
         if batch is None:
             return []
-        if self.graph_runner is not None:
-            # keep the per-slot device block tables current
-            for seq in batch.seqs:
-                if seq.synced_blocks < len(seq.block_ids):
-                    self.graph_runner.write_block_row(
-                        seq.slot, seq.block_ids, start=seq.synced_blocks)
-                    seq.synced_blocks = len(seq.block_ids)
-        if not batch.is_prefill and self.graph_runner is not None:
-            hidden_last = self.graph_runner.execute(batch.seqs)
-        else:
-            fb = self._build_forward_batch(batch)
-            hidden = self.model(fb, self.kv_caches)
-            hidden_last = hidden[fb.last_token_indices]
+        gr = self.graph_runner
+        if not batch.is_prefill and gr is not None:
+            # graph-captured decode: all input prep is device-side; the host
+            # only pushes block growths and reads back the sampled tokens
+            gr.write_block_appends(batch.new_blocks)
+            ver = self.scheduler.membership_version
+            gr.execute(batch.seqs, membership_changed=ver != self._gr_version)
+            self._gr_version = ver
+            torch.cuda.synchronize()
+            tokens = gr.tokens_pinned[: len(batch.seqs)].tolist()
+            return self._append_tokens(batch, tokens)
+
+        fb = self._build_forward_batch(batch)
+        hidden = self.model(fb, self.kv_caches)
+        hidden_last = hidden[fb.last_token_indices]
         logits = self.model.compute_logits(hidden_last)
         tokens = self.sampler.sample(logits, batch.seqs)
-        return self._append_tokens(batch, tokens)
+        outputs = self._append_tokens(batch, tokens)
+        if batch.is_prefill and gr is not None:
+            # seed the persistent decode slot state for newly admitted seqs
+            for seq, tok in zip(batch.seqs, tokens):
+                if seq.state != SeqState.FINISHED:
+                    gr.write_block_row(seq.slot, seq.block_ids)
+                    gr.init_slot(seq, tok)
+        return outputs
 
     def _build_forward_batch(self, batch: ScheduledBatch) -> ForwardBatch:
         dev = self.device

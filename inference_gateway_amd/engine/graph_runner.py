@@ -1,20 +1,29 @@
 """hipGraph-captured decode execution (SURVEY.md §7 stage 3d).
 
 Decode steps on MI355X are launch-bound in eager mode (~300 kernel launches
-per 8B step; profiles/r01: ~3.8 ms/step of gaps). This runner:
+per 8B step; profiles/r01: ~3.8 ms/step of gaps). v2 moves ALL per-step
+input preparation onto the device and inside the captured graph:
 
-  * keeps per-SLOT persistent device block tables (a sequence keeps one slot
-    for its whole life; appends are one small index_put per step, not a
-    full rebuild);
-  * captures the decode forward as a hipGraph per batch-size bucket
-    (torch.cuda.CUDAGraph == hipGraph on ROCm) over static input buffers,
-    sharing one memory pool across buckets;
-  * per step: one pinned H2D copy of the [B] scalars, one index_select to
-    assemble the bucket's block-table rows, one graph replay.
+  * persistent per-SLOT device state — last sampled token, position,
+    block-table row, sampling params — written once at admission and then
+    advanced on-device each step (one scatter + one index_add after
+    sampling), never rebuilt from Python lists;
+  * the captured graph gathers {tokens, positions, slot_mapping,
+    block_table, ctx_lens} from slot state by a per-bucket active-slot
+    index buffer, runs the forward AND the lm_head, and leaves fp32
+    logits in a static buffer (torch.cuda.CUDAGraph == hipGraph on ROCm);
+  * per step the host does: replay, one fused-sampling kernel, two tiny
+    state-advance ops, one async D2H of the sampled tokens — no per-seq
+    Python work until the bookkeeping loop.
 
-Padding rows (B < bucket) run a 1-token attention against a live slot's
-first block (read-only) and write their dummy KV to the reserved scratch
-block 0 (BlockManager never hands out block 0).
+The active-slot index buffer is refreshed only when batch membership
+changes (admission / finish / preemption); in steady-state decode the
+host's per-step device interaction is O(1) kernel launches.
+
+Padding rows (B < bucket) point at a dedicated pad slot whose position is
+0 and whose block-table row is all zeros, so they run a 1-token attention
+against reserved scratch block 0 and write their dummy KV there
+(BlockManager never hands out block 0).
 """
 from __future__ import annotations
 
@@ -22,6 +31,7 @@ from typing import Dict, List, Optional
 
 import torch
 
+from .. import ops
 from .config import EngineConfig
 from .models.batch import ForwardBatch
 from .scheduler import Sequence
@@ -43,29 +53,35 @@ class DecodeGraphRunner:
         if self.buckets[-1] != cfg.max_num_seqs:
             self.buckets.append(cfg.max_num_seqs)
         bmax = self.buckets[-1]
+        S = self.max_slots + 1          # +1: dedicated pad slot (last row)
+        self.pad_slot = S - 1
 
-        # persistent per-slot block tables
-        self.bt_slots = torch.zeros((self.max_slots, self.maxb),
-                                    dtype=torch.int32, device=device)
-        # static graph inputs
-        self.in_tokens = torch.zeros(bmax, dtype=torch.int64, device=device)
-        self.in_positions = torch.zeros(bmax, dtype=torch.int32, device=device)
-        self.in_slot_mapping = torch.zeros(bmax, dtype=torch.int32,
-                                           device=device)
-        self.in_ctx = torch.ones(bmax, dtype=torch.int32, device=device)
-        self.in_block_table = torch.zeros((bmax, self.maxb),
-                                          dtype=torch.int32, device=device)
-        # pinned staging: [4, bmax] int64 rows: tokens, positions, slots, ctx
-        self.staging = torch.zeros((4, bmax), dtype=torch.int64,
-                                   pin_memory=True)
-        self.slot_idx_staging = torch.zeros(bmax, dtype=torch.int64,
-                                            pin_memory=True)
+        # ---- persistent per-slot device state ----
+        self.slot_token = torch.zeros(S, dtype=torch.int64, device=device)
+        self.slot_pos = torch.zeros(S, dtype=torch.int32, device=device)
+        self.bt_slots = torch.zeros((S, self.maxb), dtype=torch.int32,
+                                    device=device)
+        self.slot_temp = torch.ones(S, dtype=torch.float32, device=device)
+        self.slot_topk = torch.zeros(S, dtype=torch.int32, device=device)
+        self.slot_topp = torch.ones(S, dtype=torch.float32, device=device)
 
+        # ---- per-bucket static graph inputs ----
+        self.in_slot_idx = torch.full((bmax,), self.pad_slot,
+                                      dtype=torch.int64, device=device)
+        self._ones_i32 = torch.ones(bmax, dtype=torch.int32, device=device)
+
+        # pinned host staging for state writes / token readback
+        self.tokens_pinned = torch.zeros(bmax, dtype=torch.int32,
+                                         pin_memory=True)
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
-        self._hidden: Dict[int, torch.Tensor] = {}
+        self._logits: Dict[int, torch.Tensor] = {}
         self._pool = None
+        self._cur_bucket: Optional[int] = None
+        self._cur_B = 0
+        self._gen = torch.Generator(device=device)
+        self._gen.manual_seed(cfg.seed)
 
-    # ---- block-table maintenance ---------------------------------------
+    # ---- slot-state maintenance (admission / growth) --------------------
     def write_block_row(self, slot: int, block_ids: List[int],
                         start: int = 0) -> None:
         if start >= len(block_ids):
@@ -74,31 +90,81 @@ class DecodeGraphRunner:
         self.bt_slots[slot, start:len(block_ids)] = vals.to(
             self.device, non_blocking=True)
 
+    def write_block_appends(self, items: List) -> None:
+        """Batched single-block appends: items = [(slot, col, block_id)]."""
+        if not items:
+            return
+        t = torch.tensor(items, dtype=torch.int64).to(self.device,
+                                                      non_blocking=True)
+        self.bt_slots[t[:, 0], t[:, 1]] = t[:, 2].to(torch.int32)
+
+    def init_slot(self, seq: Sequence, first_token: int) -> None:
+        """After prefill: seed the slot's decode state (token/pos/params)."""
+        s = seq.slot
+        sp = seq.sampling
+        vals = torch.tensor(
+            [float(first_token), float(seq.num_tokens - 1),
+             sp.temperature, float(sp.top_k), sp.top_p],
+            dtype=torch.float64).to(self.device, non_blocking=True)
+        self.slot_token[s] = vals[0].to(torch.int64)
+        self.slot_pos[s] = vals[1].to(torch.int32)
+        self.slot_temp[s] = vals[2].to(torch.float32)
+        self.slot_topk[s] = vals[3].to(torch.int32)
+        self.slot_topp[s] = vals[4].to(torch.float32)
+
+    def set_batch(self, seqs: List[Sequence]) -> int:
+        """Refresh the active-slot index buffer; returns the bucket size."""
+        B = len(seqs)
+        bucket = self._bucket_for(B)
+        slots = [s.slot for s in seqs] + [self.pad_slot] * (bucket - B)
+        idx = torch.tensor(slots, dtype=torch.int64).to(self.device,
+                                                        non_blocking=True)
+        self.in_slot_idx[:bucket].copy_(idx)
+        self._cur_bucket = bucket
+        self._cur_B = B
+        return bucket
+
     # ---- capture --------------------------------------------------------
-    def _capture(self, B: int) -> None:
+    def _build_fb(self, bucket: int):
+        idx = self.in_slot_idx[:bucket]
+        bs = self.cfg.block_size
+        tokens = self.slot_token.index_select(0, idx)
+        pos = self.slot_pos.index_select(0, idx)
+        bt = self.bt_slots.index_select(0, idx)
+        posl = pos.to(torch.int64)
+        blk = bt.gather(1, (posl // bs).unsqueeze(1)).squeeze(1)
+        slot_map = (blk * bs + (pos % bs)).to(torch.int32)
         fb = ForwardBatch(
-            token_ids=self.in_tokens[:B],
-            positions=self.in_positions[:B],
-            slot_mapping=self.in_slot_mapping[:B],
+            token_ids=tokens,
+            positions=pos,
+            slot_mapping=slot_map,
             is_prefill=False,
-            block_table=self.in_block_table[:B],
-            ctx_lens=self.in_ctx[:B],
+            block_table=bt,
+            ctx_lens=pos + 1,
         )
-        # warmup on a side stream (records cuBLAS/hipBLASLt workspaces etc.)
+        return fb
+
+    def _forward(self, bucket: int) -> torch.Tensor:
+        fb = self._build_fb(bucket)
+        hidden = self.model(fb, self.kv_caches)
+        return self.model.compute_logits(hidden)
+
+    def _capture(self, bucket: int) -> None:
+        # warmup on a side stream (records hipBLASLt workspaces etc.)
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
             for _ in range(2):
-                self.model(fb, self.kv_caches)
+                self._forward(bucket)
         torch.cuda.current_stream().wait_stream(side)
 
         if self._pool is None:
             self._pool = torch.cuda.graph_pool_handle()
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g, pool=self._pool):
-            hidden = self.model(fb, self.kv_caches)
-        self._graphs[B] = g
-        self._hidden[B] = hidden
+            logits = self._forward(bucket)
+        self._graphs[bucket] = g
+        self._logits[bucket] = logits
 
     def _bucket_for(self, B: int) -> int:
         for b in self.buckets:
@@ -107,39 +173,30 @@ class DecodeGraphRunner:
         return self.buckets[-1]
 
     # ---- execute --------------------------------------------------------
-    def execute(self, seqs: List[Sequence]) -> torch.Tensor:
-        """One decode step for seqs; returns hidden states [B, hidden]."""
+    def execute(self, seqs: List[Sequence],
+                membership_changed: bool) -> torch.Tensor:
+        """One decode step; returns sampled tokens [B] int32 ON DEVICE and
+        queues an async D2H into self.tokens_pinned.  Slot state (token,
+        pos) is advanced on-device for the next step."""
         B = len(seqs)
-        bucket = self._bucket_for(B)
-        bs = self.cfg.block_size
-
-        st = self.staging
-        pad_slot = seqs[0].slot
-        npad = bucket - B
-        tokens = [(s.output_token_ids[-1] if s.output_token_ids
-                   else s.prompt_token_ids[-1]) for s in seqs] + [0] * npad
-        positions = [s.num_tokens - 1 for s in seqs] + [0] * npad
-        # padding rows write their dummy KV into reserved scratch block 0
-        slot_map = [s.block_ids[(s.num_tokens - 1) // bs] * bs
-                    + (s.num_tokens - 1) % bs for s in seqs] + [0] * npad
-        ctx = [s.num_tokens for s in seqs] + [1] * npad
-        slots = [s.slot for s in seqs] + [pad_slot] * npad
-        st[:, :bucket] = torch.tensor(
-            [tokens, positions, slot_map, ctx], dtype=torch.int64)
-        self.slot_idx_staging[:bucket] = torch.tensor(slots,
-                                                      dtype=torch.int64)
-
-        dev_st = st[:, :bucket].to(self.device, non_blocking=True)
-        self.in_tokens[:bucket].copy_(dev_st[0])
-        self.in_positions[:bucket].copy_(dev_st[1].to(torch.int32))
-        self.in_slot_mapping[:bucket].copy_(dev_st[2].to(torch.int32))
-        self.in_ctx[:bucket].copy_(dev_st[3].to(torch.int32))
-        slot_idx = self.slot_idx_staging[:bucket].to(self.device,
-                                                     non_blocking=True)
-        torch.index_select(self.bt_slots, 0, slot_idx,
-                           out=self.in_block_table[:bucket])
-
+        if membership_changed or self._cur_B != B:
+            bucket = self.set_batch(seqs)
+        else:
+            bucket = self._cur_bucket
         if bucket not in self._graphs:
             self._capture(bucket)
         self._graphs[bucket].replay()
-        return self._hidden[bucket][:B]
+
+        logits = self._logits[bucket]
+        idx = self.in_slot_idx[:bucket]
+        temps = self.slot_temp.index_select(0, idx)
+        topk = self.slot_topk.index_select(0, idx)
+        topp = self.slot_topp.index_select(0, idx)
+        u = torch.rand(bucket, generator=self._gen,
+                       device=self.device).clamp_(1e-6, 1 - 1e-6)
+        toks = ops.sample(logits, temps, topk, topp, u)     # [bucket] int32
+        act = idx[:B]
+        self.slot_token.scatter_(0, act, toks[:B].to(torch.int64))
+        self.slot_pos.index_add_(0, act, self._ones_i32[:B])
+        self.tokens_pinned[:B].copy_(toks[:B], non_blocking=True)
+        return toks

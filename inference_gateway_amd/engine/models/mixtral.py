@@ -56,9 +56,55 @@ class MixtralMoE(nn.Module):
         self.down = nn.Parameter(torch.stack(downs).contiguous(),
                                  requires_grad=False)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor,
+                is_prefill: bool = False) -> torch.Tensor:
         from ...parallel.comm import tp_all_reduce
 
+        if is_prefill and x.shape[0] > 1024:
+            out = self._forward_sorted(x)
+        else:
+            out = self._forward_padded(x)
+        return tp_all_reduce(out)
+
+    def _forward_padded(self, x: torch.Tensor) -> torch.Tensor:
+        """Sync-free dispatch for decode: tokens are scattered into a
+        per-expert padded buffer [E, C=T, h] purely with device ops (no
+        .tolist()/argsort round trips), so the whole MoE layer is
+        hipGraph-capturable. The padding rows cost batched-GEMM FLOPs but
+        no extra HBM traffic — decode MoE is bound by streaming all 8
+        experts' weights (~1.4 GB/layer), so the padded rows are free."""
+        T, h = x.shape
+        E, K, C = self.num_experts, self.top_k, x.shape[0]
+        router_logits = self.router(x).float()                # [T, E]
+        weights, topk_idx = torch.topk(router_logits, K, dim=-1)
+        weights = torch.softmax(weights, dim=-1).to(x.dtype)  # [T, K]
+        flat_expert = topk_idx.reshape(-1)                    # [T*K] int64
+        flat_token = (
+            torch.arange(T, device=x.device)
+            .unsqueeze(1)
+            .expand(T, K)
+            .reshape(-1)
+        )
+        # rank of each (token, k) within its expert via a one-hot cumsum —
+        # no sort, no host sync, static shapes
+        onehot = F.one_hot(flat_expert, E)                    # [T*K, E]
+        rank = (torch.cumsum(onehot, 0) - onehot).mul(onehot).sum(-1)
+        dst = flat_expert * C + rank                          # unique rows
+        buf = x.new_zeros((E * C, h))
+        buf.index_copy_(0, dst, x[flat_token])
+        gu = torch.bmm(buf.view(E, C, h),
+                       self.gate_up.transpose(1, 2))          # [E, C, 2s]
+        act = ops.silu_mul(gu.view(E * C, -1))
+        y = torch.bmm(act.view(E, C, -1),
+                      self.down.transpose(1, 2)).view(E * C, h)
+        ys = y.index_select(0, dst)                           # [T*K, h]
+        out = torch.zeros_like(x)
+        out.index_add_(0, flat_token, ys * weights.reshape(-1, 1))
+        return out
+
+    def _forward_sorted(self, x: torch.Tensor) -> torch.Tensor:
+        """Per-expert contiguous GEMMs for prefill-size batches, where the
+        padded path's E/K-fold FLOP overhead would be compute-bound."""
         T, h = x.shape
         router_logits = self.router(x).float()            # [T, E]
         weights, topk_idx = torch.topk(router_logits, self.top_k, dim=-1)
@@ -88,7 +134,7 @@ class MixtralMoE(nn.Module):
             w = weights.reshape(-1)[order[start:start + cnt]].unsqueeze(1)
             out.index_add_(0, toks, ye * w)
             start += cnt
-        return tp_all_reduce(out)
+        return out
 
 
 class MixtralLayer(nn.Module):
@@ -114,7 +160,7 @@ class MixtralLayer(nn.Module):
         x = self.attn(x, fb, kv_cache, cos_sin)
         x, residual = ops.fused_add_rmsnorm(x, residual, self.post_norm_w,
                                             self.eps)
-        x = self.moe(x)
+        x = self.moe(x, is_prefill=fb.is_prefill)
         return x, residual
 
 

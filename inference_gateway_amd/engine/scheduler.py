@@ -50,6 +50,9 @@ class Sequence:
 class ScheduledBatch:
     is_prefill: bool
     seqs: List[Sequence]
+    # decode-only: single-block growths this step, as (slot, col, block_id)
+    # rows for DecodeGraphRunner.write_block_appends
+    new_blocks: List[tuple] = field(default_factory=list)
 
 
 class Scheduler:
@@ -62,6 +65,9 @@ class Scheduler:
         self.waiting: deque[Sequence] = deque()
         self.running: List[Sequence] = []
         self._free_slots: List[int] = list(range(max_num_seqs - 1, -1, -1))
+        # bumped whenever the running set changes (admit/finish/preempt) so
+        # the graph runner knows when to refresh its active-slot buffer
+        self.membership_version = 0
 
     def add(self, seq: Sequence) -> None:
         if seq.prompt_len > self.max_model_len:
@@ -97,6 +103,7 @@ class Scheduler:
             tokens += seq.prompt_len
         if prefills:
             self.running.extend(prefills)
+            self.membership_version += 1
             return ScheduledBatch(is_prefill=True, seqs=prefills)
 
         if not self.running:
@@ -105,29 +112,41 @@ class Scheduler:
         # Decode: every running sequence generates one token; make room first.
         # LIFO preemption — newest sequence goes back to waiting, whole-prompt
         # recompute (matches the 288GB-HBM design point where this is rare).
+        bs = self.bm.block_size
+        new_blocks: List[tuple] = []
         while True:
-            try:
-                for seq in self.running:
-                    # this step writes KV at position num_tokens-1
-                    self.bm.ensure_capacity(seq.block_ids, seq.num_tokens)
+            ok = True
+            for seq in self.running:
+                # this step writes KV at position num_tokens-1
+                if seq.num_tokens > len(seq.block_ids) * bs:
+                    try:
+                        self.bm.ensure_capacity(seq.block_ids, seq.num_tokens)
+                    except RuntimeError:
+                        ok = False
+                        break
+                    new_blocks.append(
+                        (seq.slot, len(seq.block_ids) - 1, seq.block_ids[-1]))
+            if ok:
                 break
-            except RuntimeError:
-                victim = self.running.pop()  # newest arrival
-                self.bm.free(victim.block_ids)
-                if victim.slot is not None:
-                    self._free_slots.append(victim.slot)
-                    victim.slot = None
-                # recompute from scratch later, keep generated tokens as part
-                # of the new prompt so the request resumes transparently
-                victim.prompt_token_ids = (
-                    victim.prompt_token_ids + victim.output_token_ids
-                )
-                victim.output_token_ids = []
-                victim.state = SeqState.WAITING
-                self.waiting.appendleft(victim)
-                if not self.running:
-                    return None
-        return ScheduledBatch(is_prefill=False, seqs=list(self.running))
+            victim = self.running.pop()  # newest arrival
+            self.membership_version += 1
+            new_blocks = [nb for nb in new_blocks if nb[0] != victim.slot]
+            self.bm.free(victim.block_ids)
+            if victim.slot is not None:
+                self._free_slots.append(victim.slot)
+                victim.slot = None
+            # recompute from scratch later, keep generated tokens as part
+            # of the new prompt so the request resumes transparently
+            victim.prompt_token_ids = (
+                victim.prompt_token_ids + victim.output_token_ids
+            )
+            victim.output_token_ids = []
+            victim.state = SeqState.WAITING
+            self.waiting.appendleft(victim)
+            if not self.running:
+                return None
+        return ScheduledBatch(is_prefill=False, seqs=list(self.running),
+                              new_blocks=new_blocks)
 
     def finish(self, seq: Sequence, reason: str) -> None:
         seq.state = SeqState.FINISHED
@@ -138,6 +157,7 @@ class Scheduler:
             seq.slot = None
         if seq in self.running:
             self.running.remove(seq)
+            self.membership_version += 1
 
     def abort(self, seq: Sequence) -> None:
         if seq.state == SeqState.WAITING:

@@ -149,3 +149,34 @@ def test_moe_engine_generates():
         eng.step()
     out = eng.get_output(rid)
     assert out.finished and len(out.token_ids) == 5
+
+
+def test_moe_padded_matches_sorted():
+    """The sync-free padded-dispatch MoE path (decode, hipGraph-capturable)
+    must agree with the per-expert sorted-GEMM path (prefill)."""
+    from inference_gateway_amd.engine.models.mixtral import MixtralMoE
+
+    torch.manual_seed(3)
+    cfg = MODEL_PRESETS["tiny-moe"]
+    moe = MixtralMoE(cfg, layer=0, device="cpu", dtype=torch.float32)
+    x = torch.randn(13, cfg.hidden_size)
+    a = moe._forward_padded(x)
+    b = moe._forward_sorted(x)
+    assert torch.allclose(a, b, atol=1e-5, rtol=1e-4)
+
+
+def test_moe_padded_every_expert_rank_unique():
+    """Dispatch ranks must give every (token, k) pair a unique padded row."""
+    from inference_gateway_amd.engine.models.mixtral import MixtralMoE
+
+    torch.manual_seed(4)
+    cfg = MODEL_PRESETS["tiny-moe"]
+    moe = MixtralMoE(cfg, layer=0, device="cpu", dtype=torch.float32)
+    T, E, K = 17, cfg.num_experts, cfg.num_experts_per_tok
+    logits = torch.randn(T, E)
+    _, topk = torch.topk(logits, K, dim=-1)
+    flat = topk.reshape(-1)
+    onehot = torch.nn.functional.one_hot(flat, E)
+    rank = (torch.cumsum(onehot, 0) - onehot).mul(onehot).sum(-1)
+    dst = flat * T + rank
+    assert dst.unique().numel() == T * K
