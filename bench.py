@@ -45,6 +45,8 @@ def main() -> None:
                     help="concurrent sequences per GPU")
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--model", type=str, default="llama-3-8b")
+    ap.add_argument("--eager", action="store_true",
+                    help="disable hipGraph capture (for TunableOp tuning)")
     args = ap.parse_args()
 
     from inference_gateway_amd.engine.config import (EngineConfig,
@@ -66,7 +68,8 @@ def main() -> None:
         max_num_seqs=args.batch,
         max_batched_tokens=max(8192, args.prompt_len * 8),
         max_model_len=args.prompt_len + args.steps + args.warmup + 64,
-        use_hip_graphs=True,
+        use_hip_graphs=not args.eager,
+        enforce_eager=args.eager,
     )
     eng = LLMEngine(cfg, device=device)
 

@@ -39,11 +39,39 @@ class RequestOutput:
     finish_reason: Optional[str] = None
 
 
+def _maybe_load_tuned_gemms() -> None:
+    """Load the committed TunableOp solution table for the decode/prefill
+    GEMM shapes (gfx950), if present.  Tensile's default heuristics pick
+    poor tilings for the skinny-M decode GEMMs (profiles/r01: 1.7 TB/s
+    aggregate on 13.9 GB of weights/step); the tuned table is produced by
+    a one-off PYTORCH_TUNABLEOP_TUNING=1 bench run.  IGW_TUNABLEOP=0
+    opts out."""
+    import os
+    from pathlib import Path
+
+    if os.environ.get("IGW_TUNABLEOP", "1") == "0":
+        return
+    if not torch.cuda.is_available():
+        return
+    if os.environ.get("PYTORCH_TUNABLEOP_TUNING", "0") == "1":
+        return  # a tuning run manages TunableOp itself via env
+    csv = Path(__file__).resolve().parent.parent / "data" / \
+        "tunableop_gfx950.csv"
+    if not csv.exists():
+        return
+    import torch.cuda.tunable as tunable
+
+    tunable.enable(True)
+    tunable.tuning_enable(False)
+    tunable.read_file(str(csv))
+
+
 class LLMEngine:
     def __init__(self, cfg: EngineConfig, device: str = "cuda:0"):
         self.cfg = cfg
         self.device = torch.device(
             device if torch.cuda.is_available() else "cpu")
+        _maybe_load_tuned_gemms()
         self.dtype = torch.bfloat16
         if cfg.model.is_moe:
             from .models.mixtral import MixtralForCausalLM

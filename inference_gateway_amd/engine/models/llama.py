@@ -65,11 +65,10 @@ class LlamaAttention(nn.Module):
                 cos_sin: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]
         qkv = F.linear(x, self.qkv)
+        # strided views into the fused QKV output — the HIP kernels take a
+        # row stride, so there are no .contiguous() copies on the hot path
         q, k, v = qkv.split(
             [self.nq * self.hd, self.nkv * self.hd, self.nkv * self.hd], dim=-1)
-        q = q.contiguous()
-        k = k.contiguous()
-        v = v.contiguous()
         ops.rotary_embedding(fb.positions, q, k, self.hd, cos_sin)
         kc, vc = kv_cache
         k3 = k.view(T, self.nkv, self.hd)
@@ -171,7 +170,8 @@ class LlamaForCausalLM(nn.Module):
         return x
 
     def compute_logits(self, hidden_last: torch.Tensor) -> torch.Tensor:
-        """hidden_last: [B, hidden] -> fp32 logits [B, vocab]."""
+        """hidden_last: [B, hidden] -> bf16 logits [B, vocab] (the
+        sampler's bf16 radix kernel consumes them directly)."""
         if self.lm_head is None:
-            return F.linear(hidden_last, self.embed).float()
-        return self.lm_head(hidden_last).float()
+            return F.linear(hidden_last, self.embed)
+        return self.lm_head(hidden_last)

@@ -194,4 +194,4 @@ class MixtralForCausalLM(nn.Module):
         return x
 
     def compute_logits(self, hidden_last: torch.Tensor) -> torch.Tensor:
-        return self.lm_head(hidden_last).float()
+        return self.lm_head(hidden_last)

@@ -43,7 +43,9 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
     const int32_t* __restrict__ cu_seqlens,  // [B+1]
     const int32_t* __restrict__ tile_seq,    // [ntiles]
     const int32_t* __restrict__ tile_off,    // [ntiles] (q row offset in seq)
-    int nq, int nkv, float scale) {
+    int nq, int nkv, float scale, int q_rs, int k_rs, int v_rs) {
+  // *_rs: token-row strides in elements — q/k/v may be views into the
+  // fused QKV GEMM output (no .contiguous() copies)
   constexpr int BM = 64;
   constexpr int BN = 64;
   constexpr int NWAVES = 4;
@@ -76,7 +78,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
       const int d8 = (idx % (HD / 8)) * 8;
       const int grow = min(qoff + row, L - 1);
       ushort8 val = *reinterpret_cast<const ushort8*>(
-          q + ((int64_t)(q0 + grow) * nq + qh) * HD + d8);
+          q + (int64_t)(q0 + grow) * q_rs + qh * HD + d8);
       *reinterpret_cast<ushort8*>(
           reinterpret_cast<char*>(q_s) + row * (HD * 2) +
           swz256(row, d8 * 2)) = val;
@@ -126,7 +128,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
       const int d8 = (idx % (HD / 8)) * 8;
       const int grow = min(kbase + row, L - 1);
       ushort8 val = *reinterpret_cast<const ushort8*>(
-          k + ((int64_t)(q0 + grow) * nkv + kvh) * HD + d8);
+          k + (int64_t)(q0 + grow) * k_rs + kvh * HD + d8);
       *reinterpret_cast<ushort8*>(
           reinterpret_cast<char*>(k_s) + row * (HD * 2) +
           swz256(row, d8 * 2)) = val;
@@ -138,7 +140,7 @@ __global__ __launch_bounds__(256) void flash_prefill_kernel(
       const int d8 = (idx % (HD / 8)) * 8;
       const int grow = min(kbase + key, L - 1);
       ushort8 val = *reinterpret_cast<const ushort8*>(
-          v + ((int64_t)(q0 + grow) * nkv + kvh) * HD + d8);
+          v + (int64_t)(q0 + grow) * v_rs + kvh * HD + d8);
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         *reinterpret_cast<uint16_t*>(
@@ -247,7 +249,10 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                    torch::Tensor tile_seq, torch::Tensor tile_off,
                    double scale) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  for (auto* t : {&q, &k, &v}) {
+    TORCH_CHECK(t->stride(-1) == 1 && t->stride(1) == t->size(2));
+  }
+  TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(cu_seqlens.dtype() == torch::kInt32);
   const int HD = q.size(2);
   TORCH_CHECK(HD == 128, "only head_dim=128 supported");
@@ -264,7 +269,8 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                      (const int32_t*)cu_seqlens.data_ptr(),
                      (const int32_t*)tile_seq.data_ptr(),
                      (const int32_t*)tile_off.data_ptr(), nq, nkv,
-                     (float)scale);
+                     (float)scale, (int)q.stride(0), (int)k.stride(0),
+                     (int)v.stride(0));
   HIP_CHECK_KERNEL();
 }
 

@@ -56,7 +56,9 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
     const uint16_t* __restrict__ vcache, // [NB, NKV, BS, HD]
     const int32_t* __restrict__ block_table, // [B, max_blocks]
     const int32_t* __restrict__ ctx_lens,    // [B]
-    int max_blocks, int nkv, float scale) {
+    int max_blocks, int nkv, float scale, int q_rs) {
+  // q_rs: q row stride in elements (q may be a view into the fused QKV
+  // GEMM output — no .contiguous() copy on the decode hot path)
   constexpr int BS = 16;  // tokens per cache block
   const int kvh = blockIdx.x;
   const int seq = blockIdx.y;
@@ -79,7 +81,7 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
   for (int i = tid * 8; i < G * HD; i += NWAVES * WAVE_SIZE * 8) {
     *reinterpret_cast<ushort8*>(&q_lds[0][0] + i) =
         *reinterpret_cast<const ushort8*>(
-            q + ((int64_t)seq * nq + kvh * G) * HD + i);
+            q + (int64_t)seq * q_rs + (kvh * G) * HD + i);
   }
   __syncthreads();
 
@@ -279,7 +281,8 @@ static void launch_decode(torch::Tensor& out, const torch::Tensor& q,
                      (const uint16_t*)vcache.data_ptr(),
                      (const int32_t*)block_table.data_ptr(),
                      (const int32_t*)ctx_lens.data_ptr(),
-                     (int)block_table.size(1), nkv, scale);
+                     (int)block_table.size(1), nkv, scale,
+                     (int)q.stride(0));
   HIP_CHECK_KERNEL();
 }
 
@@ -288,7 +291,8 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        torch::Tensor block_table, torch::Tensor ctx_lens,
                        double scale) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
-  TORCH_CHECK(q.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(q.stride(-1) == 1 && q.stride(1) == q.size(2));
+  TORCH_CHECK(out.is_contiguous());
   TORCH_CHECK(kcache.size(3) == 128, "only head_dim=128 is supported");
   TORCH_CHECK(kcache.size(2) == 16, "only block_size=16 is supported");
   TORCH_CHECK(block_table.dtype() == torch::kInt32 &&

@@ -20,7 +20,9 @@ namespace igw {
 __global__ void rope_kernel(uint16_t* __restrict__ q, uint16_t* __restrict__ k,
                             const int32_t* __restrict__ positions,
                             const float* __restrict__ cos_sin, int nq, int nkv,
-                            int hd) {
+                            int hd, int qs, int ks) {
+  // qs/ks: row strides in elements — q/k may be views into a fused-QKV
+  // GEMM output (no .contiguous() copies on the decode hot path)
   const int t = blockIdx.x;
   const int pos = positions[t];
   const float* cs = cos_sin + (int64_t)pos * hd;
@@ -30,8 +32,8 @@ __global__ void rope_kernel(uint16_t* __restrict__ q, uint16_t* __restrict__ k,
   for (int i = threadIdx.x; i < nheads * half; i += blockDim.x) {
     const int h = i / half;
     const int d = i % half;
-    uint16_t* base = (h < nq) ? q + ((int64_t)t * nq + h) * hd
-                              : k + ((int64_t)t * nkv + (h - nq)) * hd;
+    uint16_t* base = (h < nq) ? q + (int64_t)t * qs + h * hd
+                              : k + (int64_t)t * ks + (h - nq) * hd;
     const float c = cs[d];
     const float s = cs[half + d];
     const float x1 = bf16_to_f32(base[d]);
@@ -49,7 +51,8 @@ __global__ void append_kv_kernel(const uint16_t* __restrict__ k,
                                  uint16_t* __restrict__ kcache,
                                  uint16_t* __restrict__ vcache,
                                  const int32_t* __restrict__ slot_mapping,
-                                 int nkv, int hd, int block_size) {
+                                 int nkv, int hd, int block_size, int ks,
+                                 int vs) {
   const int t = blockIdx.x;
   const int slot = slot_mapping[t];
   if (slot < 0) return;
@@ -59,12 +62,11 @@ __global__ void append_kv_kernel(const uint16_t* __restrict__ k,
   for (int i = threadIdx.x * 8; i < nkv * hd; i += blockDim.x * 8) {
     const int h = i / hd;
     const int d = i % hd;
-    const int64_t src = ((int64_t)t * nkv + h) * hd + d;
     const int64_t dst = (((int64_t)blk * nkv + h) * block_size + off) * hd + d;
     *reinterpret_cast<ushort8*>(kcache + dst) =
-        *reinterpret_cast<const ushort8*>(k + src);
+        *reinterpret_cast<const ushort8*>(k + (int64_t)t * ks + h * hd + d);
     *reinterpret_cast<ushort8*>(vcache + dst) =
-        *reinterpret_cast<const ushort8*>(v + src);
+        *reinterpret_cast<const ushort8*>(v + (int64_t)t * vs + h * hd + d);
   }
 }
 
@@ -73,6 +75,7 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   TORCH_CHECK(positions.dtype() == torch::kInt32);
   TORCH_CHECK(cos_sin_cache.dtype() == torch::kFloat32);
+  TORCH_CHECK(q.stride(-1) == 1 && k.stride(-1) == 1);
   const int T = q.size(0);
   const int hd = head_dim;
   const int nq = q.size(-1) / hd;
@@ -82,7 +85,8 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
   hipLaunchKernelGGL(rope_kernel, dim3(T), dim3(threads), 0, stream,
                      (uint16_t*)q.data_ptr(), (uint16_t*)k.data_ptr(),
                      (const int32_t*)positions.data_ptr(),
-                     (const float*)cos_sin_cache.data_ptr(), nq, nkv, hd);
+                     (const float*)cos_sin_cache.data_ptr(), nq, nkv, hd,
+                     (int)q.stride(0), (int)k.stride(0));
   HIP_CHECK_KERNEL();
 }
 
@@ -90,7 +94,9 @@ void append_kv_cache(torch::Tensor k, torch::Tensor v, torch::Tensor kcache,
                      torch::Tensor vcache, torch::Tensor slot_mapping) {
   TORCH_CHECK(k.is_cuda() && k.dtype() == torch::kBFloat16);
   TORCH_CHECK(slot_mapping.dtype() == torch::kInt32);
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
+  // k/v: [T, NKV, HD] views with possibly-strided row dim (fused QKV out)
+  TORCH_CHECK(k.stride(-1) == 1 && v.stride(-1) == 1);
+  TORCH_CHECK(k.stride(1) == k.size(2) && v.stride(1) == v.size(2));
   TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous());
   const int T = k.size(0);
   const int nkv = kcache.size(1);
@@ -103,7 +109,7 @@ void append_kv_cache(torch::Tensor k, torch::Tensor v, torch::Tensor kcache,
                      (const uint16_t*)v.data_ptr(), (uint16_t*)kcache.data_ptr(),
                      (uint16_t*)vcache.data_ptr(),
                      (const int32_t*)slot_mapping.data_ptr(), nkv, hd,
-                     block_size);
+                     block_size, (int)k.stride(0), (int)v.stride(0));
   HIP_CHECK_KERNEL();
 }
 

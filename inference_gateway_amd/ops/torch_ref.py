@@ -154,11 +154,14 @@ def sample_ref(logits, temperature, top_k, top_p, uniforms):
         if temp <= 0.0:
             out[i] = int(torch.argmax(row))
             continue
-        x = row / temp
-        m = float(x.max())
-        probs = torch.exp(x - m)
+        # thresholds live in the UNSCALED value domain (temperature > 0
+        # preserves order; the bf16 kernel walks unscaled bf16 keys), the
+        # exp-weights use the temperature
+        inv_t = 1.0 / temp
+        m = float(row.max())
+        probs = torch.exp((row - m) * inv_t)
         Z = float(probs.sum())
-        sorted_vals, _ = torch.sort(x, descending=True)
+        sorted_vals, _ = torch.sort(row, descending=True)
         k = int(top_k[i])
         tau_k = float("-inf")
         if 0 < k < V:
@@ -166,13 +169,13 @@ def sample_ref(logits, temperature, top_k, top_p, uniforms):
         p = float(top_p[i])
         tau_p = float("-inf")
         if p < 1.0:
-            sp = torch.exp(sorted_vals - m)
+            sp = torch.exp((sorted_vals - m) * inv_t)
             cum = torch.cumsum(sp, 0)
             idx = int(torch.searchsorted(cum, p * Z))
             idx = min(idx, V - 1)
             tau_p = float(sorted_vals[idx])
         tau = max(tau_k, tau_p)
-        keep = x >= tau
+        keep = row >= tau
         masked = torch.where(keep, probs, torch.zeros(()))
         zf = float(masked.sum())
         target = float(uniforms[i]) * zf

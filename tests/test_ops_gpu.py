@@ -149,3 +149,48 @@ def test_sample_topk_topp(k, p):
         logits.cpu(), temp.cpu(), tk.cpu(), tp.cpu(), u.cpu()
     )
     assert torch.equal(out.cpu(), ref)
+
+
+def test_sample_bf16_greedy():
+    B, V = 16, 128256
+    logits = torch.randn(B, V, dtype=torch.bfloat16, device="cuda")
+    temp = torch.zeros(B, device="cuda")
+    tk = torch.zeros(B, dtype=torch.int32, device="cuda")
+    tp = torch.ones(B, device="cuda")
+    u = torch.rand(B, device="cuda")
+    out = ops.sample(logits, temp, tk, tp, u)
+    ref = logits.float().argmax(-1).to(torch.int32)
+    assert torch.equal(out.cpu(), ref.cpu())
+
+
+@pytest.mark.parametrize("k,p", [(0, 1.0), (50, 1.0), (0, 0.9), (40, 0.95),
+                                 (1, 1.0)])
+def test_sample_bf16_topk_topp(k, p):
+    torch.manual_seed(123)
+    B, V = 8, 32000
+    logits = (torch.randn(B, V, device="cuda") * 4).to(torch.bfloat16)
+    temp = torch.full((B,), 0.8, device="cuda")
+    tk = torch.full((B,), k, dtype=torch.int32, device="cuda")
+    tp = torch.full((B,), p, device="cuda")
+    u = torch.rand(B, device="cuda").clamp(0.01, 0.99)
+    out = ops.sample(logits, temp, tk, tp, u)
+    ref = torch_ref.sample_ref(
+        logits.cpu(), temp.cpu(), tk.cpu(), tp.cpu(), u.cpu()
+    )
+    assert torch.equal(out.cpu(), ref)
+
+
+def test_sample_bf16_odd_vocab_tail():
+    """V not divisible by 8 exercises the scalar tail paths."""
+    torch.manual_seed(5)
+    B, V = 4, 1003
+    logits = (torch.randn(B, V, device="cuda") * 3).to(torch.bfloat16)
+    temp = torch.full((B,), 1.0, device="cuda")
+    tk = torch.full((B,), 25, dtype=torch.int32, device="cuda")
+    tp = torch.full((B,), 0.9, device="cuda")
+    u = torch.rand(B, device="cuda").clamp(0.01, 0.99)
+    out = ops.sample(logits, temp, tk, tp, u)
+    ref = torch_ref.sample_ref(
+        logits.cpu(), temp.cpu(), tk.cpu(), tp.cpu(), u.cpu()
+    )
+    assert torch.equal(out.cpu(), ref)
