@@ -90,10 +90,13 @@ class LLMEngine:
         self.kv_caches = self._allocate_kv(num_blocks)
         self._seq_counter = itertools.count()
         self._seqs: Dict[int, Sequence] = {}
+        # +1 block of headroom: the async decode pipeline can write KV for
+        # one in-flight token past max_model_len before the stop check runs
         self.max_blocks_per_seq = (
-            cfg.max_model_len + cfg.block_size - 1) // cfg.block_size
+            cfg.max_model_len + cfg.block_size - 1) // cfg.block_size + 1
         self.graph_runner = None
         self._gr_version = -1
+        self._pending = None   # (seqs, ring, event) of the in-flight decode
         if (self.device.type == "cuda" and cfg.use_hip_graphs
                 and not cfg.enforce_eager):
             from .graph_runner import DecodeGraphRunner
@@ -150,7 +153,7 @@ class LLMEngine:
             self.scheduler.abort(seq)
 
     def is_finished(self) -> bool:
-        return not self.scheduler.has_work
+        return not self.scheduler.has_work and self._pending is None
 
     def get_output(self, seq_id: int) -> RequestOutput:
         seq = self._seqs[seq_id]
@@ -164,21 +167,29 @@ class LLMEngine:
     # ---------------- step ----------------
     @torch.inference_mode()
     def step(self) -> List[StepOutput]:
+        """One engine step.  Graph-captured decode runs an async pipeline:
+        the host schedules and submits step N, THEN drains step N-1's
+        sampled tokens (D2H overlapped with step N's GPU work), so outputs
+        lag decode by one step.  A sequence whose stop condition is met at
+        the drain gets one in-flight step's token discarded — the standard
+        async-scheduling trade."""
         batch = self.scheduler.schedule()
         if batch is None:
-            return []
+            return self._drain_pending()
         gr = self.graph_runner
         if not batch.is_prefill and gr is not None:
-            # graph-captured decode: all input prep is device-side; the host
-            # only pushes block growths and reads back the sampled tokens
             gr.write_block_appends(batch.new_blocks)
             ver = self.scheduler.membership_version
-            gr.execute(batch.seqs, membership_changed=ver != self._gr_version)
+            ring, ev = gr.execute(batch.seqs,
+                                  membership_changed=ver != self._gr_version)
             self._gr_version = ver
-            torch.cuda.synchronize()
-            tokens = gr.tokens_pinned[: len(batch.seqs)].tolist()
-            return self._append_tokens(batch, tokens)
+            prev = self._drain_pending()      # overlaps with this step's GPU
+            for seq in batch.seqs:
+                seq.in_flight = True
+            self._pending = (list(batch.seqs), ring, ev)
+            return prev
 
+        prev = self._drain_pending()
         fb = self._build_forward_batch(batch)
         hidden = self.model(fb, self.kv_caches)
         hidden_last = hidden[fb.last_token_indices]
@@ -191,6 +202,26 @@ class LLMEngine:
                 if seq.state != SeqState.FINISHED:
                     gr.write_block_row(seq.slot, seq.block_ids)
                     gr.init_slot(seq, tok)
+        return prev + outputs
+
+    def _drain_pending(self) -> List[StepOutput]:
+        if self._pending is None:
+            return []
+        seqs, ring, ev = self._pending
+        self._pending = None
+        ev.synchronize()
+        tokens = self.graph_runner.tokens_pinned[ring, : len(seqs)].tolist()
+        outputs = []
+        for seq, tok in zip(seqs, tokens):
+            seq.in_flight = False
+            if seq.state == SeqState.FINISHED:
+                continue  # stop applied while this token was in flight
+            if seq.state == SeqState.WAITING:
+                # preempted after this token was sampled: keep it as part of
+                # the rebuilt prompt so the request resumes transparently
+                seq.prompt_token_ids.append(tok)
+                continue
+            outputs.extend(self._apply_token(seq, tok))
         return outputs
 
     def _build_forward_batch(self, batch: ScheduledBatch) -> ForwardBatch:
@@ -244,28 +275,27 @@ class LLMEngine:
             last_token_indices=torch.arange(B, dtype=torch.int64, device=dev),
         )
 
-    def _append_tokens(self, batch: ScheduledBatch,
-                       tokens: List[int]) -> List[StepOutput]:
+    def _apply_token(self, seq, tok: int) -> List[StepOutput]:
         import time
 
+        seq.output_token_ids.append(tok)
+        if seq.first_token_time is None:
+            seq.first_token_time = time.monotonic()
+        reason = None
+        sp = seq.sampling
+        if not sp.ignore_eos and tok in sp.stop_token_ids:
+            reason = "stop"
+        elif len(seq.output_token_ids) >= sp.max_new_tokens:
+            reason = "length"
+        elif seq.num_tokens >= self.cfg.max_model_len:
+            reason = "length"
+        if reason:
+            self.scheduler.finish(seq, reason)
+        return [StepOutput(seq.seq_id, tok, reason is not None, reason)]
+
+    def _append_tokens(self, batch: ScheduledBatch,
+                       tokens: List[int]) -> List[StepOutput]:
         outputs = []
-        now = time.monotonic()
         for seq, tok in zip(batch.seqs, tokens):
-            # decode positions were counted before this new token: the token
-            # generated now becomes part of the sequence
-            seq.output_token_ids.append(tok)
-            if seq.first_token_time is None:
-                seq.first_token_time = now
-            reason = None
-            sp = seq.sampling
-            if not sp.ignore_eos and tok in sp.stop_token_ids:
-                reason = "stop"
-            elif len(seq.output_token_ids) >= sp.max_new_tokens:
-                reason = "length"
-            elif seq.num_tokens >= self.cfg.max_model_len:
-                reason = "length"
-            if reason:
-                self.scheduler.finish(seq, reason)
-            outputs.append(
-                StepOutput(seq.seq_id, tok, reason is not None, reason))
+            outputs.extend(self._apply_token(seq, tok))
         return outputs

@@ -70,9 +70,14 @@ class DecodeGraphRunner:
                                       dtype=torch.int64, device=device)
         self._ones_i32 = torch.ones(bmax, dtype=torch.int32, device=device)
 
-        # pinned host staging for state writes / token readback
-        self.tokens_pinned = torch.zeros(bmax, dtype=torch.int32,
+        # pinned host staging for state writes / token readback.  Two
+        # buffers + events: the engine runs one decode step ahead of the
+        # host bookkeeping (async pipeline), so step N's D2H lands while
+        # step N+1 executes.
+        self.tokens_pinned = torch.zeros((2, bmax), dtype=torch.int32,
                                          pin_memory=True)
+        self._events = [torch.cuda.Event(), torch.cuda.Event()]
+        self._ring = 0
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._logits: Dict[int, torch.Tensor] = {}
         self._pool = None
@@ -173,11 +178,11 @@ class DecodeGraphRunner:
         return self.buckets[-1]
 
     # ---- execute --------------------------------------------------------
-    def execute(self, seqs: List[Sequence],
-                membership_changed: bool) -> torch.Tensor:
-        """One decode step; returns sampled tokens [B] int32 ON DEVICE and
-        queues an async D2H into self.tokens_pinned.  Slot state (token,
-        pos) is advanced on-device for the next step."""
+    def execute(self, seqs: List[Sequence], membership_changed: bool):
+        """One decode step; advances slot state (token, pos) on-device,
+        queues an async D2H of the sampled tokens, and returns
+        (ring_index, event) — the engine drains the tokens later (one-step
+        -ahead pipeline), reading self.tokens_pinned[ring_index]."""
         B = len(seqs)
         if membership_changed or self._cur_B != B:
             bucket = self.set_batch(seqs)
@@ -198,5 +203,8 @@ class DecodeGraphRunner:
         act = idx[:B]
         self.slot_token.scatter_(0, act, toks[:B].to(torch.int64))
         self.slot_pos.index_add_(0, act, self._ones_i32[:B])
-        self.tokens_pinned[:B].copy_(toks[:B], non_blocking=True)
-        return toks
+        ring = self._ring
+        self._ring ^= 1
+        self.tokens_pinned[ring, :B].copy_(toks[:B], non_blocking=True)
+        self._events[ring].record()
+        return ring, self._events[ring]

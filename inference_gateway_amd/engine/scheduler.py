@@ -36,6 +36,7 @@ class Sequence:
     first_token_time: Optional[float] = None
     slot: Optional[int] = None        # persistent batch slot (graph runner)
     synced_blocks: int = 0            # device block-table rows written so far
+    in_flight: bool = False           # a decode step's token not yet drained
 
     @property
     def num_tokens(self) -> int:
@@ -117,10 +118,13 @@ class Scheduler:
         while True:
             ok = True
             for seq in self.running:
-                # this step writes KV at position num_tokens-1
-                if seq.num_tokens > len(seq.block_ids) * bs:
+                # this step writes KV at position num_tokens-1, plus one
+                # more when a step's token is still in flight (async
+                # pipeline: the host count lags the device position)
+                need = seq.num_tokens + (1 if seq.in_flight else 0)
+                if need > len(seq.block_ids) * bs:
                     try:
-                        self.bm.ensure_capacity(seq.block_ids, seq.num_tokens)
+                        self.bm.ensure_capacity(seq.block_ids, need)
                     except RuntimeError:
                         ok = False
                         break

@@ -264,6 +264,51 @@ DEVINLINE float bf16_val(uint16_t u) {
   return v.f;
 }
 
+// Parallel suffix-threshold search over 256 combined bins, run by ONE wave.
+// vals: combined per-bin values (descending walk from bin 255).  Returns
+// (bstar, remaining) through s_res: the first bin b (from the top) where
+// cum_above + vals[b] >= target, with remaining = target - cum_above.
+template <typename T>
+DEVINLINE void suffix_threshold_search(const T* vals, T target, int lane,
+                                       uint32_t* s_b, T* s_rem) {
+  T lsum = T(0);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) lsum += vals[lane * 4 + j];
+  // inclusive suffix sum across lanes: s[l] = sum_{l' >= l} lsum[l']
+  T s = lsum;
+#pragma unroll
+  for (int off = 1; off < 64; off <<= 1) {
+    const T t = __shfl_down(s, off, 64);
+    if (lane + off < 64) s += t;
+  }
+  const T above = s - lsum;  // bins owned by lanes > l
+  if (s >= target && (lane == 63 || above < target)) {
+    // descend within this lane's 4 bins, highest first
+    T cum = above;
+    int bstar = lane * 4;
+    T rem = target - cum;
+#pragma unroll
+    for (int j = 3; j >= 0; --j) {
+      const int b = lane * 4 + j;
+      const T c = vals[b];
+      if (cum + c >= target) {
+        bstar = b;
+        rem = target - cum;
+        break;
+      }
+      cum += c;
+    }
+    *s_b = (uint32_t)bstar;
+    *s_rem = rem;
+  }
+  // degenerate: target exceeds the total (float rounding at p ~ 1.0) —
+  // fall through to bin 0 like the serial walk
+  if (lane == 0 && s < target) {
+    *s_b = 0;
+    *s_rem = target - s;
+  }
+}
+
 template <int BLOCK>
 __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
     int32_t* __restrict__ out,             // [B]
@@ -274,10 +319,14 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
     const float* __restrict__ uniforms,    // [B]
     int V) {
   constexpr int NW = BLOCK / WAVE_SIZE;
+  constexpr int REP = 4;  // bin replicas per wave: cuts LDS-atomic
+                          // conflicts on flat logits from ~32- to ~8-way
   __shared__ float redf[NW];
-  __shared__ uint32_t ubins[NW][256];
-  __shared__ float fbins[NW][256];
-  __shared__ uint32_t s_u32[2];
+  __shared__ uint32_t ubins[NW][REP][256];
+  __shared__ float fbins[NW][REP][256];
+  __shared__ uint32_t ucomb[256];
+  __shared__ float fcomb[256];
+  __shared__ uint32_t s_u32[4];
   __shared__ float s_f32[4];
   __shared__ float seg_sums[BLOCK + 1];
 
@@ -289,7 +338,7 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
   const float temp = temperature[seq];
   const bool greedy = (temp <= 0.0f);
   const float inv_t = greedy ? 1.0f : 1.0f / temp;
-  const int V8 = V >> 3;  // V is a multiple of 8 for every model vocab here
+  const int V8 = V >> 3;
   const uint4* row8 = reinterpret_cast<const uint4*>(row);
 
   // ---- pass 1: unscaled max (order == scaled order) ----
@@ -337,19 +386,22 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
     uint32_t tk_prefix = 0, tp_prefix = 0;
     uint32_t tk_remaining = do_k ? (uint32_t)k : 0;
     float tp_target = 0.0f;
+    const int rep = (lane >> 3) & (REP - 1);
 
     for (int level = 1; level >= 0; --level) {
       for (int b = tid; b < 256; b += BLOCK) {
 #pragma unroll
-        for (int w = 0; w < NW; ++w) {
-          ubins[w][b] = 0;
-          fbins[w][b] = 0.0f;
-        }
+        for (int w = 0; w < NW; ++w)
+#pragma unroll
+          for (int r = 0; r < REP; ++r) {
+            ubins[w][r][b] = 0;
+            fbins[w][r][b] = 0.0f;
+          }
       }
       __syncthreads();
       const int shift = level * 8;
-      uint32_t* const my_u = ubins[wave];
-      float* const my_f = fbins[wave];
+      uint32_t* const my_u = ubins[wave][rep];
+      float* const my_f = fbins[wave][rep];
       for (int i = tid; i < V8; i += BLOCK) {
         const uint4 v = row8[i];
         uint32_t runb_k = 0xffffffffu, runc_k = 0;
@@ -396,54 +448,55 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
           atomicAdd(&my_f[b], __expf((bf16_val(u) - M) * inv_t));
       }
       __syncthreads();
-      if (tid == 0) {
-        if (level == 1) {
-          float z = 0.0f;
-          for (int b = 0; b < 256; ++b)
-            for (int w = 0; w < NW; ++w) z += fbins[w][b];
-          s_f32[1] = z;
-          tp_target = p * z;
-        }
-        if (do_k) {
-          uint32_t cum = 0;
-          int bstar = 0;
-          for (int b = 255; b >= 0; --b) {
-            uint32_t c = 0;
-            for (int w = 0; w < NW; ++w) c += ubins[w][b];
-            if (cum + c >= tk_remaining) {
-              bstar = b;
-              break;
-            }
-            cum += c;
+      // combine replicas (parallel: one thread per bin)
+      {
+        const int b = tid;
+        uint32_t cu = 0;
+        float cf = 0.0f;
+#pragma unroll
+        for (int w = 0; w < NW; ++w)
+#pragma unroll
+          for (int r = 0; r < REP; ++r) {
+            cu += ubins[w][r][b];
+            cf += fbins[w][r][b];
           }
-          tk_prefix = (tk_prefix << 8) | (uint32_t)bstar;
-          tk_remaining -= cum;
-          s_u32[0] = tk_prefix;
-        }
-        if (do_p) {
-          float cum = 0.0f;
-          int bstar = 0;
-          for (int b = 255; b >= 0; --b) {
-            float c = 0.0f;
-            for (int w = 0; w < NW; ++w) c += fbins[w][b];
-            if (cum + c >= tp_target) {
-              bstar = b;
-              break;
-            }
-            cum += c;
-          }
-          tp_prefix = (tp_prefix << 8) | (uint32_t)bstar;
-          tp_target -= cum;
-          s_f32[2] = tp_target;
-          s_u32[1] = tp_prefix;
-        }
+        ucomb[b] = cu;
+        fcomb[b] = cf;
       }
       __syncthreads();
-      if (do_k) tk_prefix = s_u32[0];
+      if (level == 1) {
+        // Z = sum of the weight bins (parallel block reduce)
+        float z = fcomb[tid];
+        z = wave_reduce_sum(z);
+        if (lane == 0) redf[wave] = z;
+        __syncthreads();
+        if (tid == 0) {
+          float zz = 0.0f;
+          for (int w = 0; w < NW; ++w) zz += redf[w];
+          s_f32[1] = zz;
+        }
+        __syncthreads();
+        if (do_p) tp_target = p * s_f32[1];
+      }
+      // wave 0: top-k count search; wave 1: top-p mass search
+      if (do_k && wave == 0) {
+        suffix_threshold_search<uint32_t>(ucomb, tk_remaining, lane,
+                                          &s_u32[0], &s_u32[1]);
+      }
+      if (do_p && wave == (NW > 1 ? 1 : 0)) {
+        suffix_threshold_search<float>(fcomb, tp_target, lane, &s_u32[2],
+                                       &s_f32[2]);
+      }
+      __syncthreads();
+      if (do_k) {
+        tk_prefix = (tk_prefix << 8) | (s_u32[0] & 0xffu);
+        tk_remaining = s_u32[1];
+      }
       if (do_p) {
-        tp_prefix = s_u32[1];
+        tp_prefix = (tp_prefix << 8) | (s_u32[2] & 0xffu);
         tp_target = s_f32[2];
       }
+      __syncthreads();
     }
     tau = max(do_k ? tk_prefix : 0u, do_p ? tp_prefix : 0u);
   }
