@@ -31,6 +31,8 @@
 
 namespace igw {
 
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+
 // xor-reduce over the token axis (lane bits 2..5); values stay replicated
 DEVINLINE float token_reduce_max(float x) {
   x = fmaxf(x, __shfl_xor(x, 4, WAVE_SIZE));
@@ -127,27 +129,22 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
   if (b0 < nblocks) load_k(b0, kraw);
 
   for (int b = b0; b < nblocks; b += NWAVES) {
-    // ---- scores for all G heads; K and Q stay PACKED in registers and are
-    //      unpacked pairwise in-loop (dims outer, heads inner) so live
-    //      register state stays under the 3-waves/SIMD budget ----
+    // ---- scores for all G heads via v_dot2_f32_bf16: Q and K stay PACKED
+    //      (no bf16 unpack VALU at all — the v3 pairwise-unpack loop was
+    //      ~6x the instruction count and made the kernel VALU-bound at
+    //      4.2 TB/s; profiles/r01) ----
     float s[G];
 #pragma unroll
     for (int j = 0; j < G; ++j) s[j] = 0.0f;
-    const uint32_t* kp32 = reinterpret_cast<const uint32_t*>(&kraw[0]);
+    const bf16x2* kp2 = reinterpret_cast<const bf16x2*>(&kraw[0]);
+    const bf16x2* qp2 = reinterpret_cast<const bf16x2*>(&qpk[0][0]);
 #pragma unroll
     for (int c = 0; c < 16; ++c) {
-      const uint32_t kp = kp32[c];
-      const float k0 = bf16_to_f32((uint16_t)(kp & 0xffff));
-      const float k1 = bf16_to_f32((uint16_t)(kp >> 16));
+      const bf16x2 kp = kp2[c];
 #pragma unroll
       for (int j = 0; j < G; ++j) {
-        uint32_t qp = qpk[j][c];
-        // opaque redefinition: stops the compiler hoisting the unpack of
-        // the loop-invariant packed Q out of the block loop (which would
-        // cost 128 fp32 VGPRs and collapse occupancy)
-        asm("" : "+v"(qp));
-        s[j] = fmaf(bf16_to_f32((uint16_t)(qp & 0xffff)), k0, s[j]);
-        s[j] = fmaf(bf16_to_f32((uint16_t)(qp >> 16)), k1, s[j]);
+        s[j] = __builtin_amdgcn_fdot2_f32_bf16(qp2[j * 16 + c], kp, s[j],
+                                               false);
       }
     }
 
