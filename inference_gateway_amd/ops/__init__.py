@@ -128,3 +128,22 @@ def sample(logits, temperature, top_k, top_p, uniforms):
 
 build_cos_sin_cache = torch_ref.build_cos_sin_cache
 make_prefill_tiles = torch_ref.make_prefill_tiles
+
+
+def moe_route(router_logits: torch.Tensor, top_k: int, capacity: int):
+    """Fused MoE routing: top-k + softmax + padded-dispatch row indices.
+
+    router_logits: [T, E] fp32. Returns (weights [T,K] fp32,
+    dst [T*K] int32, topk_idx [T,K] int32)."""
+    if router_logits.is_cuda:
+        _require_kernels()
+        T, E = router_logits.shape
+        weights = torch.empty((T, top_k), dtype=torch.float32,
+                              device=router_logits.device)
+        dst = torch.empty(T * top_k, dtype=torch.int32,
+                          device=router_logits.device)
+        topk_idx = torch.empty((T, top_k), dtype=torch.int32,
+                               device=router_logits.device)
+        _kernels.moe_route(weights, dst, topk_idx, router_logits, capacity)
+        return weights, dst, topk_idx
+    return torch_ref.moe_route_ref(router_logits, top_k, capacity)

@@ -28,6 +28,8 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor top_k, torch::Tensor top_p, torch::Tensor uniforms);
 void silu_mul(torch::Tensor out, torch::Tensor gate_up);
+void moe_route(torch::Tensor weights, torch::Tensor dst,
+               torch::Tensor topk_idx, torch::Tensor logits, int64_t C);
 }  // namespace igw
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -45,4 +47,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample", &igw::sample,
         "fused temperature/top-k/top-p sampling (fp32 logits, gfx950)");
   m.def("silu_mul", &igw::silu_mul, "fused SwiGLU activation (bf16, gfx950)");
+  m.def("moe_route", &igw::moe_route,
+        "fused MoE top-k routing + softmax + padded-dispatch ranks (gfx950)");
 }

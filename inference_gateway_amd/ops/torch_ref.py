@@ -186,3 +186,21 @@ def sample_ref(logits, temperature, top_k, top_p, uniforms):
             chosen -= 1
         out[i] = chosen
     return out
+
+
+def moe_route_ref(logits: torch.Tensor, top_k: int, capacity: int):
+    """Reference for the fused MoE routing kernel (moe_route.hip).
+
+    Returns (weights [T,K] fp32, dst [T*K] int32, topk_idx [T,K] int32).
+    dst = expert * capacity + rank; rank order within an expert is NOT
+    part of the contract (the kernel uses atomic counters) — tests check
+    validity, not order.
+    """
+    T, E = logits.shape
+    vals, idx = torch.topk(logits.float(), top_k, dim=-1)
+    weights = torch.softmax(vals, dim=-1)
+    flat = idx.reshape(-1)
+    onehot = torch.nn.functional.one_hot(flat, E)
+    rank = (torch.cumsum(onehot, 0) - onehot).mul(onehot).sum(-1)
+    dst = (flat * capacity + rank).to(torch.int32)
+    return weights, dst, idx.to(torch.int32)

@@ -180,3 +180,25 @@ def test_moe_padded_every_expert_rank_unique():
     rank = (torch.cumsum(onehot, 0) - onehot).mul(onehot).sum(-1)
     dst = flat * T + rank
     assert dst.unique().numel() == T * K
+
+
+def test_moe_route_ref_valid():
+    from inference_gateway_amd.ops import torch_ref
+
+    torch.manual_seed(11)
+    T, E, K, C = 33, 8, 2, 33
+    logits = torch.randn(T, E)
+    w, dst, idx = torch_ref.moe_route_ref(logits, K, C)
+    # weights are softmax over the top-k logits
+    vals, ref_idx = torch.topk(logits, K, dim=-1)
+    assert torch.equal(idx.long(), ref_idx)
+    assert torch.allclose(w, torch.softmax(vals, -1), atol=1e-6)
+    # dst rows are unique and live in the right expert segment with a
+    # rank below that expert's count
+    assert dst.unique().numel() == T * K
+    flat = ref_idx.reshape(-1)
+    counts = torch.bincount(flat, minlength=E)
+    for i in range(T * K):
+        e = int(flat[i])
+        r = int(dst[i]) - e * C
+        assert 0 <= r < int(counts[e])

@@ -194,3 +194,22 @@ def test_sample_bf16_odd_vocab_tail():
         logits.cpu(), temp.cpu(), tk.cpu(), tp.cpu(), u.cpu()
     )
     assert torch.equal(out.cpu(), ref)
+
+
+def test_moe_route_matches_ref():
+    torch.manual_seed(21)
+    T, E, K, C = 256, 8, 2, 256
+    logits = torch.randn(T, E, device="cuda")
+    w, dst, idx = ops.moe_route(logits, K, C)
+    rw, rdst, ridx = torch_ref.moe_route_ref(logits.cpu(), K, C)
+    assert torch.equal(idx.cpu(), ridx)
+    assert torch.allclose(w.cpu(), rw, atol=1e-5)
+    # dst validity (rank order is not part of the contract)
+    d = dst.cpu()
+    assert d.unique().numel() == T * K
+    flat = ridx.reshape(-1).long()
+    counts = torch.bincount(flat, minlength=E)
+    for i in range(T * K):
+        e = int(flat[i])
+        r = int(d[i]) - e * C
+        assert 0 <= r < int(counts[e])
