@@ -102,6 +102,10 @@ class Gateway:
 
         mws: List[Middleware] = []
         mws.append(logging_middleware(self.logger))
+        if self.cfg.general.environment == "development":
+            from .middlewares.dev_mw import dev_logging_middleware
+
+            mws.append(dev_logging_middleware(self.logger))
         if self.telemetry is not None:
             mws.append(telemetry_middleware(self.telemetry))
         if self.authenticator is not None:

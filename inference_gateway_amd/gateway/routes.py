@@ -73,6 +73,13 @@ class Router:
         models = routing.filter_models(
             models, self.cfg.general.allowed_models,
             self.cfg.general.disallowed_models)
+        include = [s.strip() for s in req.query.get("include", "").split(",")
+                   if s.strip()]
+        if include:
+            from ..providers import metadata
+
+            models = await metadata.enrich_models(models, include,
+                                                  self.registry)
         return Response.json(
             {"object": "list", "data": models})
 
@@ -118,10 +125,16 @@ class Router:
             return Response.error(e.status, e.message)
         body = dict(body)
         body["model"] = model
-        # vision gate (reference routes.go:710-740): when vision is
-        # disabled, strip image parts rather than erroring
+        # vision gate (reference routes.go:710-740): strip image parts when
+        # vision is disabled OR the target model is not image-capable per
+        # the community modalities table (ModelAcceptsImages)
         if not self.cfg.general.enable_vision:
             self._strip_images(body)
+        else:
+            from ..providers import metadata
+
+            if not metadata.model_accepts_images(f"{pid}/{model}"):
+                self._strip_images(body)
 
         stream = bool(body.get("stream"))
         req.state["provider_id"] = pid
@@ -253,19 +266,65 @@ class Router:
             "/v1/images/edits": OPENAI_IMAGES_EDITS_ENDPOINT,
             "/v1/images/variations": OPENAI_IMAGES_VARIATIONS_ENDPOINT,
         }[req.path]
+        ct = req.headers.get("content-type", "application/json")
+        body = req.body
+        pid = "openai"
+        if ct.startswith("multipart/form-data"):
+            # multipart (edits/variations with binary image parts): sniff
+            # the model field, strip a provider prefix, re-encode in place
+            # (reference handleImagesMultipart, routes.go:1388-1575)
+            body, pid = self._rewrite_multipart_model(body, ct) or (body, pid)
+        else:
+            try:
+                parsed = json.loads(body or b"{}")
+            except Exception:
+                parsed = {}
+            model = parsed.get("model") if isinstance(parsed, dict) else None
+            if model:
+                p, m = routing.determine_provider_and_model(model)
+                if p is not None:
+                    pid = p
+                    parsed["model"] = m
+                    body = json.dumps(parsed).encode()
         try:
-            provider = self.registry.build("openai")
+            provider = self.registry.build(pid)
         except HTTPError as e:
             return Response.error(e.status, e.message)
-        ct = req.headers.get("content-type", "application/json")
         try:
-            r = await provider.forward_raw("POST", endpoint, req.body,
+            r = await provider.forward_raw("POST", endpoint, body,
                                            {"content-type": ct})
         except Exception as e:
             return Response.error(502, f"upstream error: {e}")
         return Response(r.status_code, r.content,
                         {"content-type": r.headers.get("content-type",
                                                        "application/json")})
+
+    @staticmethod
+    def _rewrite_multipart_model(body: bytes, content_type: str):
+        """Strip a provider prefix from the multipart `model` field.
+        Returns (new_body, provider_id) or None if no prefixed model."""
+        marker = "boundary="
+        i = content_type.find(marker)
+        if i < 0:
+            return None
+        boundary = content_type[i + len(marker):].split(";")[0].strip()
+        boundary = boundary.strip('"')
+        delim = ("--" + boundary).encode()
+        parts = body.split(delim)
+        for pi, part in enumerate(parts):
+            if b'name="model"' not in part:
+                continue
+            head, sep, value = part.partition(b"\r\n\r\n")
+            if not sep:
+                continue
+            value_body = value.rstrip(b"\r\n")
+            model = value_body.decode(errors="replace")
+            pid, bare = routing.determine_provider_and_model(model)
+            if pid is None:
+                return None
+            parts[pi] = head + sep + bare.encode() + b"\r\n"
+            return delim.join(parts), pid
+        return None
 
     # ---------------- /proxy/{provider}/... ----------------
     async def proxy(self, req: Request) -> Response:

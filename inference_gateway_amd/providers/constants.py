@@ -28,58 +28,13 @@ AUTH_NONE = "none"
 
 LOCAL_PROVIDER_ID = "local"
 
+# registry data is GENERATED from spec/gateway.yaml (tools/generate.py),
+# mirroring the reference's openapi.yaml -> registry_data.go pipeline
+from .constants_gen import PROVIDER_TABLE  # noqa: E402
+
 PROVIDERS: Dict[str, ProviderMeta] = {
-    "anthropic": ProviderMeta(
-        "anthropic", "Anthropic", "https://api.anthropic.com/v1",
-        AUTH_XHEADER, "/models", "/chat/completions",
-        (("anthropic-version", "2023-06-01"),),
-    ),
-    "cloudflare": ProviderMeta(
-        "cloudflare", "Cloudflare",
-        "https://api.cloudflare.com/client/v4/accounts/{ACCOUNT_ID}/ai",
-        AUTH_BEARER, "/finetunes/public?limit=1000", "/v1/chat/completions",
-    ),
-    "cohere": ProviderMeta(
-        "cohere", "Cohere", "https://api.cohere.ai", AUTH_BEARER,
-        "/v1/models", "/compatibility/v1/chat/completions",
-    ),
-    "deepseek": ProviderMeta(
-        "deepseek", "Deepseek", "https://api.deepseek.com", AUTH_BEARER,
-    ),
-    "google": ProviderMeta(
-        "google", "Google",
-        "https://generativelanguage.googleapis.com/v1beta/openai", AUTH_BEARER,
-    ),
-    "groq": ProviderMeta(
-        "groq", "Groq", "https://api.groq.com/openai/v1", AUTH_BEARER,
-    ),
-    "llamacpp": ProviderMeta(
-        "llamacpp", "Llamacpp", "http://llamacpp:8080/v1", AUTH_BEARER,
-    ),
-    "minimax": ProviderMeta(
-        "minimax", "Minimax", "https://api.minimax.io/v1", AUTH_BEARER,
-    ),
-    "mistral": ProviderMeta(
-        "mistral", "Mistral", "https://api.mistral.ai/v1", AUTH_BEARER,
-    ),
-    "moonshot": ProviderMeta(
-        "moonshot", "Moonshot", "https://api.moonshot.ai/v1", AUTH_BEARER,
-    ),
-    "nvidia": ProviderMeta(
-        "nvidia", "Nvidia", "https://integrate.api.nvidia.com/v1", AUTH_BEARER,
-    ),
-    "ollama": ProviderMeta(
-        "ollama", "Ollama", "http://ollama:8080/v1", AUTH_NONE,
-    ),
-    "ollama_cloud": ProviderMeta(
-        "ollama_cloud", "OllamaCloud", "https://ollama.com/v1", AUTH_BEARER,
-    ),
-    "openai": ProviderMeta(
-        "openai", "Openai", "https://api.openai.com/v1", AUTH_BEARER,
-    ),
-    "zai": ProviderMeta(
-        "zai", "Zai", "https://api.z.ai/api/paas/v4", AUTH_BEARER,
-    ),
+    pid: ProviderMeta(pid, name, url, auth, models_ep, chat_ep, extra)
+    for pid, name, url, auth, models_ep, chat_ep, extra in PROVIDER_TABLE
 }
 
 # endpoints only openai supports (reference constants.go:60-63)

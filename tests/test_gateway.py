@@ -246,3 +246,71 @@ async def test_images_gated_by_flag():
     async with gw_client(gw) as c:
         r = await c.post("/v1/images/generations", json={"prompt": "x"})
         assert r.status_code == 404
+
+
+@pytest.mark.asyncio
+async def test_models_include_enrichment():
+    """?include=context_window,pricing,modalities resolves provider >
+    community precedence; unknown values render as explicit null."""
+    gw, _ = make_gateway()
+    async with gw_client(gw) as c:
+        r = await c.get("/v1/models?provider=groq"
+                        "&include=context_window,pricing,modalities")
+    assert r.status_code == 200
+    models = r.json()["data"]
+    for m in models:
+        assert "context_window" in m and "pricing" in m and "modalities" in m
+        # mock upstream models are unknown to the community tables => null
+        assert m["context_window"] is None
+
+
+@pytest.mark.asyncio
+async def test_models_include_community_table_hit():
+    from inference_gateway_amd.providers import metadata
+
+    assert metadata.community_context_window("openai/gpt-4o") == 128000
+    assert metadata.community_pricing("openai/gpt-4o")["output"] == 1e-5
+    assert metadata.model_accepts_images("openai/gpt-4o")
+    assert not metadata.model_accepts_images("openai/gpt-3.5-turbo")
+    assert metadata.model_accepts_images("totally/unknown-model")
+
+
+@pytest.mark.asyncio
+async def test_images_multipart_model_rewrite():
+    """Multipart edits: provider prefix is stripped from the model field."""
+    gw, calls = make_gateway(env={"OPENAI_API_KEY": "sk-x",
+                                  "ENABLE_IMAGES": "true"})
+    boundary = "XbOuNdArYx"
+    parts = (
+        f"--{boundary}\r\n"
+        'Content-Disposition: form-data; name="model"\r\n\r\n'
+        "openai/dall-e-2\r\n"
+        f"--{boundary}\r\n"
+        'Content-Disposition: form-data; name="image"; filename="a.png"\r\n'
+        "Content-Type: image/png\r\n\r\n"
+        "PNGBYTES\r\n"
+        f"--{boundary}--\r\n"
+    ).encode()
+    async with gw_client(gw) as c:
+        r = await c.post(
+            "/v1/images/edits", content=parts,
+            headers={"content-type":
+                     f"multipart/form-data; boundary={boundary}"})
+    # mock upstream 404s unknown paths — we only care what was SENT
+    sent = [req for req in calls if "images" in req.url.path]
+    assert sent, r.text
+    body = sent[-1].content
+    assert b"openai/dall-e-2" not in body
+    assert b"dall-e-2" in body
+    assert b"PNGBYTES" in body
+
+
+def test_dev_truncation():
+    from inference_gateway_amd.gateway.middlewares.dev_mw import (
+        truncate_chat_body)
+
+    body = {"model": "m", "messages": [
+        {"role": "user", "content": "x" * 5000}]}
+    out = truncate_chat_body(body)
+    assert len(out["messages"][0]["content"]) < 300
+    assert "5000 bytes" in out["messages"][0]["content"]
