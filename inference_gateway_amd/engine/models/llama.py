@@ -25,7 +25,12 @@ from .batch import ForwardBatch
 
 
 def _seed(name: str) -> int:
-    return hash(name) & 0x7FFFFFFF
+    # deterministic ACROSS PROCESSES: python's hash() is salted per process,
+    # which would give each TP rank different weights (caught by the gloo
+    # world_size=2 CPU test)
+    import zlib
+
+    return zlib.crc32(name.encode()) & 0x7FFFFFFF
 
 
 class LlamaAttention(nn.Module):

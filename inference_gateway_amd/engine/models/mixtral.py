@@ -29,26 +29,43 @@ from .llama import LlamaAttention, _seed
 
 
 class MixtralMoE(nn.Module):
-    def __init__(self, cfg: ModelConfig, layer: int, device, dtype):
+    def __init__(self, cfg: ModelConfig, layer: int, device, dtype,
+                 expert_parallel: bool = False):
         super().__init__()
         tp, rank = tp_world_size(), tp_rank()
         h, inter = cfg.hidden_size, cfg.intermediate_size
         self.num_experts = cfg.num_experts
         self.top_k = cfg.num_experts_per_tok
+        self.expert_parallel = expert_parallel and tp > 1
         self.router = ReplicatedLinear(h, cfg.num_experts, device, dtype,
                                        _seed(f"l{layer}.router"))
-        assert inter % tp == 0
-        shard = inter // tp
+        if self.expert_parallel:
+            # EP: experts PARTITIONED across ranks, each expert full-width;
+            # dispatch/combine go over all-to-all (xGMI point-to-point)
+            assert cfg.num_experts % tp == 0
+            self.ep_size = tp
+            self.ep_local = cfg.num_experts // tp
+            eids = range(rank * self.ep_local, (rank + 1) * self.ep_local)
+            shard_tp, shard_rank = 1, 0
+        else:
+            # default: experts TP-SHARDED like dense MLPs (every rank holds
+            # a slice of every expert) — the right fit for the
+            # fully-connected 8-GPU xGMI node at Mixtral's scale
+            self.ep_size = 1
+            self.ep_local = cfg.num_experts
+            eids = range(cfg.num_experts)
+            shard_tp, shard_rank = tp, rank
+        assert inter % shard_tp == 0
         # per-expert fused [gate|up] and down weights, stacked:
-        # gate_up: [E, 2*shard, h], down: [E, h, shard]
+        # gate_up: [E_local, 2*shard, h], down: [E_local, h, shard]
         gus, downs = [], []
-        for e in range(cfg.num_experts):
-            g = ColumnParallelLinear(h, inter, rank, tp, device, dtype,
-                                     _seed(f"l{layer}.e{e}.gate"))
-            u = ColumnParallelLinear(h, inter, rank, tp, device, dtype,
-                                     _seed(f"l{layer}.e{e}.up"))
-            d = RowParallelLinear(inter, h, rank, tp, device, dtype,
-                                  _seed(f"l{layer}.e{e}.down"))
+        for e in eids:
+            g = ColumnParallelLinear(h, inter, shard_rank, shard_tp, device,
+                                     dtype, _seed(f"l{layer}.e{e}.gate"))
+            u = ColumnParallelLinear(h, inter, shard_rank, shard_tp, device,
+                                     dtype, _seed(f"l{layer}.e{e}.up"))
+            d = RowParallelLinear(inter, h, shard_rank, shard_tp, device,
+                                  dtype, _seed(f"l{layer}.e{e}.down"))
             gus.append(torch.cat([g.weight.data, u.weight.data], 0))
             downs.append(d.weight.data)
         self.gate_up = nn.Parameter(torch.stack(gus).contiguous(),
@@ -60,11 +77,77 @@ class MixtralMoE(nn.Module):
                 is_prefill: bool = False) -> torch.Tensor:
         from ...parallel.comm import tp_all_reduce
 
+        if self.expert_parallel:
+            return self._forward_expert_parallel(x)
         if is_prefill and x.shape[0] > 1024:
             out = self._forward_sorted(x)
         else:
             out = self._forward_padded(x)
         return tp_all_reduce(out)
+
+    def _forward_expert_parallel(self, x: torch.Tensor) -> torch.Tensor:
+        """EP dispatch/combine over torch.distributed all-to-all (RCCL on
+        xGMI; gloo in the CPU tests).  Activations enter replicated (the
+        attention block ends in an all-reduce), so each rank routes its
+        own sequence shard, all-to-alls the routed token copies to the
+        expert-owning ranks, computes its local full-width experts, and
+        all-to-alls the results back; an all-gather re-replicates."""
+        import torch.distributed as dist
+
+        W, r = self.ep_size, tp_rank()
+        T, h = x.shape
+        K, El = self.top_k, self.ep_local
+        logits = self.router(x).float()
+        weights, topk = torch.topk(logits, K, dim=-1)
+        weights = torch.softmax(weights, dim=-1).to(x.dtype)
+        # this rank's sequence shard
+        t0, t1 = (T * r) // W, (T * (r + 1)) // W
+        flat_e = topk[t0:t1].reshape(-1)                     # [n]
+        flat_t = (torch.arange(t0, t1, device=x.device)
+                  .unsqueeze(1).expand(t1 - t0, K).reshape(-1))
+        flat_w = weights[t0:t1].reshape(-1)
+        dest = flat_e // El                                  # owning rank
+        order = torch.argsort(dest, stable=True)
+        send_e = flat_e[order]
+        send_x = x.index_select(0, flat_t[order])
+        out_splits = torch.bincount(dest, minlength=W)
+        in_splits = torch.zeros_like(out_splits)
+        dist.all_to_all_single(in_splits, out_splits)
+        n_in = int(in_splits.sum())
+        recv_x = x.new_empty((n_in, h))
+        recv_e = send_e.new_empty(n_in)
+        dist.all_to_all_single(recv_x, send_x,
+                               in_splits.tolist(), out_splits.tolist())
+        dist.all_to_all_single(recv_e, send_e,
+                               in_splits.tolist(), out_splits.tolist())
+        # local experts on the received rows
+        ys = torch.zeros_like(recv_x)
+        for le in range(El):
+            rows = (recv_e == r * El + le).nonzero(as_tuple=True)[0]
+            if rows.numel() == 0:
+                continue
+            xe = recv_x.index_select(0, rows)
+            gu = F.linear(xe, self.gate_up[le])
+            ys.index_copy_(0, rows,
+                           F.linear(ops.silu_mul(gu), self.down[le]))
+        # combine: send results back along the reverse routes
+        back = torch.empty_like(send_x)
+        dist.all_to_all_single(back, ys,
+                               out_splits.tolist(), in_splits.tolist())
+        out_local = x.new_zeros((t1 - t0, h))
+        out_local.index_add_(0, flat_t[order] - t0,
+                             back * flat_w[order].unsqueeze(1))
+        # re-replicate the sequence across ranks (pad to the max shard —
+        # gloo/RCCL all-gather needs equal sizes)
+        max_tl = max((T * (d + 1)) // W - (T * d) // W for d in range(W))
+        padded = x.new_zeros((max_tl, h))
+        padded[: t1 - t0] = out_local
+        chunks = [x.new_empty((max_tl, h)) for _ in range(W)]
+        dist.all_gather(chunks, padded)
+        out = torch.cat(
+            [chunks[d][: (T * (d + 1)) // W - (T * d) // W]
+             for d in range(W)], 0)
+        return out
 
     def _forward_padded(self, x: torch.Tensor) -> torch.Tensor:
         """Sync-free dispatch for decode: tokens are scattered into a
