@@ -1,0 +1,156 @@
+"""Soak + end-to-end tests:
+
+* 64 concurrent SSE clients against the in-process local engine through
+  the gateway (BASELINE config 5's soak bar, CPU-sized), asserting every
+  stream terminates with exactly one [DONE];
+* the MCP agent loop against the REAL example filesystem MCP server
+  (examples/mcp-servers/filesystem_server.py) — not a canned mock — with
+  a scripted provider that asks for a write then a read (BASELINE
+  config 4's plumbing)."""
+import asyncio
+import importlib.util
+import json
+import sys
+from pathlib import Path
+
+import httpx
+import pytest
+
+from inference_gateway_amd import config as config_mod
+from inference_gateway_amd.config import LocalConfig
+from inference_gateway_amd.gateway.app import Gateway
+from inference_gateway_amd.logger import noop_logger
+from inference_gateway_amd.providers.local import LocalProvider
+
+from test_gateway import gw_client
+
+ROOT = Path(__file__).resolve().parent.parent
+
+
+@pytest.mark.asyncio
+async def test_soak_64_concurrent_sse_clients():
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=64,
+                     max_model_len=256)
+    lc.num_gpu_blocks = 1024
+    provider = LocalProvider(lc, device="cpu")
+    cfg = config_mod.load({})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    try:
+        async with gw_client(gw) as c:
+            async def one(i: int):
+                done, chunks = 0, 0
+                async with c.stream("POST", "/v1/chat/completions", json={
+                    "model": "local/tiny-llama",
+                    "messages": [{"role": "user", "content": f"req {i}"}],
+                    "max_tokens": 4, "temperature": 0.8, "stream": True,
+                }) as r:
+                    assert r.status_code == 200
+                    async for line in r.aiter_lines():
+                        if not line.startswith("data:"):
+                            continue
+                        if line[5:].strip() == "[DONE]":
+                            done += 1
+                        else:
+                            chunks += 1
+                return done, chunks
+
+            results = await asyncio.gather(*(one(i) for i in range(64)))
+        for done, chunks in results:
+            assert done == 1, "exactly one [DONE] per stream"
+            # byte-tokenizer may merge partial UTF-8 runs: role chunk
+            # + >=1 content chunk + final usage chunk
+            assert chunks >= 3
+    finally:
+        provider.engine.shutdown()
+
+
+def _load_fs_server():
+    spec = importlib.util.spec_from_file_location(
+        "fs_server", ROOT / "examples" / "mcp-servers" /
+        "filesystem_server.py")
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    return mod
+
+
+@pytest.mark.asyncio
+async def test_mcp_agent_against_real_filesystem_server(tmp_path,
+                                                        monkeypatch):
+    monkeypatch.setenv("SANDBOX_DIR", str(tmp_path))
+    fs = _load_fs_server()
+
+    # scripted provider: turn 1 asks to write, turn 2 asks to read, turn 3
+    # answers with the read content
+    state = {"turn": 0}
+
+    def provider_handler(request: httpx.Request) -> httpx.Response:
+        path = request.url.path
+        if path.endswith("/models"):
+            return httpx.Response(200, json={"object": "list", "data": [
+                {"id": "m1", "object": "model"}]})
+        if "/mcp" in path:
+            # route MCP JSON-RPC into the real example server
+            resp = fs.handle_rpc(json.loads(request.content))
+            return httpx.Response(200, json=resp)
+        body = json.loads(request.content)
+        state["turn"] += 1
+        t = state["turn"]
+        if t == 1:
+            msg = {"role": "assistant", "content": None, "tool_calls": [
+                {"id": "c1", "type": "function", "function": {
+                    "name": "mcp_write_file",
+                    "arguments": json.dumps({"path": "note.txt",
+                                             "content": "hello-mi355x"})}}]}
+            finish = "tool_calls"
+        elif t == 2:
+            # the previous tool result must be in the history
+            tool_msgs = [m for m in body["messages"]
+                         if m.get("role") == "tool"]
+            assert tool_msgs and "wrote" in tool_msgs[-1]["content"]
+            msg = {"role": "assistant", "content": None, "tool_calls": [
+                {"id": "c2", "type": "function", "function": {
+                    "name": "mcp_read_file",
+                    "arguments": json.dumps({"path": "note.txt"})}}]}
+            finish = "tool_calls"
+        else:
+            tool_msgs = [m for m in body["messages"]
+                         if m.get("role") == "tool"]
+            assert tool_msgs and "hello-mi355x" in tool_msgs[-1]["content"]
+            msg = {"role": "assistant",
+                   "content": "the file says: hello-mi355x"}
+            finish = "stop"
+        return httpx.Response(200, json={
+            "id": "x", "object": "chat.completion", "model": body["model"],
+            "choices": [{"index": 0, "message": msg,
+                         "finish_reason": finish}],
+            "usage": {"prompt_tokens": 1, "completion_tokens": 1,
+                      "total_tokens": 2}})
+
+    env = {"GROQ_API_KEY": "tok", "MCP_ENABLED": "true",
+           "MCP_SERVERS": "http://fs-server:8083/mcp",
+           "MCP_TOOL_MODE": "direct"}
+    cfg = config_mod.load(env)
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(provider_handler))
+    from inference_gateway_amd.mcp.layer import MCPLayer
+
+    mcp = MCPLayer(cfg.mcp, client, noop_logger())
+    await mcp.initialize()
+    assert len(mcp.client.all_tools()) == 7, "filesystem server has 7 tools"
+
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 mcp_layer=mcp)
+    async with gw_client(gw) as c:
+        r = await c.post("/v1/chat/completions", json={
+            "model": "groq/m1",
+            "messages": [{"role": "user",
+                          "content": "write a note then read it back"}]})
+    assert r.status_code == 200
+    final = r.json()["choices"][0]["message"]["content"]
+    assert "hello-mi355x" in final
+    # the file really exists in the sandbox
+    assert (tmp_path / "note.txt").read_text() == "hello-mi355x"
+    await mcp.stop()
