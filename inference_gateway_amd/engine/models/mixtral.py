@@ -157,12 +157,13 @@ class MixtralMoE(nn.Module):
         no extra HBM traffic — decode MoE is bound by streaming all 8
         experts' weights (~1.4 GB/layer), so the padded rows are free."""
         T, h = x.shape
-        E, K, C = self.num_experts, self.top_k, x.shape[0]
+        E, K = self.num_experts, self.top_k
+        C = (T + 63) // 64 * 64    # grouped-GEMM row-tile granularity
         router_logits = self.router(x).float()                # [T, E]
         # fused routing kernel: top-k + softmax + atomic-counter dispatch
         # ranks in one launch (the torch glue — topk/one-hot/cumsum — was
         # ~5.6 ms/step at batch 256; profiles/r01)
-        w32, dst32, _ = ops.moe_route(router_logits, K, C)
+        w32, dst32, _, counts = ops.moe_route(router_logits, K, C)
         weights = w32.to(x.dtype)                             # [T, K]
         dst = dst32.to(torch.int64)
         flat_token = (
@@ -171,13 +172,22 @@ class MixtralMoE(nn.Module):
             .expand(T, K)
             .reshape(-1)
         )
-        buf = x.new_zeros((E * C, h))
-        buf.index_copy_(0, dst, x[flat_token])
-        gu = torch.bmm(buf.view(E, C, h),
-                       self.gate_up.transpose(1, 2))          # [E, C, 2s]
-        act = ops.silu_mul(gu.view(E * C, -1))
-        y = torch.bmm(act.view(E, C, -1),
-                      self.down.transpose(1, 2)).view(E * C, h)
+        if x.is_cuda and ops.HAS_KERNELS:
+            # grouped MFMA GEMMs compute only the real rows per expert —
+            # the padded bmm burned up to E/K x the FLOPs
+            buf = x.new_empty((E * C, h))
+            buf.index_copy_(0, dst, x[flat_token])
+            gu = ops.moe_grouped_gemm(buf, self.gate_up, counts, C)
+            act = ops.silu_mul_grouped(gu, counts, C)
+            y = ops.moe_grouped_gemm(act, self.down, counts, C)
+        else:
+            buf = x.new_zeros((E * C, h))
+            buf.index_copy_(0, dst, x[flat_token])
+            gu = torch.bmm(buf.view(E, C, h),
+                           self.gate_up.transpose(1, 2))      # [E, C, 2s]
+            act = ops.silu_mul(gu.view(E * C, -1))
+            y = torch.bmm(act.view(E, C, -1),
+                          self.down.transpose(1, 2)).view(E * C, h)
         ys = y.index_select(0, dst)                           # [T*K, h]
         out = torch.zeros_like(x)
         out.index_add_(0, flat_token, ys * weights.reshape(-1, 1))

@@ -134,7 +134,7 @@ def moe_route(router_logits: torch.Tensor, top_k: int, capacity: int):
     """Fused MoE routing: top-k + softmax + padded-dispatch row indices.
 
     router_logits: [T, E] fp32. Returns (weights [T,K] fp32,
-    dst [T*K] int32, topk_idx [T,K] int32)."""
+    dst [T*K] int32, topk_idx [T,K] int32, counts [E] int32)."""
     if router_logits.is_cuda:
         _require_kernels()
         T, E = router_logits.shape
@@ -144,6 +144,38 @@ def moe_route(router_logits: torch.Tensor, top_k: int, capacity: int):
                           device=router_logits.device)
         topk_idx = torch.empty((T, top_k), dtype=torch.int32,
                                device=router_logits.device)
-        _kernels.moe_route(weights, dst, topk_idx, router_logits, capacity)
-        return weights, dst, topk_idx
+        counts = torch.empty(E, dtype=torch.int32,
+                             device=router_logits.device)
+        _kernels.moe_route(weights, dst, topk_idx, counts, router_logits,
+                           capacity)
+        return weights, dst, topk_idx, counts
     return torch_ref.moe_route_ref(router_logits, top_k, capacity)
+
+
+def moe_grouped_gemm(a: torch.Tensor, w: torch.Tensor,
+                     counts: torch.Tensor, capacity: int) -> torch.Tensor:
+    """Per-expert GEMM over the padded dispatch buffer.
+
+    a: [E*C, K] bf16 (rows >= counts[e] in each expert segment are
+    ignored), w: [E, N, K] bf16 -> out [E*C, N] bf16 (only real rows
+    written)."""
+    if a.is_cuda:
+        _require_kernels()
+        E, N, _K = w.shape
+        out = torch.empty((a.shape[0], N), dtype=a.dtype, device=a.device)
+        _kernels.moe_grouped_gemm(out, a, w, counts, capacity)
+        return out
+    return torch_ref.moe_grouped_gemm_ref(a, w, counts, capacity)
+
+
+def silu_mul_grouped(gate_up: torch.Tensor, counts: torch.Tensor,
+                     capacity: int) -> torch.Tensor:
+    """SwiGLU over only the real rows of the padded dispatch buffer."""
+    if gate_up.is_cuda:
+        _require_kernels()
+        inter = gate_up.shape[-1] // 2
+        out = torch.empty((*gate_up.shape[:-1], inter), dtype=gate_up.dtype,
+                          device=gate_up.device)
+        _kernels.silu_mul_grouped(out, gate_up, counts, capacity)
+        return out
+    return torch_ref.silu_mul_grouped_ref(gate_up, counts, capacity)

@@ -29,7 +29,12 @@ void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor top_k, torch::Tensor top_p, torch::Tensor uniforms);
 void silu_mul(torch::Tensor out, torch::Tensor gate_up);
 void moe_route(torch::Tensor weights, torch::Tensor dst,
-               torch::Tensor topk_idx, torch::Tensor logits, int64_t C);
+               torch::Tensor topk_idx, torch::Tensor counts,
+               torch::Tensor logits, int64_t C);
+void moe_grouped_gemm(torch::Tensor out, torch::Tensor a, torch::Tensor w,
+                      torch::Tensor counts, int64_t capacity);
+void silu_mul_grouped(torch::Tensor out, torch::Tensor gate_up,
+                      torch::Tensor counts, int64_t capacity);
 }  // namespace igw
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -49,4 +54,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_mul", &igw::silu_mul, "fused SwiGLU activation (bf16, gfx950)");
   m.def("moe_route", &igw::moe_route,
         "fused MoE top-k routing + softmax + padded-dispatch ranks (gfx950)");
+  m.def("moe_grouped_gemm", &igw::moe_grouped_gemm,
+        "grouped per-expert MFMA GEMM over the padded dispatch buffer");
+  m.def("silu_mul_grouped", &igw::silu_mul_grouped,
+        "SwiGLU over only the real (non-padded) MoE dispatch rows");
 }

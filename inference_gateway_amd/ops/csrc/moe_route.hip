@@ -27,6 +27,7 @@ __global__ __launch_bounds__(BLOCK) void moe_route_kernel(
     float* __restrict__ weights,       // [T, K] fp32 softmax weights
     int32_t* __restrict__ dst,         // [T*K] padded-buffer row index
     int32_t* __restrict__ topk_idx,    // [T, K] selected experts
+    int32_t* __restrict__ counts,      // [E] per-expert row counts
     const float* __restrict__ logits,  // [T, E] fp32 router logits
     int T, int E, int C) {
   __shared__ uint32_t counters[64];
@@ -75,15 +76,20 @@ __global__ __launch_bounds__(BLOCK) void moe_route_kernel(
       dst[(int64_t)t * K + k] = e * C + (int32_t)r;
     }
   }
+  __syncthreads();
+  for (int e = threadIdx.x; e < E; e += BLOCK)
+    counts[e] = (int32_t)counters[e];
 }
 
 void moe_route(torch::Tensor weights, torch::Tensor dst,
-               torch::Tensor topk_idx, torch::Tensor logits, int64_t C) {
+               torch::Tensor topk_idx, torch::Tensor counts,
+               torch::Tensor logits, int64_t C) {
   TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kFloat32);
   TORCH_CHECK(logits.is_contiguous());
   TORCH_CHECK(weights.dtype() == torch::kFloat32);
   TORCH_CHECK(dst.dtype() == torch::kInt32 &&
-              topk_idx.dtype() == torch::kInt32);
+              topk_idx.dtype() == torch::kInt32 &&
+              counts.dtype() == torch::kInt32);
   const int T = logits.size(0);
   const int E = logits.size(1);
   const int K = weights.size(1);
@@ -94,6 +100,7 @@ void moe_route(torch::Tensor weights, torch::Tensor dst,
   hipLaunchKernelGGL((moe_route_kernel<BLOCK, KV>), dim3(1), dim3(BLOCK), 0, \
                      stream, (float*)weights.data_ptr(),                     \
                      (int32_t*)dst.data_ptr(), (int32_t*)topk_idx.data_ptr(),\
+                     (int32_t*)counts.data_ptr(),                            \
                      (const float*)logits.data_ptr(), T, E, (int)C)
   switch (K) {
     case 1: LAUNCH_ROUTE(1); break;

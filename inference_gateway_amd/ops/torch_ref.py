@@ -203,4 +203,32 @@ def moe_route_ref(logits: torch.Tensor, top_k: int, capacity: int):
     onehot = torch.nn.functional.one_hot(flat, E)
     rank = (torch.cumsum(onehot, 0) - onehot).mul(onehot).sum(-1)
     dst = (flat * capacity + rank).to(torch.int32)
-    return weights, dst, idx.to(torch.int32)
+    counts = torch.bincount(flat, minlength=E).to(torch.int32)
+    return weights, dst, idx.to(torch.int32), counts
+
+
+def moe_grouped_gemm_ref(a, w, counts, capacity):
+    """Per-expert GEMM reference: only rows < counts[e] are computed;
+    padded rows of the output stay zero."""
+    E, N, K = w.shape
+    out = torch.zeros((a.shape[0], N), dtype=a.dtype)
+    for e in range(E):
+        c = int(counts[e])
+        if c == 0:
+            continue
+        rows = a[e * capacity:e * capacity + c].float()
+        out[e * capacity:e * capacity + c] = (
+            rows @ w[e].float().T).to(a.dtype)
+    return out
+
+
+def silu_mul_grouped_ref(gate_up, counts, capacity):
+    inter = gate_up.shape[-1] // 2
+    out = torch.zeros((*gate_up.shape[:-1], inter), dtype=gate_up.dtype)
+    E = counts.shape[0]
+    for e in range(E):
+        c = int(counts[e])
+        if c:
+            seg = gate_up[e * capacity:e * capacity + c]
+            out[e * capacity:e * capacity + c] = silu_mul_ref(seg)
+    return out

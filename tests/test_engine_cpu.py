@@ -188,7 +188,7 @@ def test_moe_route_ref_valid():
     torch.manual_seed(11)
     T, E, K, C = 33, 8, 2, 33
     logits = torch.randn(T, E)
-    w, dst, idx = torch_ref.moe_route_ref(logits, K, C)
+    w, dst, idx, counts = torch_ref.moe_route_ref(logits, K, C)
     # weights are softmax over the top-k logits
     vals, ref_idx = torch.topk(logits, K, dim=-1)
     assert torch.equal(idx.long(), ref_idx)
@@ -197,7 +197,7 @@ def test_moe_route_ref_valid():
     # rank below that expert's count
     assert dst.unique().numel() == T * K
     flat = ref_idx.reshape(-1)
-    counts = torch.bincount(flat, minlength=E)
+    assert torch.equal(counts.long(), torch.bincount(flat, minlength=E))
     for i in range(T * K):
         e = int(flat[i])
         r = int(dst[i]) - e * C

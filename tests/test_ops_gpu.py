@@ -200,8 +200,9 @@ def test_moe_route_matches_ref():
     torch.manual_seed(21)
     T, E, K, C = 256, 8, 2, 256
     logits = torch.randn(T, E, device="cuda")
-    w, dst, idx = ops.moe_route(logits, K, C)
-    rw, rdst, ridx = torch_ref.moe_route_ref(logits.cpu(), K, C)
+    w, dst, idx, counts = ops.moe_route(logits, K, C)
+    rw, rdst, ridx, rcounts = torch_ref.moe_route_ref(logits.cpu(), K, C)
+    assert torch.equal(counts.cpu(), rcounts)
     assert torch.equal(idx.cpu(), ridx)
     assert torch.allclose(w.cpu(), rw, atol=1e-5)
     # dst validity (rank order is not part of the contract)
@@ -213,3 +214,48 @@ def test_moe_route_matches_ref():
         e = int(flat[i])
         r = int(d[i]) - e * C
         assert 0 <= r < int(counts[e])
+
+
+def test_moe_grouped_gemm_matches_ref():
+    torch.manual_seed(31)
+    E, C, K, N = 4, 128, 256, 256
+    a = torch.randn(E * C, K, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(E, N, K, dtype=torch.bfloat16, device="cuda") * 0.05
+    counts = torch.tensor([128, 7, 0, 65], dtype=torch.int32, device="cuda")
+    out = ops.moe_grouped_gemm(a, w, counts, C)
+    ref = torch_ref.moe_grouped_gemm_ref(a.cpu(), w.cpu(), counts.cpu(), C)
+    for e in range(E):
+        c = int(counts[e])
+        got = out[e * C:e * C + c].float().cpu()
+        want = ref[e * C:e * C + c].float()
+        assert torch.allclose(got, want, atol=0.5, rtol=0.05), \
+            f"expert {e} mismatch: max {(got - want).abs().max()}"
+
+
+def test_silu_mul_grouped_matches_ref():
+    torch.manual_seed(32)
+    E, C, inter = 4, 64, 512
+    gu = torch.randn(E * C, 2 * inter, dtype=torch.bfloat16, device="cuda")
+    counts = torch.tensor([64, 1, 0, 33], dtype=torch.int32, device="cuda")
+    out = ops.silu_mul_grouped(gu, counts, C)
+    ref = torch_ref.silu_mul_grouped_ref(gu.cpu(), counts.cpu(), C)
+    for e in range(E):
+        c = int(counts[e])
+        assert torch.allclose(out[e * C:e * C + c].float().cpu(),
+                              ref[e * C:e * C + c].float(), atol=2e-2)
+
+
+def test_moe_padded_forward_gpu_matches_sorted():
+    """Grouped-GEMM decode MoE vs the independent sorted-F.linear path."""
+    from inference_gateway_amd.engine.config import MODEL_PRESETS
+    from inference_gateway_amd.engine.models.mixtral import MixtralMoE
+
+    torch.manual_seed(33)
+    cfg = MODEL_PRESETS["tiny-moe"]
+    moe = MixtralMoE(cfg, layer=0, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(37, cfg.hidden_size, dtype=torch.bfloat16,
+                    device="cuda") * 0.5
+    a = moe._forward_padded(x).float()
+    b = moe._forward_sorted(x).float()
+    assert torch.allclose(a, b, atol=5e-2, rtol=5e-2), \
+        f"max diff {(a - b).abs().max()}"
