@@ -11,12 +11,13 @@
 //
 //   * tile BM=64 x BN=128 x BK=64, 4 waves, each wave a 32x64 sub-tile of
 //     16x16x32 bf16 MFMAs accumulating in AGPRs;
-//   * A and W staged global->reg->LDS (T14 async split: loads for tile
-//     t+2 issue while tile t computes; the reg->LDS write for tile t+1
-//     lands right after the barrier), double buffered, one barrier per
-//     K-step;
-//   * the ds_read_b128 fragment reads are column-slices of 128-B-row
-//     tiles => XOR swizzle ((row&7)<<4) on write and read (guide T2);
+//   * A staged global->reg->LDS (T14 split, double buffered, XOR
+//     swizzle on write and read — guide T2); W is NOT staged: each
+//     expert's weight tile is streamed exactly once per workgroup and a
+//     lane's 16 contiguous bytes of a W row ARE its MFMA B-fragment, so
+//     W fragments load straight to VGPRs with non-temporal hints (the
+//     guide's "operand streamed once per block" rule — an LDS round
+//     trip for W was pure overhead: 1.9 TB/s vs 4+);
 //   * grid.y = E * C/BM row-tiles; a tile whose row base is past
 //     counts[e] exits immediately, and stores are masked to real rows.
 //
@@ -63,31 +64,25 @@ __global__ __launch_bounds__(NWAVES * WAVE_SIZE) void moe_gemm_kernel(
   const int lane = tid & 63;
   const int wave = tid >> 6;
 
-  // LDS: double-buffered A [BM][BK] and W [BN][BK] images (swizzled);
-  // computed per use (a static array of LDS pointers trips hipcc's
-  // addrspacecast-in-initializer limitation)
-  __shared__ uint16_t lds[2 * (BM + BN) * BK];
-#define A_LDS(b) (lds + (b) * (BM + BN) * BK)
-#define W_LDS(b) (lds + (b) * (BM + BN) * BK + BM * BK)
+  // LDS: double-buffered A [BM][BK] image only (swizzled)
+  __shared__ uint16_t lds[2 * BM * BK];
+#define A_LDS(b) (lds + (b) * BM * BK)
 
   const uint16_t* a_base = a + ((int64_t)e * C + row0) * K;
   const uint16_t* w_base = w + ((int64_t)e * N + ntile * BN) * K;
+  // wave (wm, wn): wm in {0,1} rows, wn in {0,1} cols -> 32x64 out/wave
+  const int wm = wave >> 1;
+  const int wn = wave & 1;
 
-  // ---- reg staging: A 2 pieces/thread (8 KiB), W 4 pieces/thread
-  //      (16 KiB); 8 16-B pieces per 128-B tile row ----
-  ushort8 ra[2], rw[4];
+  // ---- A reg staging: 2 pieces/thread (8 KiB); 8 16-B pieces per
+  //      128-B tile row ----
+  ushort8 ra[2];
   auto load_tile = [&](int k0) {
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       const int p = i * 256 + tid;
       ra[i] = *reinterpret_cast<const ushort8*>(
           a_base + (int64_t)(p >> 3) * K + k0 + (p & 7) * 8);
-    }
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int p = i * 256 + tid;
-      rw[i] = *reinterpret_cast<const ushort8*>(
-          w_base + (int64_t)(p >> 3) * K + k0 + (p & 7) * 8);
     }
   };
   auto write_tile = [&](int buf) {
@@ -97,17 +92,27 @@ __global__ __launch_bounds__(NWAVES * WAVE_SIZE) void moe_gemm_kernel(
       *reinterpret_cast<ushort8*>(
           (char*)A_LDS(buf) + swz(p >> 3, p & 7)) = ra[i];
     }
+  };
+  // W fragment rows for this wave (fixed across the K loop)
+  const int frow = lane & 15;
+  const int fk0 = (lane >> 4) * 8;         // k element offset within BK=32
+  const uint16_t* wrow[4];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int p = i * 256 + tid;
-      *reinterpret_cast<ushort8*>(
-          (char*)W_LDS(buf) + swz(p >> 3, p & 7)) = rw[i];
-    }
+  for (int ni = 0; ni < 4; ++ni) {
+    wrow[ni] = w_base + (int64_t)((wave & 1) * 64 + ni * 16 + frow) * K;
+  }
+  bf16x8 wfrag[2][4];
+  auto load_wfrags = [&](int t) {
+    const int k0 = t * BK;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        wfrag[ks][ni] = __builtin_nontemporal_load(
+            reinterpret_cast<const bf16x8*>(wrow[ni] + k0 + ks * 32 + fk0));
+      }
   };
 
-  // wave (wm, wn): wm in {0,1} rows, wn in {0,1} cols -> 32x64 out/wave
-  const int wm = wave >> 1;
-  const int wn = wave & 1;
   f32x4 acc[2][4];
 #pragma unroll
   for (int i = 0; i < 2; ++i)
@@ -116,26 +121,24 @@ __global__ __launch_bounds__(NWAVES * WAVE_SIZE) void moe_gemm_kernel(
 
   load_tile(0);
   write_tile(0);
+  load_wfrags(0);
   if (K > BK) load_tile(BK);
   __syncthreads();
 
   const int nk = K / BK;
   int cur = 0;
+  const int fkp = fk0 >> 3;                  // k piece index (0..3)
   for (int t = 0; t < nk; ++t) {
-    // write tile t+1 (its loads were issued an iteration ago), then issue
+    // write A tile t+1 (loads issued an iteration ago), then issue A
     // tile t+2's loads so HBM latency hides under this step's MFMAs
     if (t + 1 < nk) {
       write_tile(cur ^ 1);
       if (t + 2 < nk) load_tile((t + 2) * BK);
     }
-    // fragment rows: A row = wm*32 + mi*16 + (lane&15); k = (lane>>4)*8
-    const int frow = lane & 15;
-    const int fk = (lane >> 4) * 8;          // k element offset
-    const int fkp = fk >> 3;                 // k piece index (0..3)
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {         // two K=32 sub-steps
-      bf16x8 afrag[2], wfrag[4];
+      bf16x8 afrag[2];
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi) {
         const int r = wm * 32 + mi * 16 + frow;
@@ -143,20 +146,15 @@ __global__ __launch_bounds__(NWAVES * WAVE_SIZE) void moe_gemm_kernel(
             (const char*)A_LDS(cur) + swz(r, ks * 4 + fkp));
       }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        const int r = wn * 64 + ni * 16 + frow;
-        wfrag[ni] = *reinterpret_cast<const bf16x8*>(
-            (const char*)W_LDS(cur) + swz(r, ks * 4 + fkp));
-      }
-#pragma unroll
       for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni) {
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag[mi], wfrag[ni], acc[mi][ni], 0, 0, 0);
+              afrag[mi], wfrag[ks][ni], acc[mi][ni], 0, 0, 0);
         }
     }
     __builtin_amdgcn_s_setprio(0);
+    if (t + 1 < nk) load_wfrags(t + 1);      // next step's W fragments
     __syncthreads();
     cur ^= 1;
   }
@@ -178,7 +176,6 @@ __global__ __launch_bounds__(NWAVES * WAVE_SIZE) void moe_gemm_kernel(
         }
       }
 #undef A_LDS
-#undef W_LDS
 }
 
 }  // namespace
