@@ -172,14 +172,29 @@ class MixtralMoE(nn.Module):
             .expand(T, K)
             .reshape(-1)
         )
-        if x.is_cuda and ops.HAS_KERNELS:
-            # grouped MFMA GEMMs compute only the real rows per expert —
-            # the padded bmm burned up to E/K x the FLOPs
+        import os
+
+        if (x.is_cuda and ops.HAS_KERNELS
+                and os.environ.get("IGW_MOE_GROUPED", "0") == "1"):
+            # grouped MFMA GEMMs compute only the real rows per expert.
+            # Opt-in while the kernel is below hipBLASLt's batched GEMM on
+            # these shapes (tools/bench_moe_gemm.py tracks the gap; the
+            # small-N down projection needs split-K to fill the chip)
             buf = x.new_empty((E * C, h))
             buf.index_copy_(0, dst, x[flat_token])
             gu = ops.moe_grouped_gemm(buf, self.gate_up, counts, C)
             act = ops.silu_mul_grouped(gu, counts, C)
             y = ops.moe_grouped_gemm(act, self.down, counts, C)
+        elif x.is_cuda and ops.HAS_KERNELS:
+            # padded batched GEMM (hipBLASLt), with the fused routing
+            # kernel and padding-skipping SwiGLU
+            buf = x.new_zeros((E * C, h))
+            buf.index_copy_(0, dst, x[flat_token])
+            gu = torch.bmm(buf.view(E, C, h),
+                           self.gate_up.transpose(1, 2))      # [E, C, 2s]
+            act = ops.silu_mul_grouped(gu.view(E * C, -1), counts, C)
+            y = torch.bmm(act.view(E, C, -1),
+                          self.down.transpose(1, 2)).view(E * C, h)
         else:
             buf = x.new_zeros((E * C, h))
             buf.index_copy_(0, dst, x[flat_token])

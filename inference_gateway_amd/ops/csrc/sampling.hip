@@ -402,11 +402,14 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
       const int shift = level * 8;
       uint32_t* const my_u = ubins[wave][rep];
       float* const my_f = fbins[wave][rep];
+      // run-merge accumulators live across the WHOLE pass: near-flat
+      // logits keep hitting the same bin, so each thread issues only a
+      // handful of LDS atomics instead of one per 8 elements
+      uint32_t runb_k = 0xffffffffu, runc_k = 0;
+      uint32_t runb_p = 0xffffffffu;
+      float runw_p = 0.0f;
       for (int i = tid; i < V8; i += BLOCK) {
         const uint4 v = row8[i];
-        uint32_t runb_k = 0xffffffffu, runc_k = 0;
-        uint32_t runb_p = 0xffffffffu;
-        float runw_p = 0.0f;
 #pragma unroll
         for (int h = 0; h < 8; ++h) {
           const uint32_t q = (&v.x)[h >> 1];
@@ -434,9 +437,9 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
             }
           }
         }
-        if (runc_k) atomicAdd(&my_u[runb_k], runc_k);
-        if (runb_p != 0xffffffffu) atomicAdd(&my_f[runb_p], runw_p);
       }
+      if (runc_k) atomicAdd(&my_u[runb_k], runc_k);
+      if (runb_p != 0xffffffffu) atomicAdd(&my_f[runb_p], runw_p);
       for (int i = (V8 << 3) + tid; i < V; i += BLOCK) {
         const uint16_t u = row[i];
         const uint32_t key = bf16_key(u);
