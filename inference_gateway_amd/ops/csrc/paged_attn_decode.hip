@@ -149,7 +149,9 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
     }
 
     // ---- issue V loads (current block) and K loads (next block) now so
-    //      their latency hides under the shfl chains below ----
+    //      their latency hides under the shfl chains below; the scores
+    //      consumed kraw above, so the next-K prefetch gets the whole
+    //      softmax + V-accumulate phase of flight time ----
     const int tlimit = min(BS, ctx - b * BS);
     ushort8 vraw[4];
     const uint16_t* vp = vcache + ((int64_t)bt_row[b] * nkv) * BS * HD +
@@ -166,6 +168,7 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
           vraw[t4] = *reinterpret_cast<const ushort8*>(
               vp + (vtok + 4 * t4) * HD);
     }
+    if (b + NWAVES < nblocks) load_k(b + NWAVES, kraw);
 
     // ---- finish scores: pairwise dot reduce + mask ----
     const bool valid = (b * BS + tok) < ctx;
@@ -195,7 +198,10 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
     for (int j = 0; j < G; ++j)
       l[j] += token_reduce_sum((lane & 3) == 0 ? p[j] : 0.0f);
 
-    // ---- V accumulate: p reaches the V lanes via bpermute ----
+    // ---- V accumulate: p reaches the V lanes via bpermute; math runs
+    //      on float2 pairs (v_pk_fma_f32: 2 FLOPs/instr — no MFMAs here
+    //      to contend with, so packed f32 is a straight 2x on the VALU
+    //      issue count of this phase) ----
 #pragma unroll
     for (int t4 = 0; t4 < 4; ++t4) {
       const int t = vtok + 4 * t4;
@@ -204,17 +210,25 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
       for (int j = 0; j < G; ++j)
         pj[j] = __shfl(p[j], 4 * t, WAVE_SIZE);
       if (t >= tlimit) continue;
-      float vf[8];
+      const uint32_t* vp32 = reinterpret_cast<const uint32_t*>(&vraw[t4]);
+      float2 vf2[4];
 #pragma unroll
-      for (int e = 0; e < 8; ++e) vf[e] = bf16_to_f32(vraw[t4].v[e]);
+      for (int e2 = 0; e2 < 4; ++e2) {
+        const uint32_t q = vp32[e2];
+        vf2[e2].x = bf16_to_f32((uint16_t)(q & 0xffff));
+        vf2[e2].y = bf16_to_f32((uint16_t)(q >> 16));
+      }
 #pragma unroll
-      for (int j = 0; j < G; ++j)
+      for (int j = 0; j < G; ++j) {
+        float2* a2 = reinterpret_cast<float2*>(&acc[j][0]);
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          acc[j][e] = fmaf(pj[j], vf[e], acc[j][e]);
+        for (int e2 = 0; e2 < 4; ++e2) {
+          a2[e2].x = fmaf(pj[j], vf2[e2].x, a2[e2].x);
+          a2[e2].y = fmaf(pj[j], vf2[e2].y, a2[e2].y);
+        }
+      }
     }
 
-    if (b + NWAVES < nblocks) load_k(b + NWAVES, kraw);
   }
 
   // fold the 4 token-subgroup partials (lanes l, l+16, l+32, l+48 share dims)
