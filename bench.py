@@ -41,12 +41,16 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--batch", type=int, default=512,
+    ap.add_argument("--batch", type=int, default=1024,
                     help="concurrent sequences per GPU")
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--model", type=str, default="llama-3-8b")
     ap.add_argument("--eager", action="store_true",
                     help="disable hipGraph capture (for TunableOp tuning)")
+    ap.add_argument("--tp", type=int, default=0,
+                    help="tensor-parallel degree (0 = DP weak scaling; "
+                         "--tp N shards the model over N ranks via "
+                         "RCCL/xGMI, requires world_size == N)")
     args = ap.parse_args()
 
     from inference_gateway_amd.engine.config import (EngineConfig,
@@ -57,6 +61,8 @@ def main() -> None:
     comm.init_distributed()
     nranks = comm.world_size()
     rank = comm.world_rank()
+    if args.tp:
+        comm.init_tp(args.tp)
     if torch.cuda.is_available():
         device = f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}"
     else:
@@ -113,7 +119,9 @@ def main() -> None:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
 
-    tokens_total = args.batch * args.steps * nranks
+    # TP ranks compute ONE model together: count the batch once
+    replicas = 1 if args.tp else nranks
+    tokens_total = args.batch * args.steps * replicas
     value = tokens_total / elapsed
     if rank == 0:
         print(json.dumps({
@@ -133,7 +141,8 @@ def main() -> None:
                 "model": args.model,
                 "global_batch": args.batch * nranks,
                 "seq_len": args.prompt_len,
-                "parallelism": f"dp{nranks}" if nranks > 1 else "tp1",
+                "parallelism": (f"tp{args.tp}" if args.tp else
+                                (f"dp{nranks}" if nranks > 1 else "tp1")),
             },
         }))
 

@@ -452,8 +452,7 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
       }
       __syncthreads();
       // combine replicas (parallel: one thread per bin)
-      {
-        const int b = tid;
+      for (int b = tid; b < 256; b += BLOCK) {
         uint32_t cu = 0;
         float cf = 0.0f;
 #pragma unroll
@@ -469,7 +468,7 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
       __syncthreads();
       if (level == 1) {
         // Z = sum of the weight bins (parallel block reduce)
-        float z = fcomb[tid];
+        float z = (tid < 256) ? fcomb[tid] : 0.0f;
         z = wave_reduce_sum(z);
         if (lane == 0) redf[wave] = z;
         __syncthreads();
@@ -504,12 +503,24 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
     tau = max(do_k ? tk_prefix : 0u, do_p ? tp_prefix : 0u);
   }
 
-  // ---- kept mass + per-thread segment sums (one fused pass) ----
-  const int seg = (V + BLOCK - 1) / BLOCK;
-  const int s0 = tid * seg;
+  // ---- kept mass + per-thread segment sums (one fused pass,
+  //      vectorized uint4 loads; segments 8-aligned) ----
+  const int seg = (((V + BLOCK - 1) / BLOCK) + 7) & ~7;
+  const int s0 = min(V, tid * seg);
   const int s1 = min(V, s0 + seg);
   float segsum = 0.0f;
-  for (int i = s0; i < s1; ++i) {
+  int i = s0;
+  for (; i + 8 <= s1; i += 8) {
+    const uint4 v = *reinterpret_cast<const uint4*>(row + i);
+#pragma unroll
+    for (int h = 0; h < 8; ++h) {
+      const uint32_t q = (&v.x)[h >> 1];
+      const uint16_t u = (uint16_t)((h & 1) ? (q >> 16) : (q & 0xffffu));
+      if (bf16_key(u) >= tau)
+        segsum += __expf((bf16_val(u) - M) * inv_t);
+    }
+  }
+  for (; i < s1; ++i) {
     const uint16_t u = row[i];
     if (bf16_key(u) >= tau) segsum += __expf((bf16_val(u) - M) * inv_t);
   }
@@ -574,7 +585,11 @@ void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
   constexpr int BLOCK = 256;
   hipStream_t stream = at::cuda::getCurrentCUDAStream();
   if (logits.dtype() == torch::kBFloat16) {
-    hipLaunchKernelGGL((sample_bf16_kernel<BLOCK>), dim3(B), dim3(BLOCK), 0,
+    // 8 waves per row: the radix walk is latency-bound at 4 waves
+    // (PMC: 50% parked / 32% issue-stalled at BLOCK=256)
+    constexpr int BLOCK_BF16 = 512;
+    hipLaunchKernelGGL((sample_bf16_kernel<BLOCK_BF16>), dim3(B),
+                       dim3(BLOCK_BF16), 0,
                        stream, (int32_t*)out.data_ptr(),
                        (const uint16_t*)logits.data_ptr(),
                        (const float*)temperature.data_ptr(),
