@@ -44,6 +44,13 @@ class Gateway:
             from ..otel.metrics import Telemetry
 
             self.telemetry = Telemetry()
+        self.tracer = None
+        if self.cfg.telemetry.tracing_enabled:
+            from ..otel.tracing import Tracer
+
+            self.tracer = Tracer(
+                endpoint=self.cfg.telemetry.tracing_otlp_endpoint,
+                client=self.client)
         self.guardrails = guardrails
         if guardrails is None and self.cfg.guardrails.enabled:
             from ..guardrails.engine import Evaluator
@@ -62,7 +69,8 @@ class Gateway:
 
             self.mcp = MCPLayer(self.cfg.mcp, self.client, self.logger,
                                 guardrails=self.guardrails,
-                                telemetry=self.telemetry)
+                                telemetry=self.telemetry,
+                                tracer=self.tracer)
         selector = None
         if self.cfg.routing.enabled:
             from ..providers import routing as routing_mod
@@ -101,6 +109,10 @@ class Gateway:
         from .middlewares.telemetry_mw import telemetry_middleware
 
         mws: List[Middleware] = []
+        if self.tracer is not None:
+            from ..otel.tracing import tracing_middleware
+
+            mws.append(tracing_middleware(self.tracer))
         mws.append(logging_middleware(self.logger))
         if self.cfg.general.environment == "development":
             from .middlewares.dev_mw import dev_logging_middleware

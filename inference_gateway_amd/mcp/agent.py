@@ -68,11 +68,13 @@ def accumulate_streaming_tool_calls(lines: List[str]) -> List[dict]:
 
 
 class Agent:
-    def __init__(self, client, logger, guardrails=None, telemetry=None):
+    def __init__(self, client, logger, guardrails=None, telemetry=None,
+                 tracer=None):
         self.client = client
         self.logger = logger
         self.guardrails = guardrails
         self.telemetry = telemetry
+        self.tracer = tracer
 
     # ---- tool execution -------------------------------------------------
     async def _dispatch(self, name: str, args: dict) -> str:
@@ -84,11 +86,27 @@ class Agent:
                 return json.dumps({
                     "error": decision.message or "tool call blocked by "
                     "guardrails"})
+        from ..otel.tracing import current_span
+
         start = time.monotonic()
+        # explicit `execute_tool <name>` span (reference agent.go:402-429)
+        parent = current_span.get()
+        span = None
+        if self.tracer is not None:
+            span = self.tracer.start_span(f"execute_tool {name}",
+                                          parent=parent, kind="internal")
+            span.set("tool.name", name)
+            token = current_span.set(span)
         try:
             output = await self.client.execute_tool(name, args)
         except Exception as e:
             output = json.dumps({"error": str(e)})
+            if span is not None:
+                span.status_ok = False
+        finally:
+            if span is not None:
+                current_span.reset(token)
+                self.tracer.finish(span)
         if self.telemetry is not None:
             self.telemetry.tool_duration.labels(name, "", "").observe(
                 time.monotonic() - start)

@@ -75,10 +75,15 @@ class MCPServerConnection:
             payload["params"] = params
         if not notification:
             payload["id"] = next(self._rpc_id)
-        headers = {
+        from ..otel.tracing import inject_headers
+
+        # W3C trace context on every MCP hop (reference transport.go:63);
+        # note caller credentials are never forwarded here (the MCP client
+        # builds its own header set — transport.go:59-61)
+        headers = inject_headers({
             "content-type": "application/json",
             "accept": "application/json, text/event-stream",
-        }
+        })
         if self.session_id:
             headers["mcp-session-id"] = self.session_id
         r = await self.client.post(self.active_url, json=payload,

@@ -16,12 +16,12 @@ BYPASS_HEADER = "x-mcp-bypass"
 
 class MCPLayer:
     def __init__(self, cfg, http_client, logger, guardrails=None,
-                 telemetry=None):
+                 telemetry=None, tracer=None):
         self.cfg = cfg
         self.logger = logger
         self.client = MCPClient(cfg, http_client, logger)
         self.agent = Agent(self.client, logger, guardrails=guardrails,
-                           telemetry=telemetry)
+                           telemetry=telemetry, tracer=tracer)
 
     async def initialize(self) -> None:
         await self.client.initialize_all()

@@ -56,7 +56,11 @@ class RemoteProvider:
 
     # ---- auth -----------------------------------------------------------
     def auth_headers(self) -> Dict[str, str]:
-        h = dict(self.meta.extra_headers)
+        # every outbound hop carries the active request's W3C traceparent
+        # (reference core/provider.go:115)
+        from ..otel.tracing import inject_headers
+
+        h = inject_headers(dict(self.meta.extra_headers))
         if self.meta.auth_type == AUTH_BEARER and self.token:
             h["Authorization"] = f"Bearer {self.token}"
         elif self.meta.auth_type == AUTH_XHEADER and self.token:
