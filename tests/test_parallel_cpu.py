@@ -30,9 +30,15 @@ def _run_worker(rank: int, world: int, port: int, fn_name: str, out_dir: str):
         dist.destroy_process_group()
 
 
+_PORT_SEQ = [0]
+
+
 def _spawn(fn_name: str, tmp_path) -> list:
     ctx = mp.get_context("spawn")
-    port = 29600 + (os.getpid() % 500)
+    # unique port per spawn group: reusing one port within a pytest run
+    # can hit TIME_WAIT from the previous group's rendezvous (flaky)
+    _PORT_SEQ[0] += 1
+    port = 29600 + (os.getpid() % 300) + _PORT_SEQ[0] * 17
     procs = [
         ctx.Process(target=_run_worker,
                     args=(r, WORLD, port, fn_name, str(tmp_path)))
