@@ -125,6 +125,10 @@ class EngineConfig:
     max_model_len: int = 8192
     use_hip_graphs: bool = True
     enforce_eager: bool = False
+    # serving mode: capture every decode bucket at engine init so no
+    # client ever pays a ~1 s first-capture stall mid-stream (the bench
+    # path leaves this off — it uses one fixed bucket)
+    capture_at_init: bool = False
     tensor_parallel_size: int = 1
     seed: int = 0
 

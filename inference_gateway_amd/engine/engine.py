@@ -104,6 +104,10 @@ class LLMEngine:
             self.graph_runner = DecodeGraphRunner(
                 self.model, self.kv_caches, cfg, self.device,
                 self.max_blocks_per_seq)
+            if cfg.capture_at_init:
+                with torch.inference_mode():
+                    for b in self.graph_runner.buckets:
+                        self.graph_runner._capture(b)
 
     # ---------------- setup ----------------
     def _auto_num_blocks(self) -> int:

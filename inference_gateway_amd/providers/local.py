@@ -33,6 +33,7 @@ class LocalProvider:
             gpu_memory_utilization=local_cfg.gpu_memory_utilization,
             enforce_eager=local_cfg.enforce_eager,
             num_gpu_blocks=getattr(local_cfg, "num_gpu_blocks", None),
+            capture_at_init=True,
         )
         self.engine = engine or AsyncEngine(ecfg, device=device)
         self.tokenizer = get_tokenizer(
