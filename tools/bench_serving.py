@@ -142,6 +142,52 @@ async def run_local(args):
         await proc.wait()
 
 
+async def run_local_multi(args):
+    import subprocess
+
+    env = dict(os.environ,
+               LOCAL_ENABLED="true",
+               LOCAL_MODEL=args.model,
+               LOCAL_MAX_NUM_SEQS=str(max(args.clients, 8)),
+               LOCAL_MAX_MODEL_LEN="2048",
+               SERVER_PORT=str(args.port))
+    proc = await asyncio.create_subprocess_exec(
+        sys.executable, "-m", "inference_gateway_amd", env=env,
+        cwd=str(Path(__file__).resolve().parent.parent))
+    base = f"http://127.0.0.1:{args.port}"
+    try:
+        async with (await make_client(base, None)) as c:
+            for _ in range(600):
+                try:
+                    if (await c.get("/health")).status_code == 200:
+                        break
+                except Exception:
+                    pass
+                await asyncio.sleep(0.5)
+            else:
+                raise RuntimeError("server never became healthy")
+        per = args.clients // args.client_procs
+        workers = [await asyncio.create_subprocess_exec(
+            sys.executable, __file__, "--worker", "--port", str(args.port),
+            "--model", args.model, "--clients", str(per),
+            "--max-tokens", str(args.max_tokens),
+            "--prompt-chars", str(args.prompt_chars),
+            stdout=subprocess.PIPE)
+            for _ in range(args.client_procs)]
+        t0 = time.perf_counter()
+        outs = await asyncio.gather(*(w.communicate() for w in workers))
+        wall = time.perf_counter() - t0
+        ttfts, tokens = [], []
+        for stdout, _ in outs:
+            d = json.loads(stdout.decode().strip().splitlines()[-1])
+            ttfts.extend(d["ttfts"])
+            tokens.extend(d["tokens"])
+        return ttfts, tokens, wall
+    finally:
+        proc.terminate()
+        await proc.wait()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--mode", choices=("local", "mock"), default="local")
@@ -150,10 +196,29 @@ def main():
     ap.add_argument("--max-tokens", type=int, default=128)
     ap.add_argument("--prompt-chars", type=int, default=512)
     ap.add_argument("--port", type=int, default=18080)
+    ap.add_argument("--client-procs", type=int, default=1,
+                    help="drive the load from N OS processes (a single "
+                         "Python client process saturates before the "
+                         "server at high concurrency)")
+    ap.add_argument("--worker", action="store_true", help=argparse.SUPPRESS)
     args = ap.parse_args()
 
-    runner = run_local if args.mode == "local" else run_mock
-    ttfts, token_counts, wall = asyncio.run(runner(args))
+    if args.worker:
+        # child: drive clients against an already-running server
+        async def child():
+            base = f"http://127.0.0.1:{args.port}"
+            return await drive(base, None, args.clients, args.max_tokens,
+                               args.prompt_chars, f"local/{args.model}")
+
+        t, c, w = asyncio.run(child())
+        print(json.dumps({"ttfts": t, "tokens": c, "wall": w}))
+        return
+
+    if args.mode == "local" and args.client_procs > 1:
+        ttfts, token_counts, wall = asyncio.run(run_local_multi(args))
+    else:
+        runner = run_local if args.mode == "local" else run_mock
+        ttfts, token_counts, wall = asyncio.run(runner(args))
     total_tokens = sum(token_counts)
     print(json.dumps({
         "metric": "http_output_tokens_per_sec",
