@@ -41,17 +41,27 @@ class AsyncEngine:
                 continue
             with self._lock:
                 outputs = self.engine.step()
-            for out in outputs:
-                self._dispatch(out)
+            self._dispatch_batch(outputs)
 
-    def _dispatch(self, out: StepOutput) -> None:
-        q = self._queues.get(out.seq_id)
-        loop = self._loops.get(out.seq_id)
-        if q is None or loop is None:
-            return
-        loop.call_soon_threadsafe(q.put_nowait, out)
-        if out.finished:
-            loop.call_soon_threadsafe(q.put_nowait, self._SENTINEL)
+    def _dispatch_batch(self, outputs) -> None:
+        # ONE event-loop wakeup per step, not one per token: at 256
+        # concurrent streams per-token call_soon_threadsafe wakeups become
+        # a serving bottleneck
+        by_loop: Dict = {}
+        for out in outputs:
+            q = self._queues.get(out.seq_id)
+            loop = self._loops.get(out.seq_id)
+            if q is None or loop is None:
+                continue
+            by_loop.setdefault(loop, []).append((q, out))
+        for loop, items in by_loop.items():
+            loop.call_soon_threadsafe(self._deliver, items)
+
+    def _deliver(self, items) -> None:
+        for q, out in items:
+            q.put_nowait(out)
+            if out.finished:
+                q.put_nowait(self._SENTINEL)
 
     # ---- async API ------------------------------------------------------
     async def generate(self, prompt_token_ids, sampling: SamplingParams

@@ -112,22 +112,21 @@ class LocalProvider:
         created = int(time.time())
         model = f"{self.id}/{self.model_name}"
 
+        # pre-rendered chunk frame: at 256 concurrent streams the
+        # per-token json.dumps shows up in the event loop (the SSE writer
+        # is the serving bottleneck long before the engine is)
+        prefix = ("data: " + json.dumps({
+            "id": cid, "object": "chat.completion.chunk",
+            "created": created, "model": model,
+        }, separators=(",", ":"))[:-1] + ',"choices":[{"index":0,"delta":')
+
         def chunk(delta: dict, finish: Optional[str] = None,
                   usage: Optional[dict] = None) -> str:
-            d = {
-                "id": cid,
-                "object": "chat.completion.chunk",
-                "created": created,
-                "model": model,
-                "choices": [{
-                    "index": 0,
-                    "delta": delta,
-                    "finish_reason": finish,
-                }],
-            }
-            if usage is not None:
-                d["usage"] = usage
-            return "data: " + json.dumps(d, separators=(",", ":"))
+            tail = ',"usage":' + json.dumps(
+                usage, separators=(",", ":")) if usage is not None else ""
+            fin = f'"{finish}"' if finish else "null"
+            return (prefix + json.dumps(delta, separators=(",", ":"))
+                    + ',"finish_reason":' + fin + "}]" + tail + "}")
 
         yield chunk({"role": "assistant", "content": ""})
         ntok = 0
