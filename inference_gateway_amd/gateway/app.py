@@ -207,6 +207,8 @@ class Gateway:
                 self.logger.error("OIDC discovery failed", "error", str(e))
         # async provider validation (reference main.go:352-382, warn-only)
         asyncio.ensure_future(self._validate_providers())
+        if self.tracer is not None and self.tracer.endpoint:
+            self._flusher = asyncio.ensure_future(self.tracer.run_flusher())
 
     async def _validate_providers(self) -> None:
         for pid in self.registry.configured_ids():
@@ -221,6 +223,11 @@ class Gateway:
     async def shutdown(self) -> None:
         if self.mcp is not None:
             await self.mcp.stop()
+        if self.tracer is not None:
+            fl = getattr(self, "_flusher", None)
+            if fl is not None:
+                fl.cancel()
+            await self.tracer.flush()
         try:
             await self.client.aclose()
         except Exception:
