@@ -92,16 +92,6 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
   const int vlane = lane & 15;      // V pass: dims [vlane*8, vlane*8+8)
   const int vtok = lane >> 4;       // V pass: tokens t = vtok + 4*t4
 
-  // this lane's Q segment, packed bf16 (16 dwords per head)
-  uint32_t qpk[G][16];
-#pragma unroll
-  for (int j = 0; j < G; ++j)
-#pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      *reinterpret_cast<uint4*>(&qpk[j][c * 4]) =
-          *reinterpret_cast<const uint4*>(&q_lds[j][seg + c * 8]);
-    }
-
   float m[G], l[G], acc[G][8];
 #pragma unroll
   for (int j = 0; j < G; ++j) {
@@ -129,23 +119,33 @@ __global__ __launch_bounds__(NWAVES* WAVE_SIZE) void paged_attn_decode_kernel(
   if (b0 < nblocks) load_k(b0, kraw);
 
   for (int b = b0; b < nblocks; b += NWAVES) {
-    // ---- scores for all G heads via v_dot2_f32_bf16: Q and K stay PACKED
-    //      (no bf16 unpack VALU at all — the v3 pairwise-unpack loop was
-    //      ~6x the instruction count and made the kernel VALU-bound at
-    //      4.2 TB/s; profiles/r01) ----
+    // ---- scores for all G heads via v_dot2_f32_bf16: Q and K stay
+    //      PACKED (no bf16 unpack VALU — v3's unpack loop made the
+    //      kernel VALU-bound).  Q is re-read from LDS per head per block
+    //      with the head loop NOT unrolled and the source pointer made
+    //      opaque: a persistent (or unroll-duplicated) packed-Q image
+    //      costs 48-64 VGPRs and pins the kernel at 2 waves/SIMD —
+    //      occupancy (PMC: 60% of wave cycles parked) is the limiter,
+    //      not the extra ds_reads ----
     float s[G];
-#pragma unroll
-    for (int j = 0; j < G; ++j) s[j] = 0.0f;
     const bf16x2* kp2 = reinterpret_cast<const bf16x2*>(&kraw[0]);
-    const bf16x2* qp2 = reinterpret_cast<const bf16x2*>(&qpk[0][0]);
+#pragma unroll 1
+    for (int j = 0; j < G; ++j) {
+      const uint16_t* qsrc = &q_lds[j][seg];
+      asm("" : "+v"(qsrc));
+      uint32_t qtmp[16];
 #pragma unroll
-    for (int c = 0; c < 16; ++c) {
-      const bf16x2 kp = kp2[c];
-#pragma unroll
-      for (int j = 0; j < G; ++j) {
-        s[j] = __builtin_amdgcn_fdot2_f32_bf16(qp2[j * 16 + c], kp, s[j],
-                                               false);
+      for (int c4 = 0; c4 < 4; ++c4) {
+        *reinterpret_cast<uint4*>(&qtmp[c4 * 4]) =
+            *reinterpret_cast<const uint4*>(qsrc + c4 * 8);
       }
+      const bf16x2* qp2 = reinterpret_cast<const bf16x2*>(&qtmp[0]);
+      float sj = 0.0f;
+#pragma unroll
+      for (int c = 0; c < 16; ++c) {
+        sj = __builtin_amdgcn_fdot2_f32_bf16(qp2[c], kp2[c], sj, false);
+      }
+      s[j] = sj;
     }
 
     // ---- issue V loads (current block) and K loads (next block) now so
