@@ -8,7 +8,6 @@ tracing/logging -> telemetry -> auth -> guardrails -> MCP -> handler.
 from __future__ import annotations
 
 import asyncio
-import time
 from typing import Callable, List, Optional
 
 from .. import config as config_mod

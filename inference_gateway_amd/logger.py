@@ -4,7 +4,6 @@ development, no-op under test)."""
 from __future__ import annotations
 
 import json
-import logging
 import sys
 import time
 from typing import Any

@@ -6,8 +6,8 @@ Auth types (constants/static.go): bearer | xheader | query | none.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Dict, Optional
+from dataclasses import dataclass
+from typing import Dict
 
 
 @dataclass(frozen=True)
