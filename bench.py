@@ -139,7 +139,7 @@ def main() -> None:
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "global_batch": args.batch * nranks,
+                "global_batch": args.batch * replicas,
                 "seq_len": args.prompt_len,
                 "parallelism": (f"tp{args.tp}" if args.tp else
                                 (f"dp{nranks}" if nranks > 1 else "tp1")),

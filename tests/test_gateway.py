@@ -314,3 +314,31 @@ def test_dev_truncation():
     out = truncate_chat_body(body)
     assert len(out["messages"][0]["content"]) < 300
     assert "5000 bytes" in out["messages"][0]["content"]
+
+
+@pytest.mark.asyncio
+async def test_runtime_context_window_lookups():
+    """llama.cpp /props and Ollama /api/show runtime lookups win over the
+    community table (reference api/context_window.go:28-137)."""
+    def handler(request: httpx.Request) -> httpx.Response:
+        path = request.url.path
+        if path.endswith("/models"):
+            return httpx.Response(200, json={"object": "list", "data": [
+                {"id": "some-gguf", "object": "model"}]})
+        if path.endswith("/props"):
+            return httpx.Response(200, json={
+                "default_generation_settings": {"n_ctx": 4096}})
+        if path.endswith("/api/show"):
+            return httpx.Response(200, json={
+                "model_info": {"llama.context_length": 32768}})
+        return httpx.Response(404)
+
+    cfg = config_mod.load({"LLAMACPP_API_KEY": "t"})  # ollama needs no key
+    client = httpx.AsyncClient(transport=httpx.MockTransport(handler))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client)
+    async with gw_client(gw) as c:
+        r = await c.get("/v1/models?provider=llamacpp&include=context_window")
+        assert r.status_code == 200
+        assert r.json()["data"][0]["context_window"] == 4096
+        r = await c.get("/v1/models?provider=ollama&include=context_window")
+        assert r.json()["data"][0]["context_window"] == 32768
