@@ -72,6 +72,16 @@ class MixtralMoE(nn.Module):
                                     requires_grad=False)
         self.down = nn.Parameter(torch.stack(downs).contiguous(),
                                  requires_grad=False)
+        # K-major copies for the batched-GEMM path: hipBLASLt/Tensile on
+        # ROCm 7.2 has a layout-dependent OOB read with a TRANSPOSED-VIEW
+        # B at [8,512,4096]x[8,4096,28672] (memory faults, reproduced in
+        # isolation — gpurun_out/iso3); contiguous-B runs clean
+        self.gate_up_t = nn.Parameter(
+            self.gate_up.data.transpose(1, 2).contiguous(),
+            requires_grad=False)
+        self.down_t = nn.Parameter(
+            self.down.data.transpose(1, 2).contiguous(),
+            requires_grad=False)
 
     def forward(self, x: torch.Tensor,
                 is_prefill: bool = False) -> torch.Tensor:
@@ -190,19 +200,17 @@ class MixtralMoE(nn.Module):
             # kernel and padding-skipping SwiGLU
             buf = x.new_zeros((E * C, h))
             buf.index_copy_(0, dst, x[flat_token])
-            gu = torch.bmm(buf.view(E, C, h),
-                           self.gate_up.transpose(1, 2))      # [E, C, 2s]
+            gu = torch.bmm(buf.view(E, C, h), self.gate_up_t)  # [E, C, 2s]
             act = ops.silu_mul_grouped(gu.view(E * C, -1), counts, C)
             y = torch.bmm(act.view(E, C, -1),
-                          self.down.transpose(1, 2)).view(E * C, h)
+                          self.down_t).view(E * C, h)
         else:
             buf = x.new_zeros((E * C, h))
             buf.index_copy_(0, dst, x[flat_token])
-            gu = torch.bmm(buf.view(E, C, h),
-                           self.gate_up.transpose(1, 2))      # [E, C, 2s]
+            gu = torch.bmm(buf.view(E, C, h), self.gate_up_t)  # [E, C, 2s]
             act = ops.silu_mul(gu.view(E * C, -1))
             y = torch.bmm(act.view(E, C, -1),
-                          self.down.transpose(1, 2)).view(E * C, h)
+                          self.down_t).view(E * C, h)
         ys = y.index_select(0, dst)                           # [T*K, h]
         out = torch.zeros_like(x)
         out.index_add_(0, flat_token, ys * weights.reshape(-1, 1))
