@@ -34,6 +34,7 @@ class AsyncEngine:
         self._lock = threading.Lock()
         self._wake = threading.Event()
         self._stop = False
+        self._fatal = None
         self._queues: Dict[int, asyncio.Queue] = {}
         self._loops: Dict[int, asyncio.AbstractEventLoop] = {}
         self._thread = threading.Thread(target=self._run, daemon=True,
@@ -42,6 +43,21 @@ class AsyncEngine:
 
     # ---- engine thread --------------------------------------------------
     def _run(self) -> None:
+        try:
+            self._run_loop()
+        except Exception as e:  # engine death must not strand SSE clients
+            self._fatal = e
+            self._close_all_queues()
+
+    def _close_all_queues(self) -> None:
+        with self._lock:
+            items = list(self._queues.items())
+        for rid, q in items:
+            loop = self._loops.get(rid)
+            if loop is not None:
+                loop.call_soon_threadsafe(q.put_nowait, self._SENTINEL)
+
+    def _run_loop(self) -> None:
         while not self._stop:
             with self._lock:
                 has_work = self.engine.scheduler.has_work
@@ -110,3 +126,4 @@ class AsyncEngine:
         self._stop = True
         self._wake.set()
         self._thread.join(timeout=5)
+        self._close_all_queues()
