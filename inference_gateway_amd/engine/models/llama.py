@@ -61,9 +61,6 @@ class LlamaAttention(nn.Module):
                        v_proj.weight.data], 0).contiguous(),
             requires_grad=False)
         del q_proj, k_proj, v_proj
-        self.o_proj = RowParallelLinear(
-            cfg.num_heads * self.hd, h, rank, tp, device, dtype,
-            _seed(f"l{layer}.o"))
         import os
 
         self._nn_gemm = os.environ.get("IGW_NN_GEMM", "0") == "1"
@@ -73,6 +70,9 @@ class LlamaAttention(nn.Module):
                                       requires_grad=False)
             self.o_t = nn.Parameter(self.o_proj.weight.data.t().contiguous(),
                                     requires_grad=False)
+        self.o_proj = RowParallelLinear(
+            cfg.num_heads * self.hd, h, rank, tp, device, dtype,
+            _seed(f"l{layer}.o"))
 
     def forward(self, x: torch.Tensor, fb: ForwardBatch,
                 kv_cache: Tuple[torch.Tensor, torch.Tensor],
