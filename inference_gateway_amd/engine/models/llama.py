@@ -61,15 +61,6 @@ class LlamaAttention(nn.Module):
                        v_proj.weight.data], 0).contiguous(),
             requires_grad=False)
         del q_proj, k_proj, v_proj
-        import os
-
-        self._nn_gemm = os.environ.get("IGW_NN_GEMM", "0") == "1"
-        if self._nn_gemm:
-            # experiment: K-major weights -> hipBLASLt NN instead of TN
-            self.qkv_t = nn.Parameter(self.qkv.data.t().contiguous(),
-                                      requires_grad=False)
-            self.o_t = nn.Parameter(self.o_proj.weight.data.t().contiguous(),
-                                    requires_grad=False)
         self.o_proj = RowParallelLinear(
             cfg.num_heads * self.hd, h, rank, tp, device, dtype,
             _seed(f"l{layer}.o"))
@@ -78,8 +69,7 @@ class LlamaAttention(nn.Module):
                 kv_cache: Tuple[torch.Tensor, torch.Tensor],
                 cos_sin: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]
-        qkv = (torch.matmul(x, self.qkv_t) if self._nn_gemm
-               else F.linear(x, self.qkv))
+        qkv = F.linear(x, self.qkv)
         # strided views into the fused QKV output — the HIP kernels take a
         # row stride, so there are no .contiguous() copies on the hot path
         q, k, v = qkv.split(
@@ -97,12 +87,7 @@ class LlamaAttention(nn.Module):
             o = ops.paged_attn_decode(
                 q.view(T, self.nq, self.hd), kc, vc, fb.block_table,
                 fb.ctx_lens, self.scale)
-        ov = o.view(T, self.nq * self.hd)
-        if self._nn_gemm:
-            from ...parallel.comm import tp_all_reduce
-
-            return tp_all_reduce(torch.matmul(ov, self.o_t))
-        return self.o_proj(ov)
+        return self.o_proj(o.view(T, self.nq * self.hd))
 
 
 class LlamaMLP(nn.Module):
@@ -122,23 +107,8 @@ class LlamaMLP(nn.Module):
         w = torch.cat([self.gate_proj.weight.data, self.up_proj.weight.data], 0)
         self.gate_up = nn.Parameter(w.contiguous(), requires_grad=False)
         del self.gate_proj, self.up_proj
-        import os
-
-        self._nn_gemm = os.environ.get("IGW_NN_GEMM", "0") == "1"
-        if self._nn_gemm:
-            self.gate_up_t = nn.Parameter(self.gate_up.data.t().contiguous(),
-                                          requires_grad=False)
-            self.down_t = nn.Parameter(
-                self.down_proj.weight.data.t().contiguous(),
-                requires_grad=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if self._nn_gemm:
-            from ...parallel.comm import tp_all_reduce
-
-            gu = torch.matmul(x, self.gate_up_t)
-            return tp_all_reduce(
-                torch.matmul(ops.silu_mul(gu), self.down_t))
         gu = F.linear(x, self.gate_up)
         return self.down_proj(ops.silu_mul(gu))
 
