@@ -155,6 +155,11 @@ class DecodeGraphRunner:
         return self.model.compute_logits(hidden)
 
     def _capture(self, bucket: int) -> None:
+        # NOTE: warmup/capture forwards run against LIVE slot state, so
+        # they write K/V for every active slot at its current position —
+        # idempotent by construction (the replay immediately rewrites the
+        # same values at the same slot_mapping), so a mid-serving capture
+        # of a new bucket is safe.
         # warmup on a side stream (records hipBLASLt workspaces etc.)
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())

@@ -91,3 +91,59 @@ def test_graph_decode_across_membership_change():
         eng2.step()
     assert eng.get_output(r1).token_ids == eng2.get_output(r1b).token_ids
     assert eng.get_output(r2).token_ids == eng2.get_output(r2b).token_ids
+
+
+def test_preemption_with_graphs_matches_unconstrained():
+    """Force KV-pool preemption mid-decode under the async hipGraph
+    pipeline: the preempted sequence must resume transparently and
+    produce the same greedy tokens as an unconstrained run."""
+    prompts = [list(range(1, 40)), list(range(50, 90)),
+               list(range(100, 140))]
+    torch.manual_seed(0)
+    eng = LLMEngine(EngineConfig.tiny_llama(use_hip_graphs=True,
+                                            num_gpu_blocks=14,
+                                            max_model_len=128),
+                    device="cuda:0")
+    rids = [eng.add_request(p, SamplingParams(max_new_tokens=24,
+                                              temperature=0.0))
+            for p in prompts]
+    steps = 0
+    while not eng.is_finished():
+        eng.step()
+        steps += 1
+        assert steps < 600
+    constrained = [eng.get_output(r).token_ids for r in rids]
+    for out in constrained:
+        assert len(out) == 24
+
+    torch.manual_seed(0)
+    eng2 = LLMEngine(EngineConfig.tiny_llama(use_hip_graphs=True,
+                                             num_gpu_blocks=256,
+                                             max_model_len=128),
+                     device="cuda:0")
+    rids2 = [eng2.add_request(p, SamplingParams(max_new_tokens=24,
+                                                temperature=0.0))
+             for p in prompts]
+    while not eng2.is_finished():
+        eng2.step()
+    assert constrained == [eng2.get_output(r).token_ids for r in rids2]
+
+
+def test_abort_mid_decode_with_graphs():
+    torch.manual_seed(0)
+    eng = LLMEngine(EngineConfig.tiny_llama(use_hip_graphs=True), "cuda:0")
+    r1 = eng.add_request(list(range(1, 30)),
+                         SamplingParams(max_new_tokens=64, temperature=0.0))
+    r2 = eng.add_request(list(range(40, 70)),
+                         SamplingParams(max_new_tokens=16, temperature=0.0))
+    for _ in range(6):
+        eng.step()
+    eng.abort_request(r1)
+    steps = 0
+    while not eng.is_finished():
+        eng.step()
+        steps += 1
+        assert steps < 200
+    assert eng.get_output(r1).finish_reason == "abort"
+    out2 = eng.get_output(r2)
+    assert out2.finished and len(out2.token_ids) == 16
