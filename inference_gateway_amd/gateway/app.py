@@ -45,11 +45,12 @@ class Gateway:
             self.telemetry = Telemetry()
         self.tracer = None
         if self.cfg.telemetry.tracing_enabled:
-            from ..otel.tracing import Tracer
+            from ..otel.tracing import Tracer, install_client_hooks
 
             self.tracer = Tracer(
                 endpoint=self.cfg.telemetry.tracing_otlp_endpoint,
                 client=self.client)
+            install_client_hooks(self.client, self.tracer)
         self.guardrails = guardrails
         if guardrails is None and self.cfg.guardrails.enabled:
             from ..guardrails.engine import Evaluator

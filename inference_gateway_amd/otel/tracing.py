@@ -200,3 +200,27 @@ def inject_headers(headers: dict) -> dict:
         headers = dict(headers)
         headers["traceparent"] = span.traceparent()
     return headers
+
+
+def install_client_hooks(client, tracer: Tracer) -> None:
+    """Client spans for every outbound HTTP call, named 'METHOD /path'
+    (the reference's otelhttp SpanNameFormatter on the shared client,
+    providers/client/client.go:23-30), parented on the active request
+    span.  Implemented as httpx event hooks so every provider/MCP call
+    is covered centrally."""
+
+    async def on_request(request) -> None:
+        span = tracer.start_span(
+            f"{request.method} {request.url.path}",
+            parent=current_span.get(), kind="client")
+        request.extensions["igw_span"] = span
+
+    async def on_response(response) -> None:
+        span = response.request.extensions.get("igw_span")
+        if span is not None:
+            span.set("http.status_code", response.status_code)
+            span.status_ok = response.status_code < 500
+            tracer.finish(span)
+
+    client.event_hooks.setdefault("request", []).append(on_request)
+    client.event_hooks.setdefault("response", []).append(on_response)

@@ -105,3 +105,25 @@ async def test_execute_tool_span(tmp_path, monkeypatch):
     names = [s.name for s in tracer.drain()]
     assert "execute_tool mcp_write_file" in names
     await mcp.stop()
+
+
+@pytest.mark.asyncio
+async def test_client_spans_on_outbound_calls():
+    """Outbound provider calls get client spans named 'METHOD /path',
+    parented on the server span (otelhttp analog)."""
+    gw, _ = make_traced_gateway()
+    async with gw_client(gw) as c:
+        r = await c.post("/v1/chat/completions",
+                         json={"model": "groq/m1",
+                               "messages": [{"role": "user",
+                                             "content": "x"}]})
+    assert r.status_code == 200
+    spans = gw.tracer.drain()
+    client_spans = [s for s in spans if s.kind == "client"]
+    server = [s for s in spans if s.kind == "server"][0]
+    assert client_spans, [s.name for s in spans]
+    cs = client_spans[-1]
+    assert cs.name.startswith("POST ") and "chat/completions" in cs.name
+    assert cs.trace_id == server.trace_id
+    assert cs.parent_id == server.span_id
+    assert cs.attributes["http.status_code"] == 200
