@@ -202,3 +202,45 @@ def test_moe_route_ref_valid():
         e = int(flat[i])
         r = int(dst[i]) - e * C
         assert 0 <= r < int(counts[e])
+
+
+def test_engine_fuzz_invariants():
+    """Randomized add/step/abort interleaving: the engine must never
+    wedge, every request must terminate, and the block pool must be fully
+    restored (no leaked KV blocks or slots)."""
+    import random
+
+    rng = random.Random(1234)
+    torch.manual_seed(0)
+    eng = _engine(num_gpu_blocks=24, max_model_len=96, max_num_seqs=4)
+    bm = eng.block_manager
+    free0 = bm.num_free
+    slots0 = len(eng.scheduler._free_slots)
+    live, done = [], []
+    for step in range(300):
+        r = rng.random()
+        if r < 0.25 and len(live) < 10:
+            rid = eng.add_request(
+                list(rng.randrange(1, 500) for _ in range(
+                    rng.randrange(4, 40))),
+                SamplingParams(max_new_tokens=rng.randrange(1, 12),
+                               temperature=rng.choice([0.0, 0.8]),
+                               top_k=rng.choice([0, 20]),
+                               top_p=rng.choice([1.0, 0.9])))
+            live.append(rid)
+        elif r < 0.32 and live:
+            victim = live.pop(rng.randrange(len(live)))
+            eng.abort_request(victim)
+            done.append(victim)
+        outs = eng.step()
+        for o in outs:
+            if o.finished and o.seq_id in live:
+                live.remove(o.seq_id)
+                done.append(o.seq_id)
+    while not eng.is_finished():
+        for o in eng.step():
+            if o.finished and o.seq_id in live:
+                live.remove(o.seq_id)
+    assert eng.is_finished()
+    assert bm.num_free == free0, "leaked KV blocks"
+    assert len(eng.scheduler._free_slots) == slots0, "leaked slots"

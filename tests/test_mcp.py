@@ -1,6 +1,7 @@
 """MCP subsystem tests: fake MCP server (JSON-RPC over MockTransport) +
 fake tool-calling provider; agent loop, selector mode, delta accumulation
 (the analog of the reference's tests/mcp_test.go suite)."""
+import asyncio
 import json
 
 import httpx
@@ -290,3 +291,56 @@ def test_tool_filter_semantics():
 def test_sse_fallback_url():
     assert build_sse_fallback_url("http://h/mcp") == "http://h/sse"
     assert build_sse_fallback_url("http://h/x") == "http://h/x/sse"
+
+
+@pytest.mark.asyncio
+async def test_reconnect_recovers_dead_server():
+    """Zero live servers at startup is tolerated when reconnect is on
+    (init.go:64-77); the background loop picks the server up once it
+    returns (init.go:330-408)."""
+    state = {"up": False}
+
+    def handler(request: httpx.Request) -> httpx.Response:
+        if not state["up"]:
+            return httpx.Response(503)
+        body = json.loads(request.content)
+        if body.get("method") == "initialize":
+            return httpx.Response(200, json={
+                "jsonrpc": "2.0", "id": body.get("id"),
+                "result": {"protocolVersion": "2024-11-05",
+                           "capabilities": {"tools": {}},
+                           "serverInfo": {"name": "s", "version": "1"}}})
+        if body.get("method") == "tools/list":
+            return httpx.Response(200, json={
+                "jsonrpc": "2.0", "id": body.get("id"),
+                "result": {"tools": [{"name": "late_tool",
+                                      "description": "d",
+                                      "inputSchema": {"type": "object"}}]}})
+        return httpx.Response(200, json={"jsonrpc": "2.0",
+                                         "id": body.get("id"),
+                                         "result": {}})
+
+    cfg = config_mod.load({
+        "MCP_ENABLED": "true",
+        "MCP_SERVERS": "http://flaky/mcp",
+        "MCP_ENABLE_RECONNECT": "true",
+        "MCP_RECONNECT_INTERVAL": "50ms",
+        "MCP_MAX_RETRIES": "1",
+        "MCP_INITIAL_BACKOFF": "10ms",
+        "MCP_RETRY_INTERVAL": "20ms",
+        "MCP_POLLING_ENABLED": "false",
+    })
+    client = httpx.AsyncClient(transport=httpx.MockTransport(handler))
+    from inference_gateway_amd.mcp.layer import MCPLayer
+
+    mcp = MCPLayer(cfg.mcp, client, noop_logger())
+    await mcp.initialize()          # all servers down: tolerated
+    assert mcp.client.all_tools() == []
+    state["up"] = True
+    for _ in range(60):
+        await asyncio.sleep(0.05)
+        if mcp.client.all_tools():
+            break
+    names = [t["name"] for t in mcp.client.all_tools()]
+    assert "late_tool" in names
+    await mcp.stop()
