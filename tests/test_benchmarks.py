@@ -78,3 +78,68 @@ def test_bench_selector_vs_direct_payload_reduction(capsys):
               f"selector_bytes={selector_bytes} reduction_x={reduction:.1f}")
     assert selector_bytes < direct_bytes
     assert reduction > 5.0
+
+
+@pytest.mark.asyncio
+async def test_bench_agent_loop_overhead(tmp_path, monkeypatch, capsys):
+    """Gateway-side overhead of one MCP agent iteration against the real
+    example filesystem server (BASELINE config 4's plumbing, CPU-only:
+    provider latency mocked to ~0)."""
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).resolve().parent))
+    from test_soak_and_e2e import _load_fs_server
+
+    monkeypatch.setenv("SANDBOX_DIR", str(tmp_path))
+    fs = _load_fs_server()
+    state = {"n": 0}
+
+    def handler(request: httpx.Request) -> httpx.Response:
+        path = request.url.path
+        if "/mcp" in path:
+            return httpx.Response(
+                200, json=fs.handle_rpc(json.loads(request.content)))
+        body = json.loads(request.content)
+        state["n"] += 1
+        if state["n"] % 2 == 1:
+            msg = {"role": "assistant", "content": None, "tool_calls": [
+                {"id": f"c{state['n']}", "type": "function", "function": {
+                    "name": "mcp_list_directory",
+                    "arguments": "{}"}}]}
+            fin = "tool_calls"
+        else:
+            msg = {"role": "assistant", "content": "done"}
+            fin = "stop"
+        return httpx.Response(200, json={
+            "id": "x", "object": "chat.completion", "model": body["model"],
+            "choices": [{"index": 0, "message": msg, "finish_reason": fin}],
+            "usage": {"prompt_tokens": 1, "completion_tokens": 1,
+                      "total_tokens": 2}})
+
+    from inference_gateway_amd import config as config_mod
+    from inference_gateway_amd.gateway.app import Gateway
+    from inference_gateway_amd.mcp.layer import MCPLayer
+
+    cfg = config_mod.load({"GROQ_API_KEY": "t", "MCP_ENABLED": "true",
+                           "MCP_SERVERS": "http://fs/mcp",
+                           "MCP_TOOL_MODE": "direct"})
+    client = httpx.AsyncClient(transport=httpx.MockTransport(handler))
+    mcp = MCPLayer(cfg.mcp, client, noop_logger())
+    await mcp.initialize()
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 mcp_layer=mcp)
+    n = 50
+    async with gw_client(gw) as c:
+        t0 = time.perf_counter()
+        for _ in range(n):
+            r = await c.post("/v1/chat/completions", json={
+                "model": "groq/m1",
+                "messages": [{"role": "user", "content": "ls"}]})
+            assert r.status_code == 200
+        wall = (time.perf_counter() - t0) / n * 1000
+    await mcp.stop()
+    with capsys.disabled():
+        print(f"\n[bench] agent loop (1 tool iteration, real MCP server): "
+              f"{wall:.2f} ms/request")
+    assert wall < 100
