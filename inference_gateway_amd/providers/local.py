@@ -71,10 +71,30 @@ class LocalProvider:
         prompt = apply_chat_template(req.get("messages") or [])
         return self.tokenizer.encode(prompt)
 
+    @staticmethod
+    def _stop_strings(req: dict):
+        stop = req.get("stop")
+        if isinstance(stop, str):
+            return [stop]
+        if isinstance(stop, list):
+            return [s for s in stop if isinstance(s, str)][:4]
+        return []
+
+    @staticmethod
+    def _find_stop(text: str, stops) -> int:
+        """Earliest stop-string match index, or -1."""
+        hit = -1
+        for s in stops:
+            i = text.find(s)
+            if i >= 0 and (hit < 0 or i < hit):
+                hit = i
+        return hit
+
     async def chat_completions(self, req: dict,
                                extra_headers=None) -> dict:
         prompt_ids = self._encode_request(req)
         sampling = self._sampling_from_request(req)
+        stops = self._stop_strings(req)
         token_ids = []
         finish_reason = "stop"
         async for out in self.engine.generate(prompt_ids, sampling):
@@ -82,12 +102,24 @@ class LocalProvider:
             if out.finished:
                 finish_reason = ("stop" if out.finish_reason == "stop"
                                  else "length")
+            if stops:
+                partial = self.tokenizer.decode(token_ids)
+                i = self._find_stop(partial, stops)
+                if i >= 0:
+                    # OpenAI semantics: stop sequence is excluded
+                    text = partial[:i]
+                    return self._completion_response(
+                        prompt_ids, token_ids, text, "stop")
         text = self.tokenizer.decode(token_ids)
-        now = int(time.time())
+        return self._completion_response(prompt_ids, token_ids, text,
+                                         finish_reason)
+
+    def _completion_response(self, prompt_ids, token_ids, text: str,
+                             finish_reason: str) -> dict:
         return {
             "id": f"chatcmpl-{uuid.uuid4().hex[:24]}",
             "object": "chat.completion",
-            "created": now,
+            "created": int(time.time()),
             "model": f"{self.id}/{self.model_name}",
             "choices": [{
                 "index": 0,
@@ -129,6 +161,10 @@ class LocalProvider:
                     + ',"finish_reason":' + fin + "}]" + tail + "}")
 
         yield chunk({"role": "assistant", "content": ""})
+        stops = self._stop_strings(req)
+        holdback = max((len(s) for s in stops), default=1) - 1
+        emitted_tail = ""          # un-emitted text (stop-prefix holdback)
+        stopped = False
         ntok = 0
         finish_reason = "stop"
         pending: list = []
@@ -137,13 +173,31 @@ class LocalProvider:
             pending.append(out.token_id)
             text = self.tokenizer.decode(pending)
             # hold back partial UTF-8 sequences (byte tokenizer)
-            if text.endswith("�"):
+            if text.endswith("\ufffd"):
                 continue
             pending = []
             if out.finished:
                 finish_reason = ("stop" if out.finish_reason == "stop"
                                  else "length")
+            if stops:
+                buf = emitted_tail + text
+                i = self._find_stop(buf, stops)
+                if i >= 0:
+                    # emit up to (excluding) the stop sequence, then end
+                    if buf[:i]:
+                        yield chunk({"content": buf[:i]})
+                    finish_reason = "stop"
+                    stopped = True
+                    break
+                # keep a stop-prefix-sized tail unemitted
+                emit = buf[:-holdback] if holdback else buf
+                emitted_tail = buf[len(emit):]
+                if emit:
+                    yield chunk({"content": emit})
+                continue
             yield chunk({"content": text})
+        if not stopped and emitted_tail:
+            yield chunk({"content": emitted_tail})
         usage = {
             "prompt_tokens": len(prompt_ids),
             "completion_tokens": ntok,

@@ -100,3 +100,49 @@ async def test_local_concurrent_streams(local_gateway):
     async with gw_client(local_gateway) as c:
         results = await asyncio.gather(*(one(c, i) for i in range(4)))
     assert results == [1, 1, 1, 1]
+
+
+@pytest.mark.asyncio
+async def test_stop_strings(local_gateway):
+    """OpenAI `stop` sequences: generation ends at (and excludes) the
+    stop string, in both modes."""
+    # pick a stop string we can force: generate once, reuse a substring
+    async with gw_client(local_gateway) as c:
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hello"}],
+            "max_tokens": 12, "temperature": 0})
+        full = r.json()["choices"][0]["message"]["content"]
+        if len(full) < 4:
+            pytest.skip("output too short to pick a stop substring")
+        stop = full[2:4]
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hello"}],
+            "max_tokens": 12, "temperature": 0, "stop": stop})
+        body = r.json()
+        assert body["choices"][0]["finish_reason"] == "stop"
+        text = body["choices"][0]["message"]["content"]
+        assert stop not in text
+        assert full.startswith(text)
+
+        # streaming: same semantics
+        got = []
+        async with c.stream("POST", "/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hello"}],
+            "max_tokens": 12, "temperature": 0, "stop": stop,
+            "stream": True,
+        }) as resp:
+            async for line in resp.aiter_lines():
+                if not line.startswith("data:"):
+                    continue
+                d = line[5:].strip()
+                if d == "[DONE]":
+                    break
+                delta = json.loads(d)["choices"][0]["delta"]
+                if delta.get("content"):
+                    got.append(delta["content"])
+        stext = "".join(got)
+        assert stop not in stext
+        assert stext == text
