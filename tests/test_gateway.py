@@ -342,3 +342,54 @@ async def test_runtime_context_window_lookups():
         assert r.json()["data"][0]["context_window"] == 4096
         r = await c.get("/v1/models?provider=ollama&include=context_window")
         assert r.json()["data"][0]["context_window"] == 32768
+
+
+@pytest.mark.asyncio
+async def test_responses_passthrough_and_sse():
+    """/v1/responses: byte passthrough (model prefix stripped) in JSON and
+    SSE modes (reference routes.go:1018-1179)."""
+    seen = {}
+
+    def handler(request: httpx.Request) -> httpx.Response:
+        seen["path"] = request.url.path
+        seen["body"] = json.loads(request.content)
+        if seen["body"].get("stream"):
+            chunks = ["event: response.output_text.delta",
+                      'data: {"type":"response.output_text.delta",'
+                      '"delta":"hi"}',
+                      "event: response.completed",
+                      'data: {"type":"response.completed"}']
+            return httpx.Response(
+                200, headers={"content-type": "text/event-stream"},
+                content="\n".join(chunks).encode())
+        return httpx.Response(200, json={"id": "resp_1",
+                                         "object": "response",
+                                         "output": []})
+
+    cfg = config_mod.load({"OPENAI_API_KEY": "sk"})
+    client = httpx.AsyncClient(transport=httpx.MockTransport(handler))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client)
+    async with gw_client(gw) as c:
+        r = await c.post("/v1/responses",
+                         json={"model": "openai/gpt-4o", "input": "hello"})
+        assert r.status_code == 200
+        assert r.json()["object"] == "response"
+        assert seen["path"].endswith("/responses")
+        assert seen["body"]["model"] == "gpt-4o"  # prefix stripped
+
+        lines = []
+        async with c.stream("POST", "/v1/responses", json={
+                "model": "openai/gpt-4o", "input": "hello",
+                "stream": True}) as resp:
+            assert resp.status_code == 200
+            assert "text/event-stream" in resp.headers["content-type"]
+            async for line in resp.aiter_lines():
+                if line:
+                    lines.append(line)
+        assert any("response.completed" in l for l in lines)
+
+        # wrong provider rejected with an OpenAI-shaped error
+        r = await c.post("/v1/responses",
+                         json={"model": "groq/m1", "input": "x"})
+        assert r.status_code == 400
+        assert "error" in r.json()
