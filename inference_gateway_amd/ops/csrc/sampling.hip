@@ -319,8 +319,12 @@ __global__ __launch_bounds__(BLOCK) void sample_bf16_kernel(
     const float* __restrict__ uniforms,    // [B]
     int V) {
   constexpr int NW = BLOCK / WAVE_SIZE;
-  constexpr int REP = 4;  // bin replicas per wave: cuts LDS-atomic
-                          // conflicts on flat logits from ~32- to ~8-way
+  constexpr int REP = 2;  // bin replicas per wave (LDS-atomic conflict
+                          // width vs footprint). Measured: REP=2 (~36 KB,
+                          // 4 blocks/CU) == REP=4 (~68 KB, 2 blocks/CU)
+                          // at B=512 — the histogram passes are
+                          // per-element VALU-bound, not LDS- or
+                          // occupancy-bound; keep the smaller footprint
   __shared__ float redf[NW];
   __shared__ uint32_t ubins[NW][REP][256];
   __shared__ float fbins[NW][REP][256];
