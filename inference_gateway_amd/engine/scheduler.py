@@ -102,8 +102,9 @@ class Scheduler:
                 break
             if not self.bm.can_allocate(seq.prompt_len):
                 break
-            if tokens + seq.prompt_len > self.max_batched_tokens:
-                break  # single over-budget prompt: wait for its own batch
+            # a single prompt longer than max_batched_tokens (legal when
+            # max_model_len > max_batched_tokens) gets its own oversized
+            # batch — breaking unconditionally would starve it forever
             self.waiting.popleft()
             seq.block_ids = self.bm.allocate(seq.prompt_len)
             seq.slot = self._free_slots.pop()

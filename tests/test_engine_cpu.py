@@ -244,3 +244,19 @@ def test_engine_fuzz_invariants():
     assert eng.is_finished()
     assert bm.num_free == free0, "leaked KV blocks"
     assert len(eng.scheduler._free_slots) == slots0, "leaked slots"
+
+
+def test_prompt_longer_than_token_budget_is_not_starved():
+    """A prompt above max_batched_tokens (but under max_model_len) must be
+    admitted as its own oversized prefill batch, not starve forever."""
+    eng = _engine(max_batched_tokens=64, max_model_len=256,
+                  num_gpu_blocks=64)
+    rid = eng.add_request(list(range(1, 101)),   # 100 > 64 budget
+                          SamplingParams(max_new_tokens=4, temperature=0.0))
+    steps = 0
+    while not eng.is_finished():
+        eng.step()
+        steps += 1
+        assert steps < 50, "long prompt starved"
+    out = eng.get_output(rid)
+    assert out.finished and len(out.token_ids) == 4
