@@ -393,3 +393,38 @@ async def test_responses_passthrough_and_sse():
                          json={"model": "groq/m1", "input": "x"})
         assert r.status_code == 400
         assert "error" in r.json()
+
+
+@pytest.mark.asyncio
+async def test_messages_passthrough_sse():
+    """Anthropic /v1/messages streaming: SSE relayed verbatim."""
+    def handler(request: httpx.Request) -> httpx.Response:
+        body = json.loads(request.content)
+        assert body["model"] == "claude-3-5-sonnet"  # prefix stripped
+        if body.get("stream"):
+            frames = ["event: message_start",
+                      'data: {"type":"message_start"}',
+                      "event: content_block_delta",
+                      'data: {"type":"content_block_delta",'
+                      '"delta":{"type":"text_delta","text":"hi"}}',
+                      "event: message_stop",
+                      'data: {"type":"message_stop"}']
+            return httpx.Response(
+                200, headers={"content-type": "text/event-stream"},
+                content="\n".join(frames).encode())
+        return httpx.Response(200, json={"type": "message"})
+
+    cfg = config_mod.load({"ANTHROPIC_API_KEY": "k"})
+    client = httpx.AsyncClient(transport=httpx.MockTransport(handler))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client)
+    async with gw_client(gw) as c:
+        lines = []
+        async with c.stream("POST", "/v1/messages", json={
+                "model": "anthropic/claude-3-5-sonnet",
+                "messages": [], "stream": True}) as r:
+            assert r.status_code == 200
+            async for line in r.aiter_lines():
+                if line:
+                    lines.append(line)
+    assert any("message_stop" in l for l in lines)
+    assert any("text_delta" in l for l in lines)
