@@ -90,11 +90,9 @@ class LocalProvider:
                 hit = i
         return hit
 
-    async def chat_completions(self, req: dict,
-                               extra_headers=None) -> dict:
-        prompt_ids = self._encode_request(req)
-        sampling = self._sampling_from_request(req)
-        stops = self._stop_strings(req)
+    async def _collect(self, prompt_ids, sampling, stops):
+        """One full completion: returns (token_ids, text, finish_reason);
+        stop sequences are excluded from the text (OpenAI semantics)."""
         token_ids = []
         finish_reason = "stop"
         async for out in self.engine.generate(prompt_ids, sampling):
@@ -106,13 +104,40 @@ class LocalProvider:
                 partial = self.tokenizer.decode(token_ids)
                 i = self._find_stop(partial, stops)
                 if i >= 0:
-                    # OpenAI semantics: stop sequence is excluded
-                    text = partial[:i]
-                    return self._completion_response(
-                        prompt_ids, token_ids, text, "stop")
-        text = self.tokenizer.decode(token_ids)
-        return self._completion_response(prompt_ids, token_ids, text,
-                                         finish_reason)
+                    return token_ids, partial[:i], "stop"
+        return token_ids, self.tokenizer.decode(token_ids), finish_reason
+
+    async def chat_completions(self, req: dict,
+                               extra_headers=None) -> dict:
+        prompt_ids = self._encode_request(req)
+        sampling = self._sampling_from_request(req)
+        stops = self._stop_strings(req)
+        n = min(int(req.get("n") or 1), 8)
+        if n <= 1:
+            token_ids, text, finish_reason = await self._collect(
+                prompt_ids, sampling, stops)
+            return self._completion_response(prompt_ids, token_ids, text,
+                                             finish_reason)
+        # n > 1: the engine batches the n sequences continuously
+        results = await asyncio.gather(
+            *(self._collect(prompt_ids, sampling, stops) for _ in range(n)))
+        total = sum(len(t) for t, _, _ in results)
+        return {
+            "id": f"chatcmpl-{uuid.uuid4().hex[:24]}",
+            "object": "chat.completion",
+            "created": int(time.time()),
+            "model": f"{self.id}/{self.model_name}",
+            "choices": [{
+                "index": i,
+                "message": {"role": "assistant", "content": text},
+                "finish_reason": fin,
+            } for i, (_, text, fin) in enumerate(results)],
+            "usage": {
+                "prompt_tokens": len(prompt_ids),
+                "completion_tokens": total,
+                "total_tokens": len(prompt_ids) + total,
+            },
+        }
 
     def _completion_response(self, prompt_ids, token_ids, text: str,
                              finish_reason: str) -> dict:

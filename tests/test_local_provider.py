@@ -146,3 +146,16 @@ async def test_stop_strings(local_gateway):
         stext = "".join(got)
         assert stop not in stext
         assert stext == text
+
+
+@pytest.mark.asyncio
+async def test_n_choices(local_gateway):
+    async with gw_client(local_gateway) as c:
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hello"}],
+            "max_tokens": 5, "temperature": 0.9, "n": 3})
+    body = r.json()
+    assert len(body["choices"]) == 3
+    assert [ch["index"] for ch in body["choices"]] == [0, 1, 2]
+    assert body["usage"]["completion_tokens"] == 15
