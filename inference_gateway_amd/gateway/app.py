@@ -43,6 +43,10 @@ class Gateway:
             from ..otel.metrics import Telemetry
 
             self.telemetry = Telemetry()
+        if (self.telemetry is not None and self.local_provider is not None
+                and hasattr(self.local_provider, "engine")):
+            self.telemetry.register_engine_gauges(
+                self.local_provider.engine.engine)
         self.tracer = None
         if self.cfg.telemetry.tracing_enabled:
             from ..otel.tracing import Tracer, install_client_hooks

@@ -86,5 +86,28 @@ class Telemetry:
         self.server_time_to_first_token.labels(provider, model).observe(
             seconds)
 
+    def register_engine_gauges(self, engine) -> None:
+        """Scrape-time gauges over the local engine's scheduler/KV state
+        (the serving-engine analog of the reference's process-level
+        footprint claims): running/waiting sequences and KV-block
+        utilization."""
+        from prometheus_client import Gauge
+
+        r = self.registry
+        g_run = Gauge("igw_engine_running_seqs",
+                      "sequences in the running batch", registry=r)
+        g_wait = Gauge("igw_engine_waiting_seqs",
+                       "sequences queued for prefill", registry=r)
+        g_free = Gauge("igw_engine_kv_blocks_free",
+                       "free paged-KV blocks", registry=r)
+        g_total = Gauge("igw_engine_kv_blocks_total",
+                        "total paged-KV blocks", registry=r)
+        sched = engine.scheduler
+        bm = engine.block_manager
+        g_run.set_function(lambda: len(sched.running))
+        g_wait.set_function(lambda: len(sched.waiting))
+        g_free.set_function(lambda: bm.num_free)
+        g_total.set_function(lambda: bm.num_blocks - 1)  # block 0 reserved
+
     def exposition(self) -> bytes:
         return generate_latest(self.registry)

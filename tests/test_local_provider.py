@@ -159,3 +159,24 @@ async def test_n_choices(local_gateway):
     assert len(body["choices"]) == 3
     assert [ch["index"] for ch in body["choices"]] == [0, 1, 2]
     assert body["usage"]["completion_tokens"] == 15
+
+
+@pytest.mark.asyncio
+async def test_engine_gauges_in_metrics_exposition():
+    from inference_gateway_amd.otel.metrics import Telemetry
+
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=4,
+                     max_model_len=256)
+    lc.num_gpu_blocks = 64
+    provider = LocalProvider(lc, device="cpu")
+    cfg = config_mod.load({"TELEMETRY_ENABLED": "true"})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    telemetry = Telemetry()
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider, telemetry=telemetry)
+    assert gw is not None
+    text = telemetry.exposition().decode()
+    assert "igw_engine_kv_blocks_total 63.0" in text
+    assert "igw_engine_running_seqs 0.0" in text
+    provider.engine.shutdown()
