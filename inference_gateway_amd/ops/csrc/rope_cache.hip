@@ -22,24 +22,39 @@ __global__ void rope_kernel(uint16_t* __restrict__ q, uint16_t* __restrict__ k,
                             const float* __restrict__ cos_sin, int nq, int nkv,
                             int hd, int qs, int ks) {
   // qs/ks: row strides in elements — q/k may be views into a fused-QKV
-  // GEMM output (no .contiguous() copies on the decode hot path)
+  // GEMM output (no .contiguous() copies on the decode hot path).
+  // Vectorized 8 pairs per thread: the rotate-half halves are each
+  // contiguous, so x1/x2 load as ushort8 and cos/sin as two float4s
+  // (scalar 2-B pair loads ran at 30% of the HBM roofline).
   const int t = blockIdx.x;
   const int pos = positions[t];
   const float* cs = cos_sin + (int64_t)pos * hd;
   const int half = hd / 2;
   const int nheads = nq + nkv;
-  // each thread handles one (head, rotary pair)
-  for (int i = threadIdx.x; i < nheads * half; i += blockDim.x) {
-    const int h = i / half;
-    const int d = i % half;
+  const int groups = half >> 3;       // 8 pairs per group (hd % 16 == 0)
+  for (int i = threadIdx.x; i < nheads * groups; i += blockDim.x) {
+    const int h = i / groups;
+    const int d0 = (i % groups) * 8;
     uint16_t* base = (h < nq) ? q + (int64_t)t * qs + h * hd
                               : k + (int64_t)t * ks + (h - nq) * hd;
-    const float c = cs[d];
-    const float s = cs[half + d];
-    const float x1 = bf16_to_f32(base[d]);
-    const float x2 = bf16_to_f32(base[half + d]);
-    base[d] = f32_to_bf16(x1 * c - x2 * s);
-    base[half + d] = f32_to_bf16(x2 * c + x1 * s);
+    const ushort8 v1 = *reinterpret_cast<const ushort8*>(base + d0);
+    const ushort8 v2 = *reinterpret_cast<const ushort8*>(base + half + d0);
+    const float4 c0 = *reinterpret_cast<const float4*>(cs + d0);
+    const float4 c1 = *reinterpret_cast<const float4*>(cs + d0 + 4);
+    const float4 s0 = *reinterpret_cast<const float4*>(cs + half + d0);
+    const float4 s1 = *reinterpret_cast<const float4*>(cs + half + d0 + 4);
+    const float cv[8] = {c0.x, c0.y, c0.z, c0.w, c1.x, c1.y, c1.z, c1.w};
+    const float sv[8] = {s0.x, s0.y, s0.z, s0.w, s1.x, s1.y, s1.z, s1.w};
+    ushort8 o1, o2;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float x1 = bf16_to_f32(v1.v[j]);
+      const float x2 = bf16_to_f32(v2.v[j]);
+      o1.v[j] = f32_to_bf16(x1 * cv[j] - x2 * sv[j]);
+      o2.v[j] = f32_to_bf16(x2 * cv[j] + x1 * sv[j]);
+    }
+    *reinterpret_cast<ushort8*>(base + d0) = o1;
+    *reinterpret_cast<ushort8*>(base + half + d0) = o2;
   }
 }
 
@@ -76,6 +91,7 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(positions.dtype() == torch::kInt32);
   TORCH_CHECK(cos_sin_cache.dtype() == torch::kFloat32);
   TORCH_CHECK(q.stride(-1) == 1 && k.stride(-1) == 1);
+  TORCH_CHECK(head_dim % 16 == 0, "rope: head_dim must be 16-aligned");
   const int T = q.size(0);
   const int hd = head_dim;
   const int nq = q.size(-1) / hd;
