@@ -35,7 +35,6 @@ class Sequence:
     finish_reason: Optional[str] = None
     first_token_time: Optional[float] = None
     slot: Optional[int] = None        # persistent batch slot (graph runner)
-    synced_blocks: int = 0            # device block-table rows written so far
     in_flight: bool = False           # a decode step's token not yet drained
 
     @property
@@ -108,7 +107,6 @@ class Scheduler:
             self.waiting.popleft()
             seq.block_ids = self.bm.allocate(seq.prompt_len)
             seq.slot = self._free_slots.pop()
-            seq.synced_blocks = 0
             seq.state = SeqState.RUNNING
             prefills.append(seq)
             tokens += seq.prompt_len
