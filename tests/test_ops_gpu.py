@@ -259,3 +259,17 @@ def test_moe_padded_forward_gpu_matches_sorted():
     b = moe._forward_sorted(x).float()
     assert torch.allclose(a, b, atol=5e-2, rtol=5e-2), \
         f"max diff {(a - b).abs().max()}"
+
+
+def test_native_extension_is_loaded_and_in_tree():
+    """On a GPU box the HIP extension must be the in-tree build — GPU
+    tests passing on a silent eager fallback would be meaningless."""
+    from pathlib import Path
+
+    import inference_gateway_amd.ops as ops_pkg
+
+    assert ops_pkg.HAS_KERNELS, "in-tree _kernels extension not loaded"
+    so = Path(ops_pkg._kernels.__file__).resolve()
+    pkg = Path(ops_pkg.__file__).resolve().parent
+    assert so.parent == pkg, f"extension loaded from {so}, not in-tree"
+    assert torch.version.hip, "expected a ROCm torch build"
