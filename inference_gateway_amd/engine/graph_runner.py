@@ -37,7 +37,7 @@ from .models.batch import ForwardBatch
 from .scheduler import Sequence
 
 _BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256,
-            320, 384, 448, 512]
+            320, 384, 448, 512, 640, 768, 896, 1024]
 
 
 class DecodeGraphRunner:
