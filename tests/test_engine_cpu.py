@@ -260,3 +260,16 @@ def test_prompt_longer_than_token_budget_is_not_starved():
         assert steps < 50, "long prompt starved"
     out = eng.get_output(rid)
     assert out.finished and len(out.token_ids) == 4
+
+
+def test_bucket_ladder_invariants():
+    """Bucket ladder: strictly increasing, and from 16 up each step grows
+    by <= 1.5x so graph-padding waste is bounded (a 600-seq batch must not
+    pad to 1024)."""
+    from inference_gateway_amd.engine.graph_runner import _BUCKETS
+
+    assert _BUCKETS == sorted(set(_BUCKETS))
+    assert _BUCKETS[0] == 1 and _BUCKETS[-1] == 1024
+    for a, b in zip(_BUCKETS, _BUCKETS[1:]):
+        if a >= 16:
+            assert b <= a * 1.5 + 1e-9, (a, b)
