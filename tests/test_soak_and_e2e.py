@@ -154,3 +154,36 @@ async def test_mcp_agent_against_real_filesystem_server(tmp_path,
     # the file really exists in the sandbox
     assert (tmp_path / "note.txt").read_text() == "hello-mi355x"
     await mcp.stop()
+
+
+def test_search_server_rpc_roundtrip():
+    """The example search MCP server answers list/call RPCs correctly."""
+    spec = importlib.util.spec_from_file_location(
+        "search_server", ROOT / "examples" / "mcp-servers" /
+        "search_server.py")
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+
+    tools = mod.handle_rpc({"jsonrpc": "2.0", "id": 1,
+                            "method": "tools/list"})["result"]["tools"]
+    assert {t["name"] for t in tools} == {"search", "get_document"}
+
+    hits = mod.handle_rpc({
+        "jsonrpc": "2.0", "id": 2, "method": "tools/call",
+        "params": {"name": "search",
+                   "arguments": {"query": "paged kv block"}}})["result"]
+    assert not hits["isError"]
+    ranked = json.loads(hits["content"][0]["text"])
+    assert ranked and ranked[0]["id"] == "paged-kv-cache"
+
+    doc = mod.handle_rpc({
+        "jsonrpc": "2.0", "id": 3, "method": "tools/call",
+        "params": {"name": "get_document",
+                   "arguments": {"id": ranked[0]["id"]}}})["result"]
+    assert "block table" in doc["content"][0]["text"]
+
+    missing = mod.handle_rpc({
+        "jsonrpc": "2.0", "id": 4, "method": "tools/call",
+        "params": {"name": "get_document",
+                   "arguments": {"id": "nope"}}})["result"]
+    assert missing["isError"]
