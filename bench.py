@@ -111,9 +111,12 @@ def main() -> None:
     elapsed = time.perf_counter() - t0
     comm.world_barrier()
 
-    # MAX over ranks (the slowest rank defines the job's step time)
+    # MAX over ranks (the slowest rank defines the job's step time).
+    # NOTE: the tensor must live on the GPU under the nccl/RCCL backend —
+    # a CPU tensor raises and would kill the run after the timed region.
     if nranks > 1:
-        t = torch.tensor([elapsed])
+        red_dev = device if torch.cuda.is_available() else "cpu"
+        t = torch.tensor([elapsed], device=red_dev)
         import torch.distributed as dist
 
         dist.all_reduce(t, op=dist.ReduceOp.MAX)

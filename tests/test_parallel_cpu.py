@@ -145,3 +145,31 @@ def test_ep2_moe_matches_single_rank(tmp_path):
     ref = _single_rank_reference(_ep_moe_worker)
     assert (abs(outs[0] - ref) < 1e-4).all(), \
         "EP=2 all-to-all MoE must match the single-rank MoE"
+
+
+def test_bench_contract_torchrun_world2(tmp_path):
+    """The driver's SCALE contract end to end: bench.py under
+    torch.distributed.run with 2 ranks (gloo on CPU) prints exactly ONE
+    JSON line from rank 0 with the whole-job aggregate — covers the
+    rank-reduction path that must not crash after the timed region."""
+    import json
+    import subprocess
+    import sys
+
+    port = 29900 + (os.getpid() % 97)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--batch", "4",
+         "--prompt-len", "32", "--model", "tiny-llama"],
+        capture_output=True, text=True, timeout=600,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["global_batch"] == 8   # 4 per rank x 2 replicas
+    assert rec["value"] > 0
