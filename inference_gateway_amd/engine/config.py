@@ -112,6 +112,9 @@ class SamplingParams:
     top_p: float = 1.0      # >=1 disables
     stop_token_ids: tuple = ()
     ignore_eos: bool = False
+    # OpenAI logprobs: -1 = off; >=0 = return the sampled token's logprob
+    # plus that many top alternatives per step (capped at 20, like OpenAI)
+    logprobs: int = -1
 
 
 @dataclass

@@ -29,6 +29,10 @@ class StepOutput:
     token_id: int
     finished: bool
     finish_reason: Optional[str] = None
+    # set when SamplingParams.logprobs >= 0: log p(sampled token) and the
+    # top-k alternatives as [(token_id, logprob), ...] (k may be 0)
+    logprob: Optional[float] = None
+    top_logprobs: Optional[list] = None
 
 
 @dataclass
@@ -37,6 +41,39 @@ class RequestOutput:
     token_ids: List[int] = field(default_factory=list)
     finished: bool = False
     finish_reason: Optional[str] = None
+
+
+MAX_TOP_LOGPROBS = 20      # OpenAI's cap on top_logprobs
+
+
+def _extract_logprobs(logits, seqs, tokens):
+    """For every batch row whose sequence requested logprobs, compute
+    (log p(sampled token), [(token_id, logprob) x k]) from this step's
+    logits.  Returns {row_index: (...)} or None when nobody asked — the
+    hot path pays nothing."""
+    rows = [i for i, s in enumerate(seqs) if s.sampling.logprobs >= 0]
+    if not rows:
+        return None
+    import torch as _t
+
+    dev = logits.device
+    rows_t = _t.tensor(rows, dtype=_t.int64, device=dev)
+    lsm = logits.index_select(0, rows_t).float().log_softmax(-1)
+    toks = _t.tensor([tokens[i] for i in rows], dtype=_t.int64, device=dev)
+    samp = lsm.gather(1, toks.unsqueeze(1)).squeeze(1)
+    kmax = min(MAX_TOP_LOGPROBS,
+               max(seqs[i].sampling.logprobs for i in rows))
+    if kmax > 0:
+        top_v, top_i = lsm.topk(kmax, dim=-1)
+        top_v, top_i = top_v.tolist(), top_i.tolist()
+    samp = samp.tolist()
+    out = {}
+    for n, i in enumerate(rows):
+        k = min(seqs[i].sampling.logprobs, MAX_TOP_LOGPROBS)
+        alts = (list(zip(top_i[n][:k], top_v[n][:k])) if kmax > 0 and k > 0
+                else [])
+        out[i] = (samp[n], alts)
+    return out
 
 
 def _maybe_load_tuned_gemms() -> None:
@@ -199,7 +236,8 @@ class LLMEngine:
         hidden_last = hidden[fb.last_token_indices]
         logits = self.model.compute_logits(hidden_last)
         tokens = self.sampler.sample(logits, batch.seqs)
-        outputs = self._append_tokens(batch, tokens)
+        lp_map = _extract_logprobs(logits, batch.seqs, tokens)
+        outputs = self._append_tokens(batch, tokens, lp_map)
         if batch.is_prefill and gr is not None:
             # seed the persistent decode slot state for newly admitted seqs
             for seq, tok in zip(batch.seqs, tokens):
@@ -215,8 +253,9 @@ class LLMEngine:
         self._pending = None
         ev.synchronize()
         tokens = self.graph_runner.tokens_pinned[ring, : len(seqs)].tolist()
+        lp_map = self.graph_runner.read_logprobs(ring)
         outputs = []
-        for seq, tok in zip(seqs, tokens):
+        for i, (seq, tok) in enumerate(zip(seqs, tokens)):
             seq.in_flight = False
             if seq.state == SeqState.FINISHED:
                 continue  # stop applied while this token was in flight
@@ -225,7 +264,8 @@ class LLMEngine:
                 # the rebuilt prompt so the request resumes transparently
                 seq.prompt_token_ids.append(tok)
                 continue
-            outputs.extend(self._apply_token(seq, tok))
+            outputs.extend(self._apply_token(
+                seq, tok, lp_map.get(i) if lp_map else None))
         return outputs
 
     def _build_forward_batch(self, batch: ScheduledBatch) -> ForwardBatch:
@@ -284,7 +324,7 @@ class LLMEngine:
             last_token_indices=torch.arange(B, dtype=torch.int64, device=dev),
         )
 
-    def _apply_token(self, seq, tok: int) -> List[StepOutput]:
+    def _apply_token(self, seq, tok: int, lp=None) -> List[StepOutput]:
         import time
 
         seq.output_token_ids.append(tok)
@@ -300,11 +340,15 @@ class LLMEngine:
             reason = "length"
         if reason:
             self.scheduler.finish(seq, reason)
-        return [StepOutput(seq.seq_id, tok, reason is not None, reason)]
+        out = StepOutput(seq.seq_id, tok, reason is not None, reason)
+        if lp is not None:
+            out.logprob, out.top_logprobs = lp
+        return [out]
 
-    def _append_tokens(self, batch: ScheduledBatch,
-                       tokens: List[int]) -> List[StepOutput]:
+    def _append_tokens(self, batch: ScheduledBatch, tokens: List[int],
+                       lp_map=None) -> List[StepOutput]:
         outputs = []
-        for seq, tok in zip(batch.seqs, tokens):
-            outputs.extend(self._apply_token(seq, tok))
+        for i, (seq, tok) in enumerate(zip(batch.seqs, tokens)):
+            outputs.extend(self._apply_token(
+                seq, tok, lp_map.get(i) if lp_map else None))
         return outputs

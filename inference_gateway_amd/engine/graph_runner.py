@@ -78,6 +78,12 @@ class DecodeGraphRunner:
                                          pin_memory=True)
         self._events = [torch.cuda.Event(), torch.cuda.Event()]
         self._ring = 0
+        # logprobs readback (lazy — only requests with logprobs pay):
+        # pinned ring buffers + per-ring (rows, kmax, per-row k) metadata
+        self._lp_meta: List = [None, None]
+        self.lp_pinned = None
+        self.lp_topv_pinned = None
+        self.lp_topi_pinned = None
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._logits: Dict[int, torch.Tensor] = {}
         self._pool = None
@@ -210,6 +216,66 @@ class DecodeGraphRunner:
         self.slot_pos.index_add_(0, act, self._ones_i32[:B])
         ring = self._ring
         self._ring ^= 1
+        self._queue_logprobs(ring, seqs, logits, toks)
         self.tokens_pinned[ring, :B].copy_(toks[:B], non_blocking=True)
         self._events[ring].record()
         return ring, self._events[ring]
+
+    # ---- logprobs (opt-in per sequence; the hot path pays nothing) ------
+    _LP_KMAX = 20        # OpenAI's top_logprobs cap
+
+    def _queue_logprobs(self, ring: int, seqs, logits, toks) -> None:
+        """For rows whose sequence asked for logprobs, compute them from
+        this step's logits (still valid: the next replay of this bucket is
+        stream-ordered after these ops) and queue an async D2H alongside
+        the token readback; read_logprobs(ring) decodes after the event."""
+        rows = [i for i, s in enumerate(seqs) if s.sampling.logprobs >= 0]
+        self._lp_meta[ring] = None
+        if not rows:
+            return
+        if self.lp_pinned is None:
+            bmax = self.tokens_pinned.shape[1]
+            self.lp_pinned = torch.zeros((2, bmax), dtype=torch.float32,
+                                         pin_memory=True)
+            self.lp_topv_pinned = torch.zeros((2, bmax, self._LP_KMAX),
+                                              dtype=torch.float32,
+                                              pin_memory=True)
+            self.lp_topi_pinned = torch.zeros((2, bmax, self._LP_KMAX),
+                                              dtype=torch.int32,
+                                              pin_memory=True)
+        n = len(rows)
+        rows_t = torch.tensor(rows, dtype=torch.int64).to(self.device,
+                                                          non_blocking=True)
+        lsm = logits.index_select(0, rows_t).float().log_softmax(-1)
+        samp = lsm.gather(1, toks.index_select(0, rows_t).to(
+            torch.int64).unsqueeze(1)).squeeze(1)
+        ks = [min(seqs[i].sampling.logprobs, self._LP_KMAX) for i in rows]
+        kmax = max(ks)
+        if kmax > 0:
+            top_v, top_i = lsm.topk(kmax, dim=-1)
+            self.lp_topv_pinned[ring, :n, :kmax].copy_(top_v,
+                                                       non_blocking=True)
+            self.lp_topi_pinned[ring, :n, :kmax].copy_(
+                top_i.to(torch.int32), non_blocking=True)
+        self.lp_pinned[ring, :n].copy_(samp, non_blocking=True)
+        self._lp_meta[ring] = (rows, kmax, ks)
+
+    def read_logprobs(self, ring: int):
+        """{batch_row: (sampled_logprob, [(token_id, logprob) x k])} for
+        the drained step, or None.  Call only after the ring's event."""
+        meta = self._lp_meta[ring]
+        if meta is None:
+            return None
+        self._lp_meta[ring] = None
+        rows, kmax, ks = meta
+        n = len(rows)
+        samp = self.lp_pinned[ring, :n].tolist()
+        if kmax > 0:
+            tv = self.lp_topv_pinned[ring, :n, :kmax].tolist()
+            ti = self.lp_topi_pinned[ring, :n, :kmax].tolist()
+        out = {}
+        for j, i in enumerate(rows):
+            alts = (list(zip(ti[j][: ks[j]], tv[j][: ks[j]]))
+                    if kmax > 0 and ks[j] > 0 else [])
+            out[i] = (samp[j], alts)
+        return out

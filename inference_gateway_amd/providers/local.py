@@ -59,13 +59,29 @@ class LocalProvider:
         stop_ids = ()
         if self.tokenizer.eos_token_id is not None:
             stop_ids = (self.tokenizer.eos_token_id,)
+        # OpenAI logprobs: bool `logprobs` + optional `top_logprobs` 0..20
+        lp = -1
+        if req.get("logprobs"):
+            lp = max(0, min(20, int(req.get("top_logprobs") or 0)))
         return SamplingParams(
             max_new_tokens=int(max_tokens),
             temperature=float(req.get("temperature", 1.0)),
             top_p=float(req.get("top_p", 1.0)),
             top_k=int(req.get("top_k", 0)),
             stop_token_ids=stop_ids,
+            logprobs=lp,
         )
+
+    def _lp_item(self, out) -> dict:
+        """One OpenAI logprobs content entry for a generated token."""
+        s = self.tokenizer.decode([out.token_id])
+        tops = []
+        for tid, v in (out.top_logprobs or []):
+            ts = self.tokenizer.decode([tid])
+            tops.append({"token": ts, "logprob": v,
+                         "bytes": list(ts.encode("utf-8"))})
+        return {"token": s, "logprob": out.logprob,
+                "bytes": list(s.encode("utf-8")), "top_logprobs": tops}
 
     def _encode_request(self, req: dict):
         prompt = apply_chat_template(req.get("messages") or [])
@@ -91,12 +107,17 @@ class LocalProvider:
         return hit
 
     async def _collect(self, prompt_ids, sampling, stops):
-        """One full completion: returns (token_ids, text, finish_reason);
-        stop sequences are excluded from the text (OpenAI semantics)."""
+        """One full completion: returns (token_ids, text, finish_reason,
+        lp_content); stop sequences are excluded from the text (OpenAI
+        semantics; logprob entries for tokens past a stop-string cut are
+        kept — the text cut falls mid-token in general)."""
         token_ids = []
+        lp_content = [] if sampling.logprobs >= 0 else None
         finish_reason = "stop"
         async for out in self.engine.generate(prompt_ids, sampling):
             token_ids.append(out.token_id)
+            if lp_content is not None:
+                lp_content.append(self._lp_item(out))
             if out.finished:
                 finish_reason = ("stop" if out.finish_reason == "stop"
                                  else "length")
@@ -104,8 +125,9 @@ class LocalProvider:
                 partial = self.tokenizer.decode(token_ids)
                 i = self._find_stop(partial, stops)
                 if i >= 0:
-                    return token_ids, partial[:i], "stop"
-        return token_ids, self.tokenizer.decode(token_ids), finish_reason
+                    return token_ids, partial[:i], "stop", lp_content
+        return (token_ids, self.tokenizer.decode(token_ids), finish_reason,
+                lp_content)
 
     async def chat_completions(self, req: dict,
                                extra_headers=None) -> dict:
@@ -114,14 +136,14 @@ class LocalProvider:
         stops = self._stop_strings(req)
         n = min(int(req.get("n") or 1), 8)
         if n <= 1:
-            token_ids, text, finish_reason = await self._collect(
+            token_ids, text, finish_reason, lp = await self._collect(
                 prompt_ids, sampling, stops)
             return self._completion_response(prompt_ids, token_ids, text,
-                                             finish_reason)
+                                             finish_reason, lp)
         # n > 1: the engine batches the n sequences continuously
         results = await asyncio.gather(
             *(self._collect(prompt_ids, sampling, stops) for _ in range(n)))
-        total = sum(len(t) for t, _, _ in results)
+        total = sum(len(t) for t, _, _, _ in results)
         return {
             "id": f"chatcmpl-{uuid.uuid4().hex[:24]}",
             "object": "chat.completion",
@@ -130,8 +152,9 @@ class LocalProvider:
             "choices": [{
                 "index": i,
                 "message": {"role": "assistant", "content": text},
+                "logprobs": ({"content": lp} if lp is not None else None),
                 "finish_reason": fin,
-            } for i, (_, text, fin) in enumerate(results)],
+            } for i, (_, text, fin, lp) in enumerate(results)],
             "usage": {
                 "prompt_tokens": len(prompt_ids),
                 "completion_tokens": total,
@@ -140,7 +163,7 @@ class LocalProvider:
         }
 
     def _completion_response(self, prompt_ids, token_ids, text: str,
-                             finish_reason: str) -> dict:
+                             finish_reason: str, lp_content=None) -> dict:
         return {
             "id": f"chatcmpl-{uuid.uuid4().hex[:24]}",
             "object": "chat.completion",
@@ -149,6 +172,8 @@ class LocalProvider:
             "choices": [{
                 "index": 0,
                 "message": {"role": "assistant", "content": text},
+                "logprobs": ({"content": lp_content}
+                             if lp_content is not None else None),
                 "finish_reason": finish_reason,
             }],
             "usage": {
@@ -178,12 +203,15 @@ class LocalProvider:
         }, separators=(",", ":"))[:-1] + ',"choices":[{"index":0,"delta":')
 
         def chunk(delta: dict, finish: Optional[str] = None,
-                  usage: Optional[dict] = None) -> str:
+                  usage: Optional[dict] = None, lp=None) -> str:
             tail = ',"usage":' + json.dumps(
                 usage, separators=(",", ":")) if usage is not None else ""
             fin = f'"{finish}"' if finish else "null"
+            lps = (',"logprobs":' + json.dumps({"content": lp},
+                                               separators=(",", ":"))
+                   if lp is not None else "")
             return (prefix + json.dumps(delta, separators=(",", ":"))
-                    + ',"finish_reason":' + fin + "}]" + tail + "}")
+                    + lps + ',"finish_reason":' + fin + "}]" + tail + "}")
 
         yield chunk({"role": "assistant", "content": ""})
         stops = self._stop_strings(req)
@@ -193,14 +221,19 @@ class LocalProvider:
         ntok = 0
         finish_reason = "stop"
         pending: list = []
+        want_lp = sampling.logprobs >= 0
+        pending_lp: list = []      # lp items for not-yet-flushed tokens
         async for out in self.engine.generate(prompt_ids, sampling):
             ntok += 1
             pending.append(out.token_id)
+            if want_lp:
+                pending_lp.append(self._lp_item(out))
             text = self.tokenizer.decode(pending)
             # hold back partial UTF-8 sequences (byte tokenizer)
             if text.endswith("\ufffd"):
                 continue
             pending = []
+            lp, pending_lp = (pending_lp, []) if want_lp else (None, [])
             if out.finished:
                 finish_reason = ("stop" if out.finish_reason == "stop"
                                  else "length")
@@ -210,7 +243,7 @@ class LocalProvider:
                 if i >= 0:
                     # emit up to (excluding) the stop sequence, then end
                     if buf[:i]:
-                        yield chunk({"content": buf[:i]})
+                        yield chunk({"content": buf[:i]}, lp=lp)
                     finish_reason = "stop"
                     stopped = True
                     break
@@ -218,9 +251,9 @@ class LocalProvider:
                 emit = buf[:-holdback] if holdback else buf
                 emitted_tail = buf[len(emit):]
                 if emit:
-                    yield chunk({"content": emit})
+                    yield chunk({"content": emit}, lp=lp)
                 continue
-            yield chunk({"content": text})
+            yield chunk({"content": text}, lp=lp)
         if not stopped and emitted_tail:
             yield chunk({"content": emitted_tail})
         usage = {

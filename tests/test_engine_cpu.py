@@ -273,3 +273,47 @@ def test_bucket_ladder_invariants():
     for a, b in zip(_BUCKETS, _BUCKETS[1:]):
         if a >= 16:
             assert b <= a * 1.5 + 1e-9, (a, b)
+
+
+def test_logprobs_eager_greedy():
+    """logprobs: greedy sampling must report the argmax as both the
+    sampled token's logprob and top-1; alternatives sorted descending;
+    sequences that did not ask must get None."""
+    eng = _engine()
+    rid_lp = eng.add_request(
+        list(range(1, 17)),
+        SamplingParams(max_new_tokens=4, temperature=0.0, logprobs=3))
+    rid_plain = eng.add_request(
+        list(range(2, 18)),
+        SamplingParams(max_new_tokens=4, temperature=0.0))
+    outs = []
+    while not eng.is_finished():
+        outs.extend(eng.step())
+    by_seq = {}
+    for o in outs:
+        by_seq.setdefault(o.seq_id, []).append(o)
+    lp_outs = by_seq[rid_lp]
+    assert len(lp_outs) == 4
+    for o in lp_outs:
+        assert o.logprob is not None and o.logprob <= 0.0
+        assert len(o.top_logprobs) == 3
+        vals = [v for _, v in o.top_logprobs]
+        assert vals == sorted(vals, reverse=True)
+        # greedy: the sampled token IS the top-1 alternative
+        assert o.top_logprobs[0][0] == o.token_id
+        assert abs(o.top_logprobs[0][1] - o.logprob) < 1e-5
+    for o in by_seq[rid_plain]:
+        assert o.logprob is None and o.top_logprobs is None
+
+
+def test_logprobs_k_zero_returns_sampled_only():
+    eng = _engine()
+    rid = eng.add_request(
+        list(range(1, 17)),
+        SamplingParams(max_new_tokens=3, temperature=0.0, logprobs=0))
+    outs = []
+    while not eng.is_finished():
+        outs.extend(eng.step())
+    assert len(outs) == 3
+    for o in outs:
+        assert o.logprob is not None and o.top_logprobs == []

@@ -147,3 +147,39 @@ def test_abort_mid_decode_with_graphs():
     assert eng.get_output(r1).finish_reason == "abort"
     out2 = eng.get_output(r2)
     assert out2.finished and len(out2.token_ids) == 16
+
+
+def _run_lp(cfg, prompts, k=3, max_new=6, seed=0):
+    torch.manual_seed(seed)
+    eng = LLMEngine(cfg, device="cuda:0")
+    rids = [eng.add_request(
+        p, SamplingParams(max_new_tokens=max_new, temperature=0.0,
+                          logprobs=k)) for p in prompts]
+    outs, steps = [], 0
+    while not eng.is_finished():
+        outs.extend(eng.step())
+        steps += 1
+        assert steps < 200
+    by_seq = {}
+    for o in outs:
+        by_seq.setdefault(o.seq_id, []).append(
+            (o.token_id, o.logprob, o.top_logprobs))
+    return [by_seq[r] for r in rids]
+
+
+def test_logprobs_graph_matches_eager():
+    """The graph runner's post-replay logprob extraction must agree with
+    the eager path (same weights, same greedy tokens, same logits)."""
+    prompts = [list(range(1, 20)), list(range(30, 47))]
+    eager = _run_lp(EngineConfig.tiny_llama(enforce_eager=True), prompts)
+    graph = _run_lp(EngineConfig.tiny_llama(use_hip_graphs=True), prompts)
+    for es, gs in zip(eager, graph):
+        assert len(es) == len(gs) == 6
+        for (et, elp, etop), (gt, glp, gtop) in zip(es, gs):
+            assert et == gt
+            assert abs(elp - glp) < 2e-3
+            assert [i for i, _ in etop] == [i for i, _ in gtop]
+            for (_, ev), (_, gv) in zip(etop, gtop):
+                assert abs(ev - gv) < 2e-3
+            # greedy: sampled token is top-1
+            assert etop[0][0] == et and glp <= 0.0

@@ -180,3 +180,58 @@ async def test_engine_gauges_in_metrics_exposition():
     assert "igw_engine_kv_blocks_total 63.0" in text
     assert "igw_engine_running_seqs 0.0" in text
     provider.engine.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_local_logprobs_non_streaming(local_gateway):
+    async with gw_client(local_gateway) as c:
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hello"}],
+            "max_tokens": 5, "temperature": 0,
+            "logprobs": True, "top_logprobs": 3,
+        })
+        assert r.status_code == 200
+        lp = r.json()["choices"][0]["logprobs"]
+        assert lp is not None and len(lp["content"]) == 5
+        for item in lp["content"]:
+            assert isinstance(item["token"], str)
+            assert item["logprob"] <= 0.0
+            assert item["bytes"] == list(item["token"].encode("utf-8"))
+            tops = item["top_logprobs"]
+            assert len(tops) == 3
+            vals = [t["logprob"] for t in tops]
+            assert vals == sorted(vals, reverse=True)
+            # greedy: sampled token is the top-1 alternative
+            assert tops[0]["token"] == item["token"]
+
+        # not requested -> null, like OpenAI
+        r2 = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hello"}],
+            "max_tokens": 2, "temperature": 0,
+        })
+        assert r2.json()["choices"][0]["logprobs"] is None
+
+
+@pytest.mark.asyncio
+async def test_local_logprobs_streaming(local_gateway):
+    async with gw_client(local_gateway) as c:
+        async with c.stream("POST", "/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 6, "temperature": 0, "stream": True,
+            "logprobs": True, "top_logprobs": 2,
+        }) as r:
+            assert r.status_code == 200
+            items = []
+            async for line in r.aiter_lines():
+                if not line.startswith("data: ") or line == "data: [DONE]":
+                    continue
+                ch = json.loads(line[6:])["choices"][0]
+                if ch.get("logprobs"):
+                    items.extend(ch["logprobs"]["content"])
+            assert 1 <= len(items) <= 6
+            for item in items:
+                assert item["logprob"] <= 0.0
+                assert len(item["top_logprobs"]) == 2
