@@ -115,6 +115,10 @@ class SamplingParams:
     # OpenAI logprobs: -1 = off; >=0 = return the sampled token's logprob
     # plus that many top alternatives per step (capped at 20, like OpenAI)
     logprobs: int = -1
+    # OpenAI seed: per-request deterministic sampling (best effort — a
+    # preempted-and-resumed request keeps its tick counter, like OpenAI's
+    # "best effort" wording). None = engine-global RNG.
+    seed: Optional[int] = None
 
 
 @dataclass

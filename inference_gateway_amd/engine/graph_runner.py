@@ -34,6 +34,7 @@ import torch
 from .. import ops
 from .config import EngineConfig
 from .models.batch import ForwardBatch
+from .sampler import apply_seeded_uniforms
 from .scheduler import Sequence
 
 _BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256,
@@ -210,6 +211,7 @@ class DecodeGraphRunner:
         topp = self.slot_topp.index_select(0, idx)
         u = torch.rand(bucket, generator=self._gen,
                        device=self.device).clamp_(1e-6, 1 - 1e-6)
+        apply_seeded_uniforms(u, seqs)
         toks = ops.sample(logits, temps, topk, topp, u)     # [bucket] int32
         act = idx[:B]
         self.slot_token.scatter_(0, act, toks[:B].to(torch.int64))

@@ -36,6 +36,7 @@ class Sequence:
     first_token_time: Optional[float] = None
     slot: Optional[int] = None        # persistent batch slot (graph runner)
     in_flight: bool = False           # a decode step's token not yet drained
+    sample_tick: int = 0              # per-seq counter for seeded sampling
 
     @property
     def num_tokens(self) -> int:

@@ -63,6 +63,7 @@ class LocalProvider:
         lp = -1
         if req.get("logprobs"):
             lp = max(0, min(20, int(req.get("top_logprobs") or 0)))
+        seed = req.get("seed")
         return SamplingParams(
             max_new_tokens=int(max_tokens),
             temperature=float(req.get("temperature", 1.0)),
@@ -70,6 +71,7 @@ class LocalProvider:
             top_k=int(req.get("top_k", 0)),
             stop_token_ids=stop_ids,
             logprobs=lp,
+            seed=int(seed) if seed is not None else None,
         )
 
     def _lp_item(self, out) -> dict:

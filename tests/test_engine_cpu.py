@@ -317,3 +317,33 @@ def test_logprobs_k_zero_returns_sampled_only():
     assert len(outs) == 3
     for o in outs:
         assert o.logprob is not None and o.top_logprobs == []
+
+
+def test_seeded_sampling_is_request_deterministic():
+    """OpenAI `seed`: a seeded request replays the same tokens regardless
+    of what else is in the batch; unseeded requests keep engine RNG."""
+    def run(extra_load: bool):
+        eng = _engine()
+        rid = eng.add_request(
+            list(range(1, 17)),
+            SamplingParams(max_new_tokens=6, temperature=1.0, seed=1234))
+        if extra_load:   # batch composition must not matter
+            eng.add_request(
+                list(range(5, 21)),
+                SamplingParams(max_new_tokens=6, temperature=1.0))
+        while not eng.is_finished():
+            eng.step()
+        return eng.get_output(rid).token_ids
+
+    a = run(extra_load=False)
+    b = run(extra_load=True)
+    assert a == b and len(a) == 6
+
+    eng = _engine()
+    rid = eng.add_request(
+        list(range(1, 17)),
+        SamplingParams(max_new_tokens=6, temperature=1.0, seed=4321))
+    while not eng.is_finished():
+        eng.step()
+    c = eng.get_output(rid).token_ids
+    assert c != a    # different seed -> (overwhelmingly) different draw

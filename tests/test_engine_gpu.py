@@ -183,3 +183,25 @@ def test_logprobs_graph_matches_eager():
                 assert abs(ev - gv) < 2e-3
             # greedy: sampled token is top-1
             assert etop[0][0] == et and glp <= 0.0
+
+
+def test_seeded_sampling_graph_matches_eager():
+    """OpenAI seed on the graph path: same seeded request must produce
+    identical tokens under hipGraph decode and eager decode (stochastic
+    temperature, uniforms from the per-request seed stream)."""
+    def run(cfg):
+        torch.manual_seed(0)
+        eng = LLMEngine(cfg, device="cuda:0")
+        rid = eng.add_request(
+            list(range(1, 20)),
+            SamplingParams(max_new_tokens=8, temperature=1.0, seed=99))
+        steps = 0
+        while not eng.is_finished():
+            eng.step()
+            steps += 1
+            assert steps < 200
+        return eng.get_output(rid).token_ids
+
+    eager = run(EngineConfig.tiny_llama(enforce_eager=True))
+    graph = run(EngineConfig.tiny_llama(use_hip_graphs=True))
+    assert eager == graph and len(eager) == 8
