@@ -119,6 +119,11 @@ class SamplingParams:
     # preempted-and-resumed request keeps its tick counter, like OpenAI's
     # "best effort" wording). None = engine-global RNG.
     seed: Optional[int] = None
+    # OpenAI repetition penalties (0 = off). Exact on the eager path; on
+    # the async graph path the host's token view lags one step, so the
+    # in-flight token's penalty is applied device-side (see sampler.py).
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
 
 
 @dataclass

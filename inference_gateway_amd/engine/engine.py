@@ -19,7 +19,7 @@ from .config import EngineConfig, SamplingParams
 from .kvcache import BlockManager
 from .models.batch import ForwardBatch
 from .models.llama import LlamaForCausalLM
-from .sampler import Sampler
+from .sampler import Sampler, apply_penalties
 from .scheduler import ScheduledBatch, Scheduler, SeqState, Sequence
 
 
@@ -235,6 +235,7 @@ class LLMEngine:
         hidden = self.model(fb, self.kv_caches)
         hidden_last = hidden[fb.last_token_indices]
         logits = self.model.compute_logits(hidden_last)
+        apply_penalties(logits, batch.seqs)
         tokens = self.sampler.sample(logits, batch.seqs)
         lp_map = _extract_logprobs(logits, batch.seqs, tokens)
         outputs = self._append_tokens(batch, tokens, lp_map)

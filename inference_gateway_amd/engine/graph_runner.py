@@ -34,7 +34,7 @@ import torch
 from .. import ops
 from .config import EngineConfig
 from .models.batch import ForwardBatch
-from .sampler import apply_seeded_uniforms
+from .sampler import apply_penalties, apply_seeded_uniforms
 from .scheduler import Sequence
 
 _BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256,
@@ -206,6 +206,8 @@ class DecodeGraphRunner:
 
         logits = self._logits[bucket]
         idx = self.in_slot_idx[:bucket]
+        apply_penalties(logits, seqs,
+                        last_tokens=self.slot_token.index_select(0, idx[:B]))
         temps = self.slot_temp.index_select(0, idx)
         topk = self.slot_topk.index_select(0, idx)
         topp = self.slot_topp.index_select(0, idx)

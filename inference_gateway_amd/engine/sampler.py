@@ -33,6 +33,49 @@ def apply_seeded_uniforms(u: torch.Tensor, seqs) -> None:
                                         device=u.device))
 
 
+def apply_penalties(logits: torch.Tensor, seqs,
+                    last_tokens: torch.Tensor = None) -> None:
+    """OpenAI presence/frequency penalties, in place on the penalized
+    rows only (zero cost when nobody asks).  Host-side token counts
+    drive the bulk of the penalty; on the async graph path the host's
+    view lags the device by exactly one token, so the caller passes
+    `last_tokens` (the device-resident last sampled token per row) and
+    that token's penalty is applied device-side — immediate repeats are
+    always penalized.  A token that repeats across both views gets its
+    presence term counted twice (documented best-effort; the frequency
+    term stays exact)."""
+    rows = [i for i, s in enumerate(seqs)
+            if s.sampling.presence_penalty or s.sampling.frequency_penalty]
+    if not rows:
+        return
+    dev, dt = logits.device, logits.dtype
+    for i in rows:
+        sp = seqs[i].sampling
+        out = seqs[i].output_token_ids
+        if out:
+            counts: dict = {}
+            for t in out:
+                counts[t] = counts.get(t, 0) + 1
+            ids = torch.tensor(list(counts), dtype=torch.int64, device=dev)
+            cnt = torch.tensor(list(counts.values()), dtype=torch.float32,
+                               device=dev)
+            logits[i, ids] -= (sp.presence_penalty
+                               + sp.frequency_penalty * cnt).to(dt)
+    if last_tokens is not None:
+        # only rows whose last token is still undrained (in_flight) — at
+        # the first decode step after prefill the host view IS current
+        # and a device-side add would double-count
+        drows = [i for i in rows if seqs[i].in_flight]
+        if drows:
+            rows_t = torch.tensor(drows, dtype=torch.int64, device=dev)
+            pen = torch.tensor(
+                [seqs[i].sampling.presence_penalty
+                 + seqs[i].sampling.frequency_penalty for i in drows],
+                dtype=torch.float32, device=dev)
+            logits[rows_t, last_tokens.index_select(0, rows_t)] -= \
+                pen.to(dt)
+
+
 class Sampler:
     def __init__(self, device, seed: int = 0):
         self.device = torch.device(device)

@@ -72,6 +72,8 @@ class LocalProvider:
             stop_token_ids=stop_ids,
             logprobs=lp,
             seed=int(seed) if seed is not None else None,
+            presence_penalty=float(req.get("presence_penalty") or 0.0),
+            frequency_penalty=float(req.get("frequency_penalty") or 0.0),
         )
 
     def _lp_item(self, out) -> dict:

@@ -205,3 +205,23 @@ def test_seeded_sampling_graph_matches_eager():
     eager = run(EngineConfig.tiny_llama(enforce_eager=True))
     graph = run(EngineConfig.tiny_llama(use_hip_graphs=True))
     assert eager == graph and len(eager) == 8
+
+
+def test_frequency_penalty_on_graph_path():
+    """Graph path: the async pipeline's host view lags one token, so the
+    device-side last-token add must close the gap — a huge frequency
+    penalty still forbids ALL repeats, including immediate ones."""
+    torch.manual_seed(0)
+    eng = LLMEngine(EngineConfig.tiny_llama(use_hip_graphs=True),
+                    device="cuda:0")
+    rid = eng.add_request(
+        list(range(1, 17)),
+        SamplingParams(max_new_tokens=12, temperature=0.0,
+                       frequency_penalty=100.0))
+    steps = 0
+    while not eng.is_finished():
+        eng.step()
+        steps += 1
+        assert steps < 200
+    toks = eng.get_output(rid).token_ids
+    assert len(toks) == 12 and len(set(toks)) == 12
