@@ -122,6 +122,36 @@ class AsyncEngine:
                 self._loops.pop(rid, None)
                 self.engine.abort_request(rid)
 
+    def generate_group(self, prompt_token_ids, sampling: SamplingParams,
+                       n: int):
+        """OpenAI n>1: one engine-side prefill, n token streams (the
+        engine forks the prompt KV — engine.add_request_group).  Returns
+        n async iterators; each consumer's exit aborts its own branch."""
+        loop = asyncio.get_running_loop()
+        qs = [asyncio.Queue() for _ in range(n)]
+        with self._lock:
+            rids = self.engine.add_request_group(prompt_token_ids,
+                                                 sampling, n)
+            for rid, q in zip(rids, qs):
+                self._queues[rid] = q
+                self._loops[rid] = loop
+        self._wake.set()
+
+        async def reader(rid, q):
+            try:
+                while True:
+                    item = await q.get()
+                    if item is self._SENTINEL:
+                        break
+                    yield item
+            finally:
+                with self._lock:
+                    self._queues.pop(rid, None)
+                    self._loops.pop(rid, None)
+                    self.engine.abort_request(rid)
+
+        return [reader(rid, q) for rid, q in zip(rids, qs)]
+
     def shutdown(self) -> None:
         self._stop = True
         self._wake.set()

@@ -127,6 +127,7 @@ class LLMEngine:
         self.kv_caches = self._allocate_kv(num_blocks)
         self._seq_counter = itertools.count()
         self._seqs: Dict[int, Sequence] = {}
+        self._fork_pending: Dict[int, List[Sequence]] = {}
         # +1 block of headroom: the async decode pipeline can write KV for
         # one in-flight token past max_model_len before the stop check runs
         self.max_blocks_per_seq = (
@@ -188,10 +189,36 @@ class LLMEngine:
         self.scheduler.add(seq)
         return seq_id
 
+    def add_request_group(self, prompt_token_ids: List[int],
+                          sampling: Optional[SamplingParams] = None,
+                          n: int = 1) -> List[int]:
+        """OpenAI n>1: ONE prefill for the whole group.  The first
+        sequence prefills normally; the other n-1 are materialized at its
+        prefill completion by forking the prompt KV (full blocks shared
+        ref-counted, partial tail block copied) — see _fork_at_prefill."""
+        sampling = sampling or SamplingParams()
+        parent_id = self.add_request(prompt_token_ids, sampling)
+        kids = []
+        for _ in range(n - 1):
+            sid = next(self._seq_counter)
+            child = Sequence(seq_id=sid,
+                             prompt_token_ids=list(prompt_token_ids),
+                             sampling=sampling)
+            self._seqs[sid] = child
+            kids.append(child)
+        if kids:
+            self._fork_pending[parent_id] = kids
+        return [parent_id] + [c.seq_id for c in kids]
+
     def abort_request(self, seq_id: int) -> None:
         seq = self._seqs.get(seq_id)
         if seq is not None and seq.state != SeqState.FINISHED:
             self.scheduler.abort(seq)
+            # a parent aborted before its fork point: finish the children
+            # too so their consumers see a sentinel instead of hanging
+            for child in self._fork_pending.pop(seq_id, []):
+                if child.state == SeqState.WAITING:
+                    self.scheduler.finish(child, "abort")
 
     def is_finished(self) -> bool:
         return not self.scheduler.has_work and self._pending is None
@@ -238,7 +265,16 @@ class LLMEngine:
         apply_penalties(logits, batch.seqs)
         tokens = self.sampler.sample(logits, batch.seqs)
         lp_map = _extract_logprobs(logits, batch.seqs, tokens)
+        # fork n>1 groups BEFORE parent bookkeeping (a parent finishing at
+        # its first token would free the blocks the fork shares)
+        fork_outputs: List[StepOutput] = []
+        if batch.is_prefill and self._fork_pending:
+            for row, seq in enumerate(batch.seqs):
+                if seq.seq_id in self._fork_pending:
+                    fork_outputs.extend(
+                        self._fork_at_prefill(seq, logits[row]))
         outputs = self._append_tokens(batch, tokens, lp_map)
+        outputs = fork_outputs + outputs
         if batch.is_prefill and gr is not None:
             # seed the persistent decode slot state for newly admitted seqs
             for seq, tok in zip(batch.seqs, tokens):
@@ -324,6 +360,56 @@ class LLMEngine:
             ctx_lens=ctx.to(dev),
             last_token_indices=torch.arange(B, dtype=torch.int64, device=dev),
         )
+
+    def _copy_block(self, src: int, dst: int) -> None:
+        for k, v in self.kv_caches:
+            k[dst].copy_(k[src])
+            v[dst].copy_(v[src])
+
+    def _fork_at_prefill(self, parent, row_logits) -> List[StepOutput]:
+        """Materialize a group's n-1 branches from the parent's freshly
+        prefilled KV: full prompt blocks are SHARED (ref-counted, decode
+        never writes them again), the partially filled tail block is
+        copied per branch, and each branch samples its own first token
+        from the parent's prefill logits.  If slots or blocks run short a
+        branch falls back to an ordinary re-prefilling request."""
+        kids = [c for c in self._fork_pending.pop(parent.seq_id, [])
+                if c.state == SeqState.WAITING]
+        if not kids:
+            return []
+        bs = self.cfg.block_size
+        plen = parent.prompt_len
+        full, partial = plen // bs, plen % bs
+        rep = row_logits.unsqueeze(0).repeat(len(kids), 1)
+        toks = self.sampler.sample(rep, kids)
+        lp_map = _extract_logprobs(rep, kids, toks)
+        gr = self.graph_runner
+        outs: List[StepOutput] = []
+        for j, child in enumerate(kids):
+            tail: List[int] = []
+            ok = self.scheduler.can_adopt()
+            if ok and partial:
+                try:
+                    tail = self.block_manager.allocate(1)
+                except RuntimeError:
+                    ok = False
+            if not ok:
+                # not enough slots/blocks right now: plain re-prefill
+                # (the sampled first token is discarded — the child's own
+                # prefill will sample one)
+                self.scheduler.add(child)
+                continue
+            if tail:
+                self._copy_block(parent.block_ids[full], tail[0])
+            child.block_ids = (self.block_manager.share(parent.block_ids[:full]) + tail)
+            self.scheduler.adopt(child)
+            if gr is not None:
+                gr.write_block_row(child.slot, child.block_ids)
+            outs.extend(self._apply_token(
+                child, toks[j], lp_map.get(j) if lp_map else None))
+            if child.state != SeqState.FINISHED and gr is not None:
+                gr.init_slot(child, toks[j])
+        return outs
 
     def _apply_token(self, seq, tok: int, lp=None) -> List[StepOutput]:
         import time

@@ -163,6 +163,18 @@ class Scheduler:
         return ScheduledBatch(is_prefill=False, seqs=list(self.running),
                               new_blocks=new_blocks)
 
+    # ---- forked sequences (OpenAI n>1: one prefill, shared prompt KV) --
+    def can_adopt(self) -> bool:
+        return bool(self._free_slots) and len(self.running) < self.max_num_seqs
+
+    def adopt(self, seq: Sequence) -> None:
+        """Place an already-prefilled forked branch straight into the
+        running set (its block table was cloned from the parent)."""
+        seq.slot = self._free_slots.pop()
+        seq.state = SeqState.RUNNING
+        self.running.append(seq)
+        self.membership_version += 1
+
     def finish(self, seq: Sequence, reason: str) -> None:
         seq.state = SeqState.FINISHED
         seq.finish_reason = reason

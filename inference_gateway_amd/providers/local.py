@@ -110,7 +110,7 @@ class LocalProvider:
                 hit = i
         return hit
 
-    async def _collect(self, prompt_ids, sampling, stops):
+    async def _collect(self, prompt_ids, sampling, stops, gen=None):
         """One full completion: returns (token_ids, text, finish_reason,
         lp_content); stop sequences are excluded from the text (OpenAI
         semantics; logprob entries for tokens past a stop-string cut are
@@ -118,7 +118,9 @@ class LocalProvider:
         token_ids = []
         lp_content = [] if sampling.logprobs >= 0 else None
         finish_reason = "stop"
-        async for out in self.engine.generate(prompt_ids, sampling):
+        if gen is None:
+            gen = self.engine.generate(prompt_ids, sampling)
+        async for out in gen:
             token_ids.append(out.token_id)
             if lp_content is not None:
                 lp_content.append(self._lp_item(out))
@@ -144,9 +146,12 @@ class LocalProvider:
                 prompt_ids, sampling, stops)
             return self._completion_response(prompt_ids, token_ids, text,
                                              finish_reason, lp)
-        # n > 1: the engine batches the n sequences continuously
+        # n > 1: ONE prefill — the engine forks the prompt KV into n
+        # branches (full blocks shared ref-counted, tail block copied)
+        gens = self.engine.generate_group(prompt_ids, sampling, n)
         results = await asyncio.gather(
-            *(self._collect(prompt_ids, sampling, stops) for _ in range(n)))
+            *(self._collect(prompt_ids, sampling, stops, gen=g)
+              for g in gens))
         total = sum(len(t) for t, _, _, _ in results)
         return {
             "id": f"chatcmpl-{uuid.uuid4().hex[:24]}",

@@ -373,3 +373,79 @@ def test_frequency_penalty_blocks_repetition():
         eng2.step()
     toks2 = eng2.get_output(rid2).token_ids
     assert len(set(toks2)) < 12
+
+
+def test_fork_group_matches_independent_requests():
+    """n>1 fork: each branch's KV (shared full blocks + copied tail)
+    must be bit-equal to a recomputed prefill — greedy branches must
+    reproduce an independent request's tokens exactly."""
+    prompt = list(range(1, 27))      # 26 tokens: 1 full block + partial
+    eng = _engine()
+    rid = eng.add_request(
+        prompt, SamplingParams(max_new_tokens=8, temperature=0.0))
+    while not eng.is_finished():
+        eng.step()
+    want = eng.get_output(rid).token_ids
+
+    eng2 = _engine()
+    rids = eng2.add_request_group(
+        prompt, SamplingParams(max_new_tokens=8, temperature=0.0), n=3)
+    while not eng2.is_finished():
+        eng2.step()
+    for r in rids:
+        out = eng2.get_output(r)
+        assert out.finished
+        assert out.token_ids == want
+
+
+def test_fork_block_accounting():
+    """All blocks (shared and private) return to the pool when every
+    branch finishes; block-aligned prompts fork with zero copies."""
+    for plen in (16, 26, 32):        # aligned and unaligned prompts
+        eng = _engine()
+        free0 = eng.block_manager.num_free
+        rids = eng.add_request_group(
+            list(range(1, plen + 1)),
+            SamplingParams(max_new_tokens=5, temperature=0.0), n=4)
+        while not eng.is_finished():
+            eng.step()
+        assert all(eng.get_output(r).finished for r in rids)
+        assert eng.block_manager.num_free == free0, f"leak at plen={plen}"
+
+
+def test_fork_seeded_branches_are_identical_and_unseeded_diverge():
+    eng = _engine()
+    rids = eng.add_request_group(
+        list(range(1, 20)),
+        SamplingParams(max_new_tokens=6, temperature=1.0, seed=7), n=2)
+    while not eng.is_finished():
+        eng.step()
+    a, b = [eng.get_output(r).token_ids for r in rids]
+    assert a == b            # same seed stream -> identical branches
+
+    eng2 = _engine()
+    rids2 = eng2.add_request_group(
+        list(range(1, 20)),
+        SamplingParams(max_new_tokens=8, temperature=1.0), n=4)
+    while not eng2.is_finished():
+        eng2.step()
+    outs = [tuple(eng2.get_output(r).token_ids) for r in rids2]
+    assert len(set(outs)) > 1    # engine RNG -> branches diverge
+
+
+def test_fork_fallback_when_slots_exhausted():
+    """More branches than slots: the overflow branches re-prefill as
+    ordinary requests and still complete correctly."""
+    eng = _engine()                    # max_num_seqs = 8
+    rids = eng.add_request_group(
+        list(range(1, 27)),
+        SamplingParams(max_new_tokens=4, temperature=0.0), n=8)
+    # a second independent request for the reference tokens
+    ref = eng.add_request(
+        list(range(1, 27)), SamplingParams(max_new_tokens=4,
+                                           temperature=0.0))
+    while not eng.is_finished():
+        eng.step()
+    want = eng.get_output(ref).token_ids
+    for r in rids:
+        assert eng.get_output(r).token_ids == want

@@ -225,3 +225,27 @@ def test_frequency_penalty_on_graph_path():
         assert steps < 200
     toks = eng.get_output(rid).token_ids
     assert len(toks) == 12 and len(set(toks)) == 12
+
+
+def test_fork_group_on_graph_path():
+    """n>1 fork through hipGraph decode: forked branches (shared prompt
+    blocks + copied tail, slot state seeded at fork) must equal an
+    independent greedy request."""
+    torch.manual_seed(0)
+    prompt = list(range(1, 27))
+    eng = LLMEngine(EngineConfig.tiny_llama(use_hip_graphs=True),
+                    device="cuda:0")
+    ref = eng.add_request(prompt, SamplingParams(max_new_tokens=8,
+                                                 temperature=0.0))
+    rids = eng.add_request_group(
+        prompt, SamplingParams(max_new_tokens=8, temperature=0.0), n=3)
+    steps = 0
+    while not eng.is_finished():
+        eng.step()
+        steps += 1
+        assert steps < 300
+    want = eng.get_output(ref).token_ids
+    assert len(want) == 8
+    for r in rids:
+        assert eng.get_output(r).token_ids == want
+    assert eng.block_manager.num_free == eng.block_manager.num_blocks - 1

@@ -235,3 +235,27 @@ async def test_local_logprobs_streaming(local_gateway):
             for item in items:
                 assert item["logprob"] <= 0.0
                 assert len(item["top_logprobs"]) == 2
+
+
+@pytest.mark.asyncio
+async def test_local_n_greater_one_forked(local_gateway):
+    """n>1 rides the engine's KV fork (one prefill); greedy branches are
+    identical and identical to n=1."""
+    async with gw_client(local_gateway) as c:
+        r1 = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "fork me"}],
+            "max_tokens": 6, "temperature": 0,
+        })
+        want = r1.json()["choices"][0]["message"]["content"]
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "fork me"}],
+            "max_tokens": 6, "temperature": 0, "n": 3,
+        })
+        assert r.status_code == 200
+        body = r.json()
+        assert [ch["index"] for ch in body["choices"]] == [0, 1, 2]
+        for ch in body["choices"]:
+            assert ch["message"]["content"] == want
+        assert body["usage"]["completion_tokens"] == 18
