@@ -449,3 +449,38 @@ def test_fork_fallback_when_slots_exhausted():
     want = eng.get_output(ref).token_ids
     for r in rids:
         assert eng.get_output(r).token_ids == want
+
+
+def test_fork_abort_one_branch_others_complete():
+    """Aborting one forked branch mid-decode must not disturb siblings
+    or leak blocks (shared refs decremented, not freed)."""
+    eng = _engine()
+    free0 = eng.block_manager.num_free
+    rids = eng.add_request_group(
+        list(range(1, 27)),
+        SamplingParams(max_new_tokens=10, temperature=0.0), n=3)
+    for _ in range(4):
+        eng.step()
+    eng.abort_request(rids[1])
+    while not eng.is_finished():
+        eng.step()
+    outs = [eng.get_output(r) for r in rids]
+    assert outs[1].finish_reason == "abort"
+    assert outs[0].finished and outs[2].finished
+    assert outs[0].token_ids == outs[2].token_ids   # greedy, undisturbed
+    assert len(outs[0].token_ids) == 10
+    assert eng.block_manager.num_free == free0
+
+
+def test_fork_parent_aborted_before_prefill_releases_children():
+    """Parent aborted while still WAITING: children must be finished too
+    (consumers get a sentinel instead of hanging)."""
+    eng = _engine()
+    # saturate the prefill queue so nothing runs yet
+    rids = eng.add_request_group(
+        list(range(1, 20)),
+        SamplingParams(max_new_tokens=4, temperature=0.0), n=3)
+    eng.abort_request(rids[0])
+    for r in rids:
+        assert eng.get_output(r).finish_reason == "abort"
+    assert not eng.scheduler.has_work
