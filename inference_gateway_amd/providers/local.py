@@ -222,6 +222,13 @@ class LocalProvider:
             return (prefix + json.dumps(delta, separators=(",", ":"))
                     + lps + ',"finish_reason":' + fin + "}]" + tail + "}")
 
+        n = min(int(req.get("n") or 1), 8)
+        if n > 1:
+            async for line in self._stream_multi(req, prompt_ids, sampling,
+                                                 cid, created, model, n):
+                yield line
+            return
+
         yield chunk({"role": "assistant", "content": ""})
         stops = self._stop_strings(req)
         holdback = max((len(s) for s in stops), default=1) - 1
@@ -272,3 +279,96 @@ class LocalProvider:
         }
         yield chunk({}, finish=finish_reason, usage=usage)
         yield "data: [DONE]"
+
+    async def _stream_multi(self, req: dict, prompt_ids, sampling,
+                            cid: str, created: int, model: str,
+                            n: int) -> AsyncIterator[str]:
+        """Streaming with n>1: the engine forks the prompt KV; branch
+        chunks interleave with per-choice `index` (OpenAI semantics).
+        One [DONE]; the final chunk carries the whole group's usage."""
+        stops = self._stop_strings(req)
+        holdback = max((len(s) for s in stops), default=1) - 1
+        want_lp = sampling.logprobs >= 0
+
+        def chunk(idx: int, delta: dict, finish=None, usage=None, lp=None):
+            ch = {"index": idx, "delta": delta,
+                  "finish_reason": finish}
+            if lp is not None:
+                ch["logprobs"] = {"content": lp}
+            frame = {"id": cid, "object": "chat.completion.chunk",
+                     "created": created, "model": model, "choices": [ch]}
+            if usage is not None:
+                frame["usage"] = usage
+            return "data: " + json.dumps(frame, separators=(",", ":"))
+
+        queue: asyncio.Queue = asyncio.Queue()
+        gens = self.engine.generate_group(prompt_ids, sampling, n)
+
+        async def pump(idx: int, gen) -> None:
+            pending, pending_lp = [], []
+            emitted_tail, ntok = "", 0
+            finish_reason = "stop"
+            try:
+                async for out in gen:
+                    ntok += 1
+                    pending.append(out.token_id)
+                    if want_lp:
+                        pending_lp.append(self._lp_item(out))
+                    text = self.tokenizer.decode(pending)
+                    if text.endswith("�"):
+                        continue
+                    pending = []
+                    lp = pending_lp if want_lp else None
+                    pending_lp = []
+                    if out.finished:
+                        finish_reason = ("stop" if out.finish_reason ==
+                                         "stop" else "length")
+                    if stops:
+                        buf = emitted_tail + text
+                        i = self._find_stop(buf, stops)
+                        if i >= 0:
+                            if buf[:i]:
+                                await queue.put(
+                                    chunk(idx, {"content": buf[:i]}, lp=lp))
+                            finish_reason = "stop"
+                            emitted_tail = ""
+                            break
+                        emit = buf[:-holdback] if holdback else buf
+                        emitted_tail = buf[len(emit):]
+                        if emit:
+                            await queue.put(
+                                chunk(idx, {"content": emit}, lp=lp))
+                        continue
+                    await queue.put(chunk(idx, {"content": text}, lp=lp))
+                if emitted_tail:
+                    await queue.put(chunk(idx,
+                                          {"content": emitted_tail}))
+            finally:
+                await queue.put(("done", idx, ntok, finish_reason))
+
+        tasks = [asyncio.ensure_future(pump(i, g))
+                 for i, g in enumerate(gens)]
+        try:
+            for i in range(n):
+                yield chunk(i, {"role": "assistant", "content": ""})
+            total, done = 0, 0
+            while done < n:
+                item = await queue.get()
+                if isinstance(item, tuple):
+                    _, idx, ntok, fin = item
+                    total += ntok
+                    done += 1
+                    usage = None
+                    if done == n:
+                        usage = {
+                            "prompt_tokens": len(prompt_ids),
+                            "completion_tokens": total,
+                            "total_tokens": len(prompt_ids) + total,
+                        }
+                    yield chunk(idx, {}, finish=fin, usage=usage)
+                else:
+                    yield item
+            yield "data: [DONE]"
+        finally:
+            for t in tasks:
+                t.cancel()

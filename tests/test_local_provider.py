@@ -259,3 +259,37 @@ async def test_local_n_greater_one_forked(local_gateway):
         for ch in body["choices"]:
             assert ch["message"]["content"] == want
         assert body["usage"]["completion_tokens"] == 18
+
+
+@pytest.mark.asyncio
+async def test_local_streaming_n_greater_one(local_gateway):
+    """Streaming with n=3: interleaved per-choice chunks, one [DONE],
+    group usage on the final chunk, greedy branches identical."""
+    async with gw_client(local_gateway) as c:
+        async with c.stream("POST", "/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "multi"}],
+            "max_tokens": 5, "temperature": 0, "stream": True, "n": 3,
+        }) as r:
+            assert r.status_code == 200
+            texts = {0: "", 1: "", 2: ""}
+            finishes, dones, usage = {}, 0, None
+            async for line in r.aiter_lines():
+                if not line.startswith("data: "):
+                    continue
+                if line == "data: [DONE]":
+                    dones += 1
+                    continue
+                body = json.loads(line[6:])
+                ch = body["choices"][0]
+                i = ch["index"]
+                texts[i] += ch["delta"].get("content") or ""
+                if ch.get("finish_reason"):
+                    finishes[i] = ch["finish_reason"]
+                if body.get("usage"):
+                    usage = body["usage"]
+            assert dones == 1
+            assert set(finishes) == {0, 1, 2}
+            assert all(f == "length" for f in finishes.values())
+            assert texts[0] == texts[1] == texts[2] != ""
+            assert usage["completion_tokens"] == 15
