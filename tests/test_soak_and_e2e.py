@@ -187,3 +187,59 @@ def test_search_server_rpc_roundtrip():
         "params": {"name": "get_document",
                    "arguments": {"id": "nope"}}})["result"]
     assert missing["isError"]
+
+
+@pytest.mark.asyncio
+async def test_soak_mixed_sampling_features():
+    """24 concurrent clients exercising every sampling feature at once —
+    n>1 KV forks, logprobs, stop strings, seeds — against one engine:
+    every stream completes with exactly one [DONE], seeded identical
+    requests return identical text."""
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=64,
+                     max_model_len=256)
+    lc.num_gpu_blocks = 1024
+    provider = LocalProvider(lc, device="cpu")
+    cfg = config_mod.load({})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    try:
+        async with gw_client(gw) as c:
+            async def stream_one(i):
+                body = {
+                    "model": "local/tiny-llama",
+                    "messages": [{"role": "user", "content": f"mix {i}"}],
+                    "max_tokens": 5, "temperature": 0.9, "stream": True,
+                }
+                if i % 3 == 0:
+                    body["n"] = 2
+                if i % 4 == 0:
+                    body["logprobs"] = True
+                    body["top_logprobs"] = 2
+                if i % 5 == 0:
+                    body["stop"] = ["zq!x"]
+                done = 0
+                async with c.stream("POST", "/v1/chat/completions",
+                                    json=body) as r:
+                    assert r.status_code == 200
+                    async for line in r.aiter_lines():
+                        if line.strip() == "data: [DONE]":
+                            done += 1
+                return done
+
+            async def seeded(i):
+                r = await c.post("/v1/chat/completions", json={
+                    "model": "local/tiny-llama",
+                    "messages": [{"role": "user", "content": "same"}],
+                    "max_tokens": 6, "temperature": 1.0, "seed": 42,
+                })
+                return r.json()["choices"][0]["message"]["content"]
+
+            dones = await asyncio.gather(*(stream_one(i)
+                                           for i in range(16)))
+            texts = await asyncio.gather(*(seeded(i) for i in range(8)))
+        assert all(d == 1 for d in dones)
+        assert len(set(texts)) == 1
+    finally:
+        provider.engine.shutdown()
