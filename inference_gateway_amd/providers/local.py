@@ -344,6 +344,12 @@ class LocalProvider:
                     await queue.put(chunk(idx,
                                           {"content": emitted_tail}))
             finally:
+                try:
+                    # a stop-string break leaves the branch generating;
+                    # close the stream now so its abort fires promptly
+                    await gen.aclose()
+                except Exception:
+                    pass
                 await queue.put(("done", idx, ntok, finish_reason))
 
         tasks = [asyncio.ensure_future(pump(i, g))
