@@ -128,6 +128,10 @@ class LLMEngine:
         self._seq_counter = itertools.count()
         self._seqs: Dict[int, Sequence] = {}
         self._fork_pending: Dict[int, List[Sequence]] = {}
+        # cumulative serving counters (Prometheus gauges read these)
+        self.num_generated_tokens = 0
+        self.num_prompt_tokens = 0
+        self.num_forks = 0
         # +1 block of headroom: the async decode pipeline can write KV for
         # one in-flight token past max_model_len before the stop check runs
         self.max_blocks_per_seq = (
@@ -258,6 +262,8 @@ class LLMEngine:
             return prev
 
         prev = self._drain_pending()
+        if batch.is_prefill:
+            self.num_prompt_tokens += sum(s.prompt_len for s in batch.seqs)
         fb = self._build_forward_batch(batch)
         hidden = self.model(fb, self.kv_caches)
         hidden_last = hidden[fb.last_token_indices]
@@ -403,6 +409,7 @@ class LLMEngine:
                 self._copy_block(parent.block_ids[full], tail[0])
             child.block_ids = (self.block_manager.share(parent.block_ids[:full]) + tail)
             self.scheduler.adopt(child)
+            self.num_forks += 1
             if gr is not None:
                 gr.write_block_row(child.slot, child.block_ids)
             outs.extend(self._apply_token(
@@ -414,6 +421,7 @@ class LLMEngine:
     def _apply_token(self, seq, tok: int, lp=None) -> List[StepOutput]:
         import time
 
+        self.num_generated_tokens += 1
         seq.output_token_ids.append(tok)
         if seq.first_token_time is None:
             seq.first_token_time = time.monotonic()

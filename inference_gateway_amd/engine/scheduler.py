@@ -74,6 +74,7 @@ class Scheduler:
         # starve in-flight streams (and in-flight streams do not starve
         # TTFT) — SURVEY.md §7 'scheduler fairness under 64 SSE clients'
         self._last_was_prefill = False
+        self.num_preemptions = 0      # cumulative (metrics)
 
     def add(self, seq: Sequence) -> None:
         if seq.prompt_len > self.max_model_len:
@@ -144,6 +145,7 @@ class Scheduler:
             if ok:
                 break
             victim = self.running.pop()  # newest arrival
+            self.num_preemptions += 1
             self.membership_version += 1
             new_blocks = [nb for nb in new_blocks if nb[0] != victim.slot]
             self.bm.free(victim.block_ids)

@@ -102,12 +102,24 @@ class Telemetry:
                        "free paged-KV blocks", registry=r)
         g_total = Gauge("igw_engine_kv_blocks_total",
                         "total paged-KV blocks", registry=r)
+        g_gen = Gauge("igw_engine_generated_tokens_total",
+                      "cumulative generated tokens", registry=r)
+        g_prompt = Gauge("igw_engine_prompt_tokens_total",
+                         "cumulative prefilled prompt tokens", registry=r)
+        g_preempt = Gauge("igw_engine_preemptions_total",
+                          "cumulative KV-pressure preemptions", registry=r)
+        g_forks = Gauge("igw_engine_forks_total",
+                        "cumulative n>1 KV-fork branches", registry=r)
         sched = engine.scheduler
         bm = engine.block_manager
         g_run.set_function(lambda: len(sched.running))
         g_wait.set_function(lambda: len(sched.waiting))
         g_free.set_function(lambda: bm.num_free)
         g_total.set_function(lambda: bm.num_blocks - 1)  # block 0 reserved
+        g_gen.set_function(lambda: engine.num_generated_tokens)
+        g_prompt.set_function(lambda: engine.num_prompt_tokens)
+        g_preempt.set_function(lambda: sched.num_preemptions)
+        g_forks.set_function(lambda: engine.num_forks)
 
     def exposition(self) -> bytes:
         return generate_latest(self.registry)

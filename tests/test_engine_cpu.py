@@ -484,3 +484,17 @@ def test_fork_parent_aborted_before_prefill_releases_children():
     for r in rids:
         assert eng.get_output(r).finish_reason == "abort"
     assert not eng.scheduler.has_work
+
+
+def test_engine_counters():
+    eng = _engine()
+    rids = eng.add_request_group(
+        list(range(1, 27)),
+        SamplingParams(max_new_tokens=4, temperature=0.0), n=2)
+    while not eng.is_finished():
+        eng.step()
+    assert all(eng.get_output(r).finished for r in rids)
+    assert eng.num_prompt_tokens == 26          # ONE prefill for the group
+    assert eng.num_generated_tokens == 8        # 4 tokens x 2 branches
+    assert eng.num_forks == 1
+    assert eng.scheduler.num_preemptions == 0

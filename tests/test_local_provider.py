@@ -179,6 +179,8 @@ async def test_engine_gauges_in_metrics_exposition():
     text = telemetry.exposition().decode()
     assert "igw_engine_kv_blocks_total 63.0" in text
     assert "igw_engine_running_seqs 0.0" in text
+    assert "igw_engine_generated_tokens_total 0.0" in text
+    assert "igw_engine_preemptions_total 0.0" in text
     provider.engine.shutdown()
 
 
