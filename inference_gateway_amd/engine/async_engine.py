@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import asyncio
 import threading
-from typing import AsyncIterator, Dict, Optional
+from typing import AsyncIterator, Dict
 
 from .config import EngineConfig, SamplingParams
 from .engine import LLMEngine, StepOutput

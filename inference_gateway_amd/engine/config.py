@@ -6,7 +6,7 @@ Mixtral 8x7B, plus tiny variants for CPU tests and smoke().
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 

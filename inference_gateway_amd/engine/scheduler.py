@@ -11,7 +11,7 @@ import enum
 import time
 from collections import deque
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from .config import SamplingParams
 from .kvcache import BlockManager

@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import json
 import time
-from typing import AsyncIterator, Dict, List, Optional
+from typing import AsyncIterator, Dict, List
 
 from .selector import (SELECTOR_TOOL_EXECUTE, SELECTOR_TOOL_GET,
                        handle_tools_get)

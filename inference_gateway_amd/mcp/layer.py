@@ -4,7 +4,6 @@ tool injection (selector meta-tools by default, every schema in direct
 mode), then the agent loop for both streaming and non-streaming chat."""
 from __future__ import annotations
 
-from typing import Optional
 
 from ..gateway.http import Request, Response
 from .agent import Agent
