@@ -498,3 +498,27 @@ def test_engine_counters():
     assert eng.num_generated_tokens == 8        # 4 tokens x 2 branches
     assert eng.num_forks == 1
     assert eng.scheduler.num_preemptions == 0
+
+
+def test_fork_under_preemption_pressure():
+    """Forks + LIFO preemption interleaved on a tiny block pool: every
+    branch still completes with the right token count, no block leaks,
+    and greedy branches (re-prefilled or not) agree."""
+    eng = _engine(num_gpu_blocks=24, max_num_seqs=8)
+    free0 = eng.block_manager.num_free
+    groups = []
+    for g in range(3):
+        groups.append(eng.add_request_group(
+            list(range(1 + g, 27 + g)),
+            SamplingParams(max_new_tokens=12, temperature=0.0), n=2))
+    steps = 0
+    while not eng.is_finished():
+        eng.step()
+        steps += 1
+        assert steps < 2000
+    for rids in groups:
+        outs = [eng.get_output(r) for r in rids]
+        for o in outs:
+            assert o.finished and len(o.token_ids) == 12
+        assert outs[0].token_ids == outs[1].token_ids
+    assert eng.block_manager.num_free == free0
