@@ -504,7 +504,7 @@ def test_fork_under_preemption_pressure():
     """Forks + LIFO preemption interleaved on a tiny block pool: every
     branch still completes with the right token count, no block leaks,
     and greedy branches (re-prefilled or not) agree."""
-    eng = _engine(num_gpu_blocks=24, max_num_seqs=8)
+    eng = _engine(num_gpu_blocks=10, max_num_seqs=8)
     free0 = eng.block_manager.num_free
     groups = []
     for g in range(3):
