@@ -231,7 +231,7 @@ class LLMEngine:
         seq = self._seqs[seq_id]
         return RequestOutput(
             seq_id=seq_id,
-            token_ids=list(seq.output_token_ids),
+            token_ids=seq.all_generated_ids,
             finished=seq.state == SeqState.FINISHED,
             finish_reason=seq.finish_reason,
         )
@@ -429,7 +429,7 @@ class LLMEngine:
         sp = seq.sampling
         if not sp.ignore_eos and tok in sp.stop_token_ids:
             reason = "stop"
-        elif len(seq.output_token_ids) >= sp.max_new_tokens:
+        elif seq.num_generated >= sp.max_new_tokens:
             reason = "length"
         elif seq.num_tokens >= self.cfg.max_model_len:
             reason = "length"

@@ -37,6 +37,23 @@ class Sequence:
     slot: Optional[int] = None        # persistent batch slot (graph runner)
     in_flight: bool = False           # a decode step's token not yet drained
     sample_tick: int = 0              # per-seq counter for seeded sampling
+    # length of the ORIGINAL prompt: preemption folds generated tokens
+    # into prompt_token_ids for the re-prefill, so generation length and
+    # output reporting must be measured against this, not prompt_len
+    orig_prompt_len: int = -1
+
+    def __post_init__(self) -> None:
+        self.orig_prompt_len = len(self.prompt_token_ids)
+
+    @property
+    def num_generated(self) -> int:
+        return (len(self.prompt_token_ids) + len(self.output_token_ids)
+                - self.orig_prompt_len)
+
+    @property
+    def all_generated_ids(self) -> List[int]:
+        return (self.prompt_token_ids
+                + self.output_token_ids)[self.orig_prompt_len:]
 
     @property
     def num_tokens(self) -> int:
