@@ -522,3 +522,32 @@ def test_fork_under_preemption_pressure():
             assert o.finished and len(o.token_ids) == 12
         assert outs[0].token_ids == outs[1].token_ids
     assert eng.block_manager.num_free == free0
+
+
+def test_preempted_request_respects_max_new_tokens():
+    """Regression: preemption folds generated tokens into the rebuilt
+    prompt — the resumed request must still stop at max_new_tokens TOTAL
+    and report every generated token, matching an unpressured run."""
+    eng = _engine()
+    rid = eng.add_request(list(range(1, 27)),
+                          SamplingParams(max_new_tokens=10,
+                                         temperature=0.0))
+    while not eng.is_finished():
+        eng.step()
+    want = eng.get_output(rid).token_ids
+
+    eng2 = _engine(num_gpu_blocks=6, max_num_seqs=4)
+    rid2 = eng2.add_request(list(range(1, 27)),
+                            SamplingParams(max_new_tokens=10,
+                                           temperature=0.0))
+    # pressure: a second long-running request forces block contention
+    eng2.add_request(list(range(30, 56)),
+                     SamplingParams(max_new_tokens=10, temperature=0.0))
+    steps = 0
+    while not eng2.is_finished():
+        eng2.step()
+        steps += 1
+        assert steps < 2000
+    assert eng2.scheduler.num_preemptions > 0
+    got = eng2.get_output(rid2).token_ids
+    assert len(got) == 10 and got == want
