@@ -115,9 +115,9 @@ class SamplingParams:
     # OpenAI logprobs: -1 = off; >=0 = return the sampled token's logprob
     # plus that many top alternatives per step (capped at 20, like OpenAI)
     logprobs: int = -1
-    # OpenAI seed: per-request deterministic sampling (best effort — a
-    # preempted-and-resumed request keeps its tick counter, like OpenAI's
-    # "best effort" wording). None = engine-global RNG.
+    # OpenAI seed: per-request deterministic sampling.  The per-seq tick
+    # keeps counting across preemption/rebuild, so a seeded request is
+    # reproducible even under KV pressure (tested). None = engine RNG.
     seed: Optional[int] = None
     # OpenAI repetition penalties (0 = off). Exact on the eager path; on
     # the async graph path the host's token view lags one step, so the
