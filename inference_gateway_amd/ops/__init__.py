@@ -100,6 +100,29 @@ def flash_prefill(q, k, v, cu_seqlens, seq_lens, scale: float):
     return torch_ref.flash_prefill_ref(q, k, v, cu_seqlens, scale)
 
 
+def flash_prefill_paged(q, kcache, vcache, block_table, prefix_lens,
+                        cu_seqlens, seq_lens, scale: float):
+    """Paged-EXTEND prefill (chunked-prefill / prefix-cache enabler):
+    q rows are a per-seq CHUNK whose own K/V plus `prefix_lens` cached
+    tokens before it are all in the paged cache; each q row attends over
+    cache positions [0, prefix + row].  seq_lens: python ints (q rows
+    per sequence)."""
+    if q.is_cuda:
+        _require_kernels()
+        out = torch.empty_like(q)
+        tile_seq, tile_off = torch_ref.make_prefill_tiles(
+            seq_lens, tile_m=64, device=q.device
+        )
+        _kernels.flash_prefill_paged(
+            out, q, kcache, vcache, block_table, prefix_lens, cu_seqlens,
+            tile_seq, tile_off, scale
+        )
+        return out
+    return torch_ref.flash_prefill_paged_ref(
+        q, kcache, vcache, block_table, prefix_lens, cu_seqlens, scale
+    )
+
+
 def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
     if gate_up.is_cuda:
         _require_kernels()

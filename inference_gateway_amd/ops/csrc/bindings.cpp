@@ -25,6 +25,11 @@ void flash_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                    torch::Tensor v, torch::Tensor cu_seqlens,
                    torch::Tensor tile_seq, torch::Tensor tile_off,
                    double scale);
+void flash_prefill_paged(torch::Tensor out, torch::Tensor q,
+                         torch::Tensor kcache, torch::Tensor vcache,
+                         torch::Tensor block_table, torch::Tensor prefix_lens,
+                         torch::Tensor cu_seqlens, torch::Tensor tile_seq,
+                         torch::Tensor tile_off, double scale);
 void sample(torch::Tensor out, torch::Tensor logits, torch::Tensor temperature,
             torch::Tensor top_k, torch::Tensor top_p, torch::Tensor uniforms);
 void silu_mul(torch::Tensor out, torch::Tensor gate_up);
@@ -49,6 +54,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "GQA paged-attention decode (bf16, gfx950)");
   m.def("flash_prefill", &igw::flash_prefill,
         "varlen causal MFMA flash-attention prefill (bf16, gfx950)");
+  m.def("flash_prefill_paged", &igw::flash_prefill_paged,
+        "paged-extend prefill: chunk queries over cached prefix+chunk K/V "
+        "via block table (bf16, gfx950)");
   m.def("sample", &igw::sample,
         "fused temperature/top-k/top-p sampling (fp32 logits, gfx950)");
   m.def("silu_mul", &igw::silu_mul, "fused SwiGLU activation (bf16, gfx950)");

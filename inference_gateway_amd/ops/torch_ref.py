@@ -114,6 +114,40 @@ def flash_prefill_ref(q, k, v, cu_seqlens, scale):
     return out
 
 
+def flash_prefill_paged_ref(q, kcache, vcache, block_table, prefix_lens,
+                            cu_seqlens, scale):
+    """Reference for the paged-extend prefill: gather each sequence's
+    full K/V (prefix + chunk) from the cache via its block table, then
+    run offset-causal attention for the chunk's q rows."""
+    Tq, NQ, HD = q.shape
+    NKV = kcache.shape[1]
+    BS = kcache.shape[2]
+    G = NQ // NKV
+    out = torch.empty_like(q)
+    for s in range(cu_seqlens.shape[0] - 1):
+        a, b = int(cu_seqlens[s]), int(cu_seqlens[s + 1])
+        Lq = b - a
+        P = int(prefix_lens[s])
+        KVL = P + Lq
+        nblk = (KVL + BS - 1) // BS
+        blocks = block_table[s, :nblk].long()
+        # [nblk*BS, NKV, HD] gathered then trimmed
+        keys = kcache[blocks].permute(0, 2, 1, 3).reshape(
+            nblk * BS, NKV, HD)[:KVL]
+        vals = vcache[blocks].permute(0, 2, 1, 3).reshape(
+            nblk * BS, NKV, HD)[:KVL]
+        qpos = torch.arange(P, KVL, device=q.device)
+        kpos = torch.arange(KVL, device=q.device)
+        mask = kpos.unsqueeze(0) <= qpos.unsqueeze(1)     # [Lq, KVL]
+        for h in range(NQ):
+            kvh = h // G
+            sc = (q[a:b, h].float() @ keys[:, kvh].float().T) * scale
+            sc = sc.masked_fill(~mask, float("-inf"))
+            p = torch.softmax(sc, dim=-1)
+            out[a:b, h] = (p @ vals[:, kvh].float()).to(q.dtype)
+    return out
+
+
 def make_prefill_tiles(seq_lens, tile_m: int = 64, device="cpu"):
     """Host-side tile map for flash_prefill: (tile_seq, tile_off) int32."""
     tseq, toff = [], []

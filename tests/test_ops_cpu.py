@@ -97,3 +97,34 @@ def test_make_prefill_tiles():
     ts, to = torch_ref.make_prefill_tiles([64, 1, 130])
     assert ts.tolist() == [0, 1, 2, 2, 2]
     assert to.tolist() == [0, 0, 0, 64, 128]
+
+
+def test_flash_prefill_paged_ref_matches_dense_ref():
+    """The paged-extend reference must agree with the dense causal
+    reference: attention of the LAST Lq rows of a sequence over the
+    whole sequence == paged-extend of that chunk with prefix P."""
+    torch.manual_seed(0)
+    NKV, NQ, HD, BS = 2, 4, 128, 16
+    for P, Lq in ((0, 33), (16, 20), (21, 13)):
+        KVL = P + Lq
+        nblk = (KVL + BS - 1) // BS
+        kcache = torch.randn(nblk + 2, NKV, BS, HD).bfloat16()
+        vcache = torch.randn(nblk + 2, NKV, BS, HD).bfloat16()
+        # blocks 1..nblk hold the sequence (block 0 reserved)
+        bt = torch.arange(1, nblk + 1, dtype=torch.int32).unsqueeze(0)
+        # dense copy of the full sequence K/V out of the cache
+        keys = kcache[bt[0].long()].permute(0, 2, 1, 3).reshape(
+            nblk * BS, NKV, HD)[:KVL]
+        vals = vcache[bt[0].long()].permute(0, 2, 1, 3).reshape(
+            nblk * BS, NKV, HD)[:KVL]
+        q_full = torch.randn(KVL, NQ, HD).bfloat16()
+        scale = HD ** -0.5
+        cu_full = torch.tensor([0, KVL], dtype=torch.int32)
+        dense = torch_ref.flash_prefill_ref(q_full, keys, vals, cu_full,
+                                            scale)
+        cu_q = torch.tensor([0, Lq], dtype=torch.int32)
+        paged = torch_ref.flash_prefill_paged_ref(
+            q_full[P:], kcache, vcache, bt,
+            torch.tensor([P], dtype=torch.int32), cu_q, scale)
+        torch.testing.assert_close(paged.float(), dense[P:].float(),
+                                   atol=2e-2, rtol=2e-2)
