@@ -148,6 +148,7 @@ class LocalConfig:
     max_model_len: int = 8192
     gpu_memory_utilization: float = 0.90
     enforce_eager: bool = False
+    prefix_caching: bool = False
 
 
 @dataclass
@@ -271,6 +272,7 @@ def load(env: Optional[Mapping[str, str]] = None) -> Config:
     lo.max_model_len = int(g("LOCAL_MAX_MODEL_LEN", "8192"))
     lo.gpu_memory_utilization = float(g("LOCAL_GPU_MEMORY_UTILIZATION", "0.90"))
     lo.enforce_eager = _bool(g("LOCAL_ENFORCE_EAGER", "false"))
+    lo.prefix_caching = _bool(g("LOCAL_PREFIX_CACHING", "false"))
 
     # provider bootstrap: registry defaults + env overrides
     from ..providers.constants import PROVIDERS

@@ -141,6 +141,9 @@ class EngineConfig:
     # client ever pays a ~1 s first-capture stall mid-stream (the bench
     # path leaves this off — it uses one fixed bucket)
     capture_at_init: bool = False
+    # prefix caching: hash-indexed reuse of full prompt blocks across
+    # requests (suffix-only extend-prefill).  Opt-in.
+    enable_prefix_caching: bool = False
     tensor_parallel_size: int = 1
     seed: int = 0
 

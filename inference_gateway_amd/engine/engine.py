@@ -119,7 +119,9 @@ class LLMEngine:
         self.model.eval()
 
         num_blocks = cfg.num_gpu_blocks or self._auto_num_blocks()
-        self.block_manager = BlockManager(num_blocks, cfg.block_size)
+        self.block_manager = BlockManager(
+            num_blocks, cfg.block_size,
+            enable_prefix_caching=cfg.enable_prefix_caching)
         self.scheduler = Scheduler(
             self.block_manager, cfg.max_num_seqs, cfg.max_batched_tokens,
             cfg.max_model_len)
@@ -316,22 +318,27 @@ class LLMEngine:
         bs = self.cfg.block_size
         if batch.is_prefill:
             # vectorized assembly: python-per-token loops cost ~30 ms on an
-            # 8k-token burst, which lands straight on TTFT
+            # 8k-token burst, which lands straight on TTFT.  With prefix
+            # caching, a sequence contributes only its SUFFIX rows (the
+            # cached prefix is block-aligned, so suffix KV writes land in
+            # private blocks) and attention takes the paged-extend path.
             token_ids, pos_parts, slot_parts = [], [], []
-            cu, seq_lens, last = [0], [], []
+            cu, seq_lens, last, prefixes = [0], [], [], []
             for seq in batch.seqs:
                 L = seq.prompt_len
-                token_ids.extend(seq.prompt_token_ids)
-                pos_parts.append(torch.arange(L, dtype=torch.int32))
+                p0 = seq.cached_prefix
+                token_ids.extend(seq.prompt_token_ids[p0:])
+                pos_parts.append(torch.arange(p0, L, dtype=torch.int32))
                 blocks = torch.tensor(seq.block_ids, dtype=torch.int32)
                 slots = (blocks.repeat_interleave(bs) * bs
                          + torch.arange(bs, dtype=torch.int32).repeat(
-                             len(seq.block_ids)))[:L]
+                             len(seq.block_ids)))[p0:L]
                 slot_parts.append(slots)
-                cu.append(cu[-1] + L)
-                seq_lens.append(L)
+                cu.append(cu[-1] + (L - p0))
+                seq_lens.append(L - p0)
                 last.append(cu[-1] - 1)
-            return ForwardBatch(
+                prefixes.append(p0)
+            fb = ForwardBatch(
                 token_ids=torch.tensor(token_ids, dtype=torch.int64, device=dev),
                 positions=torch.cat(pos_parts).to(dev),
                 slot_mapping=torch.cat(slot_parts).to(dev),
@@ -341,6 +348,16 @@ class LLMEngine:
                 last_token_indices=torch.tensor(last, dtype=torch.int64,
                                                 device=dev),
             )
+            if any(prefixes):
+                maxb = max(len(s.block_ids) for s in batch.seqs)
+                bt = torch.zeros((len(batch.seqs), maxb), dtype=torch.int32)
+                for i, s in enumerate(batch.seqs):
+                    bt[i, : len(s.block_ids)] = torch.tensor(
+                        s.block_ids, dtype=torch.int32)
+                fb.block_table = bt.to(dev)
+                fb.prefix_lens = torch.tensor(prefixes, dtype=torch.int32,
+                                              device=dev)
+            return fb
         # decode
         B = len(batch.seqs)
         token_ids = [s.output_token_ids[-1] if s.output_token_ids

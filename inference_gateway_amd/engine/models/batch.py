@@ -16,6 +16,9 @@ class ForwardBatch:
     # prefill-only
     cu_seqlens: Optional[torch.Tensor] = None   # [B+1] int32
     seq_lens: Optional[List[int]] = None
+    # prefix-cached prefill: per-seq cached tokens before the q rows
+    # (requires block_table; attention takes the paged-extend path)
+    prefix_lens: Optional[torch.Tensor] = None  # [B] int32
     # decode-only
     block_table: Optional[torch.Tensor] = None  # [B, max_blocks] int32
     ctx_lens: Optional[torch.Tensor] = None     # [B] int32

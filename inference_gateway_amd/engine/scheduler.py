@@ -37,6 +37,8 @@ class Sequence:
     slot: Optional[int] = None        # persistent batch slot (graph runner)
     in_flight: bool = False           # a decode step's token not yet drained
     sample_tick: int = 0              # per-seq counter for seeded sampling
+    cached_prefix: int = 0            # prompt tokens served from the
+                                      # prefix cache (block-aligned)
     # length of the ORIGINAL prompt: preemption folds generated tokens
     # into prompt_token_ids for the re-prefill, so generation length and
     # output reporting must be measured against this, not prompt_len
@@ -124,7 +126,8 @@ class Scheduler:
             # max_model_len > max_batched_tokens) gets its own oversized
             # batch — breaking unconditionally would starve it forever
             self.waiting.popleft()
-            seq.block_ids = self.bm.allocate(seq.prompt_len)
+            seq.block_ids, seq.cached_prefix = \
+                self.bm.allocate_for_prompt(seq.prompt_token_ids)
             seq.slot = self._free_slots.pop()
             seq.state = SeqState.RUNNING
             prefills.append(seq)

@@ -33,6 +33,8 @@ class LocalProvider:
             gpu_memory_utilization=local_cfg.gpu_memory_utilization,
             enforce_eager=local_cfg.enforce_eager,
             num_gpu_blocks=getattr(local_cfg, "num_gpu_blocks", None),
+            enable_prefix_caching=getattr(local_cfg, "prefix_caching",
+                                          False),
             capture_at_init=True,
         )
         self.engine = engine or AsyncEngine(ecfg, device=device)

@@ -575,3 +575,101 @@ def test_seeded_sampling_survives_preemption():
         assert steps < 2000
     assert eng2.scheduler.num_preemptions > 0
     assert eng2.get_output(rid2).token_ids == want
+
+
+def test_prefix_caching_equivalence_and_hits():
+    """Prefix caching must be output-invariant: identical and
+    prefix-shared prompts produce exactly the tokens a cache-less engine
+    produces, while full leading blocks are served from cache."""
+    base = list(range(1, 27))                 # 26 tokens: 1 full block
+    longer = base + list(range(100, 120))     # 46: shares base's prefix
+    sp = lambda: SamplingParams(max_new_tokens=6, temperature=0.0)
+
+    eng0 = _engine()                          # caching OFF
+    r_a0 = eng0.add_request(base, sp())
+    r_b0 = eng0.add_request(longer, sp())
+    while not eng0.is_finished():
+        eng0.step()
+    want_a = eng0.get_output(r_a0).token_ids
+    want_b = eng0.get_output(r_b0).token_ids
+
+    eng = _engine(enable_prefix_caching=True)
+    r1 = eng.add_request(base, sp())
+    while not eng.is_finished():
+        eng.step()
+    assert eng.block_manager.prefix_hit_tokens == 0
+    assert eng.get_output(r1).token_ids == want_a
+
+    # same prompt again: its 1 full block must come from the pool
+    r2 = eng.add_request(base, sp())
+    # a longer prompt sharing the 16-token prefix: 1-block hit too
+    r3 = eng.add_request(longer, sp())
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(r2).token_ids == want_a
+    assert eng.get_output(r3).token_ids == want_b
+    assert eng.block_manager.prefix_hit_tokens == 32   # 16 + 16
+
+    # the longer prompt's 2 full blocks are now cached: re-running it
+    # must hit both
+    r4 = eng.add_request(longer, sp())
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(r4).token_ids == want_b
+    assert eng.block_manager.prefix_hit_tokens == 64   # +32
+
+
+def test_prefix_caching_live_sharing_same_batch():
+    """Two identical prompts admitted together: the second shares the
+    first's LIVE blocks (KV written earlier in the same forward)."""
+    eng = _engine(enable_prefix_caching=True)
+    sp = lambda: SamplingParams(max_new_tokens=5, temperature=0.0)
+    prompt = list(range(1, 40))               # 2 full blocks + partial
+    r1 = eng.add_request(prompt, sp())
+    r2 = eng.add_request(prompt, sp())
+    while not eng.is_finished():
+        eng.step()
+    a, b = eng.get_output(r1).token_ids, eng.get_output(r2).token_ids
+    assert a == b and len(a) == 5
+    assert eng.block_manager.prefix_hit_tokens == 32
+
+
+def test_prefix_caching_block_aligned_prompt_keeps_query_row():
+    """A prompt that is an exact multiple of the block size must keep
+    >= 1 uncached token (prefill needs a query row for logits, and the
+    suffix's KV writes must stay out of shared blocks)."""
+    eng = _engine(enable_prefix_caching=True)
+    sp = lambda: SamplingParams(max_new_tokens=4, temperature=0.0)
+    prompt = list(range(1, 33))               # exactly 2 blocks
+    r1 = eng.add_request(prompt, sp())
+    while not eng.is_finished():
+        eng.step()
+    want = eng.get_output(r1).token_ids
+    r2 = eng.add_request(prompt, sp())
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(r2).token_ids == want
+    # only ONE of the two full blocks may be served from cache
+    assert eng.block_manager.prefix_hit_tokens == 16
+
+
+def test_prefix_pool_eviction_under_pressure():
+    """Retained blocks are reclaimed LRU when the allocator runs dry —
+    distinct prompts keep working and match a cache-less engine."""
+    prompts = [list(range(1 + 50 * i, 27 + 50 * i)) for i in range(6)]
+    sp = lambda: SamplingParams(max_new_tokens=4, temperature=0.0)
+
+    eng0 = _engine()
+    want = []
+    for p in prompts:
+        r = eng0.add_request(p, sp())
+        while not eng0.is_finished():
+            eng0.step()
+        want.append(eng0.get_output(r).token_ids)
+
+    eng = _engine(num_gpu_blocks=8, enable_prefix_caching=True)
+    for p, w in zip(prompts, want):
+        r = eng.add_request(p, sp())
+        while not eng.is_finished():
+            eng.step()
+        assert eng.get_output(r).token_ids == w

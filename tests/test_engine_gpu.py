@@ -249,3 +249,42 @@ def test_fork_group_on_graph_path():
     for r in rids:
         assert eng.get_output(r).token_ids == want
     assert eng.block_manager.num_free == eng.block_manager.num_blocks - 1
+
+
+def test_prefix_caching_on_gpu_matches_uncached():
+    """Prefix caching on the HIP path (paged-extend prefill kernel +
+    shared blocks + graph decode): cached repeats and prefix-shared
+    prompts must match a cache-less engine token-for-token."""
+    torch.manual_seed(0)
+    base = list(range(1, 27))
+    longer = base + list(range(100, 140))
+
+    def run(cfg, prompts):
+        torch.manual_seed(0)
+        eng = LLMEngine(cfg, device="cuda:0")
+        rids = [eng.add_request(p, SamplingParams(max_new_tokens=6,
+                                                  temperature=0.0))
+                for p in prompts]
+        steps = 0
+        while not eng.is_finished():
+            eng.step()
+            steps += 1
+            assert steps < 300
+        return eng, [eng.get_output(r).token_ids for r in rids]
+
+    _, want = run(EngineConfig.tiny_llama(use_hip_graphs=True),
+                  [base, longer])
+    eng, got1 = run(EngineConfig.tiny_llama(use_hip_graphs=True,
+                                            enable_prefix_caching=True),
+                    [base, longer])
+    assert got1 == want
+    # second wave on the SAME engine: now the pool has the blocks
+    r2 = eng.add_request(base, SamplingParams(max_new_tokens=6,
+                                              temperature=0.0))
+    r3 = eng.add_request(longer, SamplingParams(max_new_tokens=6,
+                                                temperature=0.0))
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(r2).token_ids == want[0]
+    assert eng.get_output(r3).token_ids == want[1]
+    assert eng.block_manager.prefix_hit_tokens >= 48
