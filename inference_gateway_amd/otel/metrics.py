@@ -110,6 +110,9 @@ class Telemetry:
                           "cumulative KV-pressure preemptions", registry=r)
         g_forks = Gauge("igw_engine_forks_total",
                         "cumulative n>1 KV-fork branches", registry=r)
+        g_pfx = Gauge("igw_engine_prefix_hit_tokens_total",
+                      "prompt tokens served from the prefix cache",
+                      registry=r)
         sched = engine.scheduler
         bm = engine.block_manager
         g_run.set_function(lambda: len(sched.running))
@@ -120,6 +123,7 @@ class Telemetry:
         g_prompt.set_function(lambda: engine.num_prompt_tokens)
         g_preempt.set_function(lambda: sched.num_preemptions)
         g_forks.set_function(lambda: engine.num_forks)
+        g_pfx.set_function(lambda: bm.prefix_hit_tokens)
 
     def exposition(self) -> bytes:
         return generate_latest(self.registry)
