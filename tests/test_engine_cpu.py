@@ -673,3 +673,54 @@ def test_prefix_pool_eviction_under_pressure():
         while not eng.is_finished():
             eng.step()
         assert eng.get_output(r).token_ids == w
+
+
+def test_prefix_caching_composes_with_fork_and_preemption():
+    """The three block-sharing machineries at once: a cache-hit prompt
+    forked n>1 under a pool small enough to preempt — outputs still
+    equal the plain engine's, and no blocks leak."""
+    prompt = list(range(1, 40))
+    sp = lambda n=8: SamplingParams(max_new_tokens=n, temperature=0.0)
+
+    eng0 = _engine()
+    r0 = eng0.add_request(prompt, sp())
+    while not eng0.is_finished():
+        eng0.step()
+    want = eng0.get_output(r0).token_ids
+
+    eng = _engine(num_gpu_blocks=12, max_num_seqs=8,
+                  enable_prefix_caching=True)
+    free0 = eng.block_manager.num_free
+    warm = eng.add_request(prompt, sp(2))       # seed the prefix pool
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(warm).token_ids == want[:2]
+    rids = eng.add_request_group(prompt, sp(), n=3)
+    steps = 0
+    while not eng.is_finished():
+        eng.step()
+        steps += 1
+        assert steps < 2000
+    for r in rids:
+        assert eng.get_output(r).token_ids == want
+    assert eng.block_manager.prefix_hit_tokens >= 32
+    assert eng.block_manager.num_free == free0
+
+
+def test_preempted_sequence_reuses_its_own_retained_blocks():
+    """After preemption, re-admission hash-hits the victim's own prompt
+    blocks instead of recomputing them."""
+    eng = _engine(num_gpu_blocks=6, max_num_seqs=4,
+                  enable_prefix_caching=True)
+    a = eng.add_request(list(range(1, 40)),
+                        SamplingParams(max_new_tokens=8, temperature=0.0))
+    eng.add_request(list(range(50, 76)),
+                    SamplingParams(max_new_tokens=8, temperature=0.0))
+    steps = 0
+    while not eng.is_finished():
+        eng.step()
+        steps += 1
+        assert steps < 2000
+    assert eng.scheduler.num_preemptions > 0
+    assert eng.block_manager.prefix_hit_tokens > 0
+    assert len(eng.get_output(a).token_ids) == 8
