@@ -724,3 +724,50 @@ def test_preempted_sequence_reuses_its_own_retained_blocks():
     assert eng.scheduler.num_preemptions > 0
     assert eng.block_manager.prefix_hit_tokens > 0
     assert len(eng.get_output(a).token_ids) == 8
+
+
+def test_fuzz_block_accounting_invariants():
+    """Randomized adds/forks/aborts with prefix caching on a tiny pool:
+    after every drain the block accounting must be exactly conservative
+    (every block is free, retained, or held by exactly its ref-count
+    owners) and at the end everything is reclaimable."""
+    import random as rnd
+
+    r = rnd.Random(1234)
+    eng = _engine(num_gpu_blocks=16, max_num_seqs=8,
+                  enable_prefix_caching=True)
+    bm = eng.block_manager
+    prompts = [list(range(1 + 10 * i, 24 + 10 * i)) for i in range(4)]
+    live = []
+    for round_i in range(60):
+        act = r.random()
+        if act < 0.45 and len(live) < 10:
+            p = r.choice(prompts)
+            if r.random() < 0.3:
+                live.extend(eng.add_request_group(
+                    p, SamplingParams(max_new_tokens=r.randint(2, 8),
+                                      temperature=0.8), n=2))
+            else:
+                live.append(eng.add_request(
+                    p, SamplingParams(max_new_tokens=r.randint(2, 8),
+                                      temperature=0.8)))
+        elif act < 0.55 and live:
+            eng.abort_request(r.choice(live))
+        for _ in range(r.randint(1, 4)):
+            if not eng.is_finished():
+                eng.step()
+        live = [x for x in live
+                if not eng.get_output(x).finished]
+        # invariant: refs are non-negative and every held ref is owned
+        held = {}
+        for seq in eng._seqs.values():
+            for b in seq.block_ids:
+                held[b] = held.get(b, 0) + 1
+        for b in range(1, bm.num_blocks):
+            assert bm._refs[b] >= 0
+            assert bm._refs[b] == held.get(b, 0), (round_i, b)
+    while not eng.is_finished():
+        eng.step()
+    for seq in eng._seqs.values():
+        assert seq.state.name == "FINISHED"
+    assert bm.num_free == bm.num_blocks - 1   # everything reclaimable
