@@ -84,14 +84,19 @@ class BlockManager:
 
     @staticmethod
     def chain_hashes(prompt_token_ids: List[int], block_size: int
-                     ) -> List[int]:
-        """Chained content hash per FULL block: h_i covers tokens
-        [0, (i+1)*bs) so a block's identity includes its whole prefix."""
+                     ) -> List[tuple]:
+        """Identity key per FULL block: (block index, chained crc32 over
+        the whole prefix, the block's exact tokens).  The exact last
+        chunk + position means a false hit needs a crc collision between
+        two prefixes that ALSO agree on their final block — ~2^-32 per
+        such pair instead of per any pair (a long-running server sees
+        millions of prompts; bare crc chains would eventually collide)."""
         out, h = [], 0
         for i in range(len(prompt_token_ids) // block_size):
-            chunk = prompt_token_ids[i * block_size:(i + 1) * block_size]
+            chunk = tuple(
+                prompt_token_ids[i * block_size:(i + 1) * block_size])
             h = zlib.crc32(repr(chunk).encode(), h)
-            out.append(h)
+            out.append((i, h, chunk))
         return out
 
     def allocate_for_prompt(self, prompt_token_ids: List[int]
