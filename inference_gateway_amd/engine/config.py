@@ -144,6 +144,10 @@ class EngineConfig:
     # prefix caching: hash-indexed reuse of full prompt blocks across
     # requests (suffix-only extend-prefill).  Opt-in.
     enable_prefix_caching: bool = False
+    # chunked prefill: max_batched_tokens becomes a HARD budget — a long
+    # prompt prefills in successive chunks (paged-extend attention) so a
+    # burst never stalls in-flight decodes for a whole long prefill.
+    enable_chunked_prefill: bool = False
     tensor_parallel_size: int = 1
     seed: int = 0
 

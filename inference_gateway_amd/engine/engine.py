@@ -124,7 +124,8 @@ class LLMEngine:
             enable_prefix_caching=cfg.enable_prefix_caching)
         self.scheduler = Scheduler(
             self.block_manager, cfg.max_num_seqs, cfg.max_batched_tokens,
-            cfg.max_model_len)
+            cfg.max_model_len,
+            chunked_prefill=cfg.enable_chunked_prefill)
         self.sampler = Sampler(self.device, cfg.seed)
         self.kv_caches = self._allocate_kv(num_blocks)
         self._seq_counter = itertools.count()
@@ -265,27 +266,38 @@ class LLMEngine:
 
         prev = self._drain_pending()
         if batch.is_prefill:
-            self.num_prompt_tokens += sum(s.prompt_len for s in batch.seqs)
+            self.num_prompt_tokens += sum(batch.prefill_takes)
         fb = self._build_forward_batch(batch)
         hidden = self.model(fb, self.kv_caches)
+        if batch.is_prefill:
+            for seq, take in zip(batch.seqs, batch.prefill_takes):
+                seq.prefilled += take
+            # chunked prefill: only sequences whose prompt COMPLETED this
+            # round sample a token; a partial chunk just extends KV
+            sample_seqs = [s for s in batch.seqs
+                           if s.prefilled >= s.prompt_len]
+        else:
+            sample_seqs = batch.seqs
+        if not sample_seqs:
+            return prev                  # pure partial-chunk round
         hidden_last = hidden[fb.last_token_indices]
         logits = self.model.compute_logits(hidden_last)
-        apply_penalties(logits, batch.seqs)
-        tokens = self.sampler.sample(logits, batch.seqs)
-        lp_map = _extract_logprobs(logits, batch.seqs, tokens)
+        apply_penalties(logits, sample_seqs)
+        tokens = self.sampler.sample(logits, sample_seqs)
+        lp_map = _extract_logprobs(logits, sample_seqs, tokens)
         # fork n>1 groups BEFORE parent bookkeeping (a parent finishing at
         # its first token would free the blocks the fork shares)
         fork_outputs: List[StepOutput] = []
         if batch.is_prefill and self._fork_pending:
-            for row, seq in enumerate(batch.seqs):
+            for row, seq in enumerate(sample_seqs):
                 if seq.seq_id in self._fork_pending:
                     fork_outputs.extend(
                         self._fork_at_prefill(seq, logits[row]))
-        outputs = self._append_tokens(batch, tokens, lp_map)
+        outputs = self._append_tokens(sample_seqs, tokens, lp_map)
         outputs = fork_outputs + outputs
         if batch.is_prefill and gr is not None:
             # seed the persistent decode slot state for newly admitted seqs
-            for seq, tok in zip(batch.seqs, tokens):
+            for seq, tok in zip(sample_seqs, tokens):
                 if seq.state != SeqState.FINISHED:
                     gr.write_block_row(seq.slot, seq.block_ids)
                     gr.init_slot(seq, tok)
@@ -324,19 +336,21 @@ class LLMEngine:
             # private blocks) and attention takes the paged-extend path.
             token_ids, pos_parts, slot_parts = [], [], []
             cu, seq_lens, last, prefixes = [0], [], [], []
-            for seq in batch.seqs:
+            for seq, take in zip(batch.seqs, batch.prefill_takes):
                 L = seq.prompt_len
-                p0 = seq.cached_prefix
-                token_ids.extend(seq.prompt_token_ids[p0:])
-                pos_parts.append(torch.arange(p0, L, dtype=torch.int32))
+                p0 = seq.prefilled        # cache hits + earlier chunks
+                p1 = p0 + take
+                token_ids.extend(seq.prompt_token_ids[p0:p1])
+                pos_parts.append(torch.arange(p0, p1, dtype=torch.int32))
                 blocks = torch.tensor(seq.block_ids, dtype=torch.int32)
                 slots = (blocks.repeat_interleave(bs) * bs
                          + torch.arange(bs, dtype=torch.int32).repeat(
-                             len(seq.block_ids)))[p0:L]
+                             len(seq.block_ids)))[p0:p1]
                 slot_parts.append(slots)
-                cu.append(cu[-1] + (L - p0))
-                seq_lens.append(L - p0)
-                last.append(cu[-1] - 1)
+                cu.append(cu[-1] + take)
+                seq_lens.append(take)
+                if p1 >= L:               # completes the prompt: sample
+                    last.append(cu[-1] - 1)
                 prefixes.append(p0)
             fb = ForwardBatch(
                 token_ids=torch.tensor(token_ids, dtype=torch.int64, device=dev),
@@ -457,10 +471,10 @@ class LLMEngine:
             out.logprob, out.top_logprobs = lp
         return [out]
 
-    def _append_tokens(self, batch: ScheduledBatch, tokens: List[int],
+    def _append_tokens(self, seqs: List[Sequence], tokens: List[int],
                        lp_map=None) -> List[StepOutput]:
         outputs = []
-        for i, (seq, tok) in enumerate(zip(batch.seqs, tokens)):
+        for i, (seq, tok) in enumerate(zip(seqs, tokens)):
             outputs.extend(self._apply_token(
                 seq, tok, lp_map.get(i) if lp_map else None))
         return outputs

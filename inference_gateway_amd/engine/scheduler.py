@@ -39,6 +39,9 @@ class Sequence:
     sample_tick: int = 0              # per-seq counter for seeded sampling
     cached_prefix: int = 0            # prompt tokens served from the
                                       # prefix cache (block-aligned)
+    prefilled: int = 0                # prompt tokens whose KV is already
+                                      # computed (cache hits + finished
+                                      # prefill chunks)
     # length of the ORIGINAL prompt: preemption folds generated tokens
     # into prompt_token_ids for the re-prefill, so generation length and
     # output reporting must be measured against this, not prompt_len
@@ -73,11 +76,16 @@ class ScheduledBatch:
     # decode-only: single-block growths this step, as (slot, col, block_id)
     # rows for DecodeGraphRunner.write_block_appends
     new_blocks: List[tuple] = field(default_factory=list)
+    # prefill-only: tokens each seq processes THIS round (chunked
+    # prefill: the last entry may be a partial chunk; None = whole
+    # remaining prompt for every seq)
+    prefill_takes: Optional[List[int]] = None
 
 
 class Scheduler:
     def __init__(self, block_manager: BlockManager, max_num_seqs: int,
-                 max_batched_tokens: int, max_model_len: int):
+                 max_batched_tokens: int, max_model_len: int,
+                 chunked_prefill: bool = False):
         self.bm = block_manager
         self.max_num_seqs = max_num_seqs
         self.max_batched_tokens = max_batched_tokens
@@ -93,6 +101,7 @@ class Scheduler:
         # starve in-flight streams (and in-flight streams do not starve
         # TTFT) — SURVEY.md §7 'scheduler fairness under 64 SSE clients'
         self._last_was_prefill = False
+        self.chunked_prefill = chunked_prefill
         self.num_preemptions = 0      # cumulative (metrics)
 
     def add(self, seq: Sequence) -> None:
@@ -114,29 +123,67 @@ class Scheduler:
         want_prefill = bool(self.waiting) and not (
             self._last_was_prefill and self.running)
         prefills: List[Sequence] = []
+        takes: List[int] = []
         tokens = 0
         while (want_prefill and self.waiting
                and len(self.running) + len(prefills) < self.max_num_seqs):
             seq = self.waiting[0]
-            if tokens + seq.prompt_len > self.max_batched_tokens and prefills:
+            remaining = seq.prompt_len - seq.prefilled
+            if (tokens + remaining > self.max_batched_tokens
+                    and self.chunked_prefill):
+                # CHUNKED: hard token budget — take a partial chunk of
+                # the front prompt and leave it at waiting[0]; its whole
+                # prompt's blocks were reserved at its first chunk
+                take = self.max_batched_tokens - tokens
+                if take <= 0:
+                    break
+                if seq.prefilled == 0:
+                    if not self.bm.can_allocate(seq.prompt_len):
+                        break
+                    seq.block_ids, seq.cached_prefix = \
+                        self.bm.allocate_for_prompt(seq.prompt_token_ids)
+                    seq.prefilled = seq.cached_prefix
+                remaining = seq.prompt_len - seq.prefilled
+                take = min(take, remaining)
+                if take <= 0:
+                    break
+                if take == remaining:
+                    # this chunk COMPLETES the prompt: admit now so the
+                    # same round samples its first token (a completed
+                    # seq must never sit in waiting with take 0)
+                    self.waiting.popleft()
+                    seq.slot = self._free_slots.pop()
+                    seq.state = SeqState.RUNNING
+                prefills.append(seq)
+                takes.append(take)
+                tokens += take
                 break
-            if not self.bm.can_allocate(seq.prompt_len):
+            if tokens + remaining > self.max_batched_tokens and prefills:
                 break
-            # a single prompt longer than max_batched_tokens (legal when
-            # max_model_len > max_batched_tokens) gets its own oversized
-            # batch — breaking unconditionally would starve it forever
+            if seq.prefilled == 0:
+                if not self.bm.can_allocate(seq.prompt_len):
+                    break
+                # a single prompt longer than max_batched_tokens (legal
+                # when max_model_len is larger) gets its own oversized
+                # batch when chunking is off — breaking unconditionally
+                # would starve it forever
+                seq.block_ids, seq.cached_prefix = \
+                    self.bm.allocate_for_prompt(seq.prompt_token_ids)
+                seq.prefilled = seq.cached_prefix
             self.waiting.popleft()
-            seq.block_ids, seq.cached_prefix = \
-                self.bm.allocate_for_prompt(seq.prompt_token_ids)
             seq.slot = self._free_slots.pop()
             seq.state = SeqState.RUNNING
             prefills.append(seq)
-            tokens += seq.prompt_len
+            takes.append(seq.prompt_len - seq.prefilled)
+            tokens += seq.prompt_len - seq.prefilled
         if prefills:
-            self.running.extend(prefills)
-            self.membership_version += 1
+            admitted = [q for q in prefills if q.state == SeqState.RUNNING]
+            if admitted:
+                self.running.extend(admitted)
+                self.membership_version += 1
             self._last_was_prefill = True
-            return ScheduledBatch(is_prefill=True, seqs=prefills)
+            return ScheduledBatch(is_prefill=True, seqs=prefills,
+                                  prefill_takes=takes)
         self._last_was_prefill = False
 
         if not self.running:
@@ -178,6 +225,8 @@ class Scheduler:
                 victim.prompt_token_ids + victim.output_token_ids
             )
             victim.output_token_ids = []
+            victim.prefilled = 0
+            victim.cached_prefix = 0
             victim.state = SeqState.WAITING
             self.waiting.appendleft(victim)
             if not self.running:
