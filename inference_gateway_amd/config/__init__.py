@@ -149,6 +149,7 @@ class LocalConfig:
     gpu_memory_utilization: float = 0.90
     enforce_eager: bool = False
     prefix_caching: bool = False
+    chunked_prefill: bool = False
 
 
 @dataclass
@@ -273,6 +274,7 @@ def load(env: Optional[Mapping[str, str]] = None) -> Config:
     lo.gpu_memory_utilization = float(g("LOCAL_GPU_MEMORY_UTILIZATION", "0.90"))
     lo.enforce_eager = _bool(g("LOCAL_ENFORCE_EAGER", "false"))
     lo.prefix_caching = _bool(g("LOCAL_PREFIX_CACHING", "false"))
+    lo.chunked_prefill = _bool(g("LOCAL_CHUNKED_PREFILL", "false"))
 
     # provider bootstrap: registry defaults + env overrides
     from ..providers.constants import PROVIDERS

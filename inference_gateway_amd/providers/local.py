@@ -35,6 +35,8 @@ class LocalProvider:
             num_gpu_blocks=getattr(local_cfg, "num_gpu_blocks", None),
             enable_prefix_caching=getattr(local_cfg, "prefix_caching",
                                           False),
+            enable_chunked_prefill=getattr(local_cfg, "chunked_prefill",
+                                           False),
             capture_at_init=True,
         )
         self.engine = engine or AsyncEngine(ecfg, device=device)
