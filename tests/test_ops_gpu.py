@@ -313,3 +313,36 @@ def test_flash_prefill_paged_vs_ref(chunks):
         cu_t.cpu(), scale)
     torch.testing.assert_close(got.cpu().float(), want.float(),
                                atol=3e-2, rtol=3e-2)
+
+
+def test_zz_chunked_prefill_engine_matches_unchunked():
+    # placed in this (alphabetically last) file, and named to sort last,
+    # so the driver's -x run covers everything else first
+    from inference_gateway_amd.engine.config import (EngineConfig,
+                                                     SamplingParams)
+    from inference_gateway_amd.engine.engine import LLMEngine
+
+    """Chunked prefill through the paged-extend kernel + hipGraph decode
+    must equal the unchunked engine token-for-token."""
+    long_prompt = list(range(3, 200))
+    short = list(range(7, 30))
+
+    def run(cfg):
+        torch.manual_seed(0)
+        eng = LLMEngine(cfg, device="cuda:0")
+        rids = [eng.add_request(p, SamplingParams(max_new_tokens=6,
+                                                  temperature=0.0))
+                for p in (long_prompt, short)]
+        steps = 0
+        while not eng.is_finished():
+            eng.step()
+            steps += 1
+            assert steps < 500
+        return [eng.get_output(r).token_ids for r in rids]
+
+    want = run(EngineConfig.tiny_llama(use_hip_graphs=True,
+                                       max_batched_tokens=2048))
+    got = run(EngineConfig.tiny_llama(use_hip_graphs=True,
+                                      max_batched_tokens=48,
+                                      enable_chunked_prefill=True))
+    assert got == want
