@@ -18,6 +18,7 @@ from ..engine.async_engine import AsyncEngine
 from ..engine.config import EngineConfig, MODEL_PRESETS, SamplingParams
 from ..utils.tokenizer import apply_chat_template, get_tokenizer
 from .constants import LOCAL_PROVIDER_ID
+from .core import HTTPError
 
 
 class LocalProvider:
@@ -139,9 +140,18 @@ class LocalProvider:
         return (token_ids, self.tokenizer.decode(token_ids), finish_reason,
                 lp_content)
 
+    def _check_context(self, prompt_ids) -> None:
+        limit = self.engine.engine.cfg.max_model_len
+        if len(prompt_ids) > limit:
+            # OpenAI returns 400 for context overflow, not a 5xx
+            raise HTTPError(
+                400, f"prompt is {len(prompt_ids)} tokens but the "
+                     f"model's maximum context length is {limit}")
+
     async def chat_completions(self, req: dict,
                                extra_headers=None) -> dict:
         prompt_ids = self._encode_request(req)
+        self._check_context(prompt_ids)
         sampling = self._sampling_from_request(req)
         stops = self._stop_strings(req)
         n = min(int(req.get("n") or 1), 8)
@@ -202,6 +212,7 @@ class LocalProvider:
         """Yields SSE lines ('data: {...}') matching OpenAI stream chunks;
         per-token flush (reference bar: per-line SSE flush)."""
         prompt_ids = self._encode_request(req)
+        self._check_context(prompt_ids)
         sampling = self._sampling_from_request(req)
         cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         created = int(time.time())

@@ -295,3 +295,52 @@ async def test_local_streaming_n_greater_one(local_gateway):
             assert all(f == "length" for f in finishes.values())
             assert texts[0] == texts[1] == texts[2] != ""
             assert usage["completion_tokens"] == 15
+
+
+@pytest.mark.asyncio
+async def test_local_prefix_caching_and_chunked_flags_plumbed():
+    """LOCAL_PREFIX_CACHING / LOCAL_CHUNKED_PREFILL reach the engine,
+    and serving with both on still answers correctly."""
+    cfg = config_mod.load({"LOCAL_ENABLED": "true",
+                           "LOCAL_PREFIX_CACHING": "true",
+                           "LOCAL_CHUNKED_PREFILL": "true"})
+    assert cfg.local.prefix_caching and cfg.local.chunked_prefill
+
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=4,
+                     max_model_len=256, prefix_caching=True,
+                     chunked_prefill=True)
+    lc.num_gpu_blocks = 64
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        eng = provider.engine.engine
+        assert eng.block_manager.enable_prefix_caching
+        assert eng.scheduler.chunked_prefill
+        client = httpx.AsyncClient(
+            transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+        gw = Gateway(cfg=config_mod.load({}), logger=noop_logger(),
+                     http_client=client, local_provider=provider)
+        async with gw_client(gw) as c:
+            body = {"model": "local/tiny-llama",
+                    "messages": [{"role": "user",
+                                  "content": "hello " * 20}],
+                    "max_tokens": 4, "temperature": 0}
+            r1 = await c.post("/v1/chat/completions", json=body)
+            r2 = await c.post("/v1/chat/completions", json=body)
+            assert r1.status_code == r2.status_code == 200
+            assert (r1.json()["choices"][0]["message"]["content"]
+                    == r2.json()["choices"][0]["message"]["content"])
+        assert eng.block_manager.prefix_hit_tokens > 0
+    finally:
+        provider.engine.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_local_context_overflow_is_400(local_gateway):
+    async with gw_client(local_gateway) as c:
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "x" * 5000}],
+            "max_tokens": 4,
+        })
+        assert r.status_code == 400
+        assert "maximum context length" in str(r.json()["error"])
