@@ -140,6 +140,16 @@ class Router:
         req.state["provider_id"] = pid
         req.state["model"] = model
 
+        # request-shape validation BEFORE the SSE response opens (the
+        # local engine's context-length check: OpenAI answers a plain
+        # 400, not an in-stream error frame)
+        validate = getattr(provider, "validate_request", None)
+        if validate is not None:
+            try:
+                validate(body)
+            except HTTPError as e:
+                return Response.error(e.status, e.message)
+
         if self.mcp is not None and self.mcp.applies(req, body):
             return await self.mcp.run_chat(provider, body, stream, req)
 

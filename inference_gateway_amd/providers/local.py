@@ -140,6 +140,11 @@ class LocalProvider:
         return (token_ids, self.tokenizer.decode(token_ids), finish_reason,
                 lp_content)
 
+    def validate_request(self, req: dict) -> None:
+        """Pre-stream validation hook (gateway/routes.py calls this
+        before opening the SSE response)."""
+        self._check_context(self._encode_request(req))
+
     def _check_context(self, prompt_ids) -> None:
         limit = self.engine.engine.cfg.max_model_len
         if len(prompt_ids) > limit:

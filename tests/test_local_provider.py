@@ -344,3 +344,17 @@ async def test_local_context_overflow_is_400(local_gateway):
         })
         assert r.status_code == 400
         assert "maximum context length" in str(r.json()["error"])
+
+
+@pytest.mark.asyncio
+async def test_local_stream_context_overflow_is_plain_400(local_gateway):
+    """Overflow on a STREAMING request must be a plain 400 response,
+    not an error frame inside an SSE stream."""
+    async with gw_client(local_gateway) as c:
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "x" * 5000}],
+            "max_tokens": 4, "stream": True,
+        })
+        assert r.status_code == 400
+        assert "text/event-stream" not in r.headers.get("content-type", "")
