@@ -852,3 +852,22 @@ def test_chunked_prefill_composes_with_prefix_cache_and_fork():
     for r in rids:
         assert eng.get_output(r).token_ids == want
     assert eng.block_manager.prefix_hit_tokens > 0
+
+
+def test_abort_mid_chunk_releases_blocks():
+    """Aborting a long prompt between prefill chunks (still WAITING but
+    holding its reserved blocks) must release everything."""
+    eng = _engine(max_batched_tokens=32, num_gpu_blocks=64,
+                  enable_chunked_prefill=True)
+    free0 = eng.block_manager.num_free
+    rid = eng.add_request(list(range(3, 160)),
+                          SamplingParams(max_new_tokens=4,
+                                         temperature=0.0))
+    eng.step()                                 # first chunk only
+    seq = eng._seqs[rid]
+    assert 0 < seq.prefilled < seq.prompt_len
+    assert seq.block_ids
+    eng.abort_request(rid)
+    assert eng.get_output(rid).finish_reason == "abort"
+    assert eng.block_manager.num_free == free0
+    assert not eng.scheduler.has_work
