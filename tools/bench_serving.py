@@ -28,7 +28,7 @@ sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 
 async def drive(base_url: str, transport, clients: int, max_tokens: int,
-                prompt_chars: int, model: str):
+                prompt_chars: int, model: str, shared_prefix: int = 0):
     import httpx
 
     ttfts, token_counts = [], []
@@ -39,7 +39,11 @@ async def drive(base_url: str, transport, clients: int, max_tokens: int,
         ntok = 0
         async with client.stream("POST", "/v1/chat/completions", json={
             "model": model,
-            "messages": [{"role": "user", "content": "x " * (prompt_chars // 2)}],
+            "messages": (
+                [{"role": "system", "content": "s " * (shared_prefix // 2)}]
+                if shared_prefix else []) + [
+                {"role": "user",
+                 "content": f"req {i} " + "x " * (prompt_chars // 2)}],
             "max_tokens": max_tokens,
             "temperature": 0.8,
             "top_k": 50,
@@ -117,7 +121,11 @@ async def run_local(args):
                LOCAL_ENABLED="true",
                LOCAL_MODEL=args.model,
                LOCAL_MAX_NUM_SEQS=str(max(args.clients, 8)),
-               LOCAL_MAX_MODEL_LEN="2048",
+               LOCAL_MAX_MODEL_LEN="4096",
+               LOCAL_PREFIX_CACHING=(
+                   "true" if args.prefix_caching else "false"),
+               LOCAL_CHUNKED_PREFILL=(
+                   "true" if args.chunked_prefill else "false"),
                SERVER_PORT=str(args.port))
     proc = await asyncio.create_subprocess_exec(
         sys.executable, "-m", "inference_gateway_amd", env=env,
@@ -136,7 +144,8 @@ async def run_local(args):
             else:
                 raise RuntimeError("server never became healthy")
         return await drive(base, None, args.clients, args.max_tokens,
-                           args.prompt_chars, f"local/{args.model}")
+                           args.prompt_chars, f"local/{args.model}",
+                           shared_prefix=args.shared_prefix)
     finally:
         proc.terminate()
         await proc.wait()
@@ -195,6 +204,14 @@ def main():
     ap.add_argument("--clients", type=int, default=64)
     ap.add_argument("--max-tokens", type=int, default=128)
     ap.add_argument("--prompt-chars", type=int, default=512)
+    ap.add_argument("--shared-prefix", type=int, default=0,
+                    help="chars of a SHARED system prompt per request "
+                         "(prefix-cache workload; pair with "
+                         "--prefix-caching to measure the benefit)")
+    ap.add_argument("--prefix-caching", action="store_true",
+                    help="run the engine with LOCAL_PREFIX_CACHING=true")
+    ap.add_argument("--chunked-prefill", action="store_true",
+                    help="run the engine with LOCAL_CHUNKED_PREFILL=true")
     ap.add_argument("--port", type=int, default=18080)
     ap.add_argument("--client-procs", type=int, default=1,
                     help="drive the load from N OS processes (a single "
