@@ -90,6 +90,39 @@ def _tp_engine_worker(rank: int, world: int):
     return eng.get_output(rid).token_ids
 
 
+def _tp_engine_features_worker(rank: int, world: int):
+    # TP2 with prefix caching + chunked prefill + a seeded request +
+    # an n>1 fork group: every rank must make IDENTICAL block/scheduler
+    # decisions (SPMD) and sample identical tokens
+    from inference_gateway_amd.engine.config import (EngineConfig,
+                                                     SamplingParams)
+    from inference_gateway_amd.engine.engine import LLMEngine
+    from inference_gateway_amd.parallel import comm
+
+    comm.init_tp(world)
+    torch.manual_seed(0)
+    eng = LLMEngine(EngineConfig.tiny_llama(
+        max_batched_tokens=48, enable_chunked_prefill=True,
+        enable_prefix_caching=True), device="cpu")
+    long_prompt = list(range(3, 120))
+    r1 = eng.add_request(long_prompt,
+                         SamplingParams(max_new_tokens=5, temperature=0.0))
+    r2 = eng.add_request(long_prompt,
+                         SamplingParams(max_new_tokens=5, temperature=1.0,
+                                        seed=42))
+    group = eng.add_request_group(
+        list(range(7, 40)),
+        SamplingParams(max_new_tokens=4, temperature=0.0), n=2)
+    n = 0
+    while not eng.is_finished():
+        eng.step()
+        n += 1
+        assert n < 1000
+    return [eng.get_output(r).token_ids
+            for r in [r1, r2] + group] + [
+                [eng.block_manager.prefix_hit_tokens]]
+
+
 def _ep_moe_worker(rank: int, world: int):
     from inference_gateway_amd.engine.config import MODEL_PRESETS
     from inference_gateway_amd.engine.models.mixtral import MixtralMoE
@@ -137,6 +170,14 @@ def test_tp2_engine_generation_matches_tp1(tmp_path):
     assert outs[0] == outs[1], "TP ranks must sample identical tokens"
     ref = _single_rank_reference(_tp_engine_worker)
     assert outs[0] == ref, "TP=2 greedy decode must match TP=1"
+
+
+def test_tp2_engine_features_match(tmp_path):
+    outs = _spawn("_tp_engine_features_worker", tmp_path)
+    assert outs[0] == outs[1], "TP ranks must agree on every feature path"
+    ref = _single_rank_reference(_tp_engine_features_worker)
+    assert outs[0] == ref, "TP=2 must match TP=1 across all features"
+    assert outs[0][-1][0] > 0, "prefix cache must have hit"
 
 
 def test_ep2_moe_matches_single_rank(tmp_path):
