@@ -158,7 +158,11 @@ async def run_local_multi(args):
                LOCAL_ENABLED="true",
                LOCAL_MODEL=args.model,
                LOCAL_MAX_NUM_SEQS=str(max(args.clients, 8)),
-               LOCAL_MAX_MODEL_LEN="2048",
+               LOCAL_MAX_MODEL_LEN="4096",
+               LOCAL_PREFIX_CACHING=(
+                   "true" if args.prefix_caching else "false"),
+               LOCAL_CHUNKED_PREFILL=(
+                   "true" if args.chunked_prefill else "false"),
                SERVER_PORT=str(args.port))
     proc = await asyncio.create_subprocess_exec(
         sys.executable, "-m", "inference_gateway_amd", env=env,
@@ -181,6 +185,7 @@ async def run_local_multi(args):
             "--model", args.model, "--clients", str(per),
             "--max-tokens", str(args.max_tokens),
             "--prompt-chars", str(args.prompt_chars),
+            "--shared-prefix", str(args.shared_prefix),
             stdout=subprocess.PIPE)
             for _ in range(args.client_procs)]
         t0 = time.perf_counter()
@@ -225,7 +230,8 @@ def main():
         async def child():
             base = f"http://127.0.0.1:{args.port}"
             return await drive(base, None, args.clients, args.max_tokens,
-                               args.prompt_chars, f"local/{args.model}")
+                               args.prompt_chars, f"local/{args.model}",
+                               shared_prefix=args.shared_prefix)
 
         t, c, w = asyncio.run(child())
         print(json.dumps({"ttfts": t, "tokens": c, "wall": w}))
