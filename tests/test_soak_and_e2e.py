@@ -243,3 +243,59 @@ async def test_soak_mixed_sampling_features():
         assert len(set(texts)) == 1
     finally:
         provider.engine.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_mcp_agent_over_real_local_engine(tmp_path, monkeypatch):
+    """MCP layer + the REAL in-process engine: the agent loop offers
+    tools to a random-init model; whatever it emits (never a valid tool
+    call) the loop must terminate cleanly with a normal completion and
+    exactly one [DONE] when streaming."""
+    monkeypatch.setenv("SANDBOX_DIR", str(tmp_path))
+    fs = _load_fs_server()
+
+    def handler(request: httpx.Request) -> httpx.Response:
+        if "/mcp" in request.url.path:
+            return httpx.Response(
+                200, json=fs.handle_rpc(json.loads(request.content)))
+        return httpx.Response(404)
+
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=4,
+                     max_model_len=256)
+    lc.num_gpu_blocks = 64
+    provider = LocalProvider(lc, device="cpu")
+    cfg = config_mod.load({
+        "MCP_ENABLED": "true", "MCP_EXPOSE": "true",
+        "MCP_SERVERS": "http://fs/mcp",
+    })
+    client = httpx.AsyncClient(transport=httpx.MockTransport(handler))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    try:
+        await gw.mcp.initialize()
+        assert gw.mcp.client.all_tools(), "MCP tools must discover"
+        async with gw_client(gw) as c:
+
+            r = await c.post("/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "list files"}],
+                "max_tokens": 6, "temperature": 0,
+            })
+            assert r.status_code == 200
+            body = r.json()
+            assert body["choices"][0]["finish_reason"] in ("stop",
+                                                           "length")
+
+            done = 0
+            async with c.stream("POST", "/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "read a file"}],
+                "max_tokens": 6, "temperature": 0, "stream": True,
+            }) as rs:
+                assert rs.status_code == 200
+                async for line in rs.aiter_lines():
+                    if line.strip() == "data: [DONE]":
+                        done += 1
+            assert done == 1
+    finally:
+        provider.engine.shutdown()
