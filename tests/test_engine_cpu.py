@@ -871,3 +871,29 @@ def test_abort_mid_chunk_releases_blocks():
     assert eng.get_output(rid).finish_reason == "abort"
     assert eng.block_manager.num_free == free0
     assert not eng.scheduler.has_work
+
+
+def test_moe_engine_with_chunked_prefix_and_fork():
+    """Mixtral path (MoE MLP + shared LlamaAttention): chunked prefill,
+    prefix caching and n>1 forks must be output-invariant there too."""
+    from inference_gateway_amd.engine.config import EngineConfig as EC
+
+    def run(cfg, prompt, n=1):
+        torch.manual_seed(0)
+        eng = LLMEngine(cfg, device="cpu")
+        sp = SamplingParams(max_new_tokens=4, temperature=0.0)
+        rids = (eng.add_request_group(prompt, sp, n=n) if n > 1
+                else [eng.add_request(prompt, sp)])
+        steps = 0
+        while not eng.is_finished():
+            eng.step()
+            steps += 1
+            assert steps < 1000
+        return [eng.get_output(r).token_ids for r in rids]
+
+    prompt = list(range(3, 120))
+    want = run(EC.tiny_moe(), prompt)[0]
+    got = run(EC.tiny_moe(max_batched_tokens=48,
+                          enable_chunked_prefill=True,
+                          enable_prefix_caching=True), prompt, n=2)
+    assert all(g == want for g in got)
