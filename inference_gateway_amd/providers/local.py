@@ -290,8 +290,21 @@ class LocalProvider:
                     yield chunk({"content": emit}, lp=lp)
                 continue
             yield chunk({"content": text}, lp=lp)
-        if not stopped and emitted_tail:
-            yield chunk({"content": emitted_tail})
+        if not stopped:
+            # flush whatever is still held back: the stop-prefix tail AND
+            # any pending tokens that never completed a UTF-8 sequence
+            # (emit them with replacement chars — dropping tokens would
+            # leave a stream with content missing entirely)
+            tail_text = self.tokenizer.decode(pending) if pending else ""
+            leftover = emitted_tail + tail_text
+            lp = pending_lp if want_lp and pending_lp else None
+            if stops and leftover:
+                i = self._find_stop(leftover, stops)
+                if i >= 0:
+                    leftover = leftover[:i]
+                    finish_reason = "stop"
+            if leftover:
+                yield chunk({"content": leftover}, lp=lp)
         usage = {
             "prompt_tokens": len(prompt_ids),
             "completion_tokens": ntok,
@@ -360,9 +373,19 @@ class LocalProvider:
                                 chunk(idx, {"content": emit}, lp=lp))
                         continue
                     await queue.put(chunk(idx, {"content": text}, lp=lp))
-                if emitted_tail:
-                    await queue.put(chunk(idx,
-                                          {"content": emitted_tail}))
+                # flush the held-back tail + any incomplete-UTF-8 pending
+                tail_text = (self.tokenizer.decode(pending) if pending
+                             else "")
+                leftover = emitted_tail + tail_text
+                if stops and leftover:
+                    i = self._find_stop(leftover, stops)
+                    if i >= 0:
+                        leftover = leftover[:i]
+                        finish_reason = "stop"
+                if leftover:
+                    await queue.put(chunk(
+                        idx, {"content": leftover},
+                        lp=pending_lp if want_lp and pending_lp else None))
             finally:
                 try:
                     # a stop-string break leaves the branch generating;

@@ -358,3 +358,42 @@ async def test_local_stream_context_overflow_is_plain_400(local_gateway):
         })
         assert r.status_code == 400
         assert "text/event-stream" not in r.headers.get("content-type", "")
+
+
+@pytest.mark.asyncio
+async def test_stream_flushes_incomplete_utf8_tail(local_gateway,
+                                                   monkeypatch):
+    """Regression (soak flake): tokens that never complete a UTF-8
+    sequence were held back forever and silently dropped at stream end —
+    the final flush must emit them (as replacement chars)."""
+    provider = local_gateway.local_provider
+
+    class _Out:
+        def __init__(self, tok, fin):
+            self.token_id = tok
+            self.finished = fin
+            self.finish_reason = "length" if fin else None
+            self.logprob = None
+            self.top_logprobs = None
+
+    async def fake_generate(prompt_ids, sampling):
+        # 0x80 continuation bytes (+3 special offset): every prefix of
+        # this stream ends mid-UTF-8, so nothing flushes until the end
+        toks = [0x80 + 3, 0x81 + 3, 0x82 + 3, 0x83 + 3]
+        for i, t in enumerate(toks):
+            yield _Out(t, i == len(toks) - 1)
+
+    monkeypatch.setattr(provider.engine, "generate", fake_generate)
+    content = ""
+    chunks = 0
+    async for line in provider.stream_chat_completions({
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "x"}],
+            "max_tokens": 4}):
+        if not line.startswith("data: ") or line == "data: [DONE]":
+            continue
+        ch = json.loads(line[6:])["choices"][0]
+        content += ch["delta"].get("content") or ""
+        chunks += 1
+    assert "�" in content          # tokens surfaced, not dropped
+    assert chunks >= 3                  # role + content + usage
