@@ -264,15 +264,18 @@ class LocalProvider:
             pending.append(out.token_id)
             if want_lp:
                 pending_lp.append(self._lp_item(out))
+            if out.finished:
+                # record BEFORE the holdback continue: a length-finish
+                # whose tail is an incomplete UTF-8 run must not report
+                # "stop"
+                finish_reason = ("stop" if out.finish_reason == "stop"
+                                 else "length")
             text = self.tokenizer.decode(pending)
             # hold back partial UTF-8 sequences (byte tokenizer)
             if text.endswith("\ufffd"):
                 continue
             pending = []
             lp, pending_lp = (pending_lp, []) if want_lp else (None, [])
-            if out.finished:
-                finish_reason = ("stop" if out.finish_reason == "stop"
-                                 else "length")
             if stops:
                 buf = emitted_tail + text
                 i = self._find_stop(buf, stops)
@@ -347,15 +350,15 @@ class LocalProvider:
                     pending.append(out.token_id)
                     if want_lp:
                         pending_lp.append(self._lp_item(out))
+                    if out.finished:
+                        finish_reason = ("stop" if out.finish_reason ==
+                                         "stop" else "length")
                     text = self.tokenizer.decode(pending)
                     if text.endswith("�"):
                         continue
                     pending = []
                     lp = pending_lp if want_lp else None
                     pending_lp = []
-                    if out.finished:
-                        finish_reason = ("stop" if out.finish_reason ==
-                                         "stop" else "length")
                     if stops:
                         buf = emitted_tail + text
                         i = self._find_stop(buf, stops)

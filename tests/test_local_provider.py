@@ -395,5 +395,8 @@ async def test_stream_flushes_incomplete_utf8_tail(local_gateway,
         ch = json.loads(line[6:])["choices"][0]
         content += ch["delta"].get("content") or ""
         chunks += 1
+        if ch.get("finish_reason"):
+            fin = ch["finish_reason"]
     assert "�" in content          # tokens surfaced, not dropped
     assert chunks >= 3                  # role + content + usage
+    assert fin == "length"              # not misreported as "stop"
