@@ -897,3 +897,58 @@ def test_moe_engine_with_chunked_prefix_and_fork():
                           enable_chunked_prefill=True,
                           enable_prefix_caching=True), prompt, n=2)
     assert all(g == want for g in got)
+
+
+@pytest.mark.asyncio
+async def test_async_engine_shutdown_unblocks_streams():
+    """shutdown() while a stream is mid-generation must deliver the
+    sentinel so the consumer ends instead of hanging."""
+    import asyncio
+
+    from inference_gateway_amd.engine.async_engine import AsyncEngine
+    from inference_gateway_amd.engine.config import EngineConfig
+
+    eng = AsyncEngine(EngineConfig.tiny_llama(), device="cpu")
+    got = []
+
+    async def consume():
+        async for out in eng.generate(
+                list(range(1, 20)),
+                SamplingParams(max_new_tokens=10 ** 6, temperature=0.8,
+                               ignore_eos=True)):
+            got.append(out.token_id)
+            if len(got) == 3:
+                eng.shutdown()
+    await asyncio.wait_for(consume(), timeout=30)
+    assert len(got) >= 3
+
+
+@pytest.mark.asyncio
+async def test_async_engine_death_closes_queues():
+    """A fatal error on the engine thread must not strand consumers:
+    every open stream gets its sentinel and the exception is recorded."""
+    import asyncio
+
+    from inference_gateway_amd.engine.async_engine import AsyncEngine
+    from inference_gateway_amd.engine.config import EngineConfig
+
+    eng = AsyncEngine(EngineConfig.tiny_llama(), device="cpu")
+
+    def boom():
+        raise RuntimeError("injected engine fault")
+
+    n = 0
+    gen = eng.generate(list(range(1, 20)),
+                       SamplingParams(max_new_tokens=10 ** 6,
+                                      temperature=0.8, ignore_eos=True))
+
+    async def consume():
+        nonlocal n
+        async for _ in gen:
+            n += 1
+            if n == 2:
+                eng.engine.step = boom    # next loop iteration dies
+    await asyncio.wait_for(consume(), timeout=30)
+    assert n >= 2
+    assert isinstance(eng._fatal, RuntimeError)
+    eng.shutdown()
