@@ -952,3 +952,29 @@ async def test_async_engine_death_closes_queues():
     assert n >= 2
     assert isinstance(eng._fatal, RuntimeError)
     eng.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_async_engine_batched_delivery_mode(monkeypatch):
+    """IGW_BATCH_DELIVER=1 (the measured-slower opt-in experiment path)
+    still delivers every token and the sentinel."""
+    import asyncio
+
+    from inference_gateway_amd.engine.async_engine import AsyncEngine
+    from inference_gateway_amd.engine.config import EngineConfig
+
+    monkeypatch.setenv("IGW_BATCH_DELIVER", "1")
+    eng = AsyncEngine(EngineConfig.tiny_llama(), device="cpu")
+    assert eng._batched_delivery
+    try:
+        toks = []
+
+        async def consume():
+            async for out in eng.generate(
+                    list(range(1, 20)),
+                    SamplingParams(max_new_tokens=6, temperature=0.0)):
+                toks.append(out.token_id)
+        await asyncio.wait_for(consume(), timeout=30)
+        assert len(toks) == 6
+    finally:
+        eng.shutdown()
