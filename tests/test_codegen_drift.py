@@ -73,3 +73,29 @@ def test_every_provider_routes_and_builds():
                 "Authorization", "").startswith("Bearer ")
         elif meta.auth_type == "xheader":
             assert "x-api-key" in {k.lower() for k in p.auth_headers()}
+
+
+def test_example_env_vars_are_real():
+    """Every IGW-looking env var referenced by the examples must exist
+    in the generated Configurations.md (catches stacks drifting from the
+    config contract, e.g. a renamed LOCAL_* var)."""
+    import re
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    documented = set(re.findall(r"`([A-Z][A-Z0-9_]{2,})`",
+                                (root / "Configurations.md").read_text()))
+    # provider bootstrap vars are <ID>_API_URL/KEY patterns
+    dynamic = re.compile(r".*_API_(URL|KEY)$")
+    prefixes = ("LOCAL_", "MCP_", "SERVER_", "TELEMETRY_", "GUARDRAILS_",
+                "AUTH_", "OIDC_", "ALLOWED_", "DISALLOWED_", "ENVIRONMENT")
+    missing = []
+    for f in list(root.glob("examples/**/*.yml")) + \
+            list(root.glob("examples/**/*.yaml")) + [root / ".env.example"]:
+        for var in re.findall(r"\b([A-Z][A-Z0-9_]{2,})=", f.read_text()):
+            if not var.startswith(prefixes):
+                continue
+            if var in documented or dynamic.match(var):
+                continue
+            missing.append(f"{f.relative_to(root)}: {var}")
+    assert not missing, f"undocumented env vars in examples: {missing}"
