@@ -400,3 +400,23 @@ async def test_stream_flushes_incomplete_utf8_tail(local_gateway,
     assert "�" in content          # tokens surfaced, not dropped
     assert chunks >= 3                  # role + content + usage
     assert fin == "length"              # not misreported as "stop"
+
+
+@pytest.mark.asyncio
+async def test_local_tolerates_tools_and_extra_params(local_gateway):
+    """OpenAI clients send tools/response_format/user etc. — the local
+    provider must accept and ignore what it cannot honor."""
+    async with gw_client(local_gateway) as c:
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 3,
+            "tools": [{"type": "function", "function": {
+                "name": "f", "parameters": {"type": "object"}}}],
+            "tool_choice": "auto",
+            "response_format": {"type": "text"},
+            "user": "tester",
+            "parallel_tool_calls": True,
+        })
+        assert r.status_code == 200
+        assert r.json()["usage"]["completion_tokens"] == 3
