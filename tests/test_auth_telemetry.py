@@ -224,3 +224,37 @@ async def test_otlp_json_ingestion():
     text = gw.telemetry.exposition().decode()
     assert "gen_ai_client_operation_duration_seconds_count" in text
     assert "evil_attr" not in text
+
+
+@pytest.mark.asyncio
+async def test_otlp_ingestion_caps_and_content_guards():
+    """Reference parity on the ingestion guards: body cap (413), bad
+    gzip (400), wrong content type (415), unknown metric rejected."""
+    import gzip as _gzip
+
+    gw, _ = make_telemetry_gateway()
+    async with gw_client(gw) as c:
+        big = b"{" + b" " * (6 * 1024 * 1024)
+        r = await c.post("/v1/metrics", content=big,
+                         headers={"content-type": "application/json"})
+        assert r.status_code == 413
+
+        r = await c.post("/v1/metrics", content=b"not-gzip",
+                         headers={"content-type": "application/json",
+                                  "content-encoding": "gzip"})
+        assert r.status_code == 400
+
+        r = await c.post("/v1/metrics", content=b"<xml/>",
+                         headers={"content-type": "text/xml"})
+        assert r.status_code == 415
+
+        payload = {"resourceMetrics": [{"scopeMetrics": [{"metrics": [
+            {"name": "not.a.known.metric",
+             "histogram": {"aggregationTemporality": 2,
+                           "dataPoints": []}}]}]}]}
+        body = _gzip.compress(__import__("json").dumps(payload).encode())
+        r = await c.post("/v1/metrics", content=body,
+                         headers={"content-type": "application/json",
+                                  "content-encoding": "gzip"})
+        assert r.status_code == 200
+        assert r.json()["rejected"] == 1 and r.json()["accepted"] == 0
