@@ -433,7 +433,11 @@ class LLMEngine:
             if not ok:
                 # not enough slots/blocks right now: plain re-prefill
                 # (the sampled first token is discarded — the child's own
-                # prefill will sample one)
+                # prefill will sample one, so give back the seed tick the
+                # discarded draw consumed or seeded branches would
+                # diverge under pressure)
+                if child.sampling.seed is not None:
+                    child.sample_tick = max(0, child.sample_tick - 1)
                 self.scheduler.add(child)
                 continue
             if tail:

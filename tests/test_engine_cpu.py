@@ -435,7 +435,21 @@ def test_fork_seeded_branches_are_identical_and_unseeded_diverge():
 
 def test_fork_fallback_when_slots_exhausted():
     """More branches than slots: the overflow branches re-prefill as
-    ordinary requests and still complete correctly."""
+    ordinary requests and still complete correctly; SEEDED groups stay
+    identical even when some branches fall back (the discarded draw's
+    seed tick is returned)."""
+    eng_s = _engine()
+    # an extra occupant forces at least one branch down the fallback path
+    eng_s.add_request(list(range(40, 60)),
+                      SamplingParams(max_new_tokens=12, temperature=0.0))
+    rids_s = eng_s.add_request_group(
+        list(range(1, 27)),
+        SamplingParams(max_new_tokens=4, temperature=1.0, seed=5), n=8)
+    while not eng_s.is_finished():
+        eng_s.step()
+    outs_s = [tuple(eng_s.get_output(r).token_ids) for r in rids_s]
+    assert len(set(outs_s)) == 1, "seeded branches must be identical"
+
     eng = _engine()                    # max_num_seqs = 8
     rids = eng.add_request_group(
         list(range(1, 27)),
