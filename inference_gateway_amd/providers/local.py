@@ -43,6 +43,9 @@ class LocalProvider:
         self.engine = engine or AsyncEngine(ecfg, device=device)
         self.tokenizer = get_tokenizer(
             preset.vocab_size, getattr(local_cfg, "tokenizer_path", None))
+        # OpenAI system_fingerprint: identifies the backend configuration
+        # (determinism domain for `seed`)
+        self.fingerprint = f"fp_igw-amd_{local_cfg.model}"
 
     # ---- provider interface --------------------------------------------
     async def list_models(self) -> dict:
@@ -177,6 +180,7 @@ class LocalProvider:
             "object": "chat.completion",
             "created": int(time.time()),
             "model": f"{self.id}/{self.model_name}",
+            "system_fingerprint": self.fingerprint,
             "choices": [{
                 "index": i,
                 "message": {"role": "assistant", "content": text},
@@ -197,6 +201,7 @@ class LocalProvider:
             "object": "chat.completion",
             "created": int(time.time()),
             "model": f"{self.id}/{self.model_name}",
+            "system_fingerprint": self.fingerprint,
             "choices": [{
                 "index": 0,
                 "message": {"role": "assistant", "content": text},
@@ -229,6 +234,7 @@ class LocalProvider:
         prefix = ("data: " + json.dumps({
             "id": cid, "object": "chat.completion.chunk",
             "created": created, "model": model,
+            "system_fingerprint": self.fingerprint,
         }, separators=(",", ":"))[:-1] + ',"choices":[{"index":0,"delta":')
 
         def chunk(delta: dict, finish: Optional[str] = None,
@@ -332,7 +338,9 @@ class LocalProvider:
             if lp is not None:
                 ch["logprobs"] = {"content": lp}
             frame = {"id": cid, "object": "chat.completion.chunk",
-                     "created": created, "model": model, "choices": [ch]}
+                     "created": created, "model": model,
+                     "system_fingerprint": self.fingerprint,
+                     "choices": [ch]}
             if usage is not None:
                 frame["usage"] = usage
             return "data: " + json.dumps(frame, separators=(",", ":"))
