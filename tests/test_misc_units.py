@@ -55,3 +55,27 @@ def test_module_entrypoint_version():
         capture_output=True, text=True, timeout=120)
     assert out.returncode == 0
     assert "inference-gateway-amd" in (out.stdout + out.stderr)
+
+
+def test_hf_tokenizer_with_real_tokenizer_file(tmp_path):
+    """HFTokenizer branch with an actual tokenizers-built tokenizer.json
+    (the offline wheel builds one programmatically): special tokens
+    discovered, encode/decode round trip."""
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+
+    vocab = {"<s>": 0, "</s>": 1, "[UNK]": 2,
+             "hello": 3, "world": 4, "foo": 5}
+    tok = Tokenizer(WordLevel(vocab, unk_token="[UNK]"))
+    tok.pre_tokenizer = Whitespace()
+    path = str(tmp_path / "tokenizer.json")
+    tok.save(path)
+
+    from inference_gateway_amd.utils.tokenizer import get_tokenizer
+
+    t = get_tokenizer(128256, path)
+    assert t.bos_token_id == 0 and t.eos_token_id == 1
+    ids = t.encode("hello world")
+    assert ids[0] == 0                      # BOS prepended
+    assert t.decode(ids[1:]) == "hello world"
