@@ -150,6 +150,7 @@ class LocalConfig:
     enforce_eager: bool = False
     prefix_caching: bool = False
     chunked_prefill: bool = False
+    tokenizer_path: str = ""          # tokenizers json; "" = byte fallback
 
 
 @dataclass
@@ -275,6 +276,7 @@ def load(env: Optional[Mapping[str, str]] = None) -> Config:
     lo.enforce_eager = _bool(g("LOCAL_ENFORCE_EAGER", "false"))
     lo.prefix_caching = _bool(g("LOCAL_PREFIX_CACHING", "false"))
     lo.chunked_prefill = _bool(g("LOCAL_CHUNKED_PREFILL", "false"))
+    lo.tokenizer_path = g("LOCAL_TOKENIZER_PATH", "")
 
     # provider bootstrap: registry defaults + env overrides
     from ..providers.constants import PROVIDERS

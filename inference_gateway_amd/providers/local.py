@@ -42,7 +42,8 @@ class LocalProvider:
         )
         self.engine = engine or AsyncEngine(ecfg, device=device)
         self.tokenizer = get_tokenizer(
-            preset.vocab_size, getattr(local_cfg, "tokenizer_path", None))
+            preset.vocab_size,
+            getattr(local_cfg, "tokenizer_path", None) or None)
         # OpenAI system_fingerprint: identifies the backend configuration
         # (determinism domain for `seed`)
         self.fingerprint = f"fp_igw-amd_{local_cfg.model}"

@@ -420,3 +420,43 @@ async def test_local_tolerates_tools_and_extra_params(local_gateway):
         })
         assert r.status_code == 200
         assert r.json()["usage"]["completion_tokens"] == 3
+
+
+@pytest.mark.asyncio
+async def test_local_real_tokenizer_end_to_end(tmp_path):
+    """LOCAL_TOKENIZER_PATH plumbs a real tokenizers file into the
+    provider and serving works with it."""
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+
+    vocab = {"<s>": 0, "</s>": 1, "[UNK]": 2, "hi": 3, "there": 4}
+    tok = Tokenizer(WordLevel(vocab, unk_token="[UNK]"))
+    tok.pre_tokenizer = Whitespace()
+    path = str(tmp_path / "tokenizer.json")
+    tok.save(path)
+
+    cfg = config_mod.load({"LOCAL_TOKENIZER_PATH": path})
+    assert cfg.local.tokenizer_path == path
+
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=4,
+                     max_model_len=256, tokenizer_path=path)
+    lc.num_gpu_blocks = 64
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        assert provider.tokenizer.bos_token_id == 0
+        client = httpx.AsyncClient(
+            transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+        gw = Gateway(cfg=config_mod.load({}), logger=noop_logger(),
+                     http_client=client, local_provider=provider)
+        async with gw_client(gw) as c:
+            r = await c.post("/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "hi there"}],
+                "max_tokens": 3, "temperature": 0,
+            })
+            assert r.status_code == 200
+            assert isinstance(
+                r.json()["choices"][0]["message"]["content"], str)
+    finally:
+        provider.engine.shutdown()
