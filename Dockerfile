@@ -10,6 +10,7 @@ FROM python:3.10-slim AS gateway
 WORKDIR /app
 COPY inference_gateway_amd/ inference_gateway_amd/
 COPY spec/ spec/
+COPY examples/mcp-servers/ examples/mcp-servers/
 RUN pip install --no-cache-dir httpx uvicorn pyyaml
 ENV SERVER_HOST=0.0.0.0 SERVER_PORT=8080
 EXPOSE 8080 9464
