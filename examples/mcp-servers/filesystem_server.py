@@ -5,7 +5,7 @@ main.go:192-447).  Serves the MCP `initialize` / `tools/list` /
 `tools/call` methods over plain ASGI; run with:
 
     SANDBOX_DIR=/tmp/fs python -m uvicorn \
-        examples.mcp-servers.filesystem_server:app --port 8083
+        --app-dir examples/mcp-servers filesystem_server:app --port 8083
 
 The gateway's MCP tests also mount `app` in-process (httpx ASGITransport)
 so the client/agent loop is exercised against a real server
