@@ -16,6 +16,19 @@ from .config import EngineConfig, SamplingParams
 from .engine import LLMEngine, StepOutput
 
 
+class EngineDeadError(RuntimeError):
+    """The engine thread died; in-flight and future requests fail with
+    this (the gateway maps it to an in-stream error frame / 502 — a crash
+    must be distinguishable from a clean finish_reason='stop')."""
+
+
+class _Fatal:
+    """Queue sentinel carrying the engine thread's fatal exception."""
+
+    def __init__(self, exc: BaseException):
+        self.exc = exc
+
+
 class AsyncEngine:
     _SENTINEL = object()
 
@@ -50,12 +63,16 @@ class AsyncEngine:
             self._close_all_queues()
 
     def _close_all_queues(self) -> None:
+        # engine death pushes a _Fatal (readers raise EngineDeadError);
+        # clean shutdown pushes the plain sentinel (readers just stop)
+        item = (_Fatal(self._fatal) if self._fatal is not None
+                else self._SENTINEL)
         with self._lock:
             items = list(self._queues.items())
         for rid, q in items:
             loop = self._loops.get(rid)
             if loop is not None:
-                loop.call_soon_threadsafe(q.put_nowait, self._SENTINEL)
+                loop.call_soon_threadsafe(q.put_nowait, item)
 
     def _run_loop(self) -> None:
         while not self._stop:
@@ -105,6 +122,9 @@ class AsyncEngine:
                        ) -> AsyncIterator[StepOutput]:
         loop = asyncio.get_running_loop()
         q: asyncio.Queue = asyncio.Queue()
+        if self._fatal is not None:
+            raise EngineDeadError(
+                f"engine thread died: {self._fatal!r}") from self._fatal
         with self._lock:
             rid = self.engine.add_request(prompt_token_ids, sampling)
             self._queues[rid] = q
@@ -115,6 +135,9 @@ class AsyncEngine:
                 item = await q.get()
                 if item is self._SENTINEL:
                     break
+                if isinstance(item, _Fatal):
+                    raise EngineDeadError(
+                        f"engine thread died: {item.exc!r}") from item.exc
                 yield item
         finally:
             with self._lock:
@@ -129,6 +152,9 @@ class AsyncEngine:
         n async iterators; each consumer's exit aborts its own branch."""
         loop = asyncio.get_running_loop()
         qs = [asyncio.Queue() for _ in range(n)]
+        if self._fatal is not None:
+            raise EngineDeadError(
+                f"engine thread died: {self._fatal!r}") from self._fatal
         with self._lock:
             rids = self.engine.add_request_group(prompt_token_ids,
                                                  sampling, n)
@@ -143,6 +169,10 @@ class AsyncEngine:
                     item = await q.get()
                     if item is self._SENTINEL:
                         break
+                    if isinstance(item, _Fatal):
+                        raise EngineDeadError(
+                            f"engine thread died: {item.exc!r}"
+                        ) from item.exc
                     yield item
             finally:
                 with self._lock:

@@ -143,6 +143,18 @@ class BlockManager:
             raise
         return blocks, prefix
 
+    def drop_unwritten(self, block_ids: List[int], written_tokens: int) -> None:
+        """Un-index blocks whose KV was never fully computed.  A sequence
+        aborted mid-chunked-prefill has hash-indexed ALL its full prompt
+        blocks at allocation time (allocate_for_prompt) but only written KV
+        for the first `written_tokens`; freeing it as-is would retain
+        garbage-KV blocks in the prefix pool, and a later prompt sharing
+        the prefix would silently generate from them.  Blocks below
+        written_tokens//block_size are written (cache hits or completed
+        chunks) and keep their hashes."""
+        for i in range(written_tokens // self.block_size, len(block_ids)):
+            self._drop_hash(block_ids[i])
+
     def ensure_capacity(self, block_ids: List[int], num_tokens: int) -> List[int]:
         """Grow the block list until it can hold num_tokens KV slots."""
         while len(block_ids) * self.block_size < num_tokens:

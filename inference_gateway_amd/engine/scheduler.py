@@ -249,6 +249,12 @@ class Scheduler:
     def finish(self, seq: Sequence, reason: str) -> None:
         seq.state = SeqState.FINISHED
         seq.finish_reason = reason
+        if (self.bm.enable_prefix_caching
+                and seq.prefilled < seq.prompt_len and seq.block_ids):
+            # aborted mid-chunked-prefill: blocks past the written point
+            # were hash-indexed at allocation but never got KV — un-index
+            # them so the retained pool cannot serve garbage (kvcache.py)
+            self.bm.drop_unwritten(seq.block_ids, seq.prefilled)
         self.bm.free(seq.block_ids)
         if seq.slot is not None:
             self._free_slots.append(seq.slot)

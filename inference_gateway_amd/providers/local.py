@@ -144,26 +144,48 @@ class LocalProvider:
         return (token_ids, self.tokenizer.decode(token_ids), finish_reason,
                 lp_content)
 
+    MAX_N = 8          # engine KV-fork group cap (OpenAI's own cap is 128)
+
+    def _resolve_n(self, req: dict) -> int:
+        """OpenAI semantics: an unsupported `n` is a 400, never a silent
+        clamp (a client indexing choices by its requested n would break)."""
+        n = int(req.get("n") or 1)
+        if n < 1 or n > self.MAX_N:
+            raise HTTPError(
+                400, f"n must be between 1 and {self.MAX_N}, got {n}")
+        return n
+
     def validate_request(self, req: dict) -> None:
         """Pre-stream validation hook (gateway/routes.py calls this
         before opening the SSE response)."""
-        self._check_context(self._encode_request(req))
+        self._resolve_n(req)
+        self._check_context(self._encode_request(req), req)
 
-    def _check_context(self, prompt_ids) -> None:
+    def _check_context(self, prompt_ids, req: Optional[dict] = None) -> None:
         limit = self.engine.engine.cfg.max_model_len
         if len(prompt_ids) > limit:
             # OpenAI returns 400 for context overflow, not a 5xx
             raise HTTPError(
                 400, f"prompt is {len(prompt_ids)} tokens but the "
                      f"model's maximum context length is {limit}")
+        if req is not None:
+            # OpenAI also 400s when an EXPLICIT max_tokens cannot fit the
+            # context (prompt + max_tokens > limit); a request without
+            # max_tokens keeps the truncate-at-limit semantics
+            mt = req.get("max_completion_tokens") or req.get("max_tokens")
+            if mt is not None and len(prompt_ids) + int(mt) > limit:
+                raise HTTPError(
+                    400, f"prompt ({len(prompt_ids)} tokens) plus "
+                         f"max_tokens ({int(mt)}) exceeds the model's "
+                         f"maximum context length of {limit}")
 
     async def chat_completions(self, req: dict,
                                extra_headers=None) -> dict:
         prompt_ids = self._encode_request(req)
-        self._check_context(prompt_ids)
+        self._check_context(prompt_ids, req)
         sampling = self._sampling_from_request(req)
         stops = self._stop_strings(req)
-        n = min(int(req.get("n") or 1), 8)
+        n = self._resolve_n(req)
         if n <= 1:
             token_ids, text, finish_reason, lp = await self._collect(
                 prompt_ids, sampling, stops)
@@ -223,7 +245,7 @@ class LocalProvider:
         """Yields SSE lines ('data: {...}') matching OpenAI stream chunks;
         per-token flush (reference bar: per-line SSE flush)."""
         prompt_ids = self._encode_request(req)
-        self._check_context(prompt_ids)
+        self._check_context(prompt_ids, req)
         sampling = self._sampling_from_request(req)
         cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
         created = int(time.time())
@@ -249,7 +271,7 @@ class LocalProvider:
             return (prefix + json.dumps(delta, separators=(",", ":"))
                     + lps + ',"finish_reason":' + fin + "}]" + tail + "}")
 
-        n = min(int(req.get("n") or 1), 8)
+        n = self._resolve_n(req)
         if n > 1:
             async for line in self._stream_multi(req, prompt_ids, sampling,
                                                  cid, created, model, n):

@@ -887,6 +887,40 @@ def test_abort_mid_chunk_releases_blocks():
     assert not eng.scheduler.has_work
 
 
+def test_abort_mid_chunk_does_not_poison_prefix_cache():
+    """with chunked prefill + prefix caching,
+    allocate_for_prompt hash-indexes ALL full prompt blocks up front;
+    aborting before their KV is computed must un-index the unwritten
+    ones, or a later identical prompt would claim garbage KV from the
+    retained pool and silently generate wrong tokens."""
+    long_prompt = list(range(11, 170))
+    sp = lambda: SamplingParams(max_new_tokens=5, temperature=0.0)  # noqa
+    eng0 = _engine()
+    r0 = eng0.add_request(long_prompt, sp())
+    while not eng0.is_finished():
+        eng0.step()
+    want = eng0.get_output(r0).token_ids
+
+    eng = _engine(max_batched_tokens=32, enable_chunked_prefill=True,
+                  enable_prefix_caching=True, num_gpu_blocks=64,
+                  max_num_seqs=8)
+    rid = eng.add_request(long_prompt, sp())
+    eng.step()                                  # first chunk only
+    seq = eng._seqs[rid]
+    assert 0 < seq.prefilled < seq.prompt_len
+    written = seq.prefilled
+    eng.abort_request(rid)
+    bm = eng.block_manager
+    # every surviving hash must belong to a block whose KV was computed
+    for (idx, _h, _chunk) in bm._by_hash:
+        assert idx < written // bm.block_size
+    # a fresh identical prompt must reproduce the reference tokens
+    r1 = eng.add_request(long_prompt, sp())
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(r1).token_ids == want
+
+
 def test_moe_engine_with_chunked_prefix_and_fork():
     """Mixtral path (MoE MLP + shared LlamaAttention): chunked prefill,
     prefix caching and n>1 forks must be output-invariant there too."""
@@ -938,12 +972,15 @@ async def test_async_engine_shutdown_unblocks_streams():
 
 
 @pytest.mark.asyncio
-async def test_async_engine_death_closes_queues():
-    """A fatal error on the engine thread must not strand consumers:
-    every open stream gets its sentinel and the exception is recorded."""
+async def test_async_engine_death_raises_not_clean_stop():
+    """a fatal error on the engine thread must not
+    look like a successful completion — open streams raise
+    EngineDeadError (the gateway turns it into an in-stream error frame)
+    and NEW requests fail fast."""
     import asyncio
 
-    from inference_gateway_amd.engine.async_engine import AsyncEngine
+    from inference_gateway_amd.engine.async_engine import (AsyncEngine,
+                                                           EngineDeadError)
     from inference_gateway_amd.engine.config import EngineConfig
 
     eng = AsyncEngine(EngineConfig.tiny_llama(), device="cpu")
@@ -962,9 +999,16 @@ async def test_async_engine_death_closes_queues():
             n += 1
             if n == 2:
                 eng.engine.step = boom    # next loop iteration dies
-    await asyncio.wait_for(consume(), timeout=30)
+    with pytest.raises(EngineDeadError, match="injected engine fault"):
+        await asyncio.wait_for(consume(), timeout=30)
     assert n >= 2
     assert isinstance(eng._fatal, RuntimeError)
+    # new work is rejected immediately, not queued into a dead engine
+    with pytest.raises(EngineDeadError):
+        async for _ in eng.generate([1, 2, 3], SamplingParams()):
+            pass
+    with pytest.raises(EngineDeadError):
+        eng.generate_group([1, 2, 3], SamplingParams(), 2)
     eng.shutdown()
 
 

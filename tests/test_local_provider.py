@@ -361,6 +361,40 @@ async def test_local_stream_context_overflow_is_plain_400(local_gateway):
 
 
 @pytest.mark.asyncio
+async def test_local_n_over_cap_is_400(local_gateway):
+    """n beyond the supported maximum is a 400, never a
+    silent clamp (clients index choices by their requested n)."""
+    async with gw_client(local_gateway) as c:
+        for stream in (False, True):
+            r = await c.post("/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "hello"}],
+                "max_tokens": 4, "n": 9, "stream": stream})
+            assert r.status_code == 400
+            assert "n must be between" in str(r.json()["error"])
+
+
+@pytest.mark.asyncio
+async def test_local_prompt_plus_max_tokens_overflow_is_400(local_gateway):
+    """OpenAI 400s when prompt + an explicit max_tokens
+    exceeds the context window (no silent truncation); without
+    max_tokens the request is accepted and truncates at the limit."""
+    async with gw_client(local_gateway) as c:
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hello"}],
+            "max_tokens": 100000})
+        assert r.status_code == 400
+        assert "max_tokens" in str(r.json()["error"])
+        # no explicit max_tokens: allowed, finishes with 'length'
+        r = await c.post("/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "hello"}],
+            "temperature": 0})
+        assert r.status_code == 200
+
+
+@pytest.mark.asyncio
 async def test_stream_flushes_incomplete_utf8_tail(local_gateway,
                                                    monkeypatch):
     """Regression (soak flake): tokens that never complete a UTF-8
