@@ -36,14 +36,16 @@ class AsyncEngine:
         import os
 
         self.engine = LLMEngine(cfg, device=device)
-        # measured A/B at 256 concurrent SSE streams: per-token wakeups
-        # outperform one-callback-per-step delivery (13.0k vs 9.5k tok/s,
-        # TTFT p50 3.4 vs 7.8 s) — the fine-grained wakeups let the event
-        # loop interleave sends with the engine thread instead of waking
-        # every stream task at once; keep the batched form as an opt-in
-        # experiment knob
+        # one-callback-per-step delivery (default).  r1 measured this
+        # SLOWER than per-token wakeups (9.5k vs 13.0k tok/s @256 SSE
+        # streams under uvicorn) — but that inverts once the consumer
+        # drains its queue greedily (generate_batched) and the native
+        # server coalesces writes: the r2 ceiling harness
+        # (tools/bench_sse_ceiling.py) measures per-step + batched drain
+        # at ~145k tok/s vs ~37k for per-token on the same box.
+        # IGW_BATCH_DELIVER=0 restores per-token wakeups.
         self._batched_delivery = os.environ.get(
-            "IGW_BATCH_DELIVER", "0") == "1"
+            "IGW_BATCH_DELIVER", "1") == "1"
         self._lock = threading.Lock()
         self._wake = threading.Event()
         self._stop = False
@@ -120,6 +122,20 @@ class AsyncEngine:
     # ---- async API ------------------------------------------------------
     async def generate(self, prompt_token_ids, sampling: SamplingParams
                        ) -> AsyncIterator[StepOutput]:
+        async for outs in self.generate_batched(prompt_token_ids, sampling):
+            for out in outs:
+                yield out
+
+    async def generate_batched(self, prompt_token_ids,
+                               sampling: SamplingParams
+                               ) -> AsyncIterator[list]:
+        """Yields LISTS of StepOutputs: every token already sitting in
+        the queue is drained in one batch.  In the engine-bound steady
+        state the list has one entry (per-token latency unchanged); when
+        the CONSUMER is the bottleneck (256-stream SSE: the socket write
+        path) batching amortizes the per-wakeup, per-json, per-write
+        costs and lets the wire catch up instead of queueing behind the
+        engine (the r1 serving ceiling)."""
         loop = asyncio.get_running_loop()
         q: asyncio.Queue = asyncio.Queue()
         if self._fatal is not None:
@@ -131,14 +147,26 @@ class AsyncEngine:
             self._loops[rid] = loop
         self._wake.set()
         try:
-            while True:
-                item = await q.get()
-                if item is self._SENTINEL:
-                    break
-                if isinstance(item, _Fatal):
-                    raise EngineDeadError(
-                        f"engine thread died: {item.exc!r}") from item.exc
-                yield item
+            done = False
+            while not done:
+                batch = [await q.get()]
+                while True:
+                    try:
+                        batch.append(q.get_nowait())
+                    except asyncio.QueueEmpty:
+                        break
+                outs = []
+                for item in batch:
+                    if item is self._SENTINEL:
+                        done = True
+                        break
+                    if isinstance(item, _Fatal):
+                        raise EngineDeadError(
+                            f"engine thread died: {item.exc!r}"
+                        ) from item.exc
+                    outs.append(item)
+                if outs:
+                    yield outs
         finally:
             with self._lock:
                 self._queues.pop(rid, None)
@@ -146,10 +174,12 @@ class AsyncEngine:
                 self.engine.abort_request(rid)
 
     def generate_group(self, prompt_token_ids, sampling: SamplingParams,
-                       n: int):
+                       n: int, batched: bool = False):
         """OpenAI n>1: one engine-side prefill, n token streams (the
         engine forks the prompt KV — engine.add_request_group).  Returns
-        n async iterators; each consumer's exit aborts its own branch."""
+        n async iterators; each consumer's exit aborts its own branch.
+        With batched=True each iterator yields LISTS of StepOutputs
+        (greedy queue drain — see generate_batched)."""
         loop = asyncio.get_running_loop()
         qs = [asyncio.Queue() for _ in range(n)]
         if self._fatal is not None:
@@ -163,16 +193,18 @@ class AsyncEngine:
                 self._loops[rid] = loop
         self._wake.set()
 
+        def check(item):
+            if isinstance(item, _Fatal):
+                raise EngineDeadError(
+                    f"engine thread died: {item.exc!r}") from item.exc
+
         async def reader(rid, q):
             try:
                 while True:
                     item = await q.get()
                     if item is self._SENTINEL:
                         break
-                    if isinstance(item, _Fatal):
-                        raise EngineDeadError(
-                            f"engine thread died: {item.exc!r}"
-                        ) from item.exc
+                    check(item)
                     yield item
             finally:
                 with self._lock:
@@ -180,7 +212,33 @@ class AsyncEngine:
                     self._loops.pop(rid, None)
                     self.engine.abort_request(rid)
 
-        return [reader(rid, q) for rid, q in zip(rids, qs)]
+        async def reader_batched(rid, q):
+            try:
+                done = False
+                while not done:
+                    batch = [await q.get()]
+                    while True:
+                        try:
+                            batch.append(q.get_nowait())
+                        except asyncio.QueueEmpty:
+                            break
+                    outs = []
+                    for item in batch:
+                        if item is self._SENTINEL:
+                            done = True
+                            break
+                        check(item)
+                        outs.append(item)
+                    if outs:
+                        yield outs
+            finally:
+                with self._lock:
+                    self._queues.pop(rid, None)
+                    self._loops.pop(rid, None)
+                    self.engine.abort_request(rid)
+
+        mk = reader_batched if batched else reader
+        return [mk(rid, q) for rid, q in zip(rids, qs)]
 
     def shutdown(self) -> None:
         self._stop = True

@@ -288,17 +288,22 @@ class LocalProvider:
         pending: list = []
         want_lp = sampling.logprobs >= 0
         pending_lp: list = []      # lp items for not-yet-flushed tokens
-        async for out in self.engine.generate(prompt_ids, sampling):
-            ntok += 1
-            pending.append(out.token_id)
-            if want_lp:
-                pending_lp.append(self._lp_item(out))
-            if out.finished:
-                # record BEFORE the holdback continue: a length-finish
-                # whose tail is an incomplete UTF-8 run must not report
-                # "stop"
-                finish_reason = ("stop" if out.finish_reason == "stop"
-                                 else "length")
+        # batched drain: one chunk (and one socket write) per group of
+        # already-available tokens — per-token when the engine is the
+        # bottleneck, coalesced when the wire is
+        async for outs in self.engine.generate_batched(prompt_ids,
+                                                       sampling):
+            for out in outs:
+                ntok += 1
+                pending.append(out.token_id)
+                if want_lp:
+                    pending_lp.append(self._lp_item(out))
+                if out.finished:
+                    # record BEFORE the holdback continue: a length-finish
+                    # whose tail is an incomplete UTF-8 run must not
+                    # report "stop"
+                    finish_reason = ("stop" if out.finish_reason == "stop"
+                                     else "length")
             text = self.tokenizer.decode(pending)
             # hold back partial UTF-8 sequences (byte tokenizer)
             if text.endswith("\ufffd"):
@@ -369,21 +374,23 @@ class LocalProvider:
             return "data: " + json.dumps(frame, separators=(",", ":"))
 
         queue: asyncio.Queue = asyncio.Queue()
-        gens = self.engine.generate_group(prompt_ids, sampling, n)
+        gens = self.engine.generate_group(prompt_ids, sampling, n,
+                                          batched=True)
 
         async def pump(idx: int, gen) -> None:
             pending, pending_lp = [], []
             emitted_tail, ntok = "", 0
             finish_reason = "stop"
             try:
-                async for out in gen:
-                    ntok += 1
-                    pending.append(out.token_id)
-                    if want_lp:
-                        pending_lp.append(self._lp_item(out))
-                    if out.finished:
-                        finish_reason = ("stop" if out.finish_reason ==
-                                         "stop" else "length")
+                async for outs in gen:
+                    for out in outs:
+                        ntok += 1
+                        pending.append(out.token_id)
+                        if want_lp:
+                            pending_lp.append(self._lp_item(out))
+                        if out.finished:
+                            finish_reason = ("stop" if out.finish_reason ==
+                                             "stop" else "length")
                     text = self.tokenizer.decode(pending)
                     if text.endswith("�"):
                         continue

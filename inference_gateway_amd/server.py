@@ -49,25 +49,29 @@ def _metrics_app(telemetry):
 
 
 async def _serve(cfg, gw: Gateway, logger: Logger) -> None:
-    import uvicorn
+    from .gateway.hserver import HttpServer, build_ssl_context
 
-    servers = []
-    main_cfg = uvicorn.Config(
-        gw, host=cfg.server.host, port=cfg.server.port, log_level="warning",
-        timeout_keep_alive=int(cfg.server.idle_timeout),
-        ssl_certfile=cfg.server.tls_cert_path or None,
-        ssl_keyfile=cfg.server.tls_key_path or None,
-    )
-    servers.append(uvicorn.Server(main_cfg))
+    ssl_ctx = None
+    if cfg.server.tls_cert_path:
+        ssl_ctx = build_ssl_context(cfg.server.tls_cert_path,
+                                    cfg.server.tls_key_path)
+    servers = [HttpServer(gw, cfg.server.host, cfg.server.port,
+                          ssl_context=ssl_ctx,
+                          max_body=cfg.server.max_request_body_size)]
     if cfg.telemetry.enabled and gw.telemetry is not None:
-        mcfg = uvicorn.Config(
-            _metrics_app(gw.telemetry), host=cfg.server.host,
-            port=cfg.telemetry.metrics_port, log_level="warning")
-        servers.append(uvicorn.Server(mcfg))
+        servers.append(HttpServer(_metrics_app(gw.telemetry),
+                                  cfg.server.host,
+                                  cfg.telemetry.metrics_port))
+    await gw.startup()
     logger.info("gateway listening", "host", cfg.server.host,
                 "port", cfg.server.port, "tls",
                 bool(cfg.server.tls_cert_path))
-    await asyncio.gather(*(s.serve() for s in servers))
+    try:
+        await asyncio.gather(*(s.serve_forever() for s in servers))
+    finally:
+        for s in servers:
+            await s.stop()
+        await gw.shutdown()
 
 
 def main(argv=None) -> int:

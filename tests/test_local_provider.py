@@ -415,9 +415,10 @@ async def test_stream_flushes_incomplete_utf8_tail(local_gateway,
         # this stream ends mid-UTF-8, so nothing flushes until the end
         toks = [0x80 + 3, 0x81 + 3, 0x82 + 3, 0x83 + 3]
         for i, t in enumerate(toks):
-            yield _Out(t, i == len(toks) - 1)
+            yield [_Out(t, i == len(toks) - 1)]
 
-    monkeypatch.setattr(provider.engine, "generate", fake_generate)
+    monkeypatch.setattr(provider.engine, "generate_batched",
+                        fake_generate)
     content = ""
     chunks = 0
     async for line in provider.stream_chat_completions({
@@ -434,6 +435,45 @@ async def test_stream_flushes_incomplete_utf8_tail(local_gateway,
     assert "�" in content          # tokens surfaced, not dropped
     assert chunks >= 3                  # role + content + usage
     assert fin == "length"              # not misreported as "stop"
+
+
+@pytest.mark.asyncio
+async def test_stream_merges_batched_tokens_into_one_chunk(local_gateway,
+                                                           monkeypatch):
+    """Wire-bound batching: when the engine queue has several tokens
+    ready, they are emitted as ONE content chunk (one socket write) with
+    no text lost and exact usage accounting."""
+    provider = local_gateway.local_provider
+
+    class _Out:
+        def __init__(self, tok, fin):
+            self.token_id = tok
+            self.finished = fin
+            self.finish_reason = "length" if fin else None
+            self.logprob = None
+            self.top_logprobs = None
+
+    # 'a','b' (+3 byte-tokenizer offset) then 'c','d','e' in one batch
+    async def fake_batched(prompt_ids, sampling):
+        yield [_Out(ord("a") + 3, False), _Out(ord("b") + 3, False)]
+        yield [_Out(ord("c") + 3, False), _Out(ord("d") + 3, False),
+               _Out(ord("e") + 3, True)]
+
+    monkeypatch.setattr(provider.engine, "generate_batched", fake_batched)
+    contents, usage = [], None
+    async for line in provider.stream_chat_completions({
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "x"}],
+            "max_tokens": 5}):
+        if not line.startswith("data: ") or line == "data: [DONE]":
+            continue
+        body = json.loads(line[6:])
+        d = body["choices"][0]["delta"].get("content")
+        if d:
+            contents.append(d)
+        usage = body.get("usage") or usage
+    assert contents == ["ab", "cde"]       # one chunk per ready batch
+    assert usage["completion_tokens"] == 5
 
 
 @pytest.mark.asyncio
