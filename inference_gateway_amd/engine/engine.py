@@ -135,6 +135,7 @@ class LLMEngine:
         self.num_generated_tokens = 0
         self.num_prompt_tokens = 0
         self.num_forks = 0
+        self.num_steps = 0        # scheduled engine steps (bench windows)
         # +1 block of headroom: the async decode pipeline can write KV for
         # one in-flight token past max_model_len before the stop check runs
         self.max_blocks_per_seq = (
@@ -251,6 +252,7 @@ class LLMEngine:
         batch = self.scheduler.schedule()
         if batch is None:
             return self._drain_pending()
+        self.num_steps += 1
         gr = self.graph_runner
         if not batch.is_prefill and gr is not None:
             gr.write_block_appends(batch.new_blocks)

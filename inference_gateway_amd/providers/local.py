@@ -47,6 +47,10 @@ class LocalProvider:
         # OpenAI system_fingerprint: identifies the backend configuration
         # (determinism domain for `seed`)
         self.fingerprint = f"fp_igw-amd_{local_cfg.model}"
+        # tokens that have gone THROUGH the SSE write path (incremented
+        # after the chunk's socket write completes) — bench.py's serving
+        # mode measures the wire rate from this, not the engine rate
+        self.streamed_tokens = 0
 
     # ---- provider interface --------------------------------------------
     async def list_models(self) -> dict:
@@ -308,6 +312,7 @@ class LocalProvider:
             # hold back partial UTF-8 sequences (byte tokenizer)
             if text.endswith("\ufffd"):
                 continue
+            nflush = len(pending)
             pending = []
             lp, pending_lp = (pending_lp, []) if want_lp else (None, [])
             if stops:
@@ -319,14 +324,18 @@ class LocalProvider:
                         yield chunk({"content": buf[:i]}, lp=lp)
                     finish_reason = "stop"
                     stopped = True
+                    self.streamed_tokens += nflush
                     break
                 # keep a stop-prefix-sized tail unemitted
                 emit = buf[:-holdback] if holdback else buf
                 emitted_tail = buf[len(emit):]
                 if emit:
                     yield chunk({"content": emit}, lp=lp)
+                self.streamed_tokens += nflush
                 continue
             yield chunk({"content": text}, lp=lp)
+            # post-yield: the chunk's write has been handed to the kernel
+            self.streamed_tokens += nflush
         if not stopped:
             # flush whatever is still held back: the stop-prefix tail AND
             # any pending tokens that never completed a UTF-8 sequence
