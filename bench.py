@@ -110,9 +110,13 @@ async def client_worker_main(args) -> None:
             ttft = None
             while True:
                 try:
-                    # tokens arrive every engine step; a long silence
-                    # means the server is gone (stop() raced the FIN)
-                    data = await asyncio.wait_for(reader.read(65536), 30)
+                    # before the first token the stream legitimately
+                    # sits in the prefill queue for the whole burst ramp
+                    # (minutes at batch 1024) — only time out fast once
+                    # tokens are flowing, where silence means the server
+                    # is gone (stop() raced the FIN)
+                    data = await asyncio.wait_for(
+                        reader.read(65536), 60 if ttft else 900)
                 except asyncio.TimeoutError:
                     break
                 if not data:
@@ -237,6 +241,8 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
     # ---- timed window: EXACTLY K engine steps ----
     t0 = time.perf_counter()
     s0, w0 = eng.num_steps, provider.streamed_tokens
+    g0 = eng.num_generated_tokens
+    run0 = len(eng.scheduler.running)
     target = s0 + args.steps
     while eng.num_steps < target:
         await asyncio.sleep(0.002)
@@ -244,9 +250,15 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
         torch.cuda.synchronize()
     t1 = time.perf_counter()
     w1 = provider.streamed_tokens
+    run1 = len(eng.scheduler.running)
+    g1 = eng.num_generated_tokens
     elapsed = t1 - t0
     wire_tokens = w1 - w0
     engine_tokens = args.batch * args.steps
+    if run0 < args.batch or run1 < args.batch:
+        print(f"[bench] WARNING: running set {run0}->{run1} during the "
+              f"window (want {args.batch}): streams died, the window is "
+              "not steady-state", file=sys.stderr)
 
     # ---- aggregate across ranks: MAX elapsed, SUM tokens ----
     elapsed, wire_tokens = _reduce(elapsed, wire_tokens, device)
@@ -287,6 +299,8 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
             if ttfts else None),
         "wire_vs_engine": round(
             wire_tokens / max(1, engine_tokens * nranks), 3),
+        "engine_tokens_window": g1 - g0,
+        "window_running": [run0, run1],
         "config": {
             "model": args.model,
             "global_batch": args.batch * nranks,
