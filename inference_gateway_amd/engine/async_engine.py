@@ -66,15 +66,20 @@ class AsyncEngine:
 
     def _close_all_queues(self) -> None:
         # engine death pushes a _Fatal (readers raise EngineDeadError);
-        # clean shutdown pushes the plain sentinel (readers just stop)
+        # clean shutdown pushes the plain sentinel (readers just stop).
+        # Direct sinks get fail() either way.
         item = (_Fatal(self._fatal) if self._fatal is not None
                 else self._SENTINEL)
         with self._lock:
             items = list(self._queues.items())
         for rid, q in items:
             loop = self._loops.get(rid)
-            if loop is not None:
+            if loop is None:
+                continue
+            if isinstance(q, asyncio.Queue):
                 loop.call_soon_threadsafe(q.put_nowait, item)
+            else:
+                loop.call_soon_threadsafe(q.fail, self._fatal)
 
     def _run_loop(self) -> None:
         while not self._stop:
@@ -94,13 +99,17 @@ class AsyncEngine:
                     loop = self._loops.get(out.seq_id)
                     if q is None or loop is None:
                         continue
-                    loop.call_soon_threadsafe(q.put_nowait, out)
-                    if out.finished:
-                        loop.call_soon_threadsafe(q.put_nowait,
-                                                  self._SENTINEL)
+                    if isinstance(q, asyncio.Queue):
+                        loop.call_soon_threadsafe(q.put_nowait, out)
+                        if out.finished:
+                            loop.call_soon_threadsafe(q.put_nowait,
+                                                      self._SENTINEL)
+                    else:
+                        loop.call_soon_threadsafe(self._deliver,
+                                                  [(q, out)])
 
     def _dispatch_batch(self, outputs) -> None:
-        # ONE event-loop wakeup per step, not one per token: at 256
+        # ONE event-loop wakeup per step, not one per token: at 256+
         # concurrent streams per-token call_soon_threadsafe wakeups become
         # a serving bottleneck
         by_loop: Dict = {}
@@ -114,10 +123,20 @@ class AsyncEngine:
             loop.call_soon_threadsafe(self._deliver, items)
 
     def _deliver(self, items) -> None:
+        # targets are asyncio.Queues (generator streams) or direct sinks
+        # (DirectStream sessions: feed per token, ONE flush per delivery
+        # so all of a step's tokens for a stream leave in one write)
+        touched = {}
         for q, out in items:
-            q.put_nowait(out)
-            if out.finished:
-                q.put_nowait(self._SENTINEL)
+            if isinstance(q, asyncio.Queue):
+                q.put_nowait(out)
+                if out.finished:
+                    q.put_nowait(self._SENTINEL)
+            else:
+                q.feed(out)
+                touched[id(q)] = q
+        for sink in touched.values():
+            sink.flush()
 
     # ---- async API ------------------------------------------------------
     async def generate(self, prompt_token_ids, sampling: SamplingParams
@@ -239,6 +258,30 @@ class AsyncEngine:
 
         mk = reader_batched if batched else reader
         return [mk(rid, q) for rid, q in zip(rids, qs)]
+
+    # ---- direct-sink API (SSE fast path; providers/local.py) ------------
+    def add_direct(self, prompt_token_ids, sampling: SamplingParams,
+                   sink) -> int:
+        """Register a direct sink: `sink.feed(out)` per token and
+        `sink.flush()` once per delivery, called on the sink's event
+        loop; `sink.fail(exc)` on engine death.  No queue, no reader
+        task — the sink writes SSE frames straight to the socket."""
+        if self._fatal is not None:
+            raise EngineDeadError(
+                f"engine thread died: {self._fatal!r}") from self._fatal
+        loop = asyncio.get_running_loop()
+        with self._lock:
+            rid = self.engine.add_request(prompt_token_ids, sampling)
+            self._queues[rid] = sink
+            self._loops[rid] = loop
+        self._wake.set()
+        return rid
+
+    def close_direct(self, rid: int) -> None:
+        with self._lock:
+            self._queues.pop(rid, None)
+            self._loops.pop(rid, None)
+            self.engine.abort_request(rid)
 
     def shutdown(self) -> None:
         self._stop = True

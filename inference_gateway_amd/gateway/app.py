@@ -192,6 +192,7 @@ class Gateway:
             elif msg["type"] == "http.disconnect":
                 return
         req = parse_scope(scope, body)
+        req.state["asgi_extensions"] = scope.get("extensions") or {}
         try:
             resp = await self.handle(req)
         except Exception as e:  # last-resort 500

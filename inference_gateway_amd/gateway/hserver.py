@@ -171,9 +171,60 @@ class HttpProtocol(asyncio.Protocol):
             "headers": headers,
             "client": self.transport.get_extra_info("peername"),
             "server": self.transport.get_extra_info("sockname"),
+            # direct SSE fast path: write a payload (chunked-framed) /
+            # terminate the response / probe send-backlog, bypassing
+            # per-event ASGI messages
+            "extensions": {"igw.direct_sse": (self._raw_write,
+                                              self._raw_end,
+                                              self._raw_backlogged)},
         }
         self._need_body = need
         self._keep_alive = keep
+
+    # ---- direct SSE fast path ------------------------------------------
+    _MAX_BUFFER = 4 << 20      # slow-consumer cutoff for direct streams
+
+    def _raw_write(self, payload: bytes) -> bool:
+        """Write one SSE payload (chunk-framed) straight to the socket.
+        Plain function — callable from an event-loop callback.  Returns
+        False when the client is gone or hopelessly slow (the caller
+        aborts its stream)."""
+        if self._closing or self.transport.is_closing():
+            return False
+        if self.transport.get_write_buffer_size() > self._MAX_BUFFER:
+            # slow consumer: drop the stream instead of buffering
+            # unboundedly (the engine keeps serving everyone else)
+            self.transport.close()
+            return False
+        if self._chunked:
+            data = b"%x\r\n" % len(payload) + payload + b"\r\n"
+        else:
+            data = payload
+        if self._pending_head:
+            data = self._pending_head + data
+            self._pending_head = b""
+        self.transport.write(data)
+        return True
+
+    _SOFT_BUFFER = 64 << 10
+
+    def _raw_backlogged(self) -> bool:
+        """True when unsent bytes are piling up — the direct stream then
+        SKIPS this flush and lets tokens coalesce, so a wire that falls
+        behind catches up with merged chunks instead of queueing one
+        write per token (self-clocking batching)."""
+        return (self.transport.get_write_buffer_size() > self._SOFT_BUFFER
+                or self._paused)
+
+    def _raw_end(self) -> None:
+        if self._closing or self.transport.is_closing():
+            return
+        data = b"0\r\n\r\n" if self._chunked else b""
+        if self._pending_head:
+            data = self._pending_head + data
+            self._pending_head = b""
+        if data:
+            self.transport.write(data)
 
     # ---- response path --------------------------------------------------
     def _simple_response(self, status: int, body: bytes) -> None:

@@ -32,7 +32,7 @@ class Request:
 
 
 class Response:
-    __slots__ = ("status", "headers", "body", "stream")
+    __slots__ = ("status", "headers", "body", "stream", "direct")
 
     def __init__(self, status: int = 200,
                  body: Union[bytes, str, None] = b"",
@@ -42,6 +42,12 @@ class Response:
         self.body = body.encode() if isinstance(body, str) else (body or b"")
         self.headers = headers or {}
         self.stream = stream  # async iterator of SSE event strings
+        # direct-write SSE session (local engine fast path): an object
+        # with `async run()` that writes frames straight to the socket
+        # via the server's raw-write extension — no per-token generator
+        # resumption, no per-event ASGI messages (providers/local.py
+        # DirectStream)
+        self.direct = None
 
     @staticmethod
     def json(obj, status: int = 200,
@@ -87,6 +93,11 @@ async def send_response(send, resp: Response) -> None:
         "status": resp.status,
         "headers": headers,
     })
+    if resp.direct is not None:
+        # the session writes every frame itself through the raw socket
+        # extension (including the terminating chunk)
+        await resp.direct.run()
+        return
     if resp.stream is None:
         await send({"type": "http.response.body", "body": resp.body})
         return

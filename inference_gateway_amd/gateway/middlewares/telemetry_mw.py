@@ -28,6 +28,23 @@ def telemetry_middleware(telemetry):
         provider = req.state.get("provider_id", "unknown")
         model = req.state.get("model", "unknown")
 
+        if getattr(resp, "direct", None) is not None:
+            # direct-write SSE session: it reports through hooks (there
+            # is no stream generator to tee)
+            sess = resp.direct
+            sess.on_first_chunk = lambda: telemetry.record_ttft(
+                provider, model, time.monotonic() - start)
+
+            def _on_end(prompt_toks, completion_toks, err):
+                telemetry.record_request(provider, model,
+                                         time.monotonic() - start, err)
+                if prompt_toks or completion_toks:
+                    telemetry.record_tokens(provider, model, prompt_toks,
+                                            completion_toks)
+
+            sess.on_end = _on_end
+            return resp
+
         if resp.stream is None:
             dur = time.monotonic() - start
             err = "" if resp.status < 400 else str(resp.status)

@@ -154,6 +154,22 @@ class Router:
             return await self.mcp.run_chat(provider, body, stream, req)
 
         if stream:
+            # direct-write SSE fast path: local engine + native server +
+            # n==1 — the session writes frames straight to the socket
+            # (providers/local.py DirectStream).  Any other combination
+            # takes the generator relay below.
+            direct_io = (req.state.get("asgi_extensions") or {}).get(
+                "igw.direct_sse")
+            opener = getattr(provider, "open_direct_stream", None)
+            if (direct_io is not None and opener is not None
+                    and int(body.get("n") or 1) == 1):
+                try:
+                    session = opener(body, *direct_io)
+                except HTTPError as e:
+                    return Response.error(e.status, e.message)
+                resp = Response.sse(None)
+                resp.direct = session
+                return resp
             try:
                 gen = self._relay_stream(provider, body)
                 return Response.sse(gen)

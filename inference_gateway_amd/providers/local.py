@@ -21,6 +21,208 @@ from .constants import LOCAL_PROVIDER_ID
 from .core import HTTPError
 
 
+class DirectStream:
+    """Direct-write SSE session for one local streaming completion
+    (n=1).  The engine's delivery callback calls feed()/flush() on the
+    event loop and the session writes fully framed SSE bytes straight to
+    the socket — no per-token asyncio task wakeup, no generator
+    resumption, no per-event ASGI message.  All of a step's ready tokens
+    leave in ONE socket write.  Mirrors stream_chat_completions token
+    semantics exactly (UTF-8 holdback, stop-string holdback, logprobs,
+    usage, one [DONE]); equivalence is tested in tests/test_hserver.py."""
+
+    def __init__(self, provider: "LocalProvider", req: dict, prompt_ids,
+                 sampling, raw_write, raw_end, raw_backlogged=None):
+        self.provider = provider
+        self.engine = provider.engine
+        self.tokenizer = provider.tokenizer
+        self.raw_write = raw_write
+        self.raw_end = raw_end
+        self.raw_backlogged = raw_backlogged
+        self.prompt_ids = prompt_ids
+        self.sampling = sampling
+        self.stops = provider._stop_strings(req)
+        self.holdback = max((len(s) for s in self.stops), default=1) - 1
+        self.want_lp = sampling.logprobs >= 0
+        cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+        self._prefix = ("data: " + json.dumps({
+            "id": cid, "object": "chat.completion.chunk",
+            "created": int(time.time()),
+            "model": f"{provider.id}/{provider.model_name}",
+            "system_fingerprint": provider.fingerprint,
+        }, separators=(",", ":"))[:-1] + ',"choices":[{"index":0,"delta":')
+        self.pending: list = []
+        self.pending_lp: list = []
+        self.emitted_tail = ""
+        self.ntok = 0
+        self.finish_reason = "stop"
+        self._finished = False      # engine delivered the final token
+        self._closed = False        # stream over (done or socket gone)
+        self._first_content = True
+        self.done = asyncio.Event()
+        self.rid: Optional[int] = None
+        # telemetry hooks (set by the telemetry middleware)
+        self.on_first_chunk = None
+        self.on_end = None
+
+    # ---- frame building -------------------------------------------------
+    def _chunk(self, delta: dict, finish=None, usage=None, lp=None) -> str:
+        if len(delta) == 1 and "content" in delta:
+            d = '{"content":' + json.dumps(delta["content"]) + "}"
+        else:
+            d = json.dumps(delta, separators=(",", ":"))
+        tail = (',"usage":' + json.dumps(usage, separators=(",", ":"))
+                if usage is not None else "")
+        fin = f'"{finish}"' if finish else "null"
+        lps = (',"logprobs":' + json.dumps({"content": lp},
+                                           separators=(",", ":"))
+               if lp is not None else "")
+        return (self._prefix + d + lps + ',"finish_reason":' + fin
+                + "}]" + tail + "}")
+
+    def _write(self, s: str) -> bool:
+        if not self.raw_write((s + "\n\n").encode()):
+            self._close()
+            return False
+        return True
+
+    def _emit_content(self, text: str, lp) -> bool:
+        if self._first_content:
+            self._first_content = False
+            if self.on_first_chunk is not None:
+                self.on_first_chunk()
+        return self._write(self._chunk({"content": text}, lp=lp))
+
+    # ---- engine-facing sink API (called on the event loop) -------------
+    def feed(self, out) -> None:
+        if self._closed:
+            return
+        self.ntok += 1
+        self.pending.append(out.token_id)
+        if self.want_lp:
+            self.pending_lp.append(self.provider._lp_item(out))
+        if out.finished:
+            self.finish_reason = ("stop" if out.finish_reason == "stop"
+                                  else "length")
+            self._finished = True
+
+    def flush(self) -> None:
+        if self._closed:
+            return
+        if (not self._finished and self.raw_backlogged is not None
+                and self.raw_backlogged()):
+            # socket send is behind: skip this flush and let tokens
+            # coalesce — the next delivery writes one merged chunk
+            return
+        try:
+            self._flush_inner()
+        except Exception:
+            self._close()
+
+    def _flush_inner(self) -> None:
+        if self.pending:
+            text = self.tokenizer.decode(self.pending)
+            if not text.endswith("�"):
+                nflush = len(self.pending)
+                self.pending = []
+                lp = self.pending_lp if self.want_lp else None
+                self.pending_lp = []
+                if self.stops:
+                    buf = self.emitted_tail + text
+                    i = self.provider._find_stop(buf, self.stops)
+                    if i >= 0:
+                        if buf[:i] and not self._emit_content(buf[:i], lp):
+                            return
+                        self.provider.streamed_tokens += nflush
+                        self.finish_reason = "stop"
+                        self.emitted_tail = ""
+                        self._end_stream()
+                        return
+                    emit = buf[:-self.holdback] if self.holdback else buf
+                    self.emitted_tail = buf[len(emit):]
+                    if emit and not self._emit_content(emit, lp):
+                        return
+                    self.provider.streamed_tokens += nflush
+                else:
+                    if not self._emit_content(text, lp):
+                        return
+                    self.provider.streamed_tokens += nflush
+        if self._finished:
+            self._end_stream()
+
+    def _end_stream(self) -> None:
+        # final leftover: held-back stop tail + incomplete-UTF-8 pending
+        tail_text = (self.tokenizer.decode(self.pending) if self.pending
+                     else "")
+        nflush = len(self.pending)
+        self.pending = []
+        lp = (self.pending_lp if self.want_lp and self.pending_lp
+              else None)
+        self.pending_lp = []
+        leftover = self.emitted_tail + tail_text
+        if self.stops and leftover:
+            i = self.provider._find_stop(leftover, self.stops)
+            if i >= 0:
+                leftover = leftover[:i]
+                self.finish_reason = "stop"
+        if leftover and not self._emit_content(leftover, lp):
+            return
+        self.provider.streamed_tokens += nflush
+        usage = {
+            "prompt_tokens": len(self.prompt_ids),
+            "completion_tokens": self.ntok,
+            "total_tokens": len(self.prompt_ids) + self.ntok,
+        }
+        if not self._write(self._chunk({}, finish=self.finish_reason,
+                                       usage=usage)):
+            return
+        if not self._write("data: [DONE]"):
+            return
+        self.raw_end()
+        self._complete(usage)
+
+    def fail(self, exc) -> None:
+        """Engine death (async_engine._close_all_queues): emit an
+        in-stream error frame so the crash is not a clean 'stop'."""
+        if self._closed:
+            return
+        msg = f"engine thread died: {exc!r}" if exc else "engine shutdown"
+        self._write("data: " + json.dumps(
+            {"error": {"message": msg, "type": "server_error"}}))
+        self._write("data: [DONE]")
+        self.raw_end()
+        self._close(error=msg)
+
+    # ---- lifecycle ------------------------------------------------------
+    def _complete(self, usage: dict) -> None:
+        self._closed = True
+        if self.on_end is not None:
+            self.on_end(usage["prompt_tokens"],
+                        usage["completion_tokens"], "")
+        self.done.set()
+
+    def _close(self, error: str = "client disconnected") -> None:
+        self._closed = True
+        if self.on_end is not None:
+            self.on_end(len(self.prompt_ids), self.ntok, error)
+            self.on_end = None
+        self.done.set()
+
+    async def run(self) -> None:
+        """Awaited by send_response: registers with the engine, writes
+        the role chunk, then sleeps until the session completes (the
+        engine delivery callbacks do all the writing)."""
+        if not self._write(self._chunk({"role": "assistant",
+                                        "content": ""})):
+            return
+        self.rid = self.engine.add_direct(self.prompt_ids, self.sampling,
+                                          self)
+        try:
+            await self.done.wait()
+        finally:
+            self.engine.close_direct(self.rid)
+
+
 class LocalProvider:
     def __init__(self, local_cfg, device: str = "cuda:0",
                  engine: Optional[AsyncEngine] = None):
@@ -242,6 +444,17 @@ class LocalProvider:
                 "total_tokens": len(prompt_ids) + len(token_ids),
             },
         }
+
+    def open_direct_stream(self, req: dict, raw_write, raw_end,
+                           raw_backlogged=None) -> DirectStream:
+        """SSE fast path (n=1): returns a DirectStream session that the
+        server awaits; validation errors raise HTTPError BEFORE the
+        response opens (plain 400, like the generator path)."""
+        prompt_ids = self._encode_request(req)
+        self._check_context(prompt_ids, req)
+        sampling = self._sampling_from_request(req)
+        return DirectStream(self, req, prompt_ids, sampling, raw_write,
+                            raw_end, raw_backlogged)
 
     async def stream_chat_completions(self, req: dict,
                                       extra_headers=None

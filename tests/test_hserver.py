@@ -122,6 +122,86 @@ async def test_concurrent_streams(provider):
 
 
 @pytest.mark.asyncio
+async def test_direct_stream_equivalence_and_telemetry(provider):
+    """The direct-write SSE path (DirectStream) must produce the same
+    content as the non-streaming completion for the same seeded request,
+    carry exact usage, and report tokens through the telemetry hooks."""
+    from inference_gateway_amd.otel.metrics import Telemetry
+
+    telemetry = Telemetry()
+    cfg = config_mod.load({})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider, telemetry=telemetry)
+    srv = HttpServer(gw, "127.0.0.1", 0)
+    await srv.start()
+    base = f"http://127.0.0.1:{srv.port}"
+    try:
+        req = {"model": "local/tiny-llama",
+               "messages": [{"role": "user", "content": "equiv"}],
+               "max_tokens": 8, "temperature": 0.7, "seed": 42}
+        async with httpx.AsyncClient(base_url=base, timeout=60) as c:
+            r = await c.post("/v1/chat/completions", json=req)
+            want = r.json()["choices"][0]["message"]["content"]
+            events = []
+            async with c.stream("POST", "/v1/chat/completions",
+                                json={**req, "stream": True}) as rs:
+                assert rs.status_code == 200
+                async for line in rs.aiter_lines():
+                    if line.startswith("data:"):
+                        events.append(line[5:].strip())
+        assert events[-1] == "[DONE]"
+        final = json.loads(events[-2])
+        assert final["choices"][0]["finish_reason"] == "length"
+        assert final["usage"]["completion_tokens"] == 8
+        got = "".join(
+            json.loads(e)["choices"][0]["delta"].get("content") or ""
+            for e in events[:-1])
+        assert got == want
+        expo = telemetry.exposition().decode()
+        assert "gen_ai_client_token_usage" in expo
+    finally:
+        await srv.stop()
+        await client.aclose()
+
+
+@pytest.mark.asyncio
+async def test_direct_stream_stop_string(provider):
+    """Stop-string semantics through the direct path: the stop sequence
+    never reaches the wire and finish_reason is 'stop'."""
+    async with serve(provider) as base:
+        async with httpx.AsyncClient(base_url=base, timeout=60) as c:
+            # find deterministic output first, derive a stop from it
+            r = await c.post("/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "stopteststop"}],
+                "max_tokens": 12, "temperature": 0, "seed": 1})
+            text = r.json()["choices"][0]["message"]["content"]
+            if len(text) < 4:
+                pytest.skip("output too short to derive a stop string")
+            stop = text[2:4]
+            events = []
+            async with c.stream("POST", "/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "stopteststop"}],
+                "max_tokens": 12, "temperature": 0, "seed": 1,
+                "stop": [stop], "stream": True,
+            }) as rs:
+                async for line in rs.aiter_lines():
+                    if line.startswith("data:"):
+                        events.append(line[5:].strip())
+            got = "".join(
+                json.loads(e)["choices"][0]["delta"].get("content") or ""
+                for e in events[:-1])
+            fins = [json.loads(e)["choices"][0]["finish_reason"]
+                    for e in events[:-1]]
+            assert stop not in got
+            assert got == text[:text.find(stop)]
+            assert fins[-1] == "stop"
+
+
+@pytest.mark.asyncio
 async def test_body_cap_413(provider):
     async with serve(provider) as base:
         async with httpx.AsyncClient(base_url=base, timeout=30) as c:

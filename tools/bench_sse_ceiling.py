@@ -67,7 +67,7 @@ class SyntheticEngine:
         self._thread.start()
 
     def _run(self):
-        batch_deliver = os.environ.get("IGW_BATCH_DELIVER", "0") == "1"
+        batch_deliver = os.environ.get("IGW_BATCH_DELIVER", "1") == "1"
         while not self._stop:
             with self._lock:
                 items = list(self._streams.items())
@@ -84,13 +84,10 @@ class SyntheticEngine:
                            "length" if fin else None)
                 try:
                     if batch_deliver:
-                        by_loop.setdefault(loop, []).append(
-                            (q, out, fin))
+                        by_loop.setdefault(loop, []).append((q, out))
                     else:
-                        loop.call_soon_threadsafe(q.put_nowait, out)
-                        if fin:
-                            loop.call_soon_threadsafe(q.put_nowait,
-                                                      self._SENTINEL)
+                        loop.call_soon_threadsafe(self._deliver,
+                                                  [(q, out)])
                 except RuntimeError:
                     fin = True           # loop closed
                 if fin:
@@ -105,10 +102,30 @@ class SyntheticEngine:
                     pass
 
     def _deliver(self, entries):
-        for q, out, fin in entries:
-            q.put_nowait(out)
-            if fin:
-                q.put_nowait(self._SENTINEL)
+        touched = {}
+        for q, out in entries:
+            if isinstance(q, asyncio.Queue):
+                q.put_nowait(out)
+                if out.finished:
+                    q.put_nowait(self._SENTINEL)
+            else:
+                q.feed(out)
+                touched[id(q)] = q
+        for sink in touched.values():
+            sink.flush()
+
+    # ---- direct-sink API (matches AsyncEngine) -------------------------
+    def add_direct(self, prompt_token_ids, sampling, sink) -> int:
+        loop = asyncio.get_running_loop()
+        with self._lock:
+            rid = self._rid = self._rid + 1
+            self._streams[rid] = [sink, loop, sampling.max_new_tokens]
+        self._wake.set()
+        return rid
+
+    def close_direct(self, rid: int) -> None:
+        with self._lock:
+            self._streams.pop(rid, None)
 
     async def generate(self, prompt_token_ids, sampling):
         async for outs in self.generate_batched(prompt_token_ids, sampling):
