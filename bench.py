@@ -207,13 +207,11 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
     srv = HttpServer(gw, "127.0.0.1", port)
     await srv.start()
 
-    per = max(1, args.batch // args.client_procs)
+    nprocs = min(args.client_procs, args.batch)
+    counts = [args.batch // nprocs + (1 if i < args.batch % nprocs else 0)
+              for i in range(nprocs)]
     procs = []
-    extra = args.batch - per * (args.client_procs - 1)
-    for i in range(args.client_procs):
-        n = extra if i == args.client_procs - 1 else per
-        if n <= 0:
-            continue
+    for n in counts:
         procs.append(subprocess.Popen(
             [sys.executable, os.path.abspath(__file__), "--client-worker",
              "--port", str(srv.port), "--batch", str(n),
