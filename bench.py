@@ -221,9 +221,19 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
             stdout=subprocess.PIPE, cwd=str(Path(__file__).parent)))
 
     # ---- ramp: wait for every stream to be admitted and prefilled ----
-    deadline = time.monotonic() + 900
+    t_ramp0 = time.monotonic()
+    deadline = t_ramp0 + 900
+    next_trace = t_ramp0
     while time.monotonic() < deadline:
         running = len(eng.scheduler.running)
+        now = time.monotonic()
+        if now >= next_trace:
+            print(f"[ramp] t={now - t_ramp0:6.1f}s steps={eng.num_steps} "
+                  f"prompt_toks={eng.num_prompt_tokens} "
+                  f"gen={eng.num_generated_tokens} run={running} "
+                  f"wait={len(eng.scheduler.waiting)} "
+                  f"wire={provider.streamed_tokens}", file=sys.stderr)
+            next_trace = now + 2.0
         if running >= args.batch and not eng.scheduler.waiting:
             break
         await asyncio.sleep(0.05)
@@ -231,6 +241,9 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
         raise RuntimeError(
             f"ramp timeout: {len(eng.scheduler.running)}/{args.batch} "
             "streams running")
+    print(f"[ramp] done in {time.monotonic() - t_ramp0:.1f}s: "
+          f"steps={eng.num_steps} prompt_toks={eng.num_prompt_tokens} "
+          f"gen={eng.num_generated_tokens}", file=sys.stderr)
 
     # ---- warmup W steps, then barrier + sync ----
     s_mark = eng.num_steps + args.warmup
@@ -254,6 +267,10 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
     w1 = provider.streamed_tokens
     run1 = len(eng.scheduler.running)
     g1 = eng.num_generated_tokens
+    ctxs = [s.num_tokens for s in eng.scheduler.running]
+    if ctxs:
+        print(f"[window] mean_ctx={sum(ctxs) / len(ctxs):.0f} "
+              f"max_ctx={max(ctxs)}", file=sys.stderr)
     elapsed = t1 - t0
     wire_tokens = w1 - w0
     engine_tokens = args.batch * args.steps
