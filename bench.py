@@ -175,10 +175,12 @@ def bench_serving(args) -> None:
         use_hip_graphs=not args.eager,
         enforce_eager=args.eager,
         num_gpu_blocks=args.num_gpu_blocks or None,
-        # serving posture: capture every decode bucket at init — lazy
-        # captures during the burst ramp cost ~1.7 s × 23 buckets of
-        # added TTFT (measured r2: 45 s ramp vs 6.3 s of prefill compute)
+        # serving posture: capture every decode bucket at init (no lazy
+        # capture stalls mid-stream) and strict prefill-priority
+        # admission — the burst ramp then takes ~prefill-compute time
+        # instead of 41 s of decode-starved trickle admission
         capture_at_init=True,
+        scheduler_prefill_first=True,
     )
     engine = AsyncEngine(ecfg, device=device)
     lc = LocalConfig(enabled=True, model=args.model,

@@ -150,6 +150,7 @@ class LocalConfig:
     enforce_eager: bool = False
     prefix_caching: bool = False
     chunked_prefill: bool = False
+    prefill_first: bool = True        # strict admission priority
     tokenizer_path: str = ""          # tokenizers json; "" = byte fallback
 
 
@@ -276,6 +277,7 @@ def load(env: Optional[Mapping[str, str]] = None) -> Config:
     lo.enforce_eager = _bool(g("LOCAL_ENFORCE_EAGER", "false"))
     lo.prefix_caching = _bool(g("LOCAL_PREFIX_CACHING", "false"))
     lo.chunked_prefill = _bool(g("LOCAL_CHUNKED_PREFILL", "false"))
+    lo.prefill_first = _bool(g("LOCAL_PREFILL_FIRST", "true"))
     lo.tokenizer_path = g("LOCAL_TOKENIZER_PATH", "")
 
     # provider bootstrap: registry defaults + env overrides

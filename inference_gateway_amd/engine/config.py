@@ -148,6 +148,10 @@ class EngineConfig:
     # prompt prefills in successive chunks (paged-extend attention) so a
     # burst never stalls in-flight decodes for a whole long prefill.
     enable_chunked_prefill: bool = False
+    # strict admission priority: always prefill when prompts wait
+    # (vLLM-style, TTFT-optimal under bursts) instead of alternating
+    # prefill/decode rounds (ITL-optimal).  See scheduler.py.
+    scheduler_prefill_first: bool = False
     tensor_parallel_size: int = 1
     seed: int = 0
 

@@ -125,7 +125,8 @@ class LLMEngine:
         self.scheduler = Scheduler(
             self.block_manager, cfg.max_num_seqs, cfg.max_batched_tokens,
             cfg.max_model_len,
-            chunked_prefill=cfg.enable_chunked_prefill)
+            chunked_prefill=cfg.enable_chunked_prefill,
+            prefill_first=cfg.scheduler_prefill_first)
         self.sampler = Sampler(self.device, cfg.seed)
         self.kv_caches = self._allocate_kv(num_blocks)
         self._seq_counter = itertools.count()

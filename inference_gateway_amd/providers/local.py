@@ -240,6 +240,8 @@ class LocalProvider:
                                           False),
             enable_chunked_prefill=getattr(local_cfg, "chunked_prefill",
                                            False),
+            scheduler_prefill_first=getattr(local_cfg, "prefill_first",
+                                            True),
             capture_at_init=True,
         )
         self.engine = engine or AsyncEngine(ecfg, device=device)

@@ -887,6 +887,42 @@ def test_abort_mid_chunk_releases_blocks():
     assert not eng.scheduler.has_work
 
 
+def test_prefill_first_policy_admits_before_decode():
+    """scheduler_prefill_first: waiting prompts are admitted in
+    consecutive prefill rounds (no decode alternation) until the queue
+    drains — the burst-TTFT policy; the default still alternates."""
+    from inference_gateway_amd.engine.scheduler import Scheduler, Sequence
+
+    def mk(policy):
+        bm = BlockManager(256, 16)
+        s = Scheduler(bm, max_num_seqs=16, max_batched_tokens=32,
+                      max_model_len=256, prefill_first=policy)
+        for i in range(4):
+            s.add(Sequence(seq_id=i, prompt_token_ids=list(range(32)),
+                           sampling=SamplingParams()))
+        return s
+
+    s = mk(True)
+    kinds = []
+    for _ in range(4):
+        b = s.schedule()
+        kinds.append("p" if b.is_prefill else "d")
+        if b.is_prefill:
+            for q, t in zip(b.seqs, b.prefill_takes):
+                q.prefilled += t
+    assert kinds == ["p", "p", "p", "p"]      # all admissions first
+
+    s = mk(False)
+    kinds = []
+    for _ in range(4):
+        b = s.schedule()
+        kinds.append("p" if b.is_prefill else "d")
+        if b.is_prefill:
+            for q, t in zip(b.seqs, b.prefill_takes):
+                q.prefilled += t
+    assert kinds == ["p", "d", "p", "d"]      # alternation (default)
+
+
 def test_abort_mid_chunk_does_not_poison_prefix_cache():
     """with chunked prefill + prefix caching,
     allocate_for_prompt hash-indexes ALL full prompt blocks up front;
