@@ -90,7 +90,7 @@ async def client_worker_main(args) -> None:
         body = json.dumps({
             "model": f"local/{args.model}",
             "messages": [{"role": "user", "content": content}],
-            "max_tokens": args.steps + args.warmup + 2048,
+            "max_tokens": args.steps + args.warmup + 512,
             "temperature": 0.8, "top_k": 50, "top_p": 0.95,
             "stream": True,
         }).encode()
@@ -166,7 +166,7 @@ def bench_serving(args) -> None:
     port = args.port + rank
 
     torch.manual_seed(1234 + rank)
-    max_len = args.prompt_len + 128 + args.steps + args.warmup + 2048 + 64
+    max_len = args.prompt_len + 128 + args.steps + args.warmup + 512 + 64
     ecfg = EngineConfig.for_model(
         args.model,
         max_num_seqs=args.batch,
@@ -209,6 +209,13 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
     srv = HttpServer(gw, "127.0.0.1", port)
     await srv.start()
 
+    # absorb the connect burst with the engine held: the event loop
+    # handles all handshakes/requests undisturbed, then the engine
+    # serves the whole burst in consecutive prefill rounds
+    # (prefill-first).  Without this, token delivery for early streams
+    # competes with request handling and admission trickles for minutes.
+    engine.hold()
+
     nprocs = min(args.client_procs, args.batch)
     counts = [args.batch // nprocs + (1 if i < args.batch % nprocs else 0)
               for i in range(nprocs)]
@@ -221,6 +228,19 @@ async def _serving_run(args, gw, provider, engine, port, rank, nranks,
              "--model", args.model,
              "--steps", str(args.steps), "--warmup", str(args.warmup)],
             stdout=subprocess.PIPE, cwd=str(Path(__file__).parent)))
+
+    # ---- arrival: every request registered before the engine starts ----
+    t_arr0 = time.monotonic()
+    while time.monotonic() < t_arr0 + 600:
+        if len(eng.scheduler.waiting) >= args.batch:
+            break
+        await asyncio.sleep(0.05)
+    else:
+        raise RuntimeError(
+            f"arrival timeout: {len(eng.scheduler.waiting)}/{args.batch}")
+    print(f"[arrival] {args.batch} requests in "
+          f"{time.monotonic() - t_arr0:.1f}s", file=sys.stderr)
+    engine.release()
 
     # ---- ramp: wait for every stream to be admitted and prefilled ----
     t_ramp0 = time.monotonic()

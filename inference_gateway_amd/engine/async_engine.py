@@ -49,6 +49,7 @@ class AsyncEngine:
         self._lock = threading.Lock()
         self._wake = threading.Event()
         self._stop = False
+        self._hold = False
         self._fatal = None
         self._queues: Dict[int, asyncio.Queue] = {}
         self._loops: Dict[int, asyncio.AbstractEventLoop] = {}
@@ -81,8 +82,22 @@ class AsyncEngine:
             else:
                 loop.call_soon_threadsafe(q.fail, self._fatal)
 
+    def hold(self) -> None:
+        """Pause stepping (requests still register).  Used by the bench
+        to absorb a connect burst with the event loop undisturbed, then
+        serve it in consecutive prefill rounds."""
+        self._hold = True
+
+    def release(self) -> None:
+        self._hold = False
+        self._wake.set()
+
     def _run_loop(self) -> None:
         while not self._stop:
+            if self._hold:
+                self._wake.wait(timeout=0.05)
+                self._wake.clear()
+                continue
             with self._lock:
                 has_work = self.engine.scheduler.has_work
             if not has_work:
