@@ -103,6 +103,20 @@ def _maybe_load_tuned_gemms() -> None:
     tunable.read_file(str(csv))
 
 
+def auto_num_blocks(model_cfg, block_size: int, tp: int,
+                    free_bytes: int, utilization: float) -> int:
+    """KV pool sizing from free HBM AFTER weights (the caller measures
+    free memory post-model-init).  Per-rank KV bytes shrink with TP (KV
+    heads are sharded); a flat 2 GiB is held back for graph pools,
+    activations and allocator slack.  Factored out of LLMEngine so the
+    70B TP=8 budget is testable without materializing 70B params
+    (tests/test_parallel_cpu.py::test_70b_tp8_memory_budget)."""
+    bytes_per_block = (
+        model_cfg.kv_bytes_per_token() * block_size // tp)
+    budget = int(free_bytes * utilization) - (2 << 30)
+    return max(64, budget // bytes_per_block)
+
+
 class LLMEngine:
     def __init__(self, cfg: EngineConfig, device: str = "cuda:0"):
         self.cfg = cfg
@@ -160,15 +174,13 @@ class LLMEngine:
     def _auto_num_blocks(self) -> int:
         from ..parallel.comm import tp_world_size
 
-        m = self.cfg.model
-        bytes_per_block = (
-            m.kv_bytes_per_token() * self.cfg.block_size // tp_world_size())
         if self.device.type == "cuda":
             free, _total = torch.cuda.mem_get_info(self.device)
-            budget = int(free * self.cfg.gpu_memory_utilization) - (2 << 30)
         else:
-            budget = 1 << 30
-        return max(64, budget // bytes_per_block)
+            free = (1 << 30) + (2 << 30)
+        return auto_num_blocks(self.cfg.model, self.cfg.block_size,
+                               tp_world_size(), free,
+                               self.cfg.gpu_memory_utilization)
 
     def _allocate_kv(self, num_blocks: int):
         from ..parallel.comm import tp_world_size
@@ -258,14 +270,30 @@ class LLMEngine:
         if not batch.is_prefill and gr is not None:
             gr.write_block_appends(batch.new_blocks)
             ver = self.scheduler.membership_version
-            ring, ev = gr.execute(batch.seqs,
-                                  membership_changed=ver != self._gr_version)
-            self._gr_version = ver
-            prev = self._drain_pending()      # overlaps with this step's GPU
-            for seq in batch.seqs:
-                seq.in_flight = True
-            self._pending = (list(batch.seqs), ring, ev)
-            return prev
+            try:
+                ring, ev = gr.execute(
+                    batch.seqs, membership_changed=ver != self._gr_version)
+            except RuntimeError as e:
+                # hipGraph capture can fail when a collective cannot be
+                # captured on this ROCm build (TP without the one-shot
+                # kernel: RCCL-in-graph is version-fragile).  Permanent
+                # eager fallback — correctness first, the step still
+                # runs below (eager-TP as automatic
+                # fallback).
+                import warnings
+
+                warnings.warn(
+                    f"decode graph capture failed ({e}); falling back "
+                    "to eager decode for the rest of this process")
+                self.graph_runner = None
+                gr = None
+            else:
+                self._gr_version = ver
+                prev = self._drain_pending()  # overlaps this step's GPU
+                for seq in batch.seqs:
+                    seq.in_flight = True
+                self._pending = (list(batch.seqs), ring, ev)
+                return prev
 
         prev = self._drain_pending()
         if batch.is_prefill:

@@ -1,0 +1,253 @@
+// One-shot all-reduce over xGMI peer-to-peer for decode-size tensors.
+//
+// MI355X topology: the 8-GPU node is FULLY CONNECTED —
+// 7 xGMI links per GPU at ~153 GB/s each.  A ring all-reduce moves
+// 2·(N-1)/N·B bytes over ONE link serially plus 2(N-1) latency hops; a
+// one-shot all-reduce has every rank read all peers' contributions
+// DIRECTLY, moving (N-1)/N·B bytes per link in PARALLEL with one
+// synchronization round.  For the ≤-few-MB tensors of decode-time
+// tensor parallelism this is both lower latency and higher bandwidth
+// than the ring, and it is hipGraph-capturable by construction (plain
+// kernel launches, device-side epoch counters — no RCCL in the graph).
+//
+// Protocol per call (stream-ordered, one kernel):
+//   phase 0: block b copies its slice of this rank's input into this
+//            rank's shared buffer (double-buffered by call parity);
+//   fence:   system-scope release, then block b writes its epoch into
+//            slot [b][me] of EVERY rank's signal page;
+//   spin:    block b polls its OWN signal page until all ranks' [b][r]
+//            have reached the epoch (relaxed system loads + one system
+//            acquire), with a bounded spin that sets an error word
+//            instead of hanging the device;
+//   phase 1: block b sums slice b from all ranks' shared buffers
+//            (bf16 -> fp32 accumulate) and writes the result.
+//
+// Epochs advance in device memory (one counter per block), so graph
+// replays are self-synchronizing.  Double buffering by epoch parity
+// makes a one-call skew between ranks safe: a rank can only be writing
+// parity p when every rank has finished reading parity p from the call
+// before last (the flag handshake orders it).
+//
+// Cross-GPU visibility notes (MI355X_MICROARCH.md §Workgroup dispatch):
+// peer writes land over xGMI; we use SYSTEM-scope atomics for flags and
+// an explicit s_waitcnt before the release (the ROCm 7.2 compiler can
+// drop the post-release wait when the scoreboard looks clean — guide
+// §6 G16 pitfall 12; the asm form is immune).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace igw {
+
+constexpr int AR_MAX_RANKS = 8;
+constexpr int AR_BLOCKS = 64;
+constexpr int AR_THREADS = 256;
+// spin bound: ~2.4 GHz, a few hundred ms worth of polls — long enough
+// for any real skew, short enough to never look like a hung GPU
+constexpr long long AR_SPIN_LIMIT = 400LL * 1000 * 1000;
+
+struct OneShotSig {
+  // [block][rank]: written by rank `rank` (its block `block`), on every
+  // rank's page; polled by the owner of the page
+  unsigned int flags[AR_BLOCKS][AR_MAX_RANKS];
+  unsigned int epochs[AR_BLOCKS];   // private per-owner call counters
+  unsigned int error;               // set on spin timeout
+};
+
+struct PtrTable {
+  const uint16_t* data[AR_MAX_RANKS];
+  OneShotSig* sig[AR_MAX_RANKS];
+};
+
+__global__ void __launch_bounds__(AR_THREADS)
+oneshot_allreduce_kernel(uint16_t* __restrict__ out,
+                         const uint16_t* __restrict__ inp,
+                         PtrTable tab, int rank, int world, long numel,
+                         long buf_stride /* elements per parity half */) {
+  const int b = blockIdx.x;
+  __shared__ unsigned int s_epoch;
+  OneShotSig* my_sig = tab.sig[rank];
+  if (threadIdx.x == 0) {
+    s_epoch = my_sig->epochs[b] + 1;
+    my_sig->epochs[b] = s_epoch;
+  }
+  __syncthreads();
+  const unsigned int epoch = s_epoch;
+  const long par_off = (epoch & 1) ? buf_stride : 0;
+
+  // slice owned by this block (grid-stride inside the slice keeps the
+  // phases' element->block map identical)
+  const long per_block = (numel + gridDim.x - 1) / gridDim.x;
+  const long lo = (long)b * per_block;
+  const long hi0 = lo + per_block;
+  const long hi = hi0 < numel ? hi0 : numel;
+  // vectors cover [lo, tail); the ragged remainder [tail, hi) is scalar
+  const long tail = hi > lo ? lo + ((hi - lo) / 8) * 8 : lo;
+  uint16_t* my_buf = const_cast<uint16_t*>(tab.data[rank]) + par_off;
+
+  // ---- phase 0: copy my slice into my shared buffer (vector 8xbf16) --
+  {
+    for (long i = lo + (long)threadIdx.x * 8; i + 8 <= hi;
+         i += (long)AR_THREADS * 8) {
+      *reinterpret_cast<uint4*>(my_buf + i) =
+          *reinterpret_cast<const uint4*>(inp + i);
+    }
+    if (threadIdx.x == 0) {
+      for (long j = tail; j < hi; ++j) my_buf[j] = inp[j];
+    }
+  }
+  // EVERY storing wave drains its stores to L2 before any wave can
+  // publish (guide §6 G16 R1: lane-0's release fence only writes back
+  // what has reached the L2)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  // ---- signal: release my slice, stamp epoch on every rank's page ----
+  if (threadIdx.x < (unsigned)world) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");  // system scope
+    // the compiler may drop the post-wbl2 wait when its scoreboard
+    // looks clean (guide §6 G16 pitfall 12); the asm form is immune
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __hip_atomic_store(&tab.sig[threadIdx.x]->flags[b][rank], epoch,
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+
+  // ---- spin: wait for every rank's slice b (bounded) -----------------
+  if (threadIdx.x == 0) {
+    long long spins = 0;
+    for (int r = 0; r < world; ++r) {
+      for (;;) {
+        unsigned int f = __hip_atomic_load(
+            &my_sig->flags[b][r], __ATOMIC_RELAXED,
+            __HIP_MEMORY_SCOPE_SYSTEM);
+        if ((int)(f - epoch) >= 0) break;
+        if (++spins > AR_SPIN_LIMIT) {
+          __hip_atomic_store(&my_sig->error, 1u, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_SYSTEM);
+          return;  // poisoned: host checks the error word
+        }
+        __builtin_amdgcn_s_sleep(1);
+      }
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");  // system scope
+  }
+  __syncthreads();
+
+  // ---- phase 1: sum slice b from all ranks -> out --------------------
+  {
+    for (long i = lo + (long)threadIdx.x * 8; i + 8 <= hi;
+         i += (long)AR_THREADS * 8) {
+      float acc[8];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) acc[k] = 0.f;
+      for (int r = 0; r < world; ++r) {
+        const uint4 v = *reinterpret_cast<const uint4*>(
+            tab.data[r] + par_off + i);
+        const uint16_t* h = reinterpret_cast<const uint16_t*>(&v);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) acc[k] += bf16_to_f32(h[k]);
+      }
+      uint4 o;
+      uint16_t* oh = reinterpret_cast<uint16_t*>(&o);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) oh[k] = f32_to_bf16(acc[k]);
+      *reinterpret_cast<uint4*>(out + i) = o;
+    }
+    if (threadIdx.x == 0) {
+      for (long j = tail; j < hi; ++j) {
+        float acc = 0.f;
+        for (int r = 0; r < world; ++r)
+          acc += bf16_to_f32(tab.data[r][par_off + j]);
+        out[j] = f32_to_bf16(acc);
+      }
+    }
+  }
+}
+
+// --------------------------------------------------------------------------
+// host API
+// --------------------------------------------------------------------------
+int64_t ar_alloc(int64_t bytes, bool finegrained) {
+  // signal pages are FINE-GRAINED (coherent during kernel execution —
+  // cross-device flag polling needs it; coarse-grained device memory is
+  // only coherent at dispatch boundaries).  Data buffers stay
+  // coarse-grained: the release/acquire fences order them, and coarse
+  // remote reads stream over xGMI at full rate.
+  void* p = nullptr;
+  hipError_t err;
+  if (finegrained) {
+    err = hipExtMallocWithFlags(&p, bytes, hipDeviceMallocFinegrained);
+  } else {
+    err = hipMalloc(&p, bytes);
+  }
+  TORCH_CHECK(err == hipSuccess, "ar_alloc failed: ",
+              hipGetErrorString(err));
+  TORCH_CHECK(hipMemset(p, 0, bytes) == hipSuccess);
+  return reinterpret_cast<int64_t>(p);
+}
+
+void ar_free(int64_t ptr) {
+  hipFree(reinterpret_cast<void*>(ptr));
+}
+
+py::bytes ar_get_handle(int64_t ptr) {
+  hipIpcMemHandle_t h;
+  auto err = hipIpcGetMemHandle(&h, reinterpret_cast<void*>(ptr));
+  TORCH_CHECK(err == hipSuccess, "hipIpcGetMemHandle failed: ",
+              hipGetErrorString(err));
+  return py::bytes(reinterpret_cast<const char*>(&h), sizeof(h));
+}
+
+int64_t ar_open_handle(py::bytes handle) {
+  std::string s = handle;
+  TORCH_CHECK(s.size() == sizeof(hipIpcMemHandle_t),
+              "bad IPC handle size");
+  hipIpcMemHandle_t h;
+  memcpy(&h, s.data(), sizeof(h));
+  void* p = nullptr;
+  auto err = hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess);
+  TORCH_CHECK(err == hipSuccess, "hipIpcOpenMemHandle failed: ",
+              hipGetErrorString(err));
+  return reinterpret_cast<int64_t>(p);
+}
+
+void ar_close_handle(int64_t ptr) {
+  hipIpcCloseMemHandle(reinterpret_cast<void*>(ptr));
+}
+
+int64_t ar_sig_bytes() { return (int64_t)sizeof(OneShotSig); }
+
+int64_t ar_check_error(int64_t sig_ptr) {
+  unsigned int e = 0;
+  hipMemcpy(&e, &reinterpret_cast<OneShotSig*>(sig_ptr)->error,
+            sizeof(e), hipMemcpyDeviceToHost);
+  return (int64_t)e;
+}
+
+void oneshot_allreduce(torch::Tensor inout, int64_t rank, int64_t world,
+                       std::vector<int64_t> data_ptrs,
+                       std::vector<int64_t> sig_ptrs,
+                       int64_t buf_stride_elems) {
+  TORCH_CHECK(inout.is_cuda() && inout.is_contiguous());
+  TORCH_CHECK(inout.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(world >= 1 && world <= AR_MAX_RANKS);
+  TORCH_CHECK((int64_t)data_ptrs.size() >= world);
+  const long numel = (long)inout.numel();
+  TORCH_CHECK(numel <= buf_stride_elems,
+              "tensor larger than the one-shot buffer");
+  PtrTable tab{};
+  for (int r = 0; r < world; ++r) {
+    tab.data[r] = reinterpret_cast<const uint16_t*>(data_ptrs[r]);
+    tab.sig[r] = reinterpret_cast<OneShotSig*>(sig_ptrs[r]);
+  }
+  uint16_t* ptr = reinterpret_cast<uint16_t*>(inout.data_ptr());
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(oneshot_allreduce_kernel, dim3(AR_BLOCKS),
+                     dim3(AR_THREADS), 0, stream, ptr, ptr, tab,
+                     (int)rank, (int)world, numel,
+                     (long)buf_stride_elems);
+}
+
+}  // namespace igw

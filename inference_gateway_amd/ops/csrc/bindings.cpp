@@ -40,6 +40,17 @@ void moe_grouped_gemm(torch::Tensor out, torch::Tensor a, torch::Tensor w,
                       torch::Tensor counts, int64_t capacity);
 void silu_mul_grouped(torch::Tensor out, torch::Tensor gate_up,
                       torch::Tensor counts, int64_t capacity);
+int64_t ar_alloc(int64_t bytes, bool finegrained);
+void ar_free(int64_t ptr);
+py::bytes ar_get_handle(int64_t ptr);
+int64_t ar_open_handle(py::bytes handle);
+void ar_close_handle(int64_t ptr);
+int64_t ar_sig_bytes();
+int64_t ar_check_error(int64_t sig_ptr);
+void oneshot_allreduce(torch::Tensor inout, int64_t rank, int64_t world,
+                       std::vector<int64_t> data_ptrs,
+                       std::vector<int64_t> sig_ptrs,
+                       int64_t buf_stride_elems);
 }  // namespace igw
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -66,4 +77,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped per-expert MFMA GEMM over the padded dispatch buffer");
   m.def("silu_mul_grouped", &igw::silu_mul_grouped,
         "SwiGLU over only the real (non-padded) MoE dispatch rows");
+  m.def("ar_alloc", &igw::ar_alloc,
+        "raw device allocation for one-shot all-reduce buffers "
+        "(finegrained=true for signal pages)");
+  m.def("ar_free", &igw::ar_free, "free an ar_alloc allocation");
+  m.def("ar_get_handle", &igw::ar_get_handle,
+        "hipIpcMemHandle for a raw allocation (64 bytes)");
+  m.def("ar_open_handle", &igw::ar_open_handle,
+        "open a peer rank's IPC handle -> device pointer");
+  m.def("ar_close_handle", &igw::ar_close_handle,
+        "close an opened peer IPC handle");
+  m.def("ar_sig_bytes", &igw::ar_sig_bytes,
+        "size of the one-shot signal page");
+  m.def("ar_check_error", &igw::ar_check_error,
+        "read the one-shot spin-timeout error word");
+  m.def("oneshot_allreduce", &igw::oneshot_allreduce,
+        "one-shot bf16 sum all-reduce over xGMI peer buffers "
+        "(graph-capturable, device-side epochs)");
 }

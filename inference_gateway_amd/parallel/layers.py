@@ -11,7 +11,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from .comm import tp_all_reduce
+from .comm import row_parallel_matmul_allreduce, tp_all_reduce
 
 
 def _init_weight(shape, device, dtype, seed: int) -> torch.Tensor:
@@ -56,7 +56,10 @@ class RowParallelLinear(nn.Module):
         self.weight = nn.Parameter(w, requires_grad=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return tp_all_reduce(F.linear(x, self.weight))
+        # prefill-size inputs chunk the GEMM and overlap each chunk's
+        # all-reduce (comm stream) with the next chunk's GEMM; decode
+        # sizes go straight through (one-shot xGMI kernel when active)
+        return row_parallel_matmul_allreduce(x, self.weight)
 
 
 class ReplicatedLinear(nn.Module):

@@ -346,3 +346,45 @@ def test_zz_chunked_prefill_engine_matches_unchunked():
                                       max_batched_tokens=48,
                                       enable_chunked_prefill=True))
     assert got == want
+
+
+@pytest.mark.gpu
+def test_oneshot_allreduce_loopback():
+    """One-shot all-reduce protocol on ONE GPU: 8 simulated ranks (8
+    buffer pairs, 8 concurrent streams) handshake through the real
+    epoch-flag protocol and each produce the exact rank-ordered fp32
+    sum.  The 8-GPU run differs only in the pointers coming from
+    hipIpcOpenMemHandle (parallel/oneshot.py) — this pins the kernel
+    math, the per-block slice map, ragged tails, double-buffer parity
+    and repeat-call epoch advance."""
+    from inference_gateway_amd import ops
+
+    k = ops._kernels
+    world = 8
+    max_bytes = 1 << 20
+    datas = [k.ar_alloc(2 * max_bytes, False) for _ in range(world)]
+    sigs = [k.ar_alloc(k.ar_sig_bytes(), True) for _ in range(world)]
+    try:
+        for trial, numel in enumerate([8192, 8191, 65536, 7, 524288]):
+            gens = torch.Generator(device="cuda").manual_seed(100 + trial)
+            ins = [torch.randn(numel, device="cuda", generator=gens,
+                               dtype=torch.float32).to(torch.bfloat16)
+                   for _ in range(world)]
+            acc = torch.zeros(numel, device="cuda")
+            for t in ins:
+                acc += t.float()
+            want = acc.to(torch.bfloat16)
+            streams = [torch.cuda.Stream() for _ in range(world)]
+            for r in range(world):
+                with torch.cuda.stream(streams[r]):
+                    k.oneshot_allreduce(ins[r], r, world, datas, sigs,
+                                        max_bytes // 2)
+            torch.cuda.synchronize()
+            for r in range(world):
+                assert k.ar_check_error(sigs[r]) == 0, (trial, r)
+                assert torch.equal(ins[r], want), (
+                    trial, r,
+                    (ins[r].float() - want.float()).abs().max().item())
+    finally:
+        for p in datas + sigs:
+            k.ar_free(p)
