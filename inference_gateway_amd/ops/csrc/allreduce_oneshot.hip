@@ -46,7 +46,7 @@ constexpr int AR_BLOCKS = 64;
 constexpr int AR_THREADS = 256;
 // spin bound: ~2.4 GHz, a few hundred ms worth of polls — long enough
 // for any real skew, short enough to never look like a hung GPU
-constexpr long long AR_SPIN_LIMIT = 400LL * 1000 * 1000;
+constexpr long long AR_SPIN_LIMIT = 200LL * 1000 * 1000;
 
 struct OneShotSig {
   // [block][rank]: written by rank `rank` (its block `block`), on every
@@ -70,6 +70,7 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
   __shared__ unsigned int s_epoch;
   OneShotSig* my_sig = tab.sig[rank];
   if (threadIdx.x == 0) {
+    if (b == 0) my_sig->error = 0;   // a past timeout is not this call's
     s_epoch = my_sig->epochs[b] + 1;
     my_sig->epochs[b] = s_epoch;
   }
@@ -124,8 +125,11 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
             __HIP_MEMORY_SCOPE_SYSTEM);
         if ((int)(f - epoch) >= 0) break;
         if (++spins > AR_SPIN_LIMIT) {
-          __hip_atomic_store(&my_sig->error, 1u, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_SYSTEM);
+          // encode what we were waiting for: block, rank, last value
+          __hip_atomic_store(&my_sig->error,
+                             0x80000000u | ((unsigned)b << 20)
+                                 | ((unsigned)r << 16) | (f & 0xffffu),
+                             __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
           return;  // poisoned: host checks the error word
         }
         __builtin_amdgcn_s_sleep(1);
@@ -224,6 +228,13 @@ int64_t ar_check_error(int64_t sig_ptr) {
   hipMemcpy(&e, &reinterpret_cast<OneShotSig*>(sig_ptr)->error,
             sizeof(e), hipMemcpyDeviceToHost);
   return (int64_t)e;
+}
+
+py::bytes ar_dump(int64_t ptr, int64_t nbytes) {
+  std::string s(nbytes, '\0');
+  hipMemcpy(&s[0], reinterpret_cast<void*>(ptr), nbytes,
+            hipMemcpyDeviceToHost);
+  return py::bytes(s);
 }
 
 void oneshot_allreduce(torch::Tensor inout, int64_t rank, int64_t world,

@@ -47,6 +47,7 @@ int64_t ar_open_handle(py::bytes handle);
 void ar_close_handle(int64_t ptr);
 int64_t ar_sig_bytes();
 int64_t ar_check_error(int64_t sig_ptr);
+py::bytes ar_dump(int64_t ptr, int64_t nbytes);
 void oneshot_allreduce(torch::Tensor inout, int64_t rank, int64_t world,
                        std::vector<int64_t> data_ptrs,
                        std::vector<int64_t> sig_ptrs,
@@ -91,6 +92,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "size of the one-shot signal page");
   m.def("ar_check_error", &igw::ar_check_error,
         "read the one-shot spin-timeout error word");
+  m.def("ar_dump", &igw::ar_dump, "copy raw device bytes to host (debug)");
   m.def("oneshot_allreduce", &igw::oneshot_allreduce,
         "one-shot bf16 sum all-reduce over xGMI peer buffers "
         "(graph-capturable, device-side epochs)");

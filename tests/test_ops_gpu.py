@@ -350,17 +350,19 @@ def test_zz_chunked_prefill_engine_matches_unchunked():
 
 @pytest.mark.gpu
 def test_oneshot_allreduce_loopback():
-    """One-shot all-reduce protocol on ONE GPU: 8 simulated ranks (8
-    buffer pairs, 8 concurrent streams) handshake through the real
-    epoch-flag protocol and each produce the exact rank-ordered fp32
-    sum.  The 8-GPU run differs only in the pointers coming from
-    hipIpcOpenMemHandle (parallel/oneshot.py) — this pins the kernel
-    math, the per-block slice map, ragged tails, double-buffer parity
-    and repeat-call epoch advance."""
+    """One-shot all-reduce protocol on ONE GPU: 4 simulated ranks (4
+    buffer pairs, 4 concurrent streams — ROCm's default 4 hardware
+    queues dispatch at most 4 independent kernels concurrently, a
+    loopback-only constraint) handshake through the real epoch-flag
+    protocol and each produce the exact rank-ordered fp32 sum.  The
+    8-GPU run differs only in the pointers coming from
+    hipIpcOpenMemHandle (parallel/oneshot.py).  Pins the kernel math,
+    per-block slice map, ragged tails, double-buffer parity and
+    repeat-call epoch advance."""
     from inference_gateway_amd import ops
 
     k = ops._kernels
-    world = 8
+    world = 4
     max_bytes = 1 << 20
     datas = [k.ar_alloc(2 * max_bytes, False) for _ in range(world)]
     sigs = [k.ar_alloc(k.ar_sig_bytes(), True) for _ in range(world)]
@@ -381,10 +383,28 @@ def test_oneshot_allreduce_loopback():
                                         max_bytes // 2)
             torch.cuda.synchronize()
             for r in range(world):
-                assert k.ar_check_error(sigs[r]) == 0, (trial, r)
+                assert k.ar_check_error(sigs[r]) == 0, (
+                    trial, r, hex(k.ar_check_error(sigs[r])))
                 assert torch.equal(ins[r], want), (
                     trial, r,
                     (ins[r].float() - want.float()).abs().max().item())
     finally:
         for p in datas + sigs:
             k.ar_free(p)
+
+
+@pytest.mark.gpu
+def test_oneshot_allreduce_world8_subprocess():
+    """Full 8-rank protocol in a subprocess with GPU_MAX_HW_QUEUES=8
+    (the HIP runtime reads it at init, so it needs a fresh process).
+    Exercises exactly the deployment world size."""
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "tools/debug_oneshot.py", "--world", "8"],
+        cwd=root, capture_output=True, text=True, timeout=600,
+        env=dict(os.environ, GPU_MAX_HW_QUEUES="8"))
+    assert out.returncode == 0, out.stdout + out.stderr
