@@ -28,11 +28,12 @@ def dump_sig(ptr, world, blocks=2):
     return {"flags": out, "epochs": epochs, "error": hex(err)}
 
 
-def run(world, numel=4096, repeat=2):
+def run(world, numels=(4096,), repeat=2):
     datas = [k.ar_alloc(2 * MAXB, False) for _ in range(world)]
     sigs = [k.ar_alloc(k.ar_sig_bytes(), True) for _ in range(world)]
     ok = True
     try:
+      for numel in numels:
         for it in range(repeat):
             ins = [torch.randn(numel, device="cuda").to(torch.bfloat16)
                    for _ in range(world)]
@@ -47,13 +48,20 @@ def run(world, numel=4096, repeat=2):
                                         MAXB // 2)
             torch.cuda.synchronize()
             errs = [hex(k.ar_check_error(s)) for s in sigs]
-            if any(int(e, 16) for e in errs):
-                for r in range(world):
-                    print(f"  rank {r} sig:", dump_sig(sigs[r], world),
-                          flush=True)
             match = [bool(torch.equal(ins[r], want))
                      for r in range(world)]
-            print(f"world={world} iter={it} errors={errs} match={match}",
+            bad = [int((ins[r].float() - want.float()).abs().gt(0).sum())
+                   for r in range(world)]
+            first_bad = None
+            for r in range(world):
+                if not match[r]:
+                    idx = (ins[r].float()
+                           - want.float()).abs().gt(0).nonzero()
+                    first_bad = (r, idx[:4].flatten().tolist(),
+                                 int(idx.numel()))
+                    break
+            print(f"world={world} numel={numel} iter={it} errors={errs} "
+                  f"match={match} nbad={bad} first={first_bad}",
                   flush=True)
             ok = ok and not any(int(e, 16) for e in errs) and all(match)
     finally:
@@ -73,9 +81,10 @@ if __name__ == "__main__":
                          "kernels dispatch concurrently — a loopback "
                          "artifact only: real TP is one process per GPU)")
     a = ap.parse_args()
+    numels = (4096, 8192, 8191, 65536, 7, 524288)
     if a.world:
-        sys.exit(0 if run(a.world) else 1)
-    print("world=1:", run(1))
-    print("world=2:", run(2))
-    print("world=4:", run(4))
-    print("world=8:", run(8))
+        sys.exit(0 if run(a.world, numels) else 1)
+    print("world=1:", run(1, numels))
+    print("world=2:", run(2, numels))
+    print("world=4:", run(4, numels))
+    print("world=8:", run(8, numels))
