@@ -61,6 +61,13 @@ struct PtrTable {
   OneShotSig* sig[AR_MAX_RANKS];
 };
 
+typedef int igw_v4i __attribute__((ext_vector_type(4)));
+// aux bits for buffer ops: sc1 = write-through / no-allocate — shared
+// data goes straight to the coherence point and reads bypass stale
+// per-XCD L2 lines (guide §6 G16 R1: the valid all-sc1 protocol form;
+// plain stores + fences measured intermittently stale cross-XCD here)
+constexpr int AUX_SC1 = 16;
+
 __global__ void __launch_bounds__(AR_THREADS)
 oneshot_allreduce_kernel(uint16_t* __restrict__ out,
                          const uint16_t* __restrict__ inp,
@@ -87,30 +94,33 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
   // vectors cover [lo, tail); the ragged remainder [tail, hi) is scalar
   const long tail = hi > lo ? lo + ((hi - lo) / 8) * 8 : lo;
   uint16_t* my_buf = const_cast<uint16_t*>(tab.data[rank]) + par_off;
+  // write-through descriptor over my parity half (32-bit offsets)
+  const auto my_rsrc = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)my_buf, 0, (unsigned)(numel * 2), 0x00020000);
 
-  // ---- phase 0: copy my slice into my shared buffer (vector 8xbf16) --
+  // ---- phase 0: my slice -> my shared buffer, sc1 write-through ------
   {
     for (long i = lo + (long)threadIdx.x * 8; i + 8 <= hi;
          i += (long)AR_THREADS * 8) {
-      *reinterpret_cast<uint4*>(my_buf + i) =
-          *reinterpret_cast<const uint4*>(inp + i);
+      const uint4 v = *reinterpret_cast<const uint4*>(inp + i);
+      __builtin_amdgcn_raw_buffer_store_b128(
+          *reinterpret_cast<const igw_v4i*>(&v), my_rsrc,
+          (unsigned)(i * 2), 0, AUX_SC1);
     }
     if (threadIdx.x == 0) {
-      for (long j = tail; j < hi; ++j) my_buf[j] = inp[j];
+      for (long j = tail; j < hi; ++j) {
+        __builtin_amdgcn_raw_buffer_store_b16(
+            (short)inp[j], my_rsrc, (unsigned)(j * 2), 0, AUX_SC1);
+      }
     }
   }
-  // EVERY storing wave drains its stores to L2 before any wave can
-  // publish (guide §6 G16 R1: lane-0's release fence only writes back
-  // what has reached the L2)
+  // EVERY storing wave drains its write-through stores to the coherence
+  // point before any lane publishes the flag (guide §6 G16 R1)
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
-  // ---- signal: release my slice, stamp epoch on every rank's page ----
+  // ---- signal: stamp epoch on every rank's page (sc1 atomics) --------
   if (threadIdx.x < (unsigned)world) {
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");  // system scope
-    // the compiler may drop the post-wbl2 wait when its scoreboard
-    // looks clean (guide §6 G16 pitfall 12); the asm form is immune
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __hip_atomic_store(&tab.sig[threadIdx.x]->flags[b][rank], epoch,
                        __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
   }
@@ -120,9 +130,12 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
     long long spins = 0;
     for (int r = 0; r < world; ++r) {
       for (;;) {
+        // poll MY OWN page with agent scope (the guide's measured
+        // wait_eq form: sc1 load, no L2 allocate); the remote writer's
+        // store is system-scope and lands here over the fabric
         unsigned int f = __hip_atomic_load(
             &my_sig->flags[b][r], __ATOMIC_RELAXED,
-            __HIP_MEMORY_SCOPE_SYSTEM);
+            __HIP_MEMORY_SCOPE_AGENT);
         if ((int)(f - epoch) >= 0) break;
         if (++spins > AR_SPIN_LIMIT) {
           // encode what we were waiting for: block, rank, last value
@@ -135,20 +148,25 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
         __builtin_amdgcn_s_sleep(1);
       }
     }
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");  // system scope
   }
   __syncthreads();
 
-  // ---- phase 1: sum slice b from all ranks -> out --------------------
+  // ---- phase 1: sum slice b from all ranks (sc1 loads) -> out --------
   {
+    __amdgpu_buffer_rsrc_t rsrcs[AR_MAX_RANKS];
+    for (int r = 0; r < world; ++r) {
+      rsrcs[r] = __builtin_amdgcn_make_buffer_rsrc(
+          (void*)(tab.data[r] + par_off), 0, (unsigned)(numel * 2),
+          0x00020000);
+    }
     for (long i = lo + (long)threadIdx.x * 8; i + 8 <= hi;
          i += (long)AR_THREADS * 8) {
       float acc[8];
 #pragma unroll
       for (int k = 0; k < 8; ++k) acc[k] = 0.f;
       for (int r = 0; r < world; ++r) {
-        const uint4 v = *reinterpret_cast<const uint4*>(
-            tab.data[r] + par_off + i);
+        const igw_v4i v = __builtin_amdgcn_raw_buffer_load_b128(
+            rsrcs[r], (unsigned)(i * 2), 0, AUX_SC1);
         const uint16_t* h = reinterpret_cast<const uint16_t*>(&v);
 #pragma unroll
         for (int k = 0; k < 8; ++k) acc[k] += bf16_to_f32(h[k]);
@@ -162,8 +180,11 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
     if (threadIdx.x == 0) {
       for (long j = tail; j < hi; ++j) {
         float acc = 0.f;
-        for (int r = 0; r < world; ++r)
-          acc += bf16_to_f32(tab.data[r][par_off + j]);
+        for (int r = 0; r < world; ++r) {
+          const short hv = __builtin_amdgcn_raw_buffer_load_b16(
+              rsrcs[r], (unsigned)(j * 2), 0, AUX_SC1);
+          acc += bf16_to_f32((uint16_t)hv);
+        }
         out[j] = f32_to_bf16(acc);
       }
     }
