@@ -145,9 +145,22 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
                              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
           return;  // poisoned: host checks the error word
         }
-        __builtin_amdgcn_s_sleep(1);
+        if ((spins & 1023) == 0) {
+          // a stale flag line can sit in this XCD's L2 indefinitely:
+          // per-XCD L2s are not coherent and a remote write-through
+          // store does not invalidate them (measured: rare multi-
+          // second stalls at exactly one stale epoch).  A SYSTEM-scope
+          // acquire invalidates L2 and forces a refetch; amortized
+          // once per 1024 polls.
+          __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+        }
+        __builtin_amdgcn_s_sleep(8);
       }
     }
+    // one unconditional system acquire before the sum: every rank's
+    // DATA slice must be re-fetched past any stale L2 lines left from
+    // two calls ago (the same parity half)
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
   }
   __syncthreads();
 
