@@ -46,7 +46,7 @@ constexpr int AR_BLOCKS = 64;
 constexpr int AR_THREADS = 256;
 // spin bound: ~2.4 GHz, a few hundred ms worth of polls — long enough
 // for any real skew, short enough to never look like a hung GPU
-constexpr long long AR_SPIN_LIMIT = 200LL * 1000 * 1000;
+constexpr long long AR_SPIN_LIMIT = 20LL * 1000 * 1000;
 
 struct OneShotSig {
   // [block][rank]: written by rank `rank` (its block `block`), on every
@@ -119,10 +119,13 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
-  // ---- signal: stamp epoch on every rank's page (sc1 atomics) --------
+  // ---- signal: stamp epoch on every rank's page ----------------------
+  // RELEASE stores (system scope): the conservative documented form —
+  // the payload is already write-through, but release-ordering the flag
+  // also covers any store still in flight in this lane's queue
   if (threadIdx.x < (unsigned)world) {
     __hip_atomic_store(&tab.sig[threadIdx.x]->flags[b][rank], epoch,
-                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+                       __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
   }
 
   // ---- spin: wait for every rank's slice b (bounded) -----------------
@@ -133,9 +136,14 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
         // poll MY OWN page with agent scope (the guide's measured
         // wait_eq form: sc1 load, no L2 allocate); the remote writer's
         // store is system-scope and lands here over the fabric
+        // ACQUIRE poll, system scope: "polling with ACQUIRE loads ->
+        // correct, 2-3x slower per hop" (guide §Workgroup dispatch) —
+        // the flag page is 64 words, the hop cost is irrelevant next
+        // to the payload; relaxed polls measured rare indefinite
+        // staleness here (per-XCD L2)
         unsigned int f = __hip_atomic_load(
-            &my_sig->flags[b][r], __ATOMIC_RELAXED,
-            __HIP_MEMORY_SCOPE_AGENT);
+            &my_sig->flags[b][r], __ATOMIC_ACQUIRE,
+            __HIP_MEMORY_SCOPE_SYSTEM);
         if ((int)(f - epoch) >= 0) break;
         if (++spins > AR_SPIN_LIMIT) {
           // encode what we were waiting for: block, rank, last value
@@ -145,16 +153,7 @@ oneshot_allreduce_kernel(uint16_t* __restrict__ out,
                              __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
           return;  // poisoned: host checks the error word
         }
-        if ((spins & 1023) == 0) {
-          // a stale flag line can sit in this XCD's L2 indefinitely:
-          // per-XCD L2s are not coherent and a remote write-through
-          // store does not invalidate them (measured: rare multi-
-          // second stalls at exactly one stale epoch).  A SYSTEM-scope
-          // acquire invalidates L2 and forces a refetch; amortized
-          // once per 1024 polls.
-          __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
-        }
-        __builtin_amdgcn_s_sleep(8);
+        __builtin_amdgcn_s_sleep(2);
       }
     }
     // one unconditional system acquire before the sum: every rank's
