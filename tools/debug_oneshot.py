@@ -41,7 +41,9 @@ def run(world, numels=(4096,), repeat=2):
             for t in ins:
                 acc += t.float()
             want = acc.to(torch.bfloat16)
-            streams = [torch.cuda.Stream() for _ in range(world)]
+            if not hasattr(run, "_streams") or len(run._streams) < world:
+                run._streams = [torch.cuda.Stream() for _ in range(world)]
+            streams = run._streams[:world]
             for r in range(world):
                 with torch.cuda.stream(streams[r]):
                     k.oneshot_allreduce(ins[r], r, world, datas, sigs,
