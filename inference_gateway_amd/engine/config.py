@@ -152,6 +152,10 @@ class EngineConfig:
     # (vLLM-style, TTFT-optimal under bursts) instead of alternating
     # prefill/decode rounds (ITL-optimal).  See scheduler.py.
     scheduler_prefill_first: bool = False
+    # real checkpoint: HF-format directory (config.json +
+    # *.safetensors); None keeps the seed-derived synthetic init
+    # (engine/weights.py — the no-network default)
+    weights_path: Optional[str] = None
     tensor_parallel_size: int = 1
     seed: int = 0
 

@@ -131,6 +131,10 @@ class LLMEngine:
         else:
             self.model = LlamaForCausalLM(cfg.model, self.device, self.dtype)
         self.model.eval()
+        if cfg.weights_path:
+            from .weights import load_weights
+
+            load_weights(self.model, cfg.weights_path)
 
         num_blocks = cfg.num_gpu_blocks or self._auto_num_blocks()
         self.block_manager = BlockManager(

@@ -228,7 +228,14 @@ class LocalProvider:
                  engine: Optional[AsyncEngine] = None):
         self.id = LOCAL_PROVIDER_ID
         self.model_name = local_cfg.served_model_name or local_cfg.model
-        preset = MODEL_PRESETS[local_cfg.model]
+        weights = getattr(local_cfg, "weights_path", "") or None
+        if local_cfg.model == "hf":
+            # architecture straight from the checkpoint's config.json
+            from ..engine.weights import model_config_from_hf
+
+            preset = model_config_from_hf(weights)
+        else:
+            preset = MODEL_PRESETS[local_cfg.model]
         ecfg = EngineConfig(
             model=preset,
             max_num_seqs=local_cfg.max_num_seqs,
@@ -242,6 +249,7 @@ class LocalProvider:
                                            False),
             scheduler_prefill_first=getattr(local_cfg, "prefill_first",
                                             True),
+            weights_path=weights,
             capture_at_init=True,
         )
         self.engine = engine or AsyncEngine(ecfg, device=device)

@@ -1,0 +1,197 @@
+"""Real-checkpoint loading (engine/weights.py): HF config ingestion,
+safetensors mapping onto the fused/TP-sharded engine layout, loud
+failure on mismatch, end-to-end generation from a loaded checkpoint
+(weight loading is the stateless gateway's
+checkpoint/resume analog)."""
+import json
+
+import pytest
+import torch
+from safetensors.torch import save_file
+
+from inference_gateway_amd.engine.config import (MODEL_PRESETS,
+                                                 EngineConfig,
+                                                 SamplingParams)
+from inference_gateway_amd.engine.engine import LLMEngine
+from inference_gateway_amd.engine import weights as W
+
+
+def _tiny_llama_checkpoint(dirpath, seed=0):
+    """Fabricate a tiny-llama HF-format checkpoint (random weights)."""
+    m = MODEL_PRESETS["tiny-llama"]
+    g = torch.Generator().manual_seed(seed)
+
+    def r(*shape):
+        return torch.randn(*shape, generator=g) * 0.02
+
+    h, hd = m.hidden_size, m.head_dim
+    t = {"model.embed_tokens.weight": r(m.vocab_size, h),
+         "model.norm.weight": torch.ones(h),
+         "lm_head.weight": r(m.vocab_size, h)}
+    for i in range(m.num_layers):
+        p = f"model.layers.{i}."
+        t[p + "self_attn.q_proj.weight"] = r(m.num_heads * hd, h)
+        t[p + "self_attn.k_proj.weight"] = r(m.num_kv_heads * hd, h)
+        t[p + "self_attn.v_proj.weight"] = r(m.num_kv_heads * hd, h)
+        t[p + "self_attn.o_proj.weight"] = r(h, m.num_heads * hd)
+        t[p + "mlp.gate_proj.weight"] = r(m.intermediate_size, h)
+        t[p + "mlp.up_proj.weight"] = r(m.intermediate_size, h)
+        t[p + "mlp.down_proj.weight"] = r(h, m.intermediate_size)
+        t[p + "input_layernorm.weight"] = torch.ones(h)
+        t[p + "post_attention_layernorm.weight"] = torch.ones(h)
+    save_file(t, str(dirpath / "model.safetensors"))
+    (dirpath / "config.json").write_text(json.dumps({
+        "model_type": "llama",
+        "hidden_size": h,
+        "num_hidden_layers": m.num_layers,
+        "num_attention_heads": m.num_heads,
+        "num_key_value_heads": m.num_kv_heads,
+        "head_dim": hd,
+        "intermediate_size": m.intermediate_size,
+        "vocab_size": m.vocab_size,
+        "rope_theta": m.rope_base,
+        "rms_norm_eps": m.norm_eps,
+        "max_position_embeddings": m.max_position,
+    }))
+    return t
+
+
+def test_hf_config_ingestion(tmp_path):
+    _tiny_llama_checkpoint(tmp_path)
+    cfg = W.model_config_from_hf(str(tmp_path))
+    m = MODEL_PRESETS["tiny-llama"]
+    assert cfg.hidden_size == m.hidden_size
+    assert cfg.num_layers == m.num_layers
+    assert cfg.num_kv_heads == m.num_kv_heads
+    assert cfg.head_dim == m.head_dim
+    assert cfg.rope_base == m.rope_base
+    assert not cfg.is_moe
+
+
+def test_llama_weights_load_and_generate(tmp_path):
+    t = _tiny_llama_checkpoint(tmp_path)
+    eng = LLMEngine(EngineConfig.tiny_llama(weights_path=str(tmp_path)),
+                    device="cpu")
+    model = eng.model
+    # fused qkv = [q; k; v] in row order
+    want = torch.cat([t["model.layers.0.self_attn.q_proj.weight"],
+                      t["model.layers.0.self_attn.k_proj.weight"],
+                      t["model.layers.0.self_attn.v_proj.weight"]], 0)
+    assert torch.equal(model.layers[0].attn.qkv.data.float(),
+                       want.to(torch.bfloat16).float())
+    # fused gate_up = [gate; up]
+    want = torch.cat([t["model.layers.1.mlp.gate_proj.weight"],
+                      t["model.layers.1.mlp.up_proj.weight"]], 0)
+    assert torch.equal(model.layers[1].mlp.gate_up.data.float(),
+                       want.to(torch.bfloat16).float())
+    assert torch.equal(model.embed.data.float(),
+                       t["model.embed_tokens.weight"]
+                       .to(torch.bfloat16).float())
+    # the engine actually generates from the loaded checkpoint
+    rid = eng.add_request(list(range(1, 17)),
+                          SamplingParams(max_new_tokens=4, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    out = eng.get_output(rid)
+    assert out.finished and len(out.token_ids) == 4
+
+
+def test_tp_sharded_load_matches_manual_slices(tmp_path, monkeypatch):
+    """Rank 1 of TP=2 must hold exactly the second half of the head/
+    intermediate dimensions (sliced reads, not full reads)."""
+    from inference_gateway_amd.parallel import comm
+    from inference_gateway_amd.engine.models.llama import LlamaForCausalLM
+
+    t = _tiny_llama_checkpoint(tmp_path)
+    m = MODEL_PRESETS["tiny-llama"]
+    monkeypatch.setattr(comm, "_TP_SIZE", 2)
+    monkeypatch.setattr(comm, "_TP_RANK", 1)
+    model = LlamaForCausalLM(m, "cpu", torch.bfloat16)
+    W.load_weights(model, str(tmp_path))
+    hd = m.head_dim
+    nq, nkv = m.num_heads // 2, m.num_kv_heads // 2
+    q = t["model.layers.0.self_attn.q_proj.weight"][nq * hd:]
+    k = t["model.layers.0.self_attn.k_proj.weight"][nkv * hd:]
+    v = t["model.layers.0.self_attn.v_proj.weight"][nkv * hd:]
+    assert torch.equal(model.layers[0].attn.qkv.data.float(),
+                       torch.cat([q, k, v], 0).to(torch.bfloat16).float())
+    o = t["model.layers.0.self_attn.o_proj.weight"][:, nq * hd:]
+    assert torch.equal(model.layers[0].attn.o_proj.weight.data.float(),
+                       o.to(torch.bfloat16).float())
+
+
+def test_mismatched_checkpoint_fails_loudly(tmp_path):
+    _tiny_llama_checkpoint(tmp_path)
+    # break one tensor's shape
+    bad = {"model.embed_tokens.weight": torch.zeros(7, 7)}
+    save_file(bad, str(tmp_path / "model.safetensors"))
+    with pytest.raises(Exception):
+        LLMEngine(EngineConfig.tiny_llama(weights_path=str(tmp_path)),
+                  device="cpu")
+
+
+def test_moe_weights_load_and_generate(tmp_path):
+    m = MODEL_PRESETS["tiny-moe"]
+    g = torch.Generator().manual_seed(3)
+
+    def r(*shape):
+        return torch.randn(*shape, generator=g) * 0.02
+
+    h, hd = m.hidden_size, m.head_dim
+    t = {"model.embed_tokens.weight": r(m.vocab_size, h),
+         "model.norm.weight": torch.ones(h),
+         "lm_head.weight": r(m.vocab_size, h)}
+    for i in range(m.num_layers):
+        p = f"model.layers.{i}."
+        t[p + "self_attn.q_proj.weight"] = r(m.num_heads * hd, h)
+        t[p + "self_attn.k_proj.weight"] = r(m.num_kv_heads * hd, h)
+        t[p + "self_attn.v_proj.weight"] = r(m.num_kv_heads * hd, h)
+        t[p + "self_attn.o_proj.weight"] = r(h, m.num_heads * hd)
+        t[p + "block_sparse_moe.gate.weight"] = r(m.num_experts, h)
+        for e in range(m.num_experts):
+            ep = p + f"block_sparse_moe.experts.{e}."
+            t[ep + "w1.weight"] = r(m.intermediate_size, h)
+            t[ep + "w3.weight"] = r(m.intermediate_size, h)
+            t[ep + "w2.weight"] = r(h, m.intermediate_size)
+        t[p + "input_layernorm.weight"] = torch.ones(h)
+        t[p + "post_attention_layernorm.weight"] = torch.ones(h)
+    save_file(t, str(tmp_path / "model.safetensors"))
+    eng = LLMEngine(EngineConfig.tiny_moe(weights_path=str(tmp_path)),
+                    device="cpu")
+    gu = eng.model.layers[0].moe.gate_up.data
+    want = torch.cat(
+        [t["model.layers.0.block_sparse_moe.experts.1.w1.weight"],
+         t["model.layers.0.block_sparse_moe.experts.1.w3.weight"]], 0)
+    assert torch.equal(gu[1].float(), want.to(torch.bfloat16).float())
+    rid = eng.add_request(list(range(1, 17)),
+                          SamplingParams(max_new_tokens=3, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    assert len(eng.get_output(rid).token_ids) == 3
+
+
+def test_local_provider_serves_from_checkpoint(tmp_path):
+    """LOCAL_WEIGHTS_PATH + LOCAL_MODEL=hf: the provider derives the
+    architecture from config.json and serves real-checkpoint tokens."""
+    import asyncio
+
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.local import LocalProvider
+
+    _tiny_llama_checkpoint(tmp_path)
+    lc = LocalConfig(enabled=True, model="hf", max_num_seqs=4,
+                     max_model_len=128)
+    lc.num_gpu_blocks = 64
+    lc.weights_path = str(tmp_path)
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        async def go():
+            return await provider.chat_completions({
+                "model": "local/hf",
+                "messages": [{"role": "user", "content": "hello"}],
+                "max_tokens": 4, "temperature": 0})
+
+        body = asyncio.run(go())
+        assert body["usage"]["completion_tokens"] == 4
+    finally:
+        provider.engine.shutdown()
