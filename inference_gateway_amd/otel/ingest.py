@@ -1,13 +1,13 @@
 """OTLP metrics push ingestion (reference POST /v1/metrics,
 api/metrics.go:25-99 + otel/ingest.go:22-218).
 
-Accepts OTLP/HTTP JSON (gzip-aware, 4 MiB cap). Semantics carried over:
-attribute allowlist, delta-temporality-only for sums/histograms, histogram
-replay via bucket midpoints capped at 10k observations per datapoint, and
-source spoof protection (the `source` attribute is overwritten from the
-authenticated caller when auth is enabled). OTLP protobuf payloads are
-answered with 415 (JSON is the supported encoding in this build; the
-reference accepts both).
+Accepts OTLP/HTTP in BOTH encodings — application/x-protobuf (the OTel
+SDK default; hand-rolled wire decoder in otlp_pb.py) and
+application/json — gzip-aware, 4 MiB cap. Semantics carried over:
+attribute allowlist, delta-temporality-only for sums/histograms,
+histogram replay via bucket midpoints capped at 10k observations per
+datapoint, and source spoof protection (the `source` attribute is
+overwritten from the authenticated caller when auth is enabled).
 """
 from __future__ import annotations
 
@@ -53,13 +53,25 @@ class MetricsIngestion:
             if len(body) > MAX_BODY:
                 return Response.error(413, "payload too large")
         ct = req.headers.get("content-type", "")
-        if "json" not in ct:
-            return Response.error(415, "only application/json OTLP is "
-                                  "supported")
-        try:
-            payload = json.loads(body)
-        except Exception:
-            return Response.error(400, "invalid JSON")
+        if "protobuf" in ct:
+            # OTLP/HTTP protobuf — the OTel SDK default exporter
+            # encoding (hand-rolled wire decoder, otlp_pb.py; decodes
+            # into the same dict shape the JSON path parses)
+            from .otlp_pb import ProtoError, decode_export_metrics
+
+            try:
+                payload = decode_export_metrics(body)
+            except ProtoError as e:
+                return Response.error(400, f"invalid protobuf: {e}")
+        elif "json" in ct:
+            try:
+                payload = json.loads(body)
+            except Exception:
+                return Response.error(400, "invalid JSON")
+        else:
+            return Response.error(
+                415, "unsupported content type: use application/json "
+                     "or application/x-protobuf OTLP")
 
         # source spoof protection (otel/ingest.go): authenticated callers
         # cannot claim another source

@@ -258,3 +258,83 @@ async def test_otlp_ingestion_caps_and_content_guards():
                                   "content-encoding": "gzip"})
         assert r.status_code == 200
         assert r.json()["rejected"] == 1 and r.json()["accepted"] == 0
+
+
+def _pb_varint(v: int) -> bytes:
+    out = b""
+    while True:
+        b = v & 0x7F
+        v >>= 7
+        if v:
+            out += bytes([b | 0x80])
+        else:
+            return out + bytes([b])
+
+
+def _pb_field(fno: int, wt: int, payload: bytes) -> bytes:
+    tag = _pb_varint((fno << 3) | wt)
+    if wt == 2:
+        return tag + _pb_varint(len(payload)) + payload
+    return tag + payload
+
+
+def _pb_str(fno: int, s: str) -> bytes:
+    return _pb_field(fno, 2, s.encode())
+
+
+def _build_otlp_protobuf(name="gen_ai.client.operation.duration",
+                         temporality=2) -> bytes:
+    """Hand-encode an ExportMetricsServiceRequest the way the OTel SDK
+    default exporter would (proto3 wire format)."""
+    import struct
+
+    kv_provider = _pb_str(1, "provider") + _pb_field(
+        2, 2, _pb_str(1, "groq"))
+    kv_model = _pb_str(1, "model") + _pb_field(2, 2, _pb_str(1, "m"))
+    counts = b"".join(_pb_varint(c) for c in (1, 2, 0))       # packed
+    bounds = struct.pack("<2d", 0.1, 1.0)                     # packed
+    dp = (_pb_field(9, 2, kv_provider) + _pb_field(9, 2, kv_model)
+          + _pb_field(6, 2, counts) + _pb_field(7, 2, bounds))
+    hist = (_pb_field(1, 2, dp)
+            + _pb_field(2, 0, _pb_varint(temporality)))
+    metric = _pb_str(1, name) + _pb_field(9, 2, hist)
+    sm = _pb_field(2, 2, metric)
+    rm = _pb_field(2, 2, sm)
+    return _pb_field(1, 2, rm)
+
+
+@pytest.mark.asyncio
+async def test_otlp_protobuf_ingestion():
+    """The OTel SDK DEFAULT exporter encoding (protobuf) must land in
+    the histograms exactly like the JSON path (reference accepts both,
+    api/metrics.go:25-99)."""
+    gw, _ = make_telemetry_gateway()
+    async with gw_client(gw) as c:
+        r = await c.post(
+            "/v1/metrics", content=_build_otlp_protobuf(),
+            headers={"content-type": "application/x-protobuf"})
+        assert r.status_code == 200, r.text
+        assert r.json()["accepted"] == 1
+        # cumulative temporality rejected, same as JSON
+        r = await c.post(
+            "/v1/metrics", content=_build_otlp_protobuf(temporality=1),
+            headers={"content-type": "application/x-protobuf"})
+        assert r.json()["accepted"] == 0 and r.json()["rejected"] >= 1
+        # gzip + protobuf
+        import gzip as _gzip
+
+        r = await c.post(
+            "/v1/metrics",
+            content=_gzip.compress(_build_otlp_protobuf()),
+            headers={"content-type": "application/x-protobuf",
+                     "content-encoding": "gzip"})
+        assert r.json()["accepted"] == 1
+        # malformed protobuf is a 400, not a 500
+        r = await c.post(
+            "/v1/metrics", content=b"\xff\xff\xff\xff",
+            headers={"content-type": "application/x-protobuf"})
+        assert r.status_code == 400
+    text = gw.telemetry.exposition().decode()
+    assert "gen_ai_client_operation_duration_seconds_count" in text
+    # 1 obs in bucket 0 (mid 0.05) + 2 in bucket 1 (mid 0.55), twice
+    assert 'provider="groq"' in text
