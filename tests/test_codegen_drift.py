@@ -99,3 +99,28 @@ def test_example_env_vars_are_real():
                 continue
             missing.append(f"{f.relative_to(root)}: {var}")
     assert not missing, f"undocumented env vars in examples: {missing}"
+
+
+def test_openapi_document_shape():
+    """openapi.yaml is the machine-readable gateway surface (reference
+    openapi.yaml analog): well-formed, carries the chat schema, the
+    x-config env contract and all 15 providers."""
+    import yaml
+
+    from inference_gateway_amd.providers.constants_gen import PROVIDER_TABLE
+    from inference_gateway_amd.providers.local import LocalProvider
+
+    root = Path(__file__).resolve().parent.parent
+    doc = yaml.safe_load((root / "openapi.yaml").read_text())
+    assert doc["openapi"].startswith("3.")
+    assert "/v1/chat/completions" in doc["paths"]
+    schemas = doc["components"]["schemas"]
+    req = schemas["CreateChatCompletionRequest"]
+    assert req["required"] == ["model", "messages"]
+    # the documented n cap matches the engine's fork-group cap
+    assert req["properties"]["n"]["maximum"] == LocalProvider.MAX_N
+    # every registered provider is in the x-provider-configs extension
+    for row in PROVIDER_TABLE:
+        assert row[0] in doc["x-provider-configs"], row[0]
+    # env contract carried as x-config
+    assert "LOCAL_WEIGHTS_PATH" in str(doc["x-config"])
