@@ -151,6 +151,7 @@ class LocalConfig:
     prefix_caching: bool = False
     chunked_prefill: bool = False
     prefill_first: bool = True        # strict admission priority
+    mixed_batches: bool = False       # same-step prefill+decode rounds
     weights_path: str = ""            # HF safetensors dir; "" = synthetic
     tokenizer_path: str = ""          # tokenizers json; "" = byte fallback
 
@@ -279,6 +280,7 @@ def load(env: Optional[Mapping[str, str]] = None) -> Config:
     lo.prefix_caching = _bool(g("LOCAL_PREFIX_CACHING", "false"))
     lo.chunked_prefill = _bool(g("LOCAL_CHUNKED_PREFILL", "false"))
     lo.prefill_first = _bool(g("LOCAL_PREFILL_FIRST", "true"))
+    lo.mixed_batches = _bool(g("LOCAL_MIXED_BATCHES", "false"))
     lo.weights_path = g("LOCAL_WEIGHTS_PATH", "")
     lo.tokenizer_path = g("LOCAL_TOKENIZER_PATH", "")
 

@@ -152,6 +152,11 @@ class EngineConfig:
     # (vLLM-style, TTFT-optimal under bursts) instead of alternating
     # prefill/decode rounds (ITL-optimal).  See scheduler.py.
     scheduler_prefill_first: bool = False
+    # same-step mixed batches: a round with pending prompts carries the
+    # (chunked) prefill segment AND every running decode row in one
+    # forward — no whole-round ITL bubble (scheduler.py; the mixed
+    # round runs eagerly, pure decode keeps the hipGraph)
+    enable_mixed_batches: bool = False
     # real checkpoint: HF-format directory (config.json +
     # *.safetensors); None keeps the seed-derived synthetic init
     # (engine/weights.py — the no-network default)

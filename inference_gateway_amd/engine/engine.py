@@ -144,7 +144,8 @@ class LLMEngine:
             self.block_manager, cfg.max_num_seqs, cfg.max_batched_tokens,
             cfg.max_model_len,
             chunked_prefill=cfg.enable_chunked_prefill,
-            prefill_first=cfg.scheduler_prefill_first)
+            prefill_first=cfg.scheduler_prefill_first,
+            mixed_batches=cfg.enable_mixed_batches)
         self.sampler = Sampler(self.device, cfg.seed)
         self.kv_caches = self._allocate_kv(num_blocks)
         self._seq_counter = itertools.count()
@@ -302,15 +303,28 @@ class LLMEngine:
         prev = self._drain_pending()
         if batch.is_prefill:
             self.num_prompt_tokens += sum(batch.prefill_takes)
+        dec_seqs: List[Sequence] = []
+        if batch.is_mixed:
+            # the drain above may have finished/preempted decode seqs
+            # scheduled into this mixed round — drop them now
+            dec_seqs = [s for s in batch.decode_seqs
+                        if s.state == SeqState.RUNNING]
+            batch.decode_seqs = dec_seqs
+            if self.graph_runner is not None:
+                # keep the graph's block-table rows in sync for the
+                # next pure-decode round
+                self.graph_runner.write_block_appends(batch.new_blocks)
         fb = self._build_forward_batch(batch)
         hidden = self.model(fb, self.kv_caches)
         if batch.is_prefill:
             for seq, take in zip(batch.seqs, batch.prefill_takes):
                 seq.prefilled += take
             # chunked prefill: only sequences whose prompt COMPLETED this
-            # round sample a token; a partial chunk just extends KV
+            # round sample a token; a partial chunk just extends KV.
+            # Mixed rounds additionally sample every decode row (their
+            # logit rows follow the prefill rows — _build_forward_batch)
             sample_seqs = [s for s in batch.seqs
-                           if s.prefilled >= s.prompt_len]
+                           if s.prefilled >= s.prompt_len] + dec_seqs
         else:
             sample_seqs = batch.seqs
         if not sample_seqs:
@@ -387,6 +401,32 @@ class LLMEngine:
                 if p1 >= L:               # completes the prompt: sample
                     last.append(cu[-1] - 1)
                 prefixes.append(p0)
+            P = cu[-1]
+            dseqs = batch.decode_seqs if batch.is_mixed else None
+            dec_bt = dec_ctx = None
+            if dseqs:
+                # mixed round: append one decode row per running seq
+                # after the prefill rows; GEMMs run over the union,
+                # attention splits at row P (models/llama.py)
+                for s in dseqs:
+                    token_ids.append(s.output_token_ids[-1]
+                                     if s.output_token_ids
+                                     else s.prompt_token_ids[-1])
+                dpos, dslots = [], []
+                maxb = max(len(s.block_ids) for s in dseqs)
+                dec_bt = torch.zeros((len(dseqs), maxb), dtype=torch.int32)
+                dec_ctx = torch.tensor([s.num_tokens for s in dseqs],
+                                       dtype=torch.int32)
+                for i, s in enumerate(dseqs):
+                    pos = s.num_tokens - 1
+                    dpos.append(pos)
+                    blk = s.block_ids[pos // bs]
+                    dslots.append(blk * bs + pos % bs)
+                    dec_bt[i, : len(s.block_ids)] = torch.tensor(
+                        s.block_ids, dtype=torch.int32)
+                pos_parts.append(torch.tensor(dpos, dtype=torch.int32))
+                slot_parts.append(torch.tensor(dslots, dtype=torch.int32))
+                last.extend(range(P, P + len(dseqs)))
             fb = ForwardBatch(
                 token_ids=torch.tensor(token_ids, dtype=torch.int64, device=dev),
                 positions=torch.cat(pos_parts).to(dev),
@@ -397,6 +437,10 @@ class LLMEngine:
                 last_token_indices=torch.tensor(last, dtype=torch.int64,
                                                 device=dev),
             )
+            if dseqs:
+                fb.num_prefill_tokens = P
+                fb.decode_block_table = dec_bt.to(dev)
+                fb.decode_ctx_lens = dec_ctx.to(dev)
             if any(prefixes):
                 maxb = max(len(s.block_ids) for s in batch.seqs)
                 bt = torch.zeros((len(batch.seqs), maxb), dtype=torch.int32)
@@ -478,6 +522,11 @@ class LLMEngine:
             if tail:
                 self._copy_block(parent.block_ids[full], tail[0])
             child.block_ids = (self.block_manager.share(parent.block_ids[:full]) + tail)
+            # the child's prompt KV EXISTS (shared/copied from the
+            # parent): mark it prefilled, or Scheduler.finish would
+            # un-index the shared blocks as unwritten (the mid-chunk
+            # abort guard) and kill prefix-cache reuse
+            child.prefilled = plen
             self.scheduler.adopt(child)
             self.num_forks += 1
             if gr is not None:

@@ -24,6 +24,19 @@ class ForwardBatch:
     ctx_lens: Optional[torch.Tensor] = None     # [B] int32
     # indices of each sequence's last token within token_ids
     last_token_indices: Optional[torch.Tensor] = None  # [B] int64
+    # MIXED batch (same-step prefill chunk + decode rows): token rows
+    # [0, num_prefill_tokens) are the prefill chunk (cu_seqlens/
+    # seq_lens/prefix_lens/block_table describe them); rows
+    # [num_prefill_tokens, T) are one decode token per running sequence
+    # (decode_block_table/decode_ctx_lens describe them).  GEMMs and
+    # elementwise ops run over the union; attention splits by row kind.
+    num_prefill_tokens: int = 0
+    decode_block_table: Optional[torch.Tensor] = None
+    decode_ctx_lens: Optional[torch.Tensor] = None
+
+    @property
+    def is_mixed(self) -> bool:
+        return self.decode_ctx_lens is not None
 
     @property
     def num_tokens(self) -> int:

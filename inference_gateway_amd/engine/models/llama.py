@@ -80,18 +80,28 @@ class LlamaAttention(nn.Module):
         v3 = v.view(T, self.nkv, self.hd)
         ops.append_kv_cache(k3, v3, kc, vc, fb.slot_mapping)
         if fb.is_prefill:
+            P = fb.num_prefill_tokens if fb.is_mixed else T
+            qv = q.view(T, self.nq, self.hd)
             if fb.prefix_lens is not None:
                 # prefix-cache hit in the batch: q rows are suffixes;
                 # cached prefix + the just-appended suffix K/V are read
                 # back from the paged cache (ops/csrc/flash_prefill.hip,
                 # flash_prefill_paged_kernel)
                 o = ops.flash_prefill_paged(
-                    q.view(T, self.nq, self.hd), kc, vc, fb.block_table,
+                    qv[:P], kc, vc, fb.block_table,
                     fb.prefix_lens, fb.cu_seqlens, fb.seq_lens, self.scale)
             else:
                 o = ops.flash_prefill(
-                    q.view(T, self.nq, self.hd), k3, v3, fb.cu_seqlens,
+                    qv[:P], k3[:P], v3[:P], fb.cu_seqlens,
                     fb.seq_lens, self.scale)
+            if fb.is_mixed:
+                # mixed round: rows [P, T) are one decode token per
+                # running sequence (engine._build_forward_batch)
+                od = ops.paged_attn_decode(
+                    qv[P:].contiguous(), kc, vc, fb.decode_block_table,
+                    fb.decode_ctx_lens, self.scale)
+                o = torch.cat([o.view(P, self.nq, self.hd),
+                               od.view(T - P, self.nq, self.hd)], 0)
         else:
             o = ops.paged_attn_decode(
                 q.view(T, self.nq, self.hd), kc, vc, fb.block_table,

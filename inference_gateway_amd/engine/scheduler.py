@@ -80,13 +80,22 @@ class ScheduledBatch:
     # prefill: the last entry may be a partial chunk; None = whole
     # remaining prompt for every seq)
     prefill_takes: Optional[List[int]] = None
+    # MIXED round: `seqs`/`prefill_takes` are the prefill chunk and
+    # decode_seqs each contribute one decode token to the SAME forward
+    # (engine runs it eagerly; pure-decode rounds keep the hipGraph)
+    decode_seqs: Optional[List[Sequence]] = None
+
+    @property
+    def is_mixed(self) -> bool:
+        return bool(self.decode_seqs)
 
 
 class Scheduler:
     def __init__(self, block_manager: BlockManager, max_num_seqs: int,
                  max_batched_tokens: int, max_model_len: int,
                  chunked_prefill: bool = False,
-                 prefill_first: bool = False):
+                 prefill_first: bool = False,
+                 mixed_batches: bool = False):
         self.bm = block_manager
         self.max_num_seqs = max_num_seqs
         self.max_batched_tokens = max_batched_tokens
@@ -113,6 +122,14 @@ class Scheduler:
         # spikes during admission in exchange — chunked prefill bounds
         # that spike.
         self.prefill_first = prefill_first
+        # mixed_batches: when prompts and running decodes are BOTH
+        # pending, emit ONE batch carrying a (chunked) prefill segment
+        # plus every running sequence's decode token — no whole-round
+        # ITL bubble from a prefill burst, and the union raises the
+        # skinny decode GEMMs' M for free.  The engine runs mixed
+        # rounds eagerly; pure-decode rounds keep the hipGraph.
+        self.mixed_batches = mixed_batches
+        self.num_mixed_rounds = 0     # cumulative (metrics)
         self.num_preemptions = 0      # cumulative (metrics)
 
     def add(self, seq: Sequence) -> None:
@@ -130,10 +147,17 @@ class Scheduler:
     def schedule(self) -> Optional[ScheduledBatch]:
         # Alternate prefill and decode rounds when both are pending;
         # admit as many waiting prompts as fit the token budget and the
-        # block pool.
+        # block pool.  With mixed_batches the round carries BOTH.
+        mixed = (self.mixed_batches and bool(self.waiting)
+                 and bool(self.running))
         want_prefill = bool(self.waiting) and (
-            self.prefill_first
+            mixed or self.prefill_first
             or not (self._last_was_prefill and self.running))
+        # a mixed round's token budget is shared with the decode rows
+        budget = self.max_batched_tokens - (
+            len(self.running) if mixed else 0)
+        budget = max(budget, 1)
+        chunked = self.chunked_prefill or mixed
         prefills: List[Sequence] = []
         takes: List[int] = []
         tokens = 0
@@ -141,12 +165,12 @@ class Scheduler:
                and len(self.running) + len(prefills) < self.max_num_seqs):
             seq = self.waiting[0]
             remaining = seq.prompt_len - seq.prefilled
-            if (tokens + remaining > self.max_batched_tokens
-                    and self.chunked_prefill):
+            if (tokens + remaining > budget
+                    and chunked):
                 # CHUNKED: hard token budget — take a partial chunk of
                 # the front prompt and leave it at waiting[0]; its whole
                 # prompt's blocks were reserved at its first chunk
-                take = self.max_batched_tokens - tokens
+                take = budget - tokens
                 if take <= 0:
                     break
                 if seq.prefilled == 0:
@@ -170,7 +194,7 @@ class Scheduler:
                 takes.append(take)
                 tokens += take
                 break
-            if tokens + remaining > self.max_batched_tokens and prefills:
+            if tokens + remaining > budget and prefills:
                 break
             if seq.prefilled == 0:
                 if not self.bm.can_allocate(seq.prompt_len):
@@ -190,6 +214,27 @@ class Scheduler:
             tokens += seq.prompt_len - seq.prefilled
         if prefills:
             admitted = [q for q in prefills if q.state == SeqState.RUNNING]
+            decode_seqs = None
+            if mixed:
+                # piggyback every ALREADY-running sequence's decode
+                # token onto this round (the just-admitted completers
+                # sample via the prefill side); fall back to a plain
+                # prefill round if the pool cannot grow every decode row
+                # self.running holds only PRIOR seqs here (admitted are
+                # appended below)
+                prior = list(self.running)
+                nb = self._grow_decode_capacity(prior)
+                if nb is not None and prior:
+                    decode_seqs = prior
+                    self.num_mixed_rounds += 1
+                    if admitted:
+                        self.running.extend(admitted)
+                        self.membership_version += 1
+                    self._last_was_prefill = True
+                    return ScheduledBatch(is_prefill=True, seqs=prefills,
+                                          prefill_takes=takes,
+                                          decode_seqs=decode_seqs,
+                                          new_blocks=nb)
             if admitted:
                 self.running.extend(admitted)
                 self.membership_version += 1
@@ -245,6 +290,24 @@ class Scheduler:
                 return None
         return ScheduledBatch(is_prefill=False, seqs=list(self.running),
                               new_blocks=new_blocks)
+
+    def _grow_decode_capacity(self, seqs) -> Optional[List[tuple]]:
+        """Best-effort block growth for a mixed round's decode rows.
+        Returns the (slot, col, block) appends, or None when the pool
+        cannot cover every row (the caller then runs a plain prefill
+        round; LIFO preemption stays a pure-decode-round concern)."""
+        bs = self.bm.block_size
+        new_blocks: List[tuple] = []
+        for seq in seqs:
+            need = seq.num_tokens + (1 if seq.in_flight else 0)
+            if need > len(seq.block_ids) * bs:
+                try:
+                    self.bm.ensure_capacity(seq.block_ids, need)
+                except RuntimeError:
+                    return None
+                new_blocks.append(
+                    (seq.slot, len(seq.block_ids) - 1, seq.block_ids[-1]))
+        return new_blocks
 
     # ---- forked sequences (OpenAI n>1: one prefill, shared prompt KV) --
     def can_adopt(self) -> bool:

@@ -249,6 +249,8 @@ class LocalProvider:
                                            False),
             scheduler_prefill_first=getattr(local_cfg, "prefill_first",
                                             True),
+            enable_mixed_batches=getattr(local_cfg, "mixed_batches",
+                                         False),
             weights_path=weights,
             capture_at_init=True,
         )

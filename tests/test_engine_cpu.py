@@ -1072,3 +1072,81 @@ async def test_async_engine_batched_delivery_mode(monkeypatch):
         assert len(toks) == 6
     finally:
         eng.shutdown()
+
+
+def test_mixed_batches_token_equivalence():
+    """Same-step mixed prefill+decode rounds must be token-exact vs the
+    plain engine: greedy decode of staggered arrivals produces identical
+    outputs with enable_mixed_batches on and off."""
+    def run(mixed):
+        eng = _engine(max_batched_tokens=48, num_gpu_blocks=128,
+                      max_num_seqs=8, enable_chunked_prefill=True,
+                      enable_mixed_batches=mixed)
+        sp = lambda: SamplingParams(max_new_tokens=10, temperature=0.0)  # noqa
+        ra = eng.add_request(list(range(1, 40)), sp())
+        for _ in range(3):
+            eng.step()                  # A decoding (or prefilling)
+        rb = eng.add_request(list(range(50, 170)), sp())   # long prompt
+        rc = eng.add_request(list(range(7, 30)), sp())
+        while not eng.is_finished():
+            eng.step()
+        return ([eng.get_output(r).token_ids for r in (ra, rb, rc)],
+                eng.scheduler.num_mixed_rounds)
+
+    plain, m0 = run(False)
+    mixed, m1 = run(True)
+    assert m0 == 0 and m1 > 0, (m0, m1)
+    assert mixed == plain
+
+
+def test_mixed_batches_no_itl_bubble():
+    """While a long prompt prefills in chunks, running sequences keep
+    generating EVERY round (no whole-round decode bubble)."""
+    eng = _engine(max_batched_tokens=32, num_gpu_blocks=128,
+                  max_num_seqs=8, enable_chunked_prefill=True,
+                  enable_mixed_batches=True)
+    ra = eng.add_request(list(range(1, 20)),
+                         SamplingParams(max_new_tokens=64,
+                                        temperature=0.0, ignore_eos=True))
+    eng.step()                           # prefill A
+    eng.step()                           # A decodes once
+    na = len(eng.get_output(ra).token_ids)
+    eng.add_request(list(range(50, 170)),
+                    SamplingParams(max_new_tokens=4, temperature=0.0))
+    # the long prompt needs ~4 chunked rounds at budget 32; A must gain
+    # one token in EVERY one of them
+    for i in range(4):
+        before = len(eng.get_output(ra).token_ids)
+        eng.step()
+        after = len(eng.get_output(ra).token_ids)
+        assert after == before + 1, (i, before, after)
+    assert eng.scheduler.num_mixed_rounds >= 3
+    assert len(eng.get_output(ra).token_ids) >= na + 4
+
+
+def test_mixed_batches_with_prefix_cache_and_forks():
+    """Mixed rounds compose with the other scheduler features."""
+    long_prompt = list(range(11, 150))
+    sp = lambda: SamplingParams(max_new_tokens=6, temperature=0.0)  # noqa
+    eng0 = _engine()
+    r0 = eng0.add_request(long_prompt, sp())
+    while not eng0.is_finished():
+        eng0.step()
+    want = eng0.get_output(r0).token_ids
+
+    eng = _engine(max_batched_tokens=48, num_gpu_blocks=128,
+                  max_num_seqs=8, enable_chunked_prefill=True,
+                  enable_prefix_caching=True, enable_mixed_batches=True)
+    ra = eng.add_request(list(range(200, 230)), sp())
+    eng.step()
+    rids = eng.add_request_group(long_prompt, sp(), n=2)
+    while not eng.is_finished():
+        eng.step()
+    for r in rids:
+        assert eng.get_output(r).token_ids == want
+    # repeat the prompt: prefix hits while other seqs decode mixed
+    r2 = eng.add_request(long_prompt, sp())
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(r2).token_ids == want
+    assert eng.block_manager.prefix_hit_tokens > 0
