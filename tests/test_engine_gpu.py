@@ -289,3 +289,33 @@ def test_prefix_caching_on_gpu_matches_uncached():
     assert eng.get_output(r3).token_ids == want[1]
     assert eng.block_manager.prefix_hit_tokens >= 48
 
+
+
+def test_zz_mixed_batches_gpu_token_equivalence():
+    """Same-step mixed prefill+decode rounds on the HIP kernel path
+    (paged decode kernel for the decode rows + flash/paged-extend for
+    the chunk, sharing one forward) must be token-exact vs the plain
+    engine with graphs on."""
+    def run(mixed):
+        torch.manual_seed(0)
+        cfg = EngineConfig.tiny_llama(
+            use_hip_graphs=True, enable_chunked_prefill=True,
+            enable_mixed_batches=mixed, max_batched_tokens=48,
+            num_gpu_blocks=256, max_num_seqs=8)
+        eng = LLMEngine(cfg, device="cuda:0")
+        sp = lambda: SamplingParams(max_new_tokens=10,  # noqa
+                                    temperature=0.0)
+        ra = eng.add_request(list(range(1, 40)), sp())
+        for _ in range(3):
+            eng.step()
+        rb = eng.add_request(list(range(50, 170)), sp())
+        rc = eng.add_request(list(range(7, 30)), sp())
+        while not eng.is_finished():
+            eng.step()
+        return ([eng.get_output(r).token_ids for r in (ra, rb, rc)],
+                eng.scheduler.num_mixed_rounds)
+
+    plain, m0 = run(False)
+    mixed, m1 = run(True)
+    assert m0 == 0 and m1 > 0
+    assert mixed == plain

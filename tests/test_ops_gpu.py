@@ -225,13 +225,21 @@ def test_moe_grouped_gemm_matches_ref():
     w = torch.randn(E, N, K, dtype=torch.bfloat16, device="cuda") * 0.05
     counts = torch.tensor([128, 7, 0, 65], dtype=torch.int32, device="cuda")
     out = ops.moe_grouped_gemm(a, w, counts, C)
-    ref = torch_ref.moe_grouped_gemm_ref(a.cpu(), w.cpu(), counts.cpu(), C)
+    # fp32 reference of the same op; RELATIVE error bound against the
+    # fp32 GEMM scale (a flat atol=0.5 is wide
+    # enough to hide a systematic accumulation bug at bf16 scale)
+    ref = torch_ref.moe_grouped_gemm_ref(
+        a.float().cpu(), w.float().cpu(), counts.cpu(), C)
     for e in range(E):
         c = int(counts[e])
         got = out[e * C:e * C + c].float().cpu()
         want = ref[e * C:e * C + c].float()
-        assert torch.allclose(got, want, atol=0.5, rtol=0.05), \
-            f"expert {e} mismatch: max {(got - want).abs().max()}"
+        scale = want.abs().max().clamp_min(1.0)
+        rel = (got - want).abs().max() / scale
+        # bf16 inputs, fp32 accumulate over K=256: relative error vs the
+        # fp32 reference is bounded by input rounding ~2^-8 * sqrt-ish
+        # growth; 2e-2 relative catches any systematic 1-ulp bias
+        assert rel < 2e-2, f"expert {e}: rel err {rel:.4f}"
 
 
 def test_silu_mul_grouped_matches_ref():
