@@ -202,3 +202,29 @@ def silu_mul_grouped(gate_up: torch.Tensor, counts: torch.Tensor,
         _kernels.silu_mul_grouped(out, gate_up, counts, capacity)
         return out
     return torch_ref.silu_mul_grouped_ref(gate_up, counts, capacity)
+
+
+def fused_gemm_tn(x: torch.Tensor, w: torch.Tensor, swiglu: bool = False,
+                  nt_weights: bool = False) -> torch.Tensor:
+    """TN bf16 GEMM out = x @ w.T with an optional fused SwiGLU
+    epilogue (w = [gate; up] rows, out = silu(g)*u).  Decode
+    weight-streaming shapes (M%128==0, N%64==0, K%32==0); callers fall
+    back to the library GEMM otherwise (ops/csrc/fused_gemm.hip)."""
+    if x.is_cuda:
+        _require_kernels()
+        N = w.shape[0] // 2 if swiglu else w.shape[0]
+        out = torch.empty((x.shape[0], N), dtype=x.dtype, device=x.device)
+        _kernels.fused_gemm_tn(out, x, w, swiglu, nt_weights)
+        return out
+    y = x.float() @ w.float().t()
+    if swiglu:
+        N = w.shape[0] // 2
+        g, u = y[:, :N], y[:, N:]
+        y = torch.nn.functional.silu(g) * u
+    return y.to(x.dtype)
+
+
+def fused_gemm_tn_supported(M: int, N_rows: int, K: int,
+                            swiglu: bool) -> bool:
+    N = N_rows // 2 if swiglu else N_rows
+    return M % 128 == 0 and N % 64 == 0 and K % 32 == 0

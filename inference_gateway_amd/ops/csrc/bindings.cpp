@@ -40,6 +40,8 @@ void moe_grouped_gemm(torch::Tensor out, torch::Tensor a, torch::Tensor w,
                       torch::Tensor counts, int64_t capacity);
 void silu_mul_grouped(torch::Tensor out, torch::Tensor gate_up,
                       torch::Tensor counts, int64_t capacity);
+void fused_gemm_tn(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                   bool swiglu, bool nt_weights);
 int64_t ar_alloc(int64_t bytes, bool finegrained);
 void ar_free(int64_t ptr);
 py::bytes ar_get_handle(int64_t ptr);
@@ -78,6 +80,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped per-expert MFMA GEMM over the padded dispatch buffer");
   m.def("silu_mul_grouped", &igw::silu_mul_grouped,
         "SwiGLU over only the real (non-padded) MoE dispatch rows");
+  m.def("fused_gemm_tn", &igw::fused_gemm_tn,
+        "TN bf16 MFMA GEMM with optional fused SwiGLU epilogue "
+        "(decode weight-streaming shapes, gfx950)");
   m.def("ar_alloc", &igw::ar_alloc,
         "raw device allocation for one-shot all-reduce buffers "
         "(finegrained=true for signal pages)");

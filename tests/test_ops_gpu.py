@@ -423,3 +423,38 @@ def test_oneshot_allreduce_world8_subprocess():
         cwd=root, capture_output=True, text=True, timeout=600,
         env=dict(os.environ, GPU_MAX_HW_QUEUES="8"))
     assert out.returncode == 0, out.stdout + out.stderr
+
+
+def test_fused_gemm_tn_matches_fp32_ref():
+    """Fused TN GEMM (plain + SwiGLU epilogue) vs the fp32 reference —
+    relative bound, asymmetric operands (a symmetric B would pass a
+    transposed C-write, guide §3)."""
+    torch.manual_seed(41)
+    M, N, K = 256, 128, 256
+    x = (torch.randn(M, K, device="cuda") * 0.5).to(torch.bfloat16)
+    w = (torch.randn(2 * N, K, device="cuda") * 0.2).to(torch.bfloat16)
+    w += torch.arange(2 * N, device="cuda").view(-1, 1).to(
+        torch.bfloat16) * 1e-3           # asymmetry across rows
+    # plain mode
+    got = ops.fused_gemm_tn(x, w[:N], swiglu=False).float()
+    want = x.float() @ w[:N].float().t()
+    rel = (got - want).abs().max() / want.abs().max().clamp_min(1.0)
+    assert rel < 2e-2, f"plain rel err {rel:.4f}"
+    # swiglu mode
+    got = ops.fused_gemm_tn(x, w, swiglu=True).float()
+    g = x.float() @ w[:N].float().t()
+    u = x.float() @ w[N:].float().t()
+    want = torch.nn.functional.silu(g) * u
+    rel = (got - want).abs().max() / want.abs().max().clamp_min(1.0)
+    assert rel < 2e-2, f"swiglu rel err {rel:.4f}"
+    # nt variant bitwise-matches the cached variant
+    got_nt = ops.fused_gemm_tn(x, w, swiglu=True, nt_weights=True)
+    assert torch.equal(got_nt.float(), got)
+    # flagship decode shapes (down-proj and gate_up) at batch 256
+    x2 = (torch.randn(256, 14336, device="cuda") * 0.1).to(torch.bfloat16)
+    wd = (torch.randn(4096, 14336, device="cuda") * 0.05).to(
+        torch.bfloat16)
+    got = ops.fused_gemm_tn(x2, wd).float()
+    want = x2.float() @ wd.float().t()
+    rel = (got - want).abs().max() / want.abs().max().clamp_min(1.0)
+    assert rel < 2e-2, f"down-proj rel err {rel:.4f}"
