@@ -230,6 +230,8 @@ def test_moe_grouped_gemm_matches_ref():
     # enough to hide a systematic accumulation bug at bf16 scale)
     for e in range(E):
         c = int(counts[e])
+        if c == 0:
+            continue      # expert with no routed rows (counts[2])
         got = out[e * C:e * C + c].float().cpu()
         want = (a[e * C:e * C + c].float().cpu()
                 @ w[e].float().cpu().T)
