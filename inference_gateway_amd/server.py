@@ -1,10 +1,11 @@
 """Process entrypoint (the analog of cmd/gateway/main.go:46-402).
 
 `python -m inference_gateway_amd` parses --version/--help, loads env
-config, builds the Gateway, runs the main HTTP server (uvicorn) plus the
-separate Prometheus metrics server on TELEMETRY_METRICS_PORT, warns when
-binding beyond loopback without auth (main.go:97-102), and shuts down
-gracefully on SIGINT/SIGTERM.
+config, builds the Gateway, runs the main HTTP server (the native
+asyncio server, gateway/hserver.py — direct transport writes are the
+SSE fast path) plus the separate Prometheus metrics server on
+TELEMETRY_METRICS_PORT, warns when binding beyond loopback without auth
+(main.go:97-102), and shuts down gracefully on SIGINT/SIGTERM.
 """
 from __future__ import annotations
 
