@@ -248,3 +248,88 @@ async def test_client_disconnect_aborts_engine_request(provider):
                 break
             await asyncio.sleep(0.1)
         assert not provider.engine.engine.scheduler.has_work
+
+
+@pytest.mark.asyncio
+async def test_tls_serving(provider, tmp_path):
+    """SERVER_TLS_*: the native server terminates TLS (the reference
+    serves HTTPS when TLS paths are set, main.go:334-350).  Self-signed
+    cert generated on the fly; client pins it as its CA."""
+    import ssl
+    import subprocess
+
+    cert = tmp_path / "cert.pem"
+    key = tmp_path / "key.pem"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+         "-keyout", str(key), "-out", str(cert), "-days", "1",
+         "-subj", "/CN=127.0.0.1",
+         "-addext", "subjectAltName=IP:127.0.0.1"],
+        check=True, capture_output=True)
+    from inference_gateway_amd.gateway.hserver import build_ssl_context
+
+    cfg = config_mod.load({})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    srv = HttpServer(gw, "127.0.0.1", 0,
+                     ssl_context=build_ssl_context(str(cert), str(key)))
+    await srv.start()
+    try:
+        ctx = ssl.create_default_context(cafile=str(cert))
+        async with httpx.AsyncClient(
+                base_url=f"https://127.0.0.1:{srv.port}",
+                verify=ctx, timeout=60) as c:
+            r = await c.get("/health")
+            assert r.status_code == 200
+            # SSE over TLS, direct path
+            events = 0
+            async with c.stream("POST", "/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "tls"}],
+                "max_tokens": 4, "temperature": 0, "stream": True,
+            }) as rs:
+                assert rs.status_code == 200
+                async for line in rs.aiter_lines():
+                    if line.startswith("data:"):
+                        events += 1
+            assert events >= 3          # role + content(+) + usage + DONE
+    finally:
+        await srv.stop()
+        await client.aclose()
+
+
+@pytest.mark.asyncio
+async def test_soak_64_streams_over_socket(provider):
+    """Reference soak bar over the REAL socket + direct-write path: 64
+    concurrent SSE streams, every stream completes with exact usage and
+    exactly one [DONE]."""
+    async def one(c, i):
+        dones = 0
+        usage = None
+        async with c.stream("POST", "/v1/chat/completions", json={
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": f"soak {i}"}],
+            "max_tokens": 6, "temperature": 0.9, "seed": i,
+            "stream": True,
+        }) as r:
+            assert r.status_code == 200
+            async for line in r.aiter_lines():
+                if not line.startswith("data:"):
+                    continue
+                d = line[5:].strip()
+                if d == "[DONE]":
+                    dones += 1
+                    continue
+                u = json.loads(d).get("usage")
+                usage = u or usage
+        return dones, usage
+
+    async with serve(provider) as base:
+        async with httpx.AsyncClient(base_url=base, timeout=300) as c:
+            results = await asyncio.gather(*(one(c, i)
+                                             for i in range(64)))
+    for dones, usage in results:
+        assert dones == 1
+        assert usage and usage["completion_tokens"] == 6
