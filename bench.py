@@ -356,6 +356,11 @@ def _barrier():
     import torch.distributed as dist
 
     if dist.is_initialized():
+        if torch.cuda.is_available():
+            # this runs via asyncio.to_thread: the executor thread's
+            # CUDA device defaults to 0, and an NCCL barrier from the
+            # wrong device wedges the whole SCALE job — bind it first
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
         dist.barrier()
 
 
