@@ -333,3 +333,48 @@ async def test_soak_64_streams_over_socket(provider):
     for dones, usage in results:
         assert dones == 1
         assert usage and usage["completion_tokens"] == 6
+
+
+@pytest.mark.asyncio
+async def test_engine_death_mid_stream_emits_error_frame():
+    """ADVICE r1 (medium), end to end over the socket: an engine-thread
+    crash mid-stream must surface as an in-stream error frame on the
+    direct-write path — never a clean finish_reason 'stop'."""
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=4,
+                     max_model_len=256)
+    lc.num_gpu_blocks = 64
+    p = LocalProvider(lc, device="cpu")
+    try:
+        async with serve(p) as base:
+            events = []
+
+            def boom():
+                raise RuntimeError("injected mid-stream fault")
+
+            async with httpx.AsyncClient(base_url=base, timeout=60) as c:
+                async with c.stream("POST", "/v1/chat/completions", json={
+                    "model": "local/tiny-llama",
+                    "messages": [{"role": "user", "content": "die"}],
+                    "max_tokens": 200, "temperature": 0.8,
+                    "stream": True,
+                }) as r:
+                    assert r.status_code == 200
+                    n = 0
+                    async for line in r.aiter_lines():
+                        if not line.startswith("data:"):
+                            continue
+                        events.append(line[5:].strip())
+                        n += 1
+                        if n == 3:
+                            p.engine.engine.step = boom
+            errs = [e for e in events if e != "[DONE]"
+                    and "error" in json.loads(e)]
+            assert errs, events[-3:]
+            assert "injected mid-stream fault" in errs[-1]
+            # and no clean finish: none of the frames said 'stop'
+            fins = [json.loads(e)["choices"][0].get("finish_reason")
+                    for e in events if e != "[DONE]"
+                    and "choices" in json.loads(e)]
+            assert "stop" not in fins
+    finally:
+        p.engine.shutdown()
