@@ -110,6 +110,11 @@ class Router:
                 return Response.error(400, "invalid JSON body")
         if not isinstance(body, dict) or not body.get("model"):
             return Response.error(400, "model is required")
+        from .validate import validate_chat_request
+
+        verr = validate_chat_request(body)
+        if verr is not None:
+            return Response.error(400, verr)
         try:
             pid, model = self._resolve(body["model"],
                                        req.query.get("provider"))

@@ -134,14 +134,15 @@ async def test_chat_completion_routing_and_auth_injection():
 async def test_chat_completion_unknown_provider_and_missing_key():
     gw, _ = make_gateway()
     async with gw_client(gw) as c:
+        msgs = [{"role": "user", "content": "x"}]
         r = await c.post("/v1/chat/completions",
-                         json={"model": "nosuch/m", "messages": []})
+                         json={"model": "nosuch/m", "messages": msgs})
         assert r.status_code == 400
         r = await c.post("/v1/chat/completions",
-                         json={"model": "openai/gpt-x", "messages": []})
+                         json={"model": "openai/gpt-x", "messages": msgs})
         assert r.status_code == 401  # openai has no key configured
         r = await c.post("/v1/chat/completions",
-                         json={"model": "bare-model", "messages": []})
+                         json={"model": "bare-model", "messages": msgs})
         assert r.status_code == 400  # no prefix, no ?provider=
 
 
@@ -150,7 +151,8 @@ async def test_model_deny_list():
     gw, _ = make_gateway(env={"DISALLOWED_MODELS": "m-one"})
     async with gw_client(gw) as c:
         r = await c.post("/v1/chat/completions",
-                         json={"model": "groq/m-one", "messages": []})
+                         json={"model": "groq/m-one", "messages": [
+                             {"role": "user", "content": "x"}]})
         assert r.status_code == 403
 
 
@@ -428,3 +430,46 @@ async def test_messages_passthrough_sse():
                     lines.append(line)
     assert any("message_stop" in l for l in lines)
     assert any("text_delta" in l for l in lines)
+
+
+@pytest.mark.asyncio
+async def test_chat_request_shape_validation():
+    """Typed wire validation (the reference's generated-types decode
+    400s): wrong-typed KNOWN fields are a 400; unknown fields pass."""
+    gw, _ = make_gateway()
+    bads = [
+        ({"model": "groq/m", "messages": "hi"},
+         "messages must be"),
+        ({"model": "groq/m", "messages": []}, "non-empty"),
+        ({"model": "groq/m", "messages": [{"role": "robot",
+                                           "content": "x"}]}, "role"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": 5}]}, "content"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": "x"}],
+          "temperature": "hot"}, "temperature must be a number"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": "x"}],
+          "max_tokens": 3.5}, "max_tokens must be an integer"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": "x"}],
+          "stream": "yes"}, "stream must be a boolean"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": "x"}],
+          "stop": [1, 2]}, "stop"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": "x"}],
+          "stop": ["a", "b", "c", "d", "e"]}, "at most 4"),
+    ]
+    async with gw_client(gw) as c:
+        for body, frag in bads:
+            r = await c.post("/v1/chat/completions", json=body)
+            assert r.status_code == 400, (body, r.status_code)
+            assert frag in str(r.json()["error"]), (body, r.json())
+        # unknown fields and multimodal parts still pass through
+        r = await c.post("/v1/chat/completions", json={
+            "model": "groq/m", "response_format": {"type": "json"},
+            "whatever": 1,
+            "messages": [{"role": "user", "content": [
+                {"type": "text", "text": "hi"}]}]})
+        assert r.status_code == 200
