@@ -378,3 +378,33 @@ async def test_engine_death_mid_stream_emits_error_frame():
             assert "stop" not in fins
     finally:
         p.engine.shutdown()
+
+
+@pytest.mark.asyncio
+async def test_direct_stream_logprobs_over_socket(provider):
+    """OpenAI logprobs ride the direct-write path: per-chunk
+    logprobs.content entries with top_logprobs."""
+    async with serve(provider) as base:
+        async with httpx.AsyncClient(base_url=base, timeout=60) as c:
+            entries = []
+            async with c.stream("POST", "/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "lp"}],
+                "max_tokens": 5, "temperature": 0.5, "seed": 3,
+                "logprobs": True, "top_logprobs": 3, "stream": True,
+            }) as r:
+                assert r.status_code == 200
+                async for line in r.aiter_lines():
+                    if not line.startswith("data:"):
+                        continue
+                    d = line[5:].strip()
+                    if d == "[DONE]":
+                        continue
+                    ch = json.loads(d)["choices"][0]
+                    lp = ch.get("logprobs")
+                    if lp:
+                        entries.extend(lp["content"])
+    assert len(entries) == 5
+    for e in entries:
+        assert isinstance(e["logprob"], float) and e["logprob"] <= 0
+        assert len(e["top_logprobs"]) == 3
