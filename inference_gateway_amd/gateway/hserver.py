@@ -51,8 +51,9 @@ class HttpProtocol(asyncio.Protocol):
 
     __slots__ = ("app", "loop", "transport", "_buf", "_head_done",
                  "_need_body", "_body", "_scope", "_task", "_paused",
-                 "_drain", "_keep_alive", "_chunked", "_started",
-                 "_pending_head", "_closing", "server")
+                 "_drain", "_keep_alive", "_chunked", "_chunked_req",
+                 "_started", "_pending_head", "_closing", "_idle_timer",
+                 "server")
 
     def __init__(self, app, loop, server: "HttpServer"):
         self.app = app
@@ -69,9 +70,11 @@ class HttpProtocol(asyncio.Protocol):
         self._drain: Optional[asyncio.Event] = None
         self._keep_alive = True
         self._chunked = False
+        self._chunked_req = False
         self._started = False
         self._pending_head = b""
         self._closing = False
+        self._idle_timer = None
 
     # ---- transport callbacks -------------------------------------------
     def connection_made(self, transport) -> None:
@@ -88,6 +91,9 @@ class HttpProtocol(asyncio.Protocol):
 
     def connection_lost(self, exc) -> None:
         self._closing = True
+        if self._idle_timer is not None:
+            self._idle_timer.cancel()
+            self._idle_timer = None
         if self._task is not None and not self._task.done():
             self._task.cancel()
         if self._paused and self._drain is not None:
@@ -112,7 +118,15 @@ class HttpProtocol(asyncio.Protocol):
                 if len(self._buf) > _MAX_HEAD:
                     self._simple_response(400, b"header too large")
                     self.transport.close()
+                elif self._idle_timer is None and self._buf:
+                    # header-read deadline: a connection that trickles
+                    # an incomplete request head is dropped (slowloris)
+                    self._idle_timer = self.loop.call_later(
+                        self.server.header_timeout, self._on_idle)
                 return
+            if self._idle_timer is not None:
+                self._idle_timer.cancel()
+                self._idle_timer = None
             try:
                 self._parse_head(self._buf[:end])
             except ValueError:
@@ -125,13 +139,66 @@ class HttpProtocol(asyncio.Protocol):
                 return
             self._buf = self._buf[end + 4:]
             self._head_done = True
+        if self._head_done and self._chunked_req:
+            # RFC 7230 chunked request body (curl -T / gRPC-style
+            # clients; the reference's Go server accepts these natively)
+            done = self._consume_chunked()
+            if done is None:
+                return                       # need more data
+            if done is False:
+                self._simple_response(400, b"malformed chunked body")
+                self.transport.close()
+                return
+            self._dispatch()
+            return
         if self._head_done and len(self._buf) >= self._need_body:
             self._body = self._buf[: self._need_body]
             self._buf = self._buf[self._need_body:]
-            self._head_done = False
-            scope, body = self._scope, self._body
-            self._scope = None
-            self._task = self.loop.create_task(self._run(scope, body))
+            self._dispatch()
+
+    def _dispatch(self) -> None:
+        self._head_done = False
+        scope, body = self._scope, self._body
+        self._scope = None
+        self._body = b""
+        self._task = self.loop.create_task(self._run(scope, body))
+
+    def _on_idle(self) -> None:
+        self._idle_timer = None
+        if not self._head_done and self._buf and not self._closing:
+            self._simple_response(408, b"request header timeout")
+            self.transport.close()
+
+    def _consume_chunked(self):
+        """Decode as much chunked body as buffered.  Returns True when
+        the terminal chunk arrived, False on malformed input, None when
+        more data is needed."""
+        while True:
+            i = self._buf.find(b"\r\n")
+            if i < 0:
+                return None if len(self._buf) <= 16 else False
+            size_tok = self._buf[:i].split(b";", 1)[0].strip()
+            try:
+                size = int(size_tok, 16)
+            except ValueError:
+                return False
+            if size == 0:
+                # trailer section ends with CRLF CRLF (or just CRLF
+                # when no trailers)
+                j = self._buf.find(b"\r\n\r\n", i)
+                if self._buf[i + 2: i + 4] == b"\r\n":
+                    self._buf = self._buf[i + 4:]
+                    return True
+                if j < 0:
+                    return None
+                self._buf = self._buf[j + 4:]
+                return True
+            if len(self._buf) < i + 2 + size + 2:
+                return None
+            self._body += self._buf[i + 2: i + 2 + size]
+            if len(self._body) > self.server.max_body:
+                return False
+            self._buf = self._buf[i + 2 + size + 2:]
 
     # ---- request parsing -----------------------------------------------
     def _parse_head(self, head: bytes) -> None:
@@ -143,6 +210,7 @@ class HttpProtocol(asyncio.Protocol):
         headers = []
         need = 0
         keep = version != b"HTTP/1.0"
+        chunked_req = False
         for ln in lines[1:]:
             i = ln.find(b":")
             if i < 0:
@@ -152,6 +220,8 @@ class HttpProtocol(asyncio.Protocol):
             headers.append((k, v))
             if k == b"content-length":
                 need = int(v)
+            elif k == b"transfer-encoding":
+                chunked_req = b"chunked" in v.lower()
             elif k == b"connection":
                 lv = v.lower()
                 if lv == b"close":
@@ -178,7 +248,9 @@ class HttpProtocol(asyncio.Protocol):
                                               self._raw_end,
                                               self._raw_backlogged)},
         }
-        self._need_body = need
+        self._need_body = 0 if chunked_req else need
+        self._chunked_req = chunked_req
+        self._body = b""
         self._keep_alive = keep
 
     # ---- direct SSE fast path ------------------------------------------
@@ -325,7 +397,8 @@ class HttpServer:
 
     def __init__(self, app, host: str, port: int,
                  ssl_context: Optional[ssl_mod.SSLContext] = None,
-                 reuse_port: bool = False, max_body: int = 32 << 20):
+                 reuse_port: bool = False, max_body: int = 32 << 20,
+                 header_timeout: float = 30.0):
         self.app = app
         self.host = host
         self.port = port
@@ -333,6 +406,7 @@ class HttpServer:
         self.is_tls = ssl_context is not None
         self.reuse_port = reuse_port
         self.max_body = max_body
+        self.header_timeout = header_timeout
         self._server: Optional[asyncio.AbstractServer] = None
         self._connections: set = set()
 

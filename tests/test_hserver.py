@@ -408,3 +408,66 @@ async def test_direct_stream_logprobs_over_socket(provider):
     for e in entries:
         assert isinstance(e["logprob"], float) and e["logprob"] <= 0
         assert len(e["top_logprobs"]) == 3
+
+
+@pytest.mark.asyncio
+async def test_chunked_request_body(provider):
+    """Transfer-Encoding: chunked request bodies decode correctly (the
+    reference's Go server accepts them natively)."""
+    async with serve(provider) as base:
+        host, port = base.rsplit("//", 1)[1].split(":")
+        body = json.dumps({
+            "model": "local/tiny-llama",
+            "messages": [{"role": "user", "content": "chunked"}],
+            "max_tokens": 3, "temperature": 0}).encode()
+        mid = len(body) // 2
+        frames = (b"%x\r\n" % mid + body[:mid] + b"\r\n"
+                  + b"%x\r\n" % (len(body) - mid) + body[mid:] + b"\r\n"
+                  + b"0\r\n\r\n")
+        reader, writer = await asyncio.open_connection(host, int(port))
+        writer.write(
+            b"POST /v1/chat/completions HTTP/1.1\r\nhost: x\r\n"
+            b"content-type: application/json\r\n"
+            b"transfer-encoding: chunked\r\n\r\n")
+        await writer.drain()
+        await asyncio.sleep(0.05)          # split across TCP segments
+        writer.write(frames[:10])
+        await writer.drain()
+        await asyncio.sleep(0.05)
+        writer.write(frames[10:])
+        await writer.drain()
+        data = b""
+        while b"\r\n\r\n" not in data or b"completion_tokens" not in data:
+            chunk = await asyncio.wait_for(reader.read(65536), 30)
+            if not chunk:
+                break
+            data += chunk
+        writer.close()
+        assert b"200" in data.split(b"\r\n", 1)[0]
+        assert b'"completion_tokens":3' in data.replace(b" ", b"")
+
+
+@pytest.mark.asyncio
+async def test_incomplete_header_times_out(provider):
+    """Slowloris guard: a trickled, never-completed request head gets a
+    408 and the connection is closed."""
+    cfg = config_mod.load({})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    srv = HttpServer(gw, "127.0.0.1", 0, header_timeout=0.3)
+    await srv.start()
+    try:
+        reader, writer = await asyncio.open_connection("127.0.0.1",
+                                                       srv.port)
+        writer.write(b"GET /health HTTP/1.1\r\nhost: x")  # never finishes
+        await writer.drain()
+        data = await asyncio.wait_for(reader.read(4096), 10)
+        assert b"408" in data
+        eof = await asyncio.wait_for(reader.read(4096), 10)
+        assert eof == b""
+        writer.close()
+    finally:
+        await srv.stop()
+        await client.aclose()
