@@ -112,6 +112,12 @@ class HttpProtocol(asyncio.Protocol):
 
     def data_received(self, data: bytes) -> None:
         self._buf += data
+        if self._task is not None and not self._task.done():
+            # a response is still being written (e.g. a long SSE
+            # stream): buffer the pipelined request — _run re-kicks
+            # parsing when it finishes.  Two concurrent tasks would
+            # interleave writes on one transport.
+            return
         if not self._head_done:
             end = self._buf.find(b"\r\n\r\n")
             if end < 0:
@@ -387,9 +393,13 @@ class HttpProtocol(asyncio.Protocol):
             return
         if not self._keep_alive:
             self.transport.close()
-        elif self._buf:
-            # pipelined next request already buffered
-            self.data_received(b"")
+        else:
+            # this task is finishing: clear it BEFORE re-kicking the
+            # parser or the task-active guard would swallow a pipelined
+            # request that arrived mid-response
+            self._task = None
+            if self._buf:
+                self.data_received(b"")
 
 
 class HttpServer:

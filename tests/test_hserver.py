@@ -471,3 +471,46 @@ async def test_incomplete_header_times_out(provider):
     finally:
         await srv.stop()
         await client.aclose()
+
+
+@pytest.mark.asyncio
+async def test_pipelined_requests_serialize(provider):
+    """Two requests written back-to-back on one connection (the second
+    arriving while the first streams) must produce two well-formed
+    responses in order — never interleaved writes."""
+    async with serve(provider) as base:
+        host, port = base.rsplit("//", 1)[1].split(":")
+        reader, writer = await asyncio.open_connection(host, int(port))
+        b1 = json.dumps({"model": "local/tiny-llama",
+                         "messages": [{"role": "user", "content": "a"}],
+                         "max_tokens": 4, "temperature": 0,
+                         "stream": True}).encode()
+        b2 = json.dumps({"model": "local/tiny-llama",
+                         "messages": [{"role": "user", "content": "b"}],
+                         "max_tokens": 2, "temperature": 0}).encode()
+        writer.write(
+            b"POST /v1/chat/completions HTTP/1.1\r\nhost: x\r\n"
+            b"content-type: application/json\r\n"
+            b"content-length: %d\r\n\r\n" % len(b1) + b1
+            + b"POST /v1/chat/completions HTTP/1.1\r\nhost: x\r\n"
+            b"content-type: application/json\r\n"
+            b"content-length: %d\r\n\r\n" % len(b2) + b2)
+        await writer.drain()
+        data = b""
+        deadline = asyncio.get_event_loop().time() + 60
+        while data.count(b"HTTP/1.1 200") < 2:
+            assert asyncio.get_event_loop().time() < deadline, data[-400:]
+            chunk = await asyncio.wait_for(reader.read(65536), 30)
+            if not chunk:
+                break
+            data += chunk
+            if (b"[DONE]" in data
+                    and data.count(b"HTTP/1.1 200") >= 2
+                    and b'"completion_tokens":2' in data.replace(b" ", b"")):
+                break
+        writer.close()
+        # first response: SSE with [DONE]; second: JSON completion
+        assert data.count(b"HTTP/1.1 200") == 2
+        first, second = data.split(b"HTTP/1.1 200", 2)[1:]
+        assert b"[DONE]" in first
+        assert b'"chat.completion"' in second.replace(b" ", b"")
