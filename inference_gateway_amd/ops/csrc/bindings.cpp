@@ -42,6 +42,8 @@ void silu_mul_grouped(torch::Tensor out, torch::Tensor gate_up,
                       torch::Tensor counts, int64_t capacity);
 void fused_gemm_tn(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                    bool swiglu, bool nt_weights);
+void fused_gemm_tn2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                    bool swiglu);
 int64_t ar_alloc(int64_t bytes, bool finegrained);
 void ar_free(int64_t ptr);
 py::bytes ar_get_handle(int64_t ptr);
@@ -80,6 +82,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped per-expert MFMA GEMM over the padded dispatch buffer");
   m.def("silu_mul_grouped", &igw::silu_mul_grouped,
         "SwiGLU over only the real (non-padded) MoE dispatch rows");
+  m.def("fused_gemm_tn2", &igw::fused_gemm_tn2,
+        "TN bf16 MFMA GEMM v2: block-shared LDS weight tiles "
+        "(glds + XOR swizzle), optional fused SwiGLU");
   m.def("fused_gemm_tn", &igw::fused_gemm_tn,
         "TN bf16 MFMA GEMM with optional fused SwiGLU epilogue "
         "(decode weight-streaming shapes, gfx950)");

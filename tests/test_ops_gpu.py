@@ -460,3 +460,35 @@ def test_fused_gemm_tn_matches_fp32_ref():
     want = x2.float() @ wd.float().t()
     rel = (got - want).abs().max() / want.abs().max().clamp_min(1.0)
     assert rel < 2e-2, f"down-proj rel err {rel:.4f}"
+
+
+def test_fused_gemm_tn2_matches_fp32_ref():
+    """v2 (block-shared LDS W tiles, glds + XOR swizzle) vs fp32 —
+    asymmetric operands; plain and SwiGLU modes; flagship shapes."""
+    from inference_gateway_amd.ops import _kernels
+
+    torch.manual_seed(43)
+    M, N, K = 256, 128, 256
+    x = (torch.randn(M, K, device="cuda") * 0.5).to(torch.bfloat16)
+    w = (torch.randn(2 * N, K, device="cuda") * 0.2).to(torch.bfloat16)
+    w += torch.arange(2 * N, device="cuda").view(-1, 1).to(
+        torch.bfloat16) * 1e-3
+    out = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
+    _kernels.fused_gemm_tn2(out, x, w[:N], False)
+    want = x.float() @ w[:N].float().t()
+    rel = (out.float() - want).abs().max() / want.abs().max().clamp_min(1.0)
+    assert rel < 2e-2, f"v2 plain rel {rel:.4f}"
+    _kernels.fused_gemm_tn2(out, x, w, True)
+    g = x.float() @ w[:N].float().t()
+    u = x.float() @ w[N:].float().t()
+    want = torch.nn.functional.silu(g) * u
+    rel = (out.float() - want).abs().max() / want.abs().max().clamp_min(1.0)
+    assert rel < 2e-2, f"v2 swiglu rel {rel:.4f}"
+    x2 = (torch.randn(256, 14336, device="cuda") * 0.1).to(torch.bfloat16)
+    wd = (torch.randn(4096, 14336, device="cuda") * 0.05).to(
+        torch.bfloat16)
+    out2 = torch.empty((256, 4096), dtype=torch.bfloat16, device="cuda")
+    _kernels.fused_gemm_tn2(out2, x2, wd, False)
+    want = x2.float() @ wd.float().t()
+    rel = (out2.float() - want).abs().max() / want.abs().max().clamp_min(1.0)
+    assert rel < 2e-2, f"v2 down rel {rel:.4f}"

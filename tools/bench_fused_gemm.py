@@ -65,20 +65,38 @@ def main():
             return ops.fused_gemm_tn(x, w, swiglu=swiglu,
                                      nt_weights=True)
 
+        from inference_gateway_amd.ops import _kernels
+
+        def fused2():
+            N = NR // 2 if swiglu else NR
+            out = torch.empty((M, N), dtype=x.dtype, device=x.device)
+            _kernels.fused_gemm_tn2(out, x, w, swiglu)
+            return out
+
         # correctness gate before timing
         got = fused().float()
         ref = lib().float()
         rel = (got - ref).abs().max() / ref.abs().max().clamp_min(1.0)
         assert rel < 5e-2, (name, float(rel))
+        has_v2 = M % 256 == 0
+        if has_v2:
+            got2 = fused2().float()
+            rel2 = (got2 - ref).abs().max() / ref.abs().max().clamp_min(1.0)
+            assert rel2 < 5e-2, ("v2", name, float(rel2))
 
-        for f in (lib, fused, fused_nt):
+        variants = [lib, fused, fused_nt] + ([fused2] if has_v2 else [])
+        for f in variants:
             for _ in range(3):
                 f()
-        t = {"lib": [], "fused": [], "fused_nt": []}
+        t = {"lib": [], "fused": [], "fused_nt": [], "fused2": []}
         for _ in range(ROUNDS):
             t["lib"].append(timed(lib))
             t["fused"].append(timed(fused))
             t["fused_nt"].append(timed(fused_nt))
+            if has_v2:
+                t["fused2"].append(timed(fused2))
+        if not has_v2:
+            t.pop("fused2")
         med = {k: sorted(v)[len(v) // 2] for k, v in t.items()}
         row = {
             "shape": name, "M": M, "N_rows": NR, "K": K,
@@ -88,8 +106,13 @@ def main():
             "lib_tbps": round(wbytes / med["lib"] / 1e12, 2),
             "fused_tbps": round(wbytes / med["fused"] / 1e12, 2),
             "fused_nt_tbps": round(wbytes / med["fused_nt"] / 1e12, 2),
-            "speedup": round(med["lib"] / min(med["fused"],
-                                              med["fused_nt"]), 3),
+            "fused2_us": (round(med["fused2"] * 1e6, 1)
+                          if "fused2" in med else None),
+            "fused2_tbps": (round(wbytes / med["fused2"] / 1e12, 2)
+                            if "fused2" in med else None),
+            "speedup": round(med["lib"] / min(
+                [med["fused"], med["fused_nt"]]
+                + ([med["fused2"]] if "fused2" in med else [])), 3),
         }
         results.append(row)
         print(json.dumps(row), flush=True)
