@@ -43,7 +43,8 @@ void silu_mul_grouped(torch::Tensor out, torch::Tensor gate_up,
 void fused_gemm_tn(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                    bool swiglu, bool nt_weights);
 void fused_gemm_tn2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
-                    bool swiglu);
+                    bool swiglu, bool swz);
+void glds_probe(torch::Tensor out, torch::Tensor inp, int64_t swizzle);
 int64_t ar_alloc(int64_t bytes, bool finegrained);
 void ar_free(int64_t ptr);
 py::bytes ar_get_handle(int64_t ptr);
@@ -82,6 +83,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "grouped per-expert MFMA GEMM over the padded dispatch buffer");
   m.def("silu_mul_grouped", &igw::silu_mul_grouped,
         "SwiGLU over only the real (non-padded) MoE dispatch rows");
+  m.def("glds_probe", &igw::glds_probe, "glds round-trip probe (debug)");
   m.def("fused_gemm_tn2", &igw::fused_gemm_tn2,
         "TN bf16 MFMA GEMM v2: block-shared LDS weight tiles "
         "(glds + XOR swizzle), optional fused SwiGLU");

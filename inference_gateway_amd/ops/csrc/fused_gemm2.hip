@@ -41,7 +41,44 @@ constexpr int BM2 = 256;    // rows per block (8 waves x 32)
 constexpr int BN2 = 64;     // output cols per block
 constexpr int BK2 = 64;     // k per pipeline step
 
-template <bool SWIGLU>
+// debug probe: stage a [64][64]-element bf16 matrix through the same
+// glds path (optionally source-swizzled) and read it back linearly —
+// pins the glds lane-placement and swizzle-pairing assumptions.
+__global__ void __launch_bounds__(512)
+glds_probe_kernel(uint16_t* __restrict__ out,
+                  const uint16_t* __restrict__ inp, int swizzle) {
+  __shared__ uint8_t sm[8192];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int base = wave * 1024;
+  const int o = base + lane * 16;
+  const int row = o / 128;
+  const int bir = o % 128;
+  const int src_byte = swizzle ? (bir ^ ((row & 7) << 4)) : bir;
+  const uint16_t* src = inp + row * 64 + src_byte / 2;
+  __builtin_amdgcn_global_load_lds(
+      reinterpret_cast<const unsigned int*>(src),
+      reinterpret_cast<unsigned int*>(sm + base), 16, 0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  for (int i = threadIdx.x; i < 64 * 64; i += 512) {
+    const int r = i / 64;
+    const int e = i % 64;
+    const int b = r * 128
+        + ((e * 2) ^ (swizzle ? ((r & 7) << 4) : 0));
+    out[i] = *reinterpret_cast<const uint16_t*>(sm + b);
+  }
+}
+
+void glds_probe(torch::Tensor out, torch::Tensor inp, int64_t swizzle) {
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(glds_probe_kernel, dim3(1), dim3(512), 0, stream,
+                     reinterpret_cast<uint16_t*>(out.data_ptr()),
+                     reinterpret_cast<const uint16_t*>(inp.data_ptr()),
+                     (int)swizzle);
+}
+
+template <bool SWIGLU, bool SWZ>
 __global__ void __launch_bounds__(512)
 fused_tn_gemm2_kernel(uint16_t* __restrict__ out,
                       const uint16_t* __restrict__ x,
@@ -82,7 +119,8 @@ fused_tn_gemm2_kernel(uint16_t* __restrict__ out,
       const int o = base + lane * 16;
       const int row = o / ROWB;
       const int byte_in_row = o % ROWB;
-      const int src_byte = byte_in_row ^ ((row & 7) << 4);
+      const int src_byte = SWZ ? (byte_in_row ^ ((row & 7) << 4))
+                                : byte_in_row;
       const uint16_t* src;
       if (!SWIGLU || row < BN2) {
         src = wg_base + (long)row * w_row + k0 + src_byte / 2;
@@ -101,7 +139,8 @@ fused_tn_gemm2_kernel(uint16_t* __restrict__ out,
                  int khalf) -> bf16x8_2 {
     const int row = base_row + nfrag * 16 + fr;
     const int byte_in_row = (khalf * 2 + kg) * 16;
-    const int b = row * ROWB + (byte_in_row ^ ((row & 7) << 4));
+    const int b = row * ROWB
+        + (SWZ ? (byte_in_row ^ ((row & 7) << 4)) : byte_in_row);
     return *reinterpret_cast<const bf16x8_2*>(&smem[buf][0][0] + b);
   };
 
@@ -172,7 +211,7 @@ fused_tn_gemm2_kernel(uint16_t* __restrict__ out,
 }
 
 void fused_gemm_tn2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
-                    bool swiglu) {
+                    bool swiglu, bool swz) {
   TORCH_CHECK(x.is_cuda() && w.is_cuda() && out.is_cuda());
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(w.scalar_type() == torch::kBFloat16);
@@ -185,22 +224,22 @@ void fused_gemm_tn2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
               && out.stride(1) == 1);
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(N / BN2, M / BM2);
+  auto launch = [&](auto sg, auto sz) {
+    hipLaunchKernelGGL(
+        (fused_tn_gemm2_kernel<decltype(sg)::value,
+                               decltype(sz)::value>),
+        grid, dim3(512), 0, stream,
+        reinterpret_cast<uint16_t*>(out.data_ptr()),
+        reinterpret_cast<const uint16_t*>(x.data_ptr()),
+        reinterpret_cast<const uint16_t*>(w.data_ptr()),
+        M, N, K, x.stride(0), w.stride(0), out.stride(0));
+  };
+  using T = std::true_type;
+  using F = std::false_type;
   if (swiglu) {
-    hipLaunchKernelGGL((fused_tn_gemm2_kernel<true>), grid, dim3(512),
-                       0, stream,
-                       reinterpret_cast<uint16_t*>(out.data_ptr()),
-                       reinterpret_cast<const uint16_t*>(x.data_ptr()),
-                       reinterpret_cast<const uint16_t*>(w.data_ptr()),
-                       M, N, K, x.stride(0), w.stride(0),
-                       out.stride(0));
+    swz ? launch(T{}, T{}) : launch(T{}, F{});
   } else {
-    hipLaunchKernelGGL((fused_tn_gemm2_kernel<false>), grid, dim3(512),
-                       0, stream,
-                       reinterpret_cast<uint16_t*>(out.data_ptr()),
-                       reinterpret_cast<const uint16_t*>(x.data_ptr()),
-                       reinterpret_cast<const uint16_t*>(w.data_ptr()),
-                       M, N, K, x.stride(0), w.stride(0),
-                       out.stride(0));
+    swz ? launch(F{}, T{}) : launch(F{}, F{});
   }
 }
 
