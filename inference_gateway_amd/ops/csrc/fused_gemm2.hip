@@ -138,7 +138,9 @@ fused_tn_gemm2_kernel(uint16_t* __restrict__ out,
   auto ldb = [&](int buf, int base_row, int nfrag,
                  int khalf) -> bf16x8_2 {
     const int row = base_row + nfrag * 16 + fr;
-    const int byte_in_row = (khalf * 2 + kg) * 16;
+    // a row holds BK2=64 elements (128 B); k-half h covers bytes
+    // [h*64, h*64+64), k-group kg its 16-B slot within the half
+    const int byte_in_row = khalf * 64 + kg * 16;
     const int b = row * ROWB
         + (SWZ ? (byte_in_row ^ ((row & 7) << 4)) : byte_in_row);
     return *reinterpret_cast<const bf16x8_2*>(&smem[buf][0][0] + b);
