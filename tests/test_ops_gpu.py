@@ -474,11 +474,11 @@ def test_fused_gemm_tn2_matches_fp32_ref():
     w += torch.arange(2 * N, device="cuda").view(-1, 1).to(
         torch.bfloat16) * 1e-3
     out = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
-    _kernels.fused_gemm_tn2(out, x, w[:N], False)
+    _kernels.fused_gemm_tn2(out, x, w[:N], False, True)
     want = x.float() @ w[:N].float().t()
     rel = (out.float() - want).abs().max() / want.abs().max().clamp_min(1.0)
     assert rel < 2e-2, f"v2 plain rel {rel:.4f}"
-    _kernels.fused_gemm_tn2(out, x, w, True)
+    _kernels.fused_gemm_tn2(out, x, w, True, True)
     g = x.float() @ w[:N].float().t()
     u = x.float() @ w[N:].float().t()
     want = torch.nn.functional.silu(g) * u
@@ -488,7 +488,7 @@ def test_fused_gemm_tn2_matches_fp32_ref():
     wd = (torch.randn(4096, 14336, device="cuda") * 0.05).to(
         torch.bfloat16)
     out2 = torch.empty((256, 4096), dtype=torch.bfloat16, device="cuda")
-    _kernels.fused_gemm_tn2(out2, x2, wd, False)
+    _kernels.fused_gemm_tn2(out2, x2, wd, False, True)
     want = x2.float() @ wd.float().t()
     rel = (out2.float() - want).abs().max() / want.abs().max().clamp_min(1.0)
     assert rel < 2e-2, f"v2 down rel {rel:.4f}"

@@ -70,7 +70,7 @@ def main():
         def fused2():
             N = NR // 2 if swiglu else NR
             out = torch.empty((M, N), dtype=x.dtype, device=x.device)
-            _kernels.fused_gemm_tn2(out, x, w, swiglu)
+            _kernels.fused_gemm_tn2(out, x, w, swiglu, True)
             return out
 
         # correctness gate before timing
