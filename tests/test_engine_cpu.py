@@ -246,6 +246,60 @@ def test_engine_fuzz_invariants():
     assert len(eng.scheduler._free_slots) == slots0, "leaked slots"
 
 
+@pytest.mark.parametrize("seed", [77, 1234, 9001])
+def test_engine_fuzz_invariants_mixed_mode(seed):
+    """The same randomized stress with the FULL r2 feature stack on:
+    mixed same-step batches + chunked prefill + prefix caching +
+    prefill-first, tiny block pool (preemption pressure), forks and
+    aborts.  Invariants: never wedges, pool and slots fully restored."""
+    import random
+
+    rng = random.Random(seed)
+    torch.manual_seed(0)
+    eng = _engine(num_gpu_blocks=28, max_model_len=96, max_num_seqs=4,
+                  max_batched_tokens=48, enable_chunked_prefill=True,
+                  enable_prefix_caching=True, enable_mixed_batches=True,
+                  scheduler_prefill_first=True)
+    bm = eng.block_manager
+    free0 = bm.num_free
+    slots0 = len(eng.scheduler._free_slots)
+    shared = list(range(200, 248))       # prefix-cacheable stem
+    live, done = [], []
+    for step in range(400):
+        r = rng.random()
+        if r < 0.22 and len(live) < 10:
+            prompt = (shared[: rng.choice([16, 32, 48])]
+                      + [rng.randrange(1, 500) for _ in range(
+                          rng.randrange(1, 20))])
+            sp = SamplingParams(max_new_tokens=rng.randrange(1, 10),
+                                temperature=rng.choice([0.0, 0.8]),
+                                seed=rng.choice([None, 5]))
+            if r < 0.05 and eng.scheduler.can_adopt():
+                rids = eng.add_request_group(prompt, sp, n=2)
+                live.extend(rids)
+            else:
+                live.append(eng.add_request(prompt, sp))
+        elif r < 0.30 and live:
+            victim = live.pop(rng.randrange(len(live)))
+            eng.abort_request(victim)
+            done.append(victim)
+        outs = eng.step()
+        for o in outs:
+            if o.finished and o.seq_id in live:
+                live.remove(o.seq_id)
+                done.append(o.seq_id)
+    guard = 0
+    while not eng.is_finished():
+        guard += 1
+        assert guard < 2000, "engine wedged during drain"
+        for o in eng.step():
+            if o.finished and o.seq_id in live:
+                live.remove(o.seq_id)
+    assert bm.num_free == free0, "leaked KV blocks"
+    assert len(eng.scheduler._free_slots) == slots0, "leaked slots"
+    assert eng.scheduler.num_mixed_rounds > 0, "mixed rounds never ran"
+
+
 def test_prompt_longer_than_token_budget_is_not_starved():
     """A prompt above max_batched_tokens (but under max_model_len) must be
     admitted as its own oversized prefill batch, not starve forever."""
