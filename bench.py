@@ -40,13 +40,19 @@ from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).resolve().parent))
 
-import torch  # noqa: E402
+# client workers are torch-free: at SCALE N=8 the bench spawns up to 64
+# of them per node and they only push bytes over loopback sockets
+if "--client-worker" not in sys.argv:
+    import torch  # noqa: E402
 
-# GEMM backend experiment knob: IGW_BLAS=hipblaslt routes torch GEMMs
-# through hipBLASLt instead of rocBLAS/Tensile (decode GEMMs are skinny-M;
-# see profiles/r01).
-if os.environ.get("IGW_BLAS", "") == "hipblaslt" and torch.cuda.is_available():
-    torch.backends.cuda.preferred_blas_library("cublaslt")
+    # GEMM backend experiment knob: IGW_BLAS=hipblaslt routes torch GEMMs
+    # through hipBLASLt instead of rocBLAS/Tensile (decode GEMMs are
+    # skinny-M; see profiles/r01).
+    if (os.environ.get("IGW_BLAS", "") == "hipblaslt"
+            and torch.cuda.is_available()):
+        torch.backends.cuda.preferred_blas_library("cublaslt")
+else:
+    torch = None
 
 
 def parse_args():
