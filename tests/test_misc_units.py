@@ -79,3 +79,44 @@ def test_hf_tokenizer_with_real_tokenizer_file(tmp_path):
     ids = t.encode("hello world")
     assert ids[0] == 0                      # BOS prepended
     assert t.decode(ids[1:]) == "hello world"
+
+
+def test_example_manifests_parse():
+    """Every shipped example manifest must be machine-valid: compose +
+    k8s YAML (multi-doc), Grafana dashboard + provisioning, Keycloak
+    realm export, routing pools, generated openapi.yaml."""
+    import json
+    from pathlib import Path
+
+    import yaml
+
+    root = Path(__file__).resolve().parent.parent
+    ex = root / "examples"
+    n_yaml = n_json = 0
+    for p in list(ex.rglob("*.yml")) + list(ex.rglob("*.yaml")):
+        docs = list(yaml.safe_load_all(p.read_text()))
+        assert docs, p
+        n_yaml += 1
+    for p in ex.rglob("*.json"):
+        json.loads(p.read_text())
+        n_json += 1
+    assert n_yaml >= 12 and n_json >= 2, (n_yaml, n_json)
+    # the dashboard targets only metrics this build actually exports
+    dash = json.loads((ex / "docker-compose" / "monitoring" / "grafana"
+                       / "dashboards" / "inference-gateway.json"
+                       ).read_text())
+    exprs = " ".join(t["expr"] for pnl in dash["panels"]
+                     for t in pnl["targets"])
+    from inference_gateway_amd.otel import metrics as m
+    import inspect
+
+    src = inspect.getsource(m)
+    for name in ("igw_engine_running_seqs", "igw_engine_waiting_seqs",
+                 "igw_engine_prefix_hit_tokens_total"):
+        assert name in exprs and name in src, name
+    # openapi.yaml parses and the realm export is importable JSON
+    yaml.safe_load((root / "openapi.yaml").read_text().split("\n", 1)[1])
+    realm = json.loads((ex / "docker-compose" / "auth" / "keycloak"
+                        / "realm-export.json").read_text())
+    assert realm["realm"] == "inference-gateway"
+    assert realm["clients"][0]["clientId"] == "inference-gateway"
