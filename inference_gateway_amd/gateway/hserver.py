@@ -437,10 +437,22 @@ class HttpServer:
         async with self._server:
             await self._server.serve_forever()
 
-    async def stop(self) -> None:
+    async def stop(self, grace: float = 0.0) -> None:
+        """Stop accepting, optionally drain in-flight responses for up
+        to `grace` seconds (the reference's server.Shutdown(5s),
+        main.go:384-402), then close every connection."""
         if self._server is not None:
             self._server.close()
             await self._server.wait_closed()
+        if grace > 0:
+            deadline = asyncio.get_running_loop().time() + grace
+            while asyncio.get_running_loop().time() < deadline:
+                busy = any(
+                    p._task is not None and not p._task.done()
+                    for p in self._connections)
+                if not busy:
+                    break
+                await asyncio.sleep(0.05)
         for proto in list(self._connections):
             if proto.transport is not None:
                 proto.transport.close()

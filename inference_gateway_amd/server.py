@@ -71,7 +71,7 @@ async def _serve(cfg, gw: Gateway, logger: Logger) -> None:
         await asyncio.gather(*(s.serve_forever() for s in servers))
     finally:
         for s in servers:
-            await s.stop()
+            await s.stop(grace=5.0)
         await gw.shutdown()
 
 

@@ -514,3 +514,39 @@ async def test_pipelined_requests_serialize(provider):
         first, second = data.split(b"HTTP/1.1 200", 2)[1:]
         assert b"[DONE]" in first
         assert b'"chat.completion"' in second.replace(b" ", b"")
+
+
+@pytest.mark.asyncio
+async def test_graceful_stop_drains_inflight(provider):
+    """stop(grace=N) lets an in-flight stream finish before closing
+    (reference server.Shutdown(5s) parity)."""
+    cfg = config_mod.load({})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    srv = HttpServer(gw, "127.0.0.1", 0)
+    await srv.start()
+    base = f"http://127.0.0.1:{srv.port}"
+    events = []
+
+    async def consume():
+        async with httpx.AsyncClient(base_url=base, timeout=60) as c:
+            async with c.stream("POST", "/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "drain"}],
+                "max_tokens": 8, "temperature": 0, "stream": True,
+            }) as r:
+                async for line in r.aiter_lines():
+                    if line.startswith("data:"):
+                        events.append(line[5:].strip())
+
+    task = asyncio.create_task(consume())
+    # let the stream start, then stop with grace
+    while not events:
+        await asyncio.sleep(0.01)
+    await srv.stop(grace=10.0)
+    await asyncio.wait_for(task, 30)
+    await client.aclose()
+    assert events[-1] == "[DONE]"
+    assert json.loads(events[-2])["usage"]["completion_tokens"] == 8
