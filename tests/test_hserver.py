@@ -550,3 +550,54 @@ async def test_graceful_stop_drains_inflight(provider):
     await client.aclose()
     assert events[-1] == "[DONE]"
     assert json.loads(events[-2])["usage"]["completion_tokens"] == 8
+
+
+@pytest.mark.asyncio
+async def test_remote_provider_sse_relay_over_socket():
+    """The generator relay path (remote providers — NOT the local
+    direct-write path) over the real socket: upstream SSE chunks arrive
+    framed, exactly one [DONE]."""
+    def upstream(request: httpx.Request) -> httpx.Response:
+        if request.url.path.endswith("/chat/completions"):
+            chunks = [
+                'data: {"id":"u1","object":"chat.completion.chunk",'
+                '"choices":[{"index":0,"delta":{"content":"re"},'
+                '"finish_reason":null}]}',
+                'data: {"id":"u1","object":"chat.completion.chunk",'
+                '"choices":[{"index":0,"delta":{"content":"mote"},'
+                '"finish_reason":"stop"}]}',
+                "data: [DONE]",
+            ]
+            return httpx.Response(
+                200, headers={"content-type": "text/event-stream"},
+                content="\n\n".join(chunks).encode())
+        return httpx.Response(404)
+
+    cfg = config_mod.load({"GROQ_API_KEY": "tok"})
+    client = httpx.AsyncClient(transport=httpx.MockTransport(upstream))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client)
+    srv = HttpServer(gw, "127.0.0.1", 0)
+    await srv.start()
+    try:
+        events = []
+        async with httpx.AsyncClient(
+                base_url=f"http://127.0.0.1:{srv.port}",
+                timeout=30) as c:
+            async with c.stream("POST", "/v1/chat/completions", json={
+                "model": "groq/m", "stream": True,
+                "messages": [{"role": "user", "content": "x"}],
+            }) as r:
+                assert r.status_code == 200
+                assert r.headers.get("transfer-encoding") == "chunked"
+                async for line in r.aiter_lines():
+                    if line.startswith("data:"):
+                        events.append(line[5:].strip())
+        assert events[-1] == "[DONE]"
+        assert sum(1 for e in events if e == "[DONE]") == 1
+        text = "".join(
+            json.loads(e)["choices"][0]["delta"].get("content") or ""
+            for e in events[:-1])
+        assert text == "remote"
+    finally:
+        await srv.stop()
+        await client.aclose()
