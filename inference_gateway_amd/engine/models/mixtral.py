@@ -88,7 +88,11 @@ class MixtralMoE(nn.Module):
         from ...parallel.comm import tp_all_reduce
 
         if self.expert_parallel:
-            return self._forward_expert_parallel(x)
+            if is_prefill:
+                # dynamic splits: exact traffic for big token counts
+                return self._forward_expert_parallel(x)
+            # decode: static-capacity, sync-free (graph-capturable)
+            return self._forward_expert_parallel_padded(x)
         if is_prefill and x.shape[0] > 1024:
             out = self._forward_sorted(x)
         else:
@@ -158,6 +162,83 @@ class MixtralMoE(nn.Module):
             [chunks[d][: (T * (d + 1)) // W - (T * d) // W]
              for d in range(W)], 0)
         return out
+
+    def _forward_expert_parallel_padded(self, x: torch.Tensor
+                                        ) -> torch.Tensor:
+        """Sync-free EP for DECODE (the dynamic-split
+        EP path host-syncs on .tolist() and cannot be graph-captured).
+        Every (src, dst) pair exchanges a STATIC capacity of
+        cap = ceil(Tl)*K rows — the worst case, so routing is EXACT
+        (zero token drops) — which makes both all-to-alls equal-split
+        (no host-known split lists), every shape static, and the whole
+        layer hipGraph-capturable.  Decode-size tensors keep the padding
+        traffic trivial (T=256, K=2, W=8: 4 MB/layer send buffer); the
+        dynamic path remains for prefill where padding would be W-fold
+        HBM traffic.  Numerics: identical to the dynamic EP path
+        (tested rank-for-rank on gloo world=2)."""
+        import torch.distributed as dist
+
+        W, r = self.ep_size, tp_rank()
+        T, h = x.shape
+        K, El = self.top_k, self.ep_local
+        dev = x.device
+        logits = self.router(x).float()
+        weights, topk = torch.topk(logits, K, dim=-1)
+        weights = torch.softmax(weights, dim=-1).to(x.dtype)
+        # this rank's sequence shard (same split as the dynamic path)
+        t0, t1 = (T * r) // W, (T * (r + 1)) // W
+        tl_max = max((T * (d + 1)) // W - (T * d) // W for d in range(W))
+        cap = tl_max * K                   # per-pair slot budget (exact)
+        n = (t1 - t0) * K
+        flat_e = topk[t0:t1].reshape(-1)                     # [n]
+        flat_t = (torch.arange(t0, t1, device=dev)
+                  .unsqueeze(1).expand(t1 - t0, K).reshape(-1))
+        flat_w = weights[t0:t1].reshape(-1)
+        dest = flat_e // El                                  # [n]
+        # slot within the destination pair: stable rank among same-dest
+        onehot = torch.nn.functional.one_hot(dest, W)
+        rank_in = (torch.cumsum(onehot, 0) - onehot).mul(onehot).sum(-1)
+        slot = dest * cap + rank_in                          # [n] < W*cap
+        sx = x.new_zeros((W * cap, h))
+        sx.index_copy_(0, slot, x.index_select(0, flat_t))
+        se = torch.full((W * cap,), -1, dtype=torch.int64, device=dev)
+        se.index_copy_(0, slot, flat_e)
+        # equal-split all-to-all: cap rows per pair, no split lists
+        rx = torch.empty_like(sx)
+        re = torch.empty_like(se)
+        dist.all_to_all_single(rx, sx)
+        dist.all_to_all_single(re, se)
+        # local experts over the received padded rows (same padded-bmm
+        # shape as the TP decode path; invalid rows route to a scrap row)
+        local_e = re - r * El
+        valid = (local_e >= 0) & (local_e < El)
+        le_safe = local_e.clamp(0, El - 1)
+        oh = torch.nn.functional.one_hot(le_safe, El) * valid.unsqueeze(1)
+        rk = (torch.cumsum(oh, 0) - oh).mul(oh).sum(-1)
+        C2 = W * cap
+        # invalid (padding) rows get DISTINCT scrap slots past the real
+        # region: index_copy_ with duplicate indices is undefined
+        scrap = El * C2 + torch.arange(C2, device=dev)
+        dst2 = torch.where(valid, le_safe * C2 + rk, scrap)
+        buf = x.new_zeros((El * C2 + C2, h))
+        buf.index_copy_(0, dst2, rx)
+        gu = torch.bmm(buf[: El * C2].view(El, C2, h), self.gate_up_t)
+        act = ops.silu_mul(gu.view(El * C2, -1))
+        y = torch.bmm(act.view(El, C2, -1), self.down_t).view(El * C2, h)
+        ys = torch.cat([y, y.new_zeros((C2, h))], 0).index_select(0, dst2)
+        # combine back along the same static routes
+        back = torch.empty_like(ys)
+        dist.all_to_all_single(back, ys)
+        out_local = x.new_zeros((tl_max, h))
+        out_local.index_add_(
+            0, flat_t - t0,
+            back.index_select(0, slot) * flat_w.unsqueeze(1))
+        # re-replicate (equal shards of tl_max rows)
+        chunks = [x.new_empty((tl_max, h)) for _ in range(W)]
+        dist.all_gather(chunks, out_local)
+        return torch.cat(
+            [chunks[d][: (T * (d + 1)) // W - (T * d) // W]
+             for d in range(W)], 0)
 
     def _forward_padded(self, x: torch.Tensor) -> torch.Tensor:
         """Sync-free dispatch for decode: tokens are scattered into a

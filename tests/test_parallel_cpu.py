@@ -134,7 +134,10 @@ def _ep_moe_worker(rank: int, world: int):
     moe = MixtralMoE(cfg, layer=0, device="cpu", dtype=torch.float32,
                      expert_parallel=True)
     x = torch.randn(9, cfg.hidden_size)
-    return moe(x).numpy()
+    # decode path = static-capacity sync-free EP; prefill path =
+    # dynamic-split EP.  Both must agree with each other and with TP=1.
+    return (moe(x, is_prefill=False).numpy(),
+            moe(x, is_prefill=True).numpy())
 
 
 # ---- tests ---------------------------------------------------------------
@@ -182,10 +185,16 @@ def test_tp2_engine_features_match(tmp_path):
 
 def test_ep2_moe_matches_single_rank(tmp_path):
     outs = _spawn("_ep_moe_worker", tmp_path)
-    assert (abs(outs[0] - outs[1]) < 1e-5).all()
-    ref = _single_rank_reference(_ep_moe_worker)
-    assert (abs(outs[0] - ref) < 1e-4).all(), \
-        "EP=2 all-to-all MoE must match the single-rank MoE"
+    dec0, pre0 = outs[0]
+    dec1, pre1 = outs[1]
+    assert (abs(dec0 - dec1) < 1e-5).all()
+    assert (abs(pre0 - pre1) < 1e-5).all()
+    # padded (decode) EP == dynamic (prefill) EP, rank-for-rank
+    assert (abs(dec0 - pre0) < 1e-5).all(), \
+        "static-capacity EP must equal the dynamic-split EP"
+    ref_dec, ref_pre = _single_rank_reference(_ep_moe_worker)
+    assert (abs(dec0 - ref_dec) < 1e-4).all(), \
+        "EP=2 MoE must match the single-rank MoE"
 
 
 def test_bench_contract_torchrun_world2(tmp_path):
