@@ -601,3 +601,49 @@ async def test_remote_provider_sse_relay_over_socket():
     finally:
         await srv.stop()
         await client.aclose()
+
+
+@pytest.mark.asyncio
+async def test_messages_passthrough_sse_over_socket():
+    """Anthropic /v1/messages streaming passthrough over the native
+    server: upstream event/data lines relayed verbatim, chunk-framed."""
+    def upstream(request: httpx.Request) -> httpx.Response:
+        if request.url.path.endswith("/messages"):
+            frames = (
+                "event: message_start\n"
+                'data: {"type":"message_start"}\n\n'
+                "event: content_block_delta\n"
+                'data: {"type":"content_block_delta","delta":'
+                '{"type":"text_delta","text":"hi"}}\n\n'
+                "event: message_stop\n"
+                'data: {"type":"message_stop"}\n\n')
+            return httpx.Response(
+                200, headers={"content-type": "text/event-stream"},
+                content=frames.encode())
+        return httpx.Response(404)
+
+    cfg = config_mod.load({"ANTHROPIC_API_KEY": "k"})
+    client = httpx.AsyncClient(transport=httpx.MockTransport(upstream))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client)
+    srv = HttpServer(gw, "127.0.0.1", 0)
+    await srv.start()
+    try:
+        lines = []
+        async with httpx.AsyncClient(
+                base_url=f"http://127.0.0.1:{srv.port}",
+                timeout=30) as c:
+            async with c.stream("POST", "/v1/messages", json={
+                "model": "anthropic/claude-x", "stream": True,
+                "max_tokens": 8,
+                "messages": [{"role": "user", "content": "x"}],
+            }) as r:
+                assert r.status_code == 200
+                async for line in r.aiter_lines():
+                    if line.strip():
+                        lines.append(line.strip())
+        assert "event: message_start" in lines
+        assert any("text_delta" in ln for ln in lines)
+        assert "event: message_stop" in lines
+    finally:
+        await srv.stop()
+        await client.aclose()
