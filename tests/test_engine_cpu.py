@@ -1204,3 +1204,29 @@ def test_mixed_batches_with_prefix_cache_and_forks():
         eng.step()
     assert eng.get_output(r2).token_ids == want
     assert eng.block_manager.prefix_hit_tokens > 0
+
+
+def test_mixed_batches_moe_token_equivalence():
+    """Mixed same-step rounds on the MoE engine (Mixtral path): the
+    union forward routes decode rows through the MoE with the prefill
+    flag — must stay token-exact vs the plain MoE engine."""
+    def run(mixed):
+        torch.manual_seed(0)
+        cfg = EngineConfig.tiny_moe(
+            max_batched_tokens=48, num_gpu_blocks=128, max_num_seqs=8,
+            enable_chunked_prefill=True, enable_mixed_batches=mixed)
+        eng = LLMEngine(cfg, device="cpu")
+        sp = lambda: SamplingParams(max_new_tokens=8, temperature=0.0)  # noqa
+        ra = eng.add_request(list(range(1, 30)), sp())
+        for _ in range(3):
+            eng.step()
+        rb = eng.add_request(list(range(40, 150)), sp())
+        while not eng.is_finished():
+            eng.step()
+        return ([eng.get_output(r).token_ids for r in (ra, rb)],
+                eng.scheduler.num_mixed_rounds)
+
+    plain, m0 = run(False)
+    mixed, m1 = run(True)
+    assert m0 == 0 and m1 > 0
+    assert mixed == plain
