@@ -140,6 +140,40 @@ def _ep_moe_worker(rank: int, world: int):
             moe(x, is_prefill=True).numpy())
 
 
+def _ckpt_tp_worker(rank: int, world: int):
+    """Load the SAME fabricated HF checkpoint TP-sharded and greedy-
+    generate — ranks must agree and match the TP=1 load (end-to-end
+    coverage for engine/weights.py's sliced sharding)."""
+    import tempfile
+    from pathlib import Path
+
+    from inference_gateway_amd.engine.config import (EngineConfig,
+                                                     SamplingParams)
+    from inference_gateway_amd.engine.engine import LLMEngine
+    from inference_gateway_amd.parallel import comm
+    from test_weights import _tiny_llama_checkpoint
+
+    comm.init_tp(world)
+    d = Path(tempfile.gettempdir()) / "igw_ckpt_tp_test"
+    if world > 1:
+        import torch.distributed as dist
+
+        if rank == 0:
+            d.mkdir(exist_ok=True)
+            _tiny_llama_checkpoint(d, seed=9)
+        dist.barrier()
+    else:
+        d.mkdir(exist_ok=True)
+        _tiny_llama_checkpoint(d, seed=9)
+    eng = LLMEngine(EngineConfig.tiny_llama(weights_path=str(d)),
+                    device="cpu")
+    rid = eng.add_request(list(range(1, 25)),
+                          SamplingParams(max_new_tokens=6, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    return eng.get_output(rid).token_ids
+
+
 # ---- tests ---------------------------------------------------------------
 
 def _single_rank_reference(fn):
@@ -195,6 +229,13 @@ def test_ep2_moe_matches_single_rank(tmp_path):
     ref_dec, ref_pre = _single_rank_reference(_ep_moe_worker)
     assert (abs(dec0 - ref_dec) < 1e-4).all(), \
         "EP=2 MoE must match the single-rank MoE"
+
+
+def test_checkpoint_load_tp2_matches_tp1(tmp_path):
+    outs = _spawn("_ckpt_tp_worker", tmp_path)
+    assert outs[0] == outs[1], "TP ranks must sample identical tokens"
+    ref = _single_rank_reference(_ckpt_tp_worker)
+    assert outs[0] == ref, "TP=2 checkpoint load must match TP=1"
 
 
 def test_bench_contract_torchrun_world2(tmp_path):
