@@ -124,15 +124,9 @@ class HttpProtocol(asyncio.Protocol):
                 if len(self._buf) > _MAX_HEAD:
                     self._simple_response(400, b"header too large")
                     self.transport.close()
-                elif self._idle_timer is None and self._buf:
-                    # header-read deadline: a connection that trickles
-                    # an incomplete request head is dropped (slowloris)
-                    self._idle_timer = self.loop.call_later(
-                        self.server.header_timeout, self._on_idle)
+                else:
+                    self._arm_read_deadline()
                 return
-            if self._idle_timer is not None:
-                self._idle_timer.cancel()
-                self._idle_timer = None
             try:
                 self._parse_head(self._buf[:end])
             except ValueError:
@@ -150,7 +144,8 @@ class HttpProtocol(asyncio.Protocol):
             # clients; the reference's Go server accepts these natively)
             done = self._consume_chunked()
             if done is None:
-                return                       # need more data
+                self._arm_read_deadline()    # need more data
+                return
             if done is False:
                 self._simple_response(400, b"malformed chunked body")
                 self.transport.close()
@@ -161,8 +156,22 @@ class HttpProtocol(asyncio.Protocol):
             self._body = self._buf[: self._need_body]
             self._buf = self._buf[self._need_body:]
             self._dispatch()
+        elif self._head_done:
+            self._arm_read_deadline()        # incomplete body
+
+    def _arm_read_deadline(self) -> None:
+        """One deadline per request READ (head + body): a connection
+        that trickles an incomplete request is dropped with 408
+        (slowloris — the Go reference gets this from net/http's
+        ReadHeaderTimeout plus its body deadline)."""
+        if self._idle_timer is None and (self._buf or self._head_done):
+            self._idle_timer = self.loop.call_later(
+                self.server.header_timeout, self._on_idle)
 
     def _dispatch(self) -> None:
+        if self._idle_timer is not None:
+            self._idle_timer.cancel()
+            self._idle_timer = None
         self._head_done = False
         scope, body = self._scope, self._body
         self._scope = None
@@ -171,8 +180,11 @@ class HttpProtocol(asyncio.Protocol):
 
     def _on_idle(self) -> None:
         self._idle_timer = None
-        if not self._head_done and self._buf and not self._closing:
-            self._simple_response(408, b"request header timeout")
+        if self._closing or (self._task is not None
+                             and not self._task.done()):
+            return
+        if self._buf or self._head_done:
+            self._simple_response(408, b"request timeout")
             self.transport.close()
 
     def _consume_chunked(self):
@@ -182,7 +194,9 @@ class HttpProtocol(asyncio.Protocol):
         while True:
             i = self._buf.find(b"\r\n")
             if i < 0:
-                return None if len(self._buf) <= 16 else False
+                # a size line longer than 64 bytes (even with chunk
+                # extensions) is not a real client
+                return None if len(self._buf) <= 64 else False
             size_tok = self._buf[:i].split(b";", 1)[0].strip()
             try:
                 size = int(size_tok, 16)
@@ -225,7 +239,14 @@ class HttpProtocol(asyncio.Protocol):
             v = ln[i + 1:].strip()
             headers.append((k, v))
             if k == b"content-length":
-                need = int(v)
+                n = int(v)
+                if n < 0:
+                    raise ValueError("negative content-length")
+                if need and n != need:
+                    # conflicting duplicate Content-Length headers are a
+                    # smuggling vector (RFC 7230 §3.3.3) — reject
+                    raise ValueError("conflicting content-length")
+                need = n
             elif k == b"transfer-encoding":
                 chunked_req = b"chunked" in v.lower()
             elif k == b"connection":

@@ -474,6 +474,52 @@ async def test_incomplete_header_times_out(provider):
 
 
 @pytest.mark.asyncio
+async def test_incomplete_body_times_out(provider):
+    """The read deadline covers the BODY too: a complete head with a
+    trickled, never-completed Content-Length body gets a 408 (slow-body
+    slowloris variant)."""
+    cfg = config_mod.load({})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    srv = HttpServer(gw, "127.0.0.1", 0, header_timeout=0.3)
+    await srv.start()
+    try:
+        reader, writer = await asyncio.open_connection("127.0.0.1",
+                                                       srv.port)
+        writer.write(b"POST /health HTTP/1.1\r\nhost: x\r\n"
+                     b"content-length: 100\r\n\r\nonly-a-few-bytes")
+        await writer.drain()
+        data = await asyncio.wait_for(reader.read(4096), 10)
+        assert b"408" in data
+        writer.close()
+    finally:
+        await srv.stop()
+        await client.aclose()
+
+
+@pytest.mark.asyncio
+async def test_bad_content_length_rejected(provider):
+    """Negative and conflicting-duplicate Content-Length headers are
+    400-rejected (RFC 7230 request-smuggling defenses)."""
+    async with serve(provider) as base:
+        host, port = base[len("http://"):].split(":")
+        for head in (b"POST /health HTTP/1.1\r\nhost: x\r\n"
+                     b"content-length: -5\r\n\r\n",
+                     b"POST /health HTTP/1.1\r\nhost: x\r\n"
+                     b"content-length: 4\r\ncontent-length: 9\r\n"
+                     b"\r\nabcd"):
+            reader, writer = await asyncio.open_connection(host,
+                                                           int(port))
+            writer.write(head)
+            await writer.drain()
+            data = await asyncio.wait_for(reader.read(4096), 10)
+            assert b"400" in data.split(b"\r\n", 1)[0]
+            writer.close()
+
+
+@pytest.mark.asyncio
 async def test_pipelined_requests_serialize(provider):
     """Two requests written back-to-back on one connection (the second
     arriving while the first streams) must produce two well-formed
