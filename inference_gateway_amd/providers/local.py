@@ -291,11 +291,17 @@ class LocalProvider:
         if req.get("logprobs"):
             lp = max(0, min(20, int(req.get("top_logprobs") or 0)))
         seed = req.get("seed")
+        # explicit JSON null means "use the default" (OpenAI semantics;
+        # validate.py deliberately lets null through)
+        def _num(key, dflt):
+            v = req.get(key)
+            return dflt if v is None else v
+
         return SamplingParams(
             max_new_tokens=int(max_tokens),
-            temperature=float(req.get("temperature", 1.0)),
-            top_p=float(req.get("top_p", 1.0)),
-            top_k=int(req.get("top_k", 0)),
+            temperature=float(_num("temperature", 1.0)),
+            top_p=float(_num("top_p", 1.0)),
+            top_k=int(_num("top_k", 0)),
             stop_token_ids=stop_ids,
             logprobs=lp,
             seed=int(seed) if seed is not None else None,

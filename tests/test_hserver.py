@@ -73,6 +73,22 @@ async def test_chat_completion_over_socket(provider):
 
 
 @pytest.mark.asyncio
+async def test_explicit_null_sampling_fields_use_defaults(provider):
+    """OpenAI clients send `"temperature": null` to mean "default";
+    validation lets null through and the local provider must not 500."""
+    async with serve(provider) as base:
+        async with httpx.AsyncClient(base_url=base, timeout=60) as c:
+            r = await c.post("/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "hi"}],
+                "max_tokens": 3, "temperature": None, "top_p": None,
+                "top_k": None, "seed": None, "stop": None,
+                "presence_penalty": None})
+            assert r.status_code == 200
+            assert r.json()["usage"]["completion_tokens"] == 3
+
+
+@pytest.mark.asyncio
 async def test_sse_streams_incrementally(provider):
     """Chunks must arrive one event at a time (chunked framing, per-event
     flush), ending with [DONE]."""
