@@ -24,6 +24,9 @@ class ModelConfig:
     max_position: int = 8192
     norm_eps: float = 1e-5
     tie_embeddings: bool = False
+    # Qwen2 family: bias on the fused qkv projection only (HF Qwen2
+    # hardcodes q/k/v bias True, o/mlp bias False)
+    qkv_bias: bool = False
     # MoE (Mixtral): 0 experts => dense MLP
     num_experts: int = 0
     num_experts_per_tok: int = 2
@@ -74,6 +77,20 @@ MODEL_PRESETS = {
         num_experts=8,
         num_experts_per_tok=2,
     ),
+    "qwen2.5-7b": ModelConfig(
+        name="qwen2.5-7b",
+        hidden_size=3584,
+        num_layers=28,
+        num_heads=28,
+        num_kv_heads=4,
+        head_dim=128,
+        intermediate_size=18944,
+        vocab_size=152064,
+        rope_base=1000000.0,
+        max_position=32768,
+        norm_eps=1e-6,
+        qkv_bias=True,
+    ),
     # tiny models for CPU tests / smoke
     "tiny-llama": ModelConfig(
         name="tiny-llama",
@@ -86,6 +103,20 @@ MODEL_PRESETS = {
         vocab_size=512,
         rope_base=10000.0,
         max_position=2048,
+    ),
+    "tiny-qwen2": ModelConfig(
+        name="tiny-qwen2",
+        hidden_size=512,
+        num_layers=2,
+        num_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        intermediate_size=1024,
+        vocab_size=512,
+        rope_base=10000.0,
+        max_position=2048,
+        qkv_bias=True,
+        tie_embeddings=True,
     ),
     "tiny-moe": ModelConfig(
         name="tiny-moe",

@@ -61,6 +61,16 @@ class LlamaAttention(nn.Module):
                        v_proj.weight.data], 0).contiguous(),
             requires_grad=False)
         del q_proj, k_proj, v_proj
+        if cfg.qkv_bias:
+            # Qwen2 family: bias on qkv only, sharded with the head dims
+            # (zeros under synthetic init; real values come from the
+            # checkpoint loader)
+            self.qkv_b = nn.Parameter(
+                torch.zeros((self.nq + 2 * self.nkv) * self.hd,
+                            dtype=dtype, device=device),
+                requires_grad=False)
+        else:
+            self.qkv_b = None
         self.o_proj = RowParallelLinear(
             cfg.num_heads * self.hd, h, rank, tp, device, dtype,
             _seed(f"l{layer}.o"))
@@ -69,7 +79,7 @@ class LlamaAttention(nn.Module):
                 kv_cache: Tuple[torch.Tensor, torch.Tensor],
                 cos_sin: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]
-        qkv = F.linear(x, self.qkv)
+        qkv = F.linear(x, self.qkv, self.qkv_b)
         # strided views into the fused QKV output — the HIP kernels take a
         # row stride, so there are no .contiguous() copies on the hot path
         q, k, v = qkv.split(

@@ -39,8 +39,14 @@ def model_config_from_hf(path: str, name: str = "hf-model"
     """Build a ModelConfig from a HuggingFace config.json."""
     cfg = json.loads((Path(path) / "config.json").read_text())
     num_experts = int(cfg.get("num_local_experts") or 0)
+    mtype = cfg.get("model_type", name)
+    # HF Qwen2 hardcodes qkv bias True and carries no attention_bias
+    # field; Llama-family configs carry attention_bias (default False)
+    qkv_bias = bool(cfg.get("attention_bias", mtype in ("qwen2",
+                                                        "qwen2_moe")))
     return ModelConfig(
-        name=cfg.get("model_type", name),
+        name=mtype,
+        qkv_bias=qkv_bias,
         hidden_size=int(cfg["hidden_size"]),
         num_layers=int(cfg["num_hidden_layers"]),
         num_heads=int(cfg["num_attention_heads"]),
@@ -118,6 +124,24 @@ def load_llama_weights(model, path: str) -> int:
         v = sh.rows(p + "self_attn.v_proj.weight",
                     rank * nkv * hd, (rank + 1) * nkv * hd)
         _copy(layer.attn.qkv, torch.cat([q, kk, v], 0), f"l{i}.qkv")
+        has_bias = (p + "self_attn.q_proj.bias") in sh.index
+        if layer.attn.qkv_b is not None:
+            if not has_bias:
+                raise ValueError(
+                    f"l{i}: model expects qkv bias (qkv_bias=True) but "
+                    "the checkpoint has no self_attn.*.bias tensors")
+            qb = sh.rows(p + "self_attn.q_proj.bias",
+                         rank * nq * hd, (rank + 1) * nq * hd)
+            kb = sh.rows(p + "self_attn.k_proj.bias",
+                         rank * nkv * hd, (rank + 1) * nkv * hd)
+            vb = sh.rows(p + "self_attn.v_proj.bias",
+                         rank * nkv * hd, (rank + 1) * nkv * hd)
+            _copy(layer.attn.qkv_b, torch.cat([qb, kb, vb], 0),
+                  f"l{i}.qkv_bias")
+        elif has_bias:
+            raise ValueError(
+                f"l{i}: checkpoint carries self_attn.*.bias but the "
+                "model config has qkv_bias=False — architecture mismatch")
         o = sh.cols(p + "self_attn.o_proj.weight",
                     rank * nq * hd, (rank + 1) * nq * hd)
         _copy(layer.attn.o_proj.weight, o, f"l{i}.o")
