@@ -228,10 +228,25 @@ def load_mixtral_weights(model, path: str) -> int:
     return loaded
 
 
+def _is_gguf(path: str) -> bool:
+    p = Path(path)
+    if p.is_file():
+        return p.suffix == ".gguf"
+    # a directory with .gguf files and no safetensors is a GGUF
+    # checkpoint (llama.cpp download layouts)
+    return (not any(p.glob("*.safetensors"))
+            and any(p.glob("*.gguf")))
+
+
 def load_weights(model, path: str) -> int:
-    """Dispatch on architecture; raises on shape/name mismatches (a
-    checkpoint that does not match the configured architecture must
-    fail loudly at startup, not generate garbage)."""
+    """Dispatch on format (safetensors dir vs .gguf) and architecture;
+    raises on shape/name mismatches (a checkpoint that does not match
+    the configured architecture must fail loudly at startup, not
+    generate garbage)."""
+    if _is_gguf(path):
+        from .gguf import load_gguf_weights
+
+        return load_gguf_weights(model, path)
     if getattr(model.cfg, "is_moe", False):
         return load_mixtral_weights(model, path)
     return load_llama_weights(model, path)

@@ -234,6 +234,11 @@ class LocalProvider:
             from ..engine.weights import model_config_from_hf
 
             preset = model_config_from_hf(weights)
+        elif local_cfg.model == "gguf":
+            # architecture from GGUF metadata (llama.cpp checkpoints)
+            from ..engine.gguf import model_config_from_gguf
+
+            preset = model_config_from_gguf(weights)
         else:
             preset = MODEL_PRESETS[local_cfg.model]
         ecfg = EngineConfig(
