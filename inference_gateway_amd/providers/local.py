@@ -260,9 +260,18 @@ class LocalProvider:
             capture_at_init=True,
         )
         self.engine = engine or AsyncEngine(ecfg, device=device)
-        self.tokenizer = get_tokenizer(
-            preset.vocab_size,
-            getattr(local_cfg, "tokenizer_path", None) or None)
+        tok_path = getattr(local_cfg, "tokenizer_path", None) or None
+        if tok_path is None and weights:
+            # HF checkpoint dirs ship tokenizer.json next to the
+            # safetensors — use it without extra configuration
+            from pathlib import Path
+
+            cand = Path(weights)
+            cand = (cand if cand.is_dir() else cand.parent) / \
+                "tokenizer.json"
+            if cand.is_file():
+                tok_path = str(cand)
+        self.tokenizer = get_tokenizer(preset.vocab_size, tok_path)
         # OpenAI system_fingerprint: identifies the backend configuration
         # (determinism domain for `seed`)
         self.fingerprint = f"fp_igw-amd_{local_cfg.model}"

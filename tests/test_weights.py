@@ -313,3 +313,31 @@ def test_qwen2_tp_sharded_bias_slices(tmp_path, monkeypatch):
     assert torch.equal(model.layers[0].attn.qkv_b.data.float(),
                        torch.cat([qb, kb, vb], 0)
                        .to(torch.bfloat16).float())
+
+
+def test_tokenizer_json_autodetected_next_to_checkpoint(tmp_path):
+    """LOCAL_MODEL=hf with a tokenizer.json in the checkpoint dir uses
+    the real tokenizer without LOCAL_TOKENIZER_PATH."""
+    from tokenizers import Tokenizer, models, pre_tokenizers
+
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.local import LocalProvider
+    from inference_gateway_amd.utils.tokenizer import HFTokenizer
+
+    _tiny_llama_checkpoint(tmp_path)
+    vocab = {"<unk>": 0, "<s>": 1, "</s>": 2, "hello": 3, "world": 4}
+    tok = Tokenizer(models.WordLevel(vocab, unk_token="<unk>"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    tok.save(str(tmp_path / "tokenizer.json"))
+
+    lc = LocalConfig(enabled=True, model="hf", max_num_seqs=4,
+                     max_model_len=128)
+    lc.num_gpu_blocks = 64
+    lc.weights_path = str(tmp_path)
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        assert isinstance(provider.tokenizer, HFTokenizer)
+        ids = provider.tokenizer.encode("hello world")
+        assert 3 in ids and 4 in ids
+    finally:
+        provider.engine.shutdown()
