@@ -250,6 +250,39 @@ def model_config_from_gguf(path: str) -> ModelConfig:
     )
 
 
+def tokenizer_from_gguf(path: str):
+    """Rebuild the embedded tokenizer from GGUF metadata.
+
+    Supports tokenizer.ggml.model == "gpt2" (byte-level BPE — what
+    Llama-3 and Qwen GGUFs carry): tokens + merges arrays become a
+    `tokenizers` BPE with ByteLevel pre-tokenizer/decoder.  Returns an
+    HFTokenizer or None when the metadata has no reconstructable
+    tokenizer (caller falls back to the synthetic ByteTokenizer)."""
+    g = path if isinstance(path, GGUFFile) else GGUFFile(path)
+    md = g.metadata
+    tokens: List[str] = md.get("tokenizer.ggml.tokens") or []
+    if not tokens or md.get("tokenizer.ggml.model") != "gpt2":
+        return None
+    from tokenizers import Tokenizer, decoders, models, pre_tokenizers
+
+    from ..utils.tokenizer import HFTokenizer
+
+    vocab = {t: i for i, t in enumerate(tokens)}
+    merges = []
+    for mstr in md.get("tokenizer.ggml.merges") or []:
+        a, _, b = mstr.partition(" ")
+        if b:
+            merges.append((a, b))
+    tok = Tokenizer(models.BPE(vocab, merges, fuse_unk=False))
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tok.decoder = decoders.ByteLevel()
+    bos = md.get("tokenizer.ggml.bos_token_id")
+    eos = md.get("tokenizer.ggml.eos_token_id")
+    return HFTokenizer(tok=tok,
+                       bos_token_id=int(bos) if bos is not None else None,
+                       eos_token_id=int(eos) if eos is not None else None)
+
+
 def load_gguf_weights(model, path: str) -> int:
     """Load a GGUF llama/qwen2-family checkpoint into LlamaForCausalLM.
 

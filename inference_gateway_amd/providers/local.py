@@ -271,7 +271,15 @@ class LocalProvider:
                 "tokenizer.json"
             if cand.is_file():
                 tok_path = str(cand)
-        self.tokenizer = get_tokenizer(preset.vocab_size, tok_path)
+        self.tokenizer = None
+        if tok_path is None and local_cfg.model == "gguf":
+            # GGUF embeds its tokenizer in metadata — fully
+            # self-contained checkpoint
+            from ..engine.gguf import tokenizer_from_gguf
+
+            self.tokenizer = tokenizer_from_gguf(weights)
+        if self.tokenizer is None:
+            self.tokenizer = get_tokenizer(preset.vocab_size, tok_path)
         # OpenAI system_fingerprint: identifies the backend configuration
         # (determinism domain for `seed`)
         self.fingerprint = f"fp_igw-amd_{local_cfg.model}"

@@ -47,23 +47,31 @@ class ByteTokenizer:
 
 
 class HFTokenizer:
-    def __init__(self, path: str):
-        from tokenizers import Tokenizer
+    def __init__(self, path: str = None, tok=None, bos_token_id=None,
+                 eos_token_id=None):
+        """From a tokenizer.json path, or an already-built `tokenizers`
+        Tokenizer (GGUF metadata reconstruction — engine/gguf.py)."""
+        if tok is None:
+            from tokenizers import Tokenizer
 
-        self._tok = Tokenizer.from_file(path)
+            tok = Tokenizer.from_file(path)
+        self._tok = tok
         self.vocab_size = self._tok.get_vocab_size()
-        self.eos_token_id = None
-        for cand in ("</s>", "<|end_of_text|>", "<|endoftext|>", "<eos>"):
-            tid = self._tok.token_to_id(cand)
-            if tid is not None:
-                self.eos_token_id = tid
-                break
-        self.bos_token_id = None
-        for cand in ("<s>", "<|begin_of_text|>", "<bos>"):
-            tid = self._tok.token_to_id(cand)
-            if tid is not None:
-                self.bos_token_id = tid
-                break
+        self.eos_token_id = eos_token_id
+        if self.eos_token_id is None:
+            for cand in ("</s>", "<|end_of_text|>", "<|endoftext|>",
+                         "<eos>"):
+                tid = self._tok.token_to_id(cand)
+                if tid is not None:
+                    self.eos_token_id = tid
+                    break
+        self.bos_token_id = bos_token_id
+        if self.bos_token_id is None:
+            for cand in ("<s>", "<|begin_of_text|>", "<bos>"):
+                tid = self._tok.token_to_id(cand)
+                if tid is not None:
+                    self.bos_token_id = tid
+                    break
 
     def encode(self, text: str) -> List[int]:
         ids = self._tok.encode(text).ids

@@ -334,3 +334,47 @@ def test_local_provider_serves_gguf(tmp_path):
         assert body["usage"]["completion_tokens"] == 4
     finally:
         provider.engine.shutdown()
+
+
+def _kv_strarr(key, vals):
+    return _kv(key, 9, struct.pack("<IQ", 8, len(vals))
+               + b"".join(_s(v.encode()) for v in vals))
+
+
+def test_tokenizer_from_gguf_metadata(tmp_path):
+    """GPT-2-style byte-level BPE rebuilt from tokenizer.ggml.* arrays
+    (what Llama-3/Qwen GGUFs embed)."""
+    tokens = ["<bos>", "<eos>", "h", "i", "e", "l", "o", "hi"]
+    meta = [_kv_str("general.architecture", "llama"),
+            _kv_str("tokenizer.ggml.model", "gpt2"),
+            _kv_strarr("tokenizer.ggml.tokens", tokens),
+            _kv_strarr("tokenizer.ggml.merges", ["h i"]),
+            _kv_u32("tokenizer.ggml.bos_token_id", 0),
+            _kv_u32("tokenizer.ggml.eos_token_id", 1)]
+    write_gguf(tmp_path / "t.gguf", meta, {})
+    tok = G.tokenizer_from_gguf(str(tmp_path / "t.gguf"))
+    assert tok is not None
+    assert tok.bos_token_id == 0 and tok.eos_token_id == 1
+    ids = tok.encode("hi")
+    # bos prepended; "h"+"i" merged into the single "hi" token
+    assert ids == [0, tokens.index("hi")]
+    assert tok.decode([tokens.index("hi")]) == "hi"
+
+
+def test_gguf_without_tokenizer_metadata_falls_back(tmp_path):
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.local import LocalProvider
+    from inference_gateway_amd.utils.tokenizer import ByteTokenizer
+
+    m = MODEL_PRESETS["tiny-llama"]
+    t = _tiny_weights(m)
+    write_gguf(tmp_path / "m.gguf", _meta_for(m), _gguf_tensors(m, t))
+    lc = LocalConfig(enabled=True, model="gguf", max_num_seqs=4,
+                     max_model_len=128)
+    lc.num_gpu_blocks = 64
+    lc.weights_path = str(tmp_path / "m.gguf")
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        assert isinstance(provider.tokenizer, ByteTokenizer)
+    finally:
+        provider.engine.shutdown()
