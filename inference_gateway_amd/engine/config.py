@@ -91,6 +91,18 @@ MODEL_PRESETS = {
         norm_eps=1e-6,
         qkv_bias=True,
     ),
+    "phi-3-mini": ModelConfig(
+        name="phi-3-mini",
+        hidden_size=3072,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=32,
+        head_dim=96,
+        intermediate_size=8192,
+        vocab_size=32064,
+        rope_base=10000.0,
+        max_position=4096,
+    ),
     # tiny models for CPU tests / smoke
     "tiny-llama": ModelConfig(
         name="tiny-llama",

@@ -9,6 +9,10 @@ TP-sharded parameter layout:
   model.embed_tokens.weight                  -> embed
   layers.N.self_attn.{q,k,v}_proj.weight     -> layer.attn.qkv   (fused,
                                                 column-sharded per rank)
+  layers.N.self_attn.qkv_proj.weight         -> same (HF Phi-3 ships the
+                                                fusion pre-done; sections
+                                                are sliced per rank)
+  layers.N.mlp.gate_up_proj.weight           -> layer.mlp.gate_up (Phi-3)
   layers.N.self_attn.o_proj.weight           -> layer.attn.o_proj (row)
   layers.N.mlp.{gate,up}_proj.weight         -> layer.mlp.gate_up (fused)
   layers.N.mlp.down_proj.weight              -> layer.mlp.down_proj (row)
@@ -115,14 +119,27 @@ def load_llama_weights(model, path: str) -> int:
 
     _copy(model.embed, sh.full("model.embed_tokens.weight"), "embed")
     loaded += 1
+    # HF Phi-3 ships PRE-FUSED tensors (qkv_proj / gate_up_proj); the
+    # sections inside still need per-rank slicing
+    fused_ckpt = "model.layers.0.self_attn.qkv_proj.weight" in sh.index
     for i, layer in enumerate(model.layers):
         p = f"model.layers.{i}."
-        q = sh.rows(p + "self_attn.q_proj.weight",
-                    rank * nq * hd, (rank + 1) * nq * hd)
-        kk = sh.rows(p + "self_attn.k_proj.weight",
-                     rank * nkv * hd, (rank + 1) * nkv * hd)
-        v = sh.rows(p + "self_attn.v_proj.weight",
-                    rank * nkv * hd, (rank + 1) * nkv * hd)
+        if fused_ckpt:
+            qkv_name = p + "self_attn.qkv_proj.weight"
+            Q = cfg.num_heads * hd          # full q section rows
+            K = cfg.num_kv_heads * hd
+            q = sh.rows(qkv_name, rank * nq * hd, (rank + 1) * nq * hd)
+            kk = sh.rows(qkv_name, Q + rank * nkv * hd,
+                         Q + (rank + 1) * nkv * hd)
+            v = sh.rows(qkv_name, Q + K + rank * nkv * hd,
+                        Q + K + (rank + 1) * nkv * hd)
+        else:
+            q = sh.rows(p + "self_attn.q_proj.weight",
+                        rank * nq * hd, (rank + 1) * nq * hd)
+            kk = sh.rows(p + "self_attn.k_proj.weight",
+                         rank * nkv * hd, (rank + 1) * nkv * hd)
+            v = sh.rows(p + "self_attn.v_proj.weight",
+                        rank * nkv * hd, (rank + 1) * nkv * hd)
         _copy(layer.attn.qkv, torch.cat([q, kk, v], 0), f"l{i}.qkv")
         has_bias = (p + "self_attn.q_proj.bias") in sh.index
         if layer.attn.qkv_b is not None:
@@ -145,10 +162,16 @@ def load_llama_weights(model, path: str) -> int:
         o = sh.cols(p + "self_attn.o_proj.weight",
                     rank * nq * hd, (rank + 1) * nq * hd)
         _copy(layer.attn.o_proj.weight, o, f"l{i}.o")
-        g = sh.rows(p + "mlp.gate_proj.weight",
-                    rank * inter, (rank + 1) * inter)
-        u = sh.rows(p + "mlp.up_proj.weight",
-                    rank * inter, (rank + 1) * inter)
+        if fused_ckpt:
+            gu_name = p + "mlp.gate_up_proj.weight"
+            I = cfg.intermediate_size       # full gate section rows
+            g = sh.rows(gu_name, rank * inter, (rank + 1) * inter)
+            u = sh.rows(gu_name, I + rank * inter, I + (rank + 1) * inter)
+        else:
+            g = sh.rows(p + "mlp.gate_proj.weight",
+                        rank * inter, (rank + 1) * inter)
+            u = sh.rows(p + "mlp.up_proj.weight",
+                        rank * inter, (rank + 1) * inter)
         _copy(layer.mlp.gate_up, torch.cat([g, u], 0), f"l{i}.gate_up")
         d = sh.cols(p + "mlp.down_proj.weight",
                     rank * inter, (rank + 1) * inter)

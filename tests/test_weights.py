@@ -341,3 +341,109 @@ def test_tokenizer_json_autodetected_next_to_checkpoint(tmp_path):
         assert 3 in ids and 4 in ids
     finally:
         provider.engine.shutdown()
+
+
+def _fused_phi3_style_checkpoint(dirpath, seed=7):
+    """Same tiny-llama architecture but with HF Phi-3's PRE-FUSED
+    tensor naming (self_attn.qkv_proj, mlp.gate_up_proj).  Returns the
+    equivalent UNFUSED tensor dict for cross-checking."""
+    m = MODEL_PRESETS["tiny-llama"]
+    g = torch.Generator().manual_seed(seed)
+
+    def r(*shape):
+        return torch.randn(*shape, generator=g) * 0.02
+
+    h, hd = m.hidden_size, m.head_dim
+    unfused = {"model.embed_tokens.weight": r(m.vocab_size, h),
+               "model.norm.weight": torch.ones(h),
+               "lm_head.weight": r(m.vocab_size, h)}
+    fused = dict(unfused)
+    for i in range(m.num_layers):
+        p = f"model.layers.{i}."
+        q = r(m.num_heads * hd, h)
+        k = r(m.num_kv_heads * hd, h)
+        v = r(m.num_kv_heads * hd, h)
+        gate = r(m.intermediate_size, h)
+        up = r(m.intermediate_size, h)
+        unfused[p + "self_attn.q_proj.weight"] = q
+        unfused[p + "self_attn.k_proj.weight"] = k
+        unfused[p + "self_attn.v_proj.weight"] = v
+        unfused[p + "mlp.gate_proj.weight"] = gate
+        unfused[p + "mlp.up_proj.weight"] = up
+        fused[p + "self_attn.qkv_proj.weight"] = torch.cat([q, k, v], 0)
+        fused[p + "mlp.gate_up_proj.weight"] = torch.cat([gate, up], 0)
+        for name, t in ((p + "self_attn.o_proj.weight",
+                         r(h, m.num_heads * hd)),
+                        (p + "mlp.down_proj.weight",
+                         r(h, m.intermediate_size)),
+                        (p + "input_layernorm.weight", torch.ones(h)),
+                        (p + "post_attention_layernorm.weight",
+                         torch.ones(h))):
+            unfused[name] = t
+            fused[name] = t
+    save_file(fused, str(dirpath / "model.safetensors"))
+    (dirpath / "config.json").write_text(json.dumps({
+        "model_type": "phi3",
+        "hidden_size": h,
+        "num_hidden_layers": m.num_layers,
+        "num_attention_heads": m.num_heads,
+        "num_key_value_heads": m.num_kv_heads,
+        "head_dim": hd,
+        "intermediate_size": m.intermediate_size,
+        "vocab_size": m.vocab_size,
+        "rope_theta": m.rope_base,
+        "rms_norm_eps": m.norm_eps,
+        "max_position_embeddings": m.max_position,
+        "attention_bias": False,
+    }))
+    return unfused
+
+
+def test_phi3_fused_checkpoint_loads(tmp_path):
+    """Pre-fused qkv_proj/gate_up_proj tensors (HF Phi-3 layout) land
+    in the same engine parameters as the unfused layout."""
+    t = _fused_phi3_style_checkpoint(tmp_path)
+    eng = LLMEngine(EngineConfig.tiny_llama(weights_path=str(tmp_path)),
+                    device="cpu")
+    want = torch.cat([t["model.layers.0.self_attn.q_proj.weight"],
+                      t["model.layers.0.self_attn.k_proj.weight"],
+                      t["model.layers.0.self_attn.v_proj.weight"]], 0)
+    assert torch.equal(eng.model.layers[0].attn.qkv.data.float(),
+                       want.to(torch.bfloat16).float())
+    want = torch.cat([t["model.layers.1.mlp.gate_proj.weight"],
+                      t["model.layers.1.mlp.up_proj.weight"]], 0)
+    assert torch.equal(eng.model.layers[1].mlp.gate_up.data.float(),
+                       want.to(torch.bfloat16).float())
+    rid = eng.add_request(list(range(1, 17)),
+                          SamplingParams(max_new_tokens=3, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    assert len(eng.get_output(rid).token_ids) == 3
+
+
+def test_phi3_fused_tp_rank1_sections(tmp_path, monkeypatch):
+    """TP=2 rank 1 takes the SECOND HALF OF EACH SECTION inside the
+    pre-fused tensors (not the second half of the whole tensor)."""
+    from inference_gateway_amd.parallel import comm
+    from inference_gateway_amd.engine.models.llama import LlamaForCausalLM
+
+    t = _fused_phi3_style_checkpoint(tmp_path)
+    m = MODEL_PRESETS["tiny-llama"]
+    monkeypatch.setattr(comm, "_TP_SIZE", 2)
+    monkeypatch.setattr(comm, "_TP_RANK", 1)
+    model = LlamaForCausalLM(m, "cpu", torch.bfloat16)
+    W.load_weights(model, str(tmp_path))
+    hd = m.head_dim
+    nq, nkv = m.num_heads // 2, m.num_kv_heads // 2
+    want = torch.cat(
+        [t["model.layers.0.self_attn.q_proj.weight"][nq * hd:],
+         t["model.layers.0.self_attn.k_proj.weight"][nkv * hd:],
+         t["model.layers.0.self_attn.v_proj.weight"][nkv * hd:]], 0)
+    assert torch.equal(model.layers[0].attn.qkv.data.float(),
+                       want.to(torch.bfloat16).float())
+    inter = m.intermediate_size // 2
+    want = torch.cat(
+        [t["model.layers.0.mlp.gate_proj.weight"][inter:],
+         t["model.layers.0.mlp.up_proj.weight"][inter:]], 0)
+    assert torch.equal(model.layers[0].mlp.gate_up.data.float(),
+                       want.to(torch.bfloat16).float())
