@@ -280,6 +280,11 @@ class LocalProvider:
             self.tokenizer = tokenizer_from_gguf(weights)
         if self.tokenizer is None:
             self.tokenizer = get_tokenizer(preset.vocab_size, tok_path)
+        # the checkpoint's own prompt format (tokenizer_config.json)
+        from ..utils.tokenizer import load_chat_template
+
+        self.chat_template = (load_chat_template(weights) if weights
+                              else None)
         # OpenAI system_fingerprint: identifies the backend configuration
         # (determinism domain for `seed`)
         self.fingerprint = f"fp_igw-amd_{local_cfg.model}"
@@ -343,7 +348,13 @@ class LocalProvider:
                 "bytes": list(s.encode("utf-8")), "top_logprobs": tops}
 
     def _encode_request(self, req: dict):
-        prompt = apply_chat_template(req.get("messages") or [])
+        msgs = req.get("messages") or []
+        if self.chat_template:
+            from ..utils.tokenizer import render_chat_template
+
+            prompt = render_chat_template(self.chat_template, msgs)
+        else:
+            prompt = apply_chat_template(msgs)
         return self.tokenizer.encode(prompt)
 
     @staticmethod

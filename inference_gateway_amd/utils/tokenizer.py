@@ -89,6 +89,49 @@ def get_tokenizer(vocab_size: int, path: Optional[str] = None):
     return ByteTokenizer(vocab_size)
 
 
+def load_chat_template(weights_path: str) -> Optional[str]:
+    """The checkpoint's own jinja chat template, from
+    tokenizer_config.json next to the weights (HF layout), or None."""
+    import json
+    from pathlib import Path
+
+    p = Path(weights_path)
+    cand = (p if p.is_dir() else p.parent) / "tokenizer_config.json"
+    if not cand.is_file():
+        return None
+    try:
+        tpl = json.loads(cand.read_text()).get("chat_template")
+    except (ValueError, OSError):
+        return None
+    return tpl if isinstance(tpl, str) else None
+
+
+def render_chat_template(template: str, messages: List[dict],
+                         bos_token: str = "", eos_token: str = "") -> str:
+    """Render an HF-style jinja chat template (the model's own prompt
+    format — without it a real checkpoint sees out-of-distribution
+    role tags and generates garbage)."""
+    import jinja2
+
+    def raise_exception(msg):
+        raise jinja2.TemplateError(msg)
+
+    env = jinja2.Environment(trim_blocks=True, lstrip_blocks=True,
+                             keep_trailing_newline=True)
+    env.globals["raise_exception"] = raise_exception
+    # flatten OpenAI content-part unions to text for the template
+    msgs = []
+    for m in messages:
+        c = m.get("content", "")
+        if isinstance(c, list):
+            c = "".join(p.get("text", "") for p in c
+                        if isinstance(p, dict) and p.get("type") == "text")
+        msgs.append({**m, "content": c})
+    return env.from_string(template).render(
+        messages=msgs, add_generation_prompt=True,
+        bos_token=bos_token, eos_token=eos_token)
+
+
 def apply_chat_template(messages: List[dict]) -> str:
     """Minimal chat template (role-tagged turns). Real checkpoints would use
     the model's own template; with random-init weights the exact template is

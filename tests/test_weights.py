@@ -447,3 +447,39 @@ def test_phi3_fused_tp_rank1_sections(tmp_path, monkeypatch):
          t["model.layers.0.mlp.up_proj.weight"][inter:]], 0)
     assert torch.equal(model.layers[0].mlp.gate_up.data.float(),
                        want.to(torch.bfloat16).float())
+
+
+def test_chat_template_from_checkpoint(tmp_path):
+    """tokenizer_config.json chat_template (jinja) drives prompt
+    construction — a Llama-3-style template renders role headers."""
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.local import LocalProvider
+    from inference_gateway_amd.utils.tokenizer import (
+        load_chat_template, render_chat_template)
+
+    _tiny_llama_checkpoint(tmp_path)
+    tpl = ("{% for m in messages %}<|start|>{{ m['role'] }}\n"
+           "{{ m['content'] }}<|end|>{% endfor %}"
+           "{% if add_generation_prompt %}<|start|>assistant\n"
+           "{% endif %}")
+    (tmp_path / "tokenizer_config.json").write_text(
+        json.dumps({"chat_template": tpl}))
+    assert load_chat_template(str(tmp_path)) == tpl
+    out = render_chat_template(tpl, [
+        {"role": "user", "content": [{"type": "text", "text": "hi"}]}])
+    assert out == "<|start|>user\nhi<|end|><|start|>assistant\n"
+
+    lc = LocalConfig(enabled=True, model="hf", max_num_seqs=4,
+                     max_model_len=128)
+    lc.num_gpu_blocks = 64
+    lc.weights_path = str(tmp_path)
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        assert provider.chat_template == tpl
+        ids = provider._encode_request(
+            {"messages": [{"role": "user", "content": "hi"}]})
+        want = provider.tokenizer.encode(
+            "<|start|>user\nhi<|end|><|start|>assistant\n")
+        assert ids == want
+    finally:
+        provider.engine.shutdown()
