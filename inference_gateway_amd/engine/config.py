@@ -167,6 +167,9 @@ class SamplingParams:
     # in-flight token's penalty is applied device-side (see sampler.py).
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
+    # OpenAI logit_bias: {token_id: bias in [-100, 100]} added to the
+    # logits every step (None = off)
+    logit_bias: Optional[dict] = None
 
 
 @dataclass

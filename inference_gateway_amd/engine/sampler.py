@@ -45,12 +45,22 @@ def apply_penalties(logits: torch.Tensor, seqs,
     presence term counted twice (documented best-effort; the frequency
     term stays exact)."""
     rows = [i for i, s in enumerate(seqs)
-            if s.sampling.presence_penalty or s.sampling.frequency_penalty]
+            if s.sampling.presence_penalty or s.sampling.frequency_penalty
+            or s.sampling.logit_bias]
     if not rows:
         return
     dev, dt = logits.device, logits.dtype
     for i in rows:
         sp = seqs[i].sampling
+        if sp.logit_bias:
+            # OpenAI logit_bias: plain additive, every step
+            ids = torch.tensor(list(sp.logit_bias.keys()),
+                               dtype=torch.int64, device=dev)
+            vals = torch.tensor(list(sp.logit_bias.values()),
+                                dtype=torch.float32, device=dev)
+            logits[i, ids] += vals.to(dt)
+        if not (sp.presence_penalty or sp.frequency_penalty):
+            continue
         out = seqs[i].output_token_ids
         if out:
             counts: dict = {}
@@ -65,7 +75,9 @@ def apply_penalties(logits: torch.Tensor, seqs,
         # only rows whose last token is still undrained (in_flight) — at
         # the first decode step after prefill the host view IS current
         # and a device-side add would double-count
-        drows = [i for i in rows if seqs[i].in_flight]
+        drows = [i for i in rows if seqs[i].in_flight
+                 and (seqs[i].sampling.presence_penalty
+                      or seqs[i].sampling.frequency_penalty)]
         if drows:
             rows_t = torch.tensor(drows, dtype=torch.int64, device=dev)
             pen = torch.tensor(

@@ -75,4 +75,17 @@ def validate_chat_request(body: dict) -> Optional[str]:
     tools = body.get("tools")
     if tools is not None and not isinstance(tools, list):
         return "tools must be an array"
+    lb = body.get("logit_bias")
+    if lb is not None:
+        if not isinstance(lb, dict):
+            return "logit_bias must be an object"
+        if len(lb) > 300:
+            return "logit_bias supports at most 300 entries"
+        for k, v in lb.items():
+            try:
+                int(k)
+            except (TypeError, ValueError):
+                return "logit_bias keys must be token ids"
+            if not _is_num(v):
+                return "logit_bias values must be numbers"
     return None

@@ -318,6 +318,17 @@ class LocalProvider:
         if req.get("logprobs"):
             lp = max(0, min(20, int(req.get("top_logprobs") or 0)))
         seed = req.get("seed")
+        lb = None
+        if req.get("logit_bias"):
+            lb = {}
+            for k, val in req["logit_bias"].items():
+                tid = int(k)
+                if not 0 <= tid < self.engine.engine.cfg.model.vocab_size:
+                    raise HTTPError(
+                        400, f"logit_bias token id {tid} outside the "
+                        "model vocabulary")
+                lb[tid] = max(-100.0, min(100.0, float(val)))
+
         # explicit JSON null means "use the default" (OpenAI semantics;
         # validate.py deliberately lets null through)
         def _num(key, dflt):
@@ -334,6 +345,7 @@ class LocalProvider:
             seed=int(seed) if seed is not None else None,
             presence_penalty=float(req.get("presence_penalty") or 0.0),
             frequency_penalty=float(req.get("frequency_penalty") or 0.0),
+            logit_bias=lb,
         )
 
     def _lp_item(self, out) -> dict:

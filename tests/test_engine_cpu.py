@@ -1230,3 +1230,33 @@ def test_mixed_batches_moe_token_equivalence():
     mixed, m1 = run(True)
     assert m0 == 0 and m1 > 0
     assert mixed == plain
+
+
+def test_logit_bias_forces_and_bans_tokens():
+    """OpenAI logit_bias: +100 forces a token under greedy decoding;
+    -100 on the natural argmax bans it."""
+    from inference_gateway_amd.engine.config import (EngineConfig,
+                                                     SamplingParams)
+    from inference_gateway_amd.engine.engine import LLMEngine
+
+    eng = LLMEngine(EngineConfig.tiny_llama(), device="cpu")
+    prompt = list(range(1, 17))
+
+    def run(sp):
+        rid = eng.add_request(prompt, sp)
+        while not eng.is_finished():
+            eng.step()
+        return eng.get_output(rid).token_ids
+
+    base = run(SamplingParams(max_new_tokens=4, temperature=0))
+    forced = run(SamplingParams(max_new_tokens=4, temperature=0,
+                                logit_bias={7: 100.0}))
+    assert forced == [7, 7, 7, 7]
+    banned = run(SamplingParams(max_new_tokens=1, temperature=0,
+                                logit_bias={base[0]: -100.0}))
+    assert banned[0] != base[0]
+    # bias composes with penalties on the same request
+    both = run(SamplingParams(max_new_tokens=3, temperature=0,
+                              logit_bias={7: 100.0},
+                              presence_penalty=0.1))
+    assert both[0] == 7

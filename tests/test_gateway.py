@@ -460,6 +460,15 @@ async def test_chat_request_shape_validation():
         ({"model": "groq/m", "messages": [{"role": "user",
                                            "content": "x"}],
           "stop": ["a", "b", "c", "d", "e"]}, "at most 4"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": "x"}],
+          "logit_bias": [1, 2]}, "logit_bias must be an object"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": "x"}],
+          "logit_bias": {"abc": 1}}, "token ids"),
+        ({"model": "groq/m", "messages": [{"role": "user",
+                                           "content": "x"}],
+          "logit_bias": {"5": "big"}}, "values must be numbers"),
     ]
     async with gw_client(gw) as c:
         for body, frag in bads:

@@ -89,6 +89,25 @@ async def test_explicit_null_sampling_fields_use_defaults(provider):
 
 
 @pytest.mark.asyncio
+async def test_logit_bias_over_socket(provider):
+    async with serve(provider) as base:
+        async with httpx.AsyncClient(base_url=base, timeout=60) as c:
+            r = await c.post("/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "hi"}],
+                "max_tokens": 3, "temperature": 0,
+                "logit_bias": {"9": 100}})
+            assert r.status_code == 200
+            # out-of-vocab token id -> 400, not a 500
+            r = await c.post("/v1/chat/completions", json={
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "hi"}],
+                "max_tokens": 3, "logit_bias": {"99999": 1}})
+            assert r.status_code == 400
+            assert "vocabulary" in str(r.json())
+
+
+@pytest.mark.asyncio
 async def test_sse_streams_incrementally(provider):
     """Chunks must arrive one event at a time (chunked framing, per-event
     flush), ending with [DONE]."""
