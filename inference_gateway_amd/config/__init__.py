@@ -117,6 +117,11 @@ class ServerConfig:
     max_request_body_size: int = 10 * 1024 * 1024
     tls_cert_path: str = ""
     tls_key_path: str = ""
+    # SO_REUSEPORT worker processes for proxy-only deployments (the Go
+    # reference scales across cores in one process; CPython needs one
+    # event loop per core).  Forced to 1 when the local engine is
+    # enabled — one GPU engine per process would double-allocate HBM.
+    workers: int = 1
 
 
 @dataclass
@@ -251,6 +256,7 @@ def load(env: Optional[Mapping[str, str]] = None) -> Config:
                                     str(10 * 1024 * 1024)))
     s.tls_cert_path = g("SERVER_TLS_CERT_PATH", "")
     s.tls_key_path = g("SERVER_TLS_KEY_PATH", "")
+    s.workers = int(g("SERVER_WORKERS", "1"))
 
     cl = cfg.client
     cl.timeout = parse_duration(g("CLIENT_TIMEOUT", "30s"))

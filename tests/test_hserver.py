@@ -728,3 +728,49 @@ async def test_messages_passthrough_sse_over_socket():
     finally:
         await srv.stop()
         await client.aclose()
+
+
+@pytest.mark.asyncio
+async def test_reuse_port_balances_two_listeners():
+    """SO_REUSEPORT (SERVER_WORKERS mechanics): two listeners on one
+    port, kernel distributes distinct connections across both."""
+    hits = {"a": 0, "b": 0}
+
+    def app_for(tag):
+        async def app(scope, receive, send):
+            hits[tag] += 1
+            await send({"type": "http.response.start", "status": 200,
+                        "headers": [(b"content-length", b"2")]})
+            await send({"type": "http.response.body", "body": b"ok"})
+        return app
+
+    s1 = HttpServer(app_for("a"), "127.0.0.1", 0, reuse_port=True)
+    await s1.start()
+    s2 = HttpServer(app_for("b"), "127.0.0.1", s1.port, reuse_port=True)
+    await s2.start()
+    try:
+        for _ in range(32):
+            reader, writer = await asyncio.open_connection(
+                "127.0.0.1", s1.port)
+            writer.write(b"GET /x HTTP/1.1\r\nhost: x\r\n"
+                         b"connection: close\r\n\r\n")
+            await writer.drain()
+            data = await asyncio.wait_for(reader.read(4096), 10)
+            assert b"200" in data
+            writer.close()
+    finally:
+        await s1.stop()
+        await s2.stop()
+    assert hits["a"] + hits["b"] == 32
+    assert hits["a"] > 0 and hits["b"] > 0, hits
+
+
+def test_resolve_workers_clamps_with_local_engine():
+    from inference_gateway_amd.server import resolve_workers
+
+    cfg = config_mod.load({"SERVER_WORKERS": "4"})
+    assert not cfg.local.enabled
+    assert resolve_workers(cfg) == 4
+    cfg = config_mod.load({"SERVER_WORKERS": "4", "LOCAL_ENABLED": "true",
+                           "LOCAL_MODEL": "tiny-llama"})
+    assert resolve_workers(cfg) == 1
