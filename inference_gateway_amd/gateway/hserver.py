@@ -53,7 +53,7 @@ class HttpProtocol(asyncio.Protocol):
                  "_need_body", "_body", "_scope", "_task", "_paused",
                  "_drain", "_keep_alive", "_chunked", "_chunked_req",
                  "_started", "_pending_head", "_closing", "_idle_timer",
-                 "server")
+                 "_ka_timer", "server")
 
     def __init__(self, app, loop, server: "HttpServer"):
         self.app = app
@@ -75,6 +75,7 @@ class HttpProtocol(asyncio.Protocol):
         self._pending_head = b""
         self._closing = False
         self._idle_timer = None
+        self._ka_timer = None
 
     # ---- transport callbacks -------------------------------------------
     def connection_made(self, transport) -> None:
@@ -88,12 +89,16 @@ class HttpProtocol(asyncio.Protocol):
             except OSError:
                 pass
         self.server._connections.add(self)
+        self._arm_keepalive()
 
     def connection_lost(self, exc) -> None:
         self._closing = True
         if self._idle_timer is not None:
             self._idle_timer.cancel()
             self._idle_timer = None
+        if self._ka_timer is not None:
+            self._ka_timer.cancel()
+            self._ka_timer = None
         if self._task is not None and not self._task.done():
             self._task.cancel()
         if self._paused and self._drain is not None:
@@ -111,6 +116,9 @@ class HttpProtocol(asyncio.Protocol):
             self._drain.set()
 
     def data_received(self, data: bytes) -> None:
+        if self._ka_timer is not None:
+            self._ka_timer.cancel()
+            self._ka_timer = None
         self._buf += data
         if self._task is not None and not self._task.done():
             # a response is still being written (e.g. a long SSE
@@ -158,6 +166,23 @@ class HttpProtocol(asyncio.Protocol):
             self._dispatch()
         elif self._head_done:
             self._arm_read_deadline()        # incomplete body
+
+    def _arm_keepalive(self) -> None:
+        """Idle keep-alive reaper (the Go reference gets this from
+        net/http IdleTimeout, main.go server config): a connection
+        sitting between requests past SERVER_IDLE_TIMEOUT is closed
+        silently — it is not an error, just resource reclamation."""
+        t = self.server.idle_timeout
+        if t and self._ka_timer is None:
+            self._ka_timer = self.loop.call_later(t, self._on_ka_idle)
+
+    def _on_ka_idle(self) -> None:
+        self._ka_timer = None
+        if self._closing:
+            return
+        if ((self._task is None or self._task.done())
+                and not self._buf and not self._head_done):
+            self.transport.close()
 
     def _arm_read_deadline(self) -> None:
         """One deadline per request READ (head + body): a connection
@@ -421,6 +446,8 @@ class HttpProtocol(asyncio.Protocol):
             self._task = None
             if self._buf:
                 self.data_received(b"")
+            else:
+                self._arm_keepalive()
 
 
 class HttpServer:
@@ -429,7 +456,8 @@ class HttpServer:
     def __init__(self, app, host: str, port: int,
                  ssl_context: Optional[ssl_mod.SSLContext] = None,
                  reuse_port: bool = False, max_body: int = 32 << 20,
-                 header_timeout: float = 30.0):
+                 header_timeout: float = 30.0,
+                 idle_timeout: float = 120.0):
         self.app = app
         self.host = host
         self.port = port
@@ -438,6 +466,7 @@ class HttpServer:
         self.reuse_port = reuse_port
         self.max_body = max_body
         self.header_timeout = header_timeout
+        self.idle_timeout = idle_timeout
         self._server: Optional[asyncio.AbstractServer] = None
         self._connections: set = set()
 

@@ -69,7 +69,9 @@ async def _serve(cfg, gw: Gateway, logger: Logger,
                                     cfg.server.tls_key_path)
     servers = [HttpServer(gw, cfg.server.host, cfg.server.port,
                           ssl_context=ssl_ctx, reuse_port=reuse_port,
-                          max_body=cfg.server.max_request_body_size)]
+                          max_body=cfg.server.max_request_body_size,
+                          header_timeout=cfg.server.read_timeout,
+                          idle_timeout=cfg.server.idle_timeout)]
     if cfg.telemetry.enabled and gw.telemetry is not None:
         # with workers > 1 each process answers /metrics with its own
         # counters (SO_REUSEPORT picks one per scrape connection) —

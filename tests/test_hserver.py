@@ -774,3 +774,30 @@ def test_resolve_workers_clamps_with_local_engine():
     cfg = config_mod.load({"SERVER_WORKERS": "4", "LOCAL_ENABLED": "true",
                            "LOCAL_MODEL": "tiny-llama"})
     assert resolve_workers(cfg) == 1
+
+
+@pytest.mark.asyncio
+async def test_idle_keepalive_connection_reaped(provider):
+    """A keep-alive connection idle past SERVER_IDLE_TIMEOUT is closed
+    silently (no 408 — between-requests idling is not an error)."""
+    cfg = config_mod.load({})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    srv = HttpServer(gw, "127.0.0.1", 0, idle_timeout=0.3)
+    await srv.start()
+    try:
+        reader, writer = await asyncio.open_connection("127.0.0.1",
+                                                       srv.port)
+        writer.write(b"GET /health HTTP/1.1\r\nhost: x\r\n\r\n")
+        await writer.drain()
+        data = await asyncio.wait_for(reader.read(4096), 10)
+        assert b"200" in data.split(b"\r\n", 1)[0]
+        # now idle past the timeout: server closes, NOT with a 408
+        rest = await asyncio.wait_for(reader.read(4096), 10)
+        assert rest == b""
+        writer.close()
+    finally:
+        await srv.stop()
+        await client.aclose()
