@@ -378,3 +378,41 @@ def test_gguf_without_tokenizer_metadata_falls_back(tmp_path):
         assert isinstance(provider.tokenizer, ByteTokenizer)
     finally:
         provider.engine.shutdown()
+
+
+def test_gguf_qwen2_bias_and_tied_embeddings(tmp_path):
+    """Qwen2-shaped GGUF: attn_q/k/v.bias tensors land in the fused
+    qkv bias; a missing output.weight means tied embeddings."""
+    m = MODEL_PRESETS["tiny-qwen2"]
+    t = _tiny_weights(m, seed=3)
+    ts = _gguf_tensors(m, t)
+    del ts["output.weight"]                    # tied embeddings
+    g = torch.Generator().manual_seed(99)
+    hd = m.head_dim
+    qb = (torch.randn(m.num_heads * hd, generator=g) * 0.02).to(
+        torch.float16).float()
+    kb = (torch.randn(m.num_kv_heads * hd, generator=g) * 0.02).to(
+        torch.float16).float()
+    vb = (torch.randn(m.num_kv_heads * hd, generator=g) * 0.02).to(
+        torch.float16).float()
+    for i in range(m.num_layers):
+        ts[f"blk.{i}.attn_q.bias"] = ((m.num_heads * hd,), G.GGML_F32,
+                                      _f32(qb))
+        ts[f"blk.{i}.attn_k.bias"] = ((m.num_kv_heads * hd,),
+                                      G.GGML_F32, _f32(kb))
+        ts[f"blk.{i}.attn_v.bias"] = ((m.num_kv_heads * hd,),
+                                      G.GGML_F32, _f32(vb))
+    write_gguf(tmp_path / "m.gguf", _meta_for(m, arch="qwen2"), ts)
+
+    cfg = G.model_config_from_gguf(str(tmp_path / "m.gguf"))
+    assert cfg.qkv_bias and cfg.tie_embeddings
+
+    from inference_gateway_amd.engine.models.llama import LlamaForCausalLM
+
+    model = LlamaForCausalLM(m, "cpu", torch.bfloat16)
+    n = G.load_gguf_weights(model, str(tmp_path / "m.gguf"))
+    assert n > 0
+    want = torch.cat([qb, kb, vb], 0)
+    assert torch.equal(model.layers[0].attn.qkv_b.data.float(),
+                       want.to(torch.bfloat16).float())
+    assert model.lm_head is None
