@@ -254,6 +254,10 @@ def load(env: Optional[Mapping[str, str]] = None) -> Config:
     s.idle_timeout = parse_duration(g("SERVER_IDLE_TIMEOUT", "120s"))
     s.max_request_body_size = int(g("SERVER_MAX_REQUEST_BODY_SIZE",
                                     str(10 * 1024 * 1024)))
+    if s.max_request_body_size <= 0:
+        # non-positive falls back to the default (reference
+        # config/server.go ResolveMaxRequestBodySize semantics)
+        s.max_request_body_size = 10 * 1024 * 1024
     s.tls_cert_path = g("SERVER_TLS_CERT_PATH", "")
     s.tls_key_path = g("SERVER_TLS_KEY_PATH", "")
     s.workers = int(g("SERVER_WORKERS", "1"))

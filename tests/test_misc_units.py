@@ -120,3 +120,16 @@ def test_example_manifests_parse():
                         / "realm-export.json").read_text())
     assert realm["realm"] == "inference-gateway"
     assert realm["clients"][0]["clientId"] == "inference-gateway"
+
+
+def test_nonpositive_body_cap_falls_back_to_default():
+    """Reference config/server.go ResolveMaxRequestBodySize: zero or
+    negative SERVER_MAX_REQUEST_BODY_SIZE means the 10 MiB default,
+    not "reject every request body"."""
+    from inference_gateway_amd import config as config_mod
+
+    for bad in ("0", "-1"):
+        cfg = config_mod.load({"SERVER_MAX_REQUEST_BODY_SIZE": bad})
+        assert cfg.server.max_request_body_size == 10 * 1024 * 1024
+    cfg = config_mod.load({"SERVER_MAX_REQUEST_BODY_SIZE": "1234"})
+    assert cfg.server.max_request_body_size == 1234
