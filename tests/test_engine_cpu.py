@@ -273,7 +273,9 @@ def test_engine_fuzz_invariants_mixed_mode(seed):
                           rng.randrange(1, 20))])
             sp = SamplingParams(max_new_tokens=rng.randrange(1, 10),
                                 temperature=rng.choice([0.0, 0.8]),
-                                seed=rng.choice([None, 5]))
+                                seed=rng.choice([None, 5]),
+                                logit_bias=rng.choice(
+                                    [None, {7: 50.0}, {3: -40.0}]))
             if r < 0.05 and eng.scheduler.can_adopt():
                 rids = eng.add_request_group(prompt, sp, n=2)
                 live.extend(rids)
@@ -1260,3 +1262,50 @@ def test_logit_bias_forces_and_bans_tokens():
                               logit_bias={7: 100.0},
                               presence_penalty=0.1))
     assert both[0] == 7
+
+
+def test_engine_fuzz_qwen2_bias_path():
+    """The mixed-mode fuzz on the QKV-BIAS architecture (tiny-qwen2,
+    tied embeddings): the bias path must hold the same pool/slot
+    invariants under preemption, forks, prefix hits and aborts."""
+    import random
+
+    rng = random.Random(4242)
+    torch.manual_seed(0)
+    eng = LLMEngine(
+        EngineConfig(model=MODEL_PRESETS["tiny-qwen2"], num_gpu_blocks=28,
+                     max_model_len=96, max_num_seqs=4,
+                     max_batched_tokens=48, enable_chunked_prefill=True,
+                     enable_prefix_caching=True, enable_mixed_batches=True,
+                     scheduler_prefill_first=True),
+        device="cpu")
+    bm = eng.block_manager
+    free0 = bm.num_free
+    slots0 = len(eng.scheduler._free_slots)
+    shared = list(range(200, 248))
+    live = []
+    for step in range(250):
+        r = rng.random()
+        if r < 0.25 and len(live) < 8:
+            prompt = (shared[: rng.choice([16, 32])]
+                      + [rng.randrange(1, 500) for _ in range(
+                          rng.randrange(1, 16))])
+            live.append(eng.add_request(
+                prompt,
+                SamplingParams(max_new_tokens=rng.randrange(1, 8),
+                               temperature=rng.choice([0.0, 0.8]),
+                               logit_bias=rng.choice([None, {9: 60.0}]))))
+        elif r < 0.32 and live:
+            eng.abort_request(live.pop(rng.randrange(len(live))))
+        for o in eng.step():
+            if o.finished and o.seq_id in live:
+                live.remove(o.seq_id)
+    guard = 0
+    while not eng.is_finished():
+        guard += 1
+        assert guard < 2000, "engine wedged during drain"
+        for o in eng.step():
+            if o.finished and o.seq_id in live:
+                live.remove(o.seq_id)
+    assert bm.num_free == free0, "leaked KV blocks"
+    assert len(eng.scheduler._free_slots) == slots0, "leaked slots"
