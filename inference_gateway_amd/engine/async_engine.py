@@ -53,9 +53,22 @@ class AsyncEngine:
         self._fatal = None
         self._queues: Dict[int, asyncio.Queue] = {}
         self._loops: Dict[int, asyncio.AbstractEventLoop] = {}
+        self._step_started = None      # monotonic ts of the running step
+        # Step watchdog (SURVEY §5 failure detection): a WEDGED engine
+        # step — a GPU fault that hangs instead of raising — would leave
+        # every SSE client waiting forever; exceptions are already
+        # handled by _run.  IGW_STEP_WATCHDOG=<seconds> fails all
+        # streams with EngineDeadError when one step exceeds the bound.
+        # Off by default: a legitimate huge prefill must not be killed
+        # by a paranoid default in the bench/driver environment.
+        self._watchdog_s = float(os.environ.get("IGW_STEP_WATCHDOG",
+                                                "0") or 0)
         self._thread = threading.Thread(target=self._run, daemon=True,
                                         name="igw-engine")
         self._thread.start()
+        if self._watchdog_s > 0:
+            threading.Thread(target=self._watchdog, daemon=True,
+                             name="igw-watchdog").start()
 
     # ---- engine thread --------------------------------------------------
     def _run(self) -> None:
@@ -71,8 +84,12 @@ class AsyncEngine:
         # Direct sinks get fail() either way.
         item = (_Fatal(self._fatal) if self._fatal is not None
                 else self._SENTINEL)
-        with self._lock:
-            items = list(self._queues.items())
+        # no lock: the watchdog calls this while the WEDGED engine
+        # thread still holds self._lock inside step() — taking it here
+        # would deadlock.  A GIL-atomic snapshot of the dict is enough:
+        # a racing add_request either sees _fatal first (raises) or its
+        # queue gets the sentinel on the next close.
+        items = list(self._queues.items())
         for rid, q in items:
             loop = self._loops.get(rid)
             if loop is None:
@@ -105,7 +122,15 @@ class AsyncEngine:
                 self._wake.clear()
                 continue
             with self._lock:
-                outputs = self.engine.step()
+                import time as _time
+
+                self._step_started = _time.monotonic()
+                try:
+                    outputs = self.engine.step()
+                finally:
+                    self._step_started = None
+            if self._fatal is not None:
+                return          # watchdog fired mid-step; queues closed
             if self._batched_delivery:
                 self._dispatch_batch(outputs)
             else:
@@ -122,6 +147,21 @@ class AsyncEngine:
                     else:
                         loop.call_soon_threadsafe(self._deliver,
                                                   [(q, out)])
+
+    def _watchdog(self) -> None:
+        import time as _time
+
+        poll = max(0.05, min(1.0, self._watchdog_s / 4))
+        while not self._stop and self._fatal is None:
+            _time.sleep(poll)
+            t0 = self._step_started
+            if (t0 is not None
+                    and _time.monotonic() - t0 > self._watchdog_s):
+                self._fatal = EngineDeadError(
+                    f"engine step exceeded the {self._watchdog_s:.0f}s "
+                    "watchdog (wedged GPU?)")
+                self._close_all_queues()
+                return
 
     def _dispatch_batch(self, outputs) -> None:
         # ONE event-loop wakeup per step, not one per token: at 256+
@@ -202,10 +242,7 @@ class AsyncEngine:
                 if outs:
                     yield outs
         finally:
-            with self._lock:
-                self._queues.pop(rid, None)
-                self._loops.pop(rid, None)
-                self.engine.abort_request(rid)
+            self._cleanup_request(rid)
 
     def generate_group(self, prompt_token_ids, sampling: SamplingParams,
                        n: int, batched: bool = False):
@@ -241,10 +278,7 @@ class AsyncEngine:
                     check(item)
                     yield item
             finally:
-                with self._lock:
-                    self._queues.pop(rid, None)
-                    self._loops.pop(rid, None)
-                    self.engine.abort_request(rid)
+                self._cleanup_request(rid)
 
         async def reader_batched(rid, q):
             try:
@@ -266,10 +300,7 @@ class AsyncEngine:
                     if outs:
                         yield outs
             finally:
-                with self._lock:
-                    self._queues.pop(rid, None)
-                    self._loops.pop(rid, None)
-                    self.engine.abort_request(rid)
+                self._cleanup_request(rid)
 
         mk = reader_batched if batched else reader
         return [mk(rid, q) for rid, q in zip(rids, qs)]
@@ -293,6 +324,18 @@ class AsyncEngine:
         return rid
 
     def close_direct(self, rid: int) -> None:
+        self._cleanup_request(rid)
+
+    def _cleanup_request(self, rid: int) -> None:
+        """Reader/sink teardown.  When the engine is DEAD (fatal set —
+        incl. a wedged step that still HOLDS self._lock) skip the
+        engine-side abort: taking the lock would deadlock the event
+        loop behind the hung thread, and there is nothing left to
+        abort."""
+        if self._fatal is not None:
+            self._queues.pop(rid, None)
+            self._loops.pop(rid, None)
+            return
         with self._lock:
             self._queues.pop(rid, None)
             self._loops.pop(rid, None)

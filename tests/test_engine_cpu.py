@@ -1309,3 +1309,42 @@ def test_engine_fuzz_qwen2_bias_path():
                 live.remove(o.seq_id)
     assert bm.num_free == free0, "leaked KV blocks"
     assert len(eng.scheduler._free_slots) == slots0, "leaked slots"
+
+
+def test_step_watchdog_fails_streams_on_wedged_step(monkeypatch):
+    """IGW_STEP_WATCHDOG: a step that WEDGES (no exception — e.g. a
+    hung GPU queue) must fail waiting clients with EngineDeadError
+    instead of hanging them forever."""
+    import asyncio
+    import threading
+    import time
+
+    from inference_gateway_amd.engine.async_engine import (AsyncEngine,
+                                                           EngineDeadError)
+
+    monkeypatch.setenv("IGW_STEP_WATCHDOG", "0.4")
+    eng = AsyncEngine(EngineConfig.tiny_llama(num_gpu_blocks=16,
+                                              max_model_len=64,
+                                              max_num_seqs=2),
+                      device="cpu")
+    release = threading.Event()
+
+    def wedged_step():
+        release.wait(timeout=10)      # simulates a hung hipLaunch
+        return []
+
+    eng.engine.step = wedged_step
+    try:
+        async def go():
+            with pytest.raises(EngineDeadError, match="watchdog"):
+                async for _ in eng.generate(
+                        list(range(1, 9)),
+                        SamplingParams(max_new_tokens=2)):
+                    pass
+
+        t0 = time.monotonic()
+        asyncio.run(go())
+        assert time.monotonic() - t0 < 5, "watchdog took too long"
+    finally:
+        release.set()
+        eng.shutdown()
