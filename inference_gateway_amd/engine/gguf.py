@@ -230,7 +230,7 @@ def model_config_from_gguf(path: str) -> ModelConfig:
     hidden = int(m("embedding_length"))
     tied = "output.weight" not in g.infos
     return ModelConfig(
-        name=a,
+        name=str(md.get("general.name") or a),
         hidden_size=hidden,
         num_layers=int(m("block_count")),
         num_heads=n_heads,

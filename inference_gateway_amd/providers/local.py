@@ -241,6 +241,11 @@ class LocalProvider:
             preset = model_config_from_gguf(weights)
         else:
             preset = MODEL_PRESETS[local_cfg.model]
+        if (not local_cfg.served_model_name
+                and local_cfg.model in ("hf", "gguf")):
+            # surface the checkpoint's own name in /v1/models and
+            # response `model` fields instead of the literal "hf"/"gguf"
+            self.model_name = preset.name
         ecfg = EngineConfig(
             model=preset,
             max_num_seqs=local_cfg.max_num_seqs,

@@ -416,3 +416,27 @@ def test_gguf_qwen2_bias_and_tied_embeddings(tmp_path):
     assert torch.equal(model.layers[0].attn.qkv_b.data.float(),
                        want.to(torch.bfloat16).float())
     assert model.lm_head is None
+
+
+def test_checkpoint_name_surfaces_in_models_list(tmp_path):
+    """LOCAL_MODEL=gguf lists the checkpoint's general.name, not the
+    literal 'gguf'."""
+    import asyncio
+
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.local import LocalProvider
+
+    m = MODEL_PRESETS["tiny-llama"]
+    t = _tiny_weights(m)
+    write_gguf(tmp_path / "m.gguf", _meta_for(m), _gguf_tensors(m, t))
+    lc = LocalConfig(enabled=True, model="gguf", max_num_seqs=4,
+                     max_model_len=128)
+    lc.num_gpu_blocks = 64
+    lc.weights_path = str(tmp_path / "m.gguf")
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        assert provider.model_name == "tiny"   # general.name metadata
+        data = asyncio.run(provider.list_models())
+        assert data["data"][0]["id"] == "local/tiny"
+    finally:
+        provider.engine.shutdown()
