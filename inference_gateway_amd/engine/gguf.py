@@ -203,8 +203,12 @@ class GGUFFile:
                 f"{self.path}: tensor {name!r} not in checkpoint")
         shape, ggml_type, offset, n_elem = self.infos[name]
         start = self._data_start + offset
-        raw = np.asarray(
-            self._mm[start: start + _type_bytes(ggml_type, n_elem)])
+        nbytes = _type_bytes(ggml_type, n_elem)
+        if start + nbytes > self._mm.size or start < 0:
+            raise ValueError(
+                f"{self.path}: tensor {name!r} data [{start}, "
+                f"{start + nbytes}) outside the file — corrupt GGUF")
+        raw = np.asarray(self._mm[start: start + nbytes])
         return _dequant(raw, ggml_type, n_elem).reshape(shape)
 
 
