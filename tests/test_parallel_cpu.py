@@ -174,6 +174,47 @@ def _ckpt_tp_worker(rank: int, world: int):
     return eng.get_output(rid).token_ids
 
 
+def _gguf_tp_worker(rank: int, world: int):
+    """Same shape as _ckpt_tp_worker but through the GGUF loader:
+    TP-sharded slices of a dequantized GGUF must generate identically
+    across ranks and match TP=1 (engine/gguf.py sharding)."""
+    import tempfile
+    from pathlib import Path
+
+    from inference_gateway_amd.engine.config import (EngineConfig,
+                                                     SamplingParams)
+    from inference_gateway_amd.engine.engine import LLMEngine
+    from inference_gateway_amd.parallel import comm
+    from test_gguf import (MODEL_PRESETS, _gguf_tensors, _meta_for,
+                           _tiny_weights, write_gguf)
+
+    comm.init_tp(world)
+    d = Path(tempfile.gettempdir()) / "igw_gguf_tp_test"
+    m = MODEL_PRESETS["tiny-llama"]
+
+    def fab():
+        d.mkdir(exist_ok=True)
+        t = _tiny_weights(m, seed=21)
+        write_gguf(d / "m.gguf", _meta_for(m), _gguf_tensors(m, t))
+
+    if world > 1:
+        import torch.distributed as dist
+
+        if rank == 0:
+            fab()
+        dist.barrier()
+    else:
+        fab()
+    eng = LLMEngine(
+        EngineConfig.tiny_llama(weights_path=str(d / "m.gguf")),
+        device="cpu")
+    rid = eng.add_request(list(range(1, 25)),
+                          SamplingParams(max_new_tokens=6, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    return eng.get_output(rid).token_ids
+
+
 # ---- tests ---------------------------------------------------------------
 
 def _single_rank_reference(fn):
@@ -236,6 +277,13 @@ def test_checkpoint_load_tp2_matches_tp1(tmp_path):
     assert outs[0] == outs[1], "TP ranks must sample identical tokens"
     ref = _single_rank_reference(_ckpt_tp_worker)
     assert outs[0] == ref, "TP=2 checkpoint load must match TP=1"
+
+
+def test_gguf_load_tp2_matches_tp1(tmp_path):
+    outs = _spawn("_gguf_tp_worker", tmp_path)
+    assert outs[0] == outs[1], "TP ranks must sample identical tokens"
+    ref = _single_rank_reference(_gguf_tp_worker)
+    assert outs[0] == ref, "TP=2 GGUF load must match TP=1"
 
 
 def test_bench_contract_torchrun_world2(tmp_path):
