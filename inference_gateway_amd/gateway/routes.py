@@ -274,7 +274,8 @@ class Router:
                 async with provider.stream_raw("POST", endpoint, raw,
                                                headers) as r:
                     if r.status_code >= 400:
-                        payload = (await r.aread()).decode()
+                        payload = (await r.aread()).decode(
+                            errors="replace")
                         yield "data: " + payload
                         return
                     async for line in r.aiter_lines():

@@ -106,7 +106,8 @@ class RemoteProvider:
                                       params=self.auth_params()) as r:
             if r.status_code >= 400:
                 body = await r.aread()
-                raise HTTPError(r.status_code, body.decode()[:2048])
+                raise HTTPError(r.status_code,
+                                body.decode(errors="replace")[:2048])
             async for line in r.aiter_lines():
                 yield line
 
