@@ -309,6 +309,9 @@ class LocalProvider:
                 "created": int(time.time()),
                 "owned_by": self.id,
                 "served_by": self.id,
+                # the engine KNOWS its context limit — enrichment
+                # preserves provider-supplied values (metadata.py)
+                "context_window": self.engine.engine.cfg.max_model_len,
             }],
         }
 

@@ -251,6 +251,28 @@ async def test_images_gated_by_flag():
 
 
 @pytest.mark.asyncio
+async def test_local_models_report_context_window():
+    """The local provider's own context limit survives enrichment
+    (provider-supplied values are preserved, not nulled)."""
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.local import LocalProvider
+
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=2,
+                     max_model_len=256)
+    lc.num_gpu_blocks = 32
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        gw, _ = make_gateway(local_provider=provider)
+        async with gw_client(gw) as c:
+            r = await c.get("/v1/models", params={
+                "provider": "local", "include": "context_window"})
+            assert r.status_code == 200
+            assert r.json()["data"][0]["context_window"] == 256
+    finally:
+        provider.engine.shutdown()
+
+
+@pytest.mark.asyncio
 async def test_models_include_enrichment():
     """?include=context_window,pricing,modalities resolves provider >
     community precedence; unknown values render as explicit null."""
