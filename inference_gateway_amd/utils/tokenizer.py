@@ -122,7 +122,7 @@ def render_chat_template(template: str, messages: List[dict],
     # flatten OpenAI content-part unions to text for the template
     msgs = []
     for m in messages:
-        c = m.get("content", "")
+        c = m.get("content") or ""
         if isinstance(c, list):
             c = "".join(p.get("text", "") for p in c
                         if isinstance(p, dict) and p.get("type") == "text")
@@ -139,7 +139,7 @@ def apply_chat_template(messages: List[dict]) -> str:
     parts = []
     for m in messages:
         role = m.get("role", "user")
-        content = m.get("content", "")
+        content = m.get("content") or ""
         if isinstance(content, list):  # OpenAI content-parts union
             content = "".join(
                 p.get("text", "") for p in content if isinstance(p, dict)
