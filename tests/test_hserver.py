@@ -801,3 +801,49 @@ async def test_idle_keepalive_connection_reaped(provider):
     finally:
         await srv.stop()
         await client.aclose()
+
+
+@pytest.mark.asyncio
+async def test_guardrails_on_direct_sse_path(provider, tmp_path):
+    """Guardrails + the direct-write fast path: pre_call still blocks
+    (403, no stream), clean requests stream via DirectStream, and the
+    middleware never runs post_call on a direct session."""
+    (tmp_path / "policy.yaml").write_text(
+        "rules:\n"
+        "  - name: block-bad\n"
+        "    phase: pre_call\n"
+        "    action: block\n"
+        "    contains: [forbidden]\n")
+    cfg = config_mod.load({"GUARDRAILS_ENABLED": "true",
+                           "GUARDRAILS_POLICY_DIR": str(tmp_path)})
+    client = httpx.AsyncClient(
+        transport=httpx.MockTransport(lambda r: httpx.Response(404)))
+    gw = Gateway(cfg=cfg, logger=noop_logger(), http_client=client,
+                 local_provider=provider)
+    srv = HttpServer(gw, "127.0.0.1", 0)
+    await srv.start()
+    base = f"http://127.0.0.1:{srv.port}"
+    try:
+        async with httpx.AsyncClient(base_url=base, timeout=60) as c:
+            r = await c.post("/v1/chat/completions", json={
+                "model": "local/tiny-llama", "stream": True,
+                "messages": [{"role": "user",
+                              "content": "forbidden stuff"}]})
+            assert r.status_code == 403
+            assert r.json()["error"]["type"] == "guardrail_blocked"
+            events = []
+            async with c.stream("POST", "/v1/chat/completions", json={
+                "model": "local/tiny-llama", "stream": True,
+                "max_tokens": 4, "temperature": 0,
+                "messages": [{"role": "user", "content": "clean"}],
+            }) as r:
+                assert r.status_code == 200
+                async for line in r.aiter_lines():
+                    if line.startswith("data:"):
+                        events.append(line[5:].strip())
+            assert events[-1] == "[DONE]"
+            body = json.loads(events[-2])
+            assert body["usage"]["completion_tokens"] == 4
+    finally:
+        await srv.stop()
+        await client.aclose()
