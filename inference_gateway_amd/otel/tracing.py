@@ -178,17 +178,45 @@ def tracing_middleware(tracer: Tracer, skip=("/health", "/v1/metrics")):
             traceparent=req.headers.get("traceparent"), kind="server")
         req.state["span"] = span
         token = current_span.set(span)
+        deferred = False
         try:
             resp = await nxt(req)
             span.set("http.status_code", resp.status)
             span.status_ok = resp.status < 500
+            # streaming responses: the span must cover the WHOLE
+            # stream (otelgin semantics — the Go handler blocks until
+            # the stream ends), not just time-to-headers
+            if resp.direct is not None:
+                prev = resp.direct.on_end
+                sess = resp.direct
+
+                def _end(*a, **kw):
+                    if prev is not None:
+                        prev(*a, **kw)
+                    tracer.finish(span)
+
+                sess.on_end = _end
+                deferred = True
+            elif resp.stream is not None:
+                orig = resp.stream
+
+                async def _wrapped():
+                    try:
+                        async for chunk in orig:
+                            yield chunk
+                    finally:
+                        tracer.finish(span)
+
+                resp.stream = _wrapped()
+                deferred = True
             return resp
         except Exception:
             span.status_ok = False
             raise
         finally:
             current_span.reset(token)
-            tracer.finish(span)
+            if not deferred:
+                tracer.finish(span)
 
     return mw
 

@@ -127,3 +127,32 @@ async def test_client_spans_on_outbound_calls():
     assert cs.trace_id == server.trace_id
     assert cs.parent_id == server.span_id
     assert cs.attributes["http.status_code"] == 200
+
+
+@pytest.mark.asyncio
+async def test_server_span_covers_whole_stream():
+    """For SSE the server span must finish when the STREAM ends (otelgin
+    semantics), not at response-start — and exactly once."""
+    from inference_gateway_amd.gateway.http import Request, Response
+    from inference_gateway_amd.otel.tracing import tracing_middleware
+
+    tracer = Tracer()
+    mw = tracing_middleware(tracer)
+
+    async def gen():
+        yield "data: a\n\n"
+        yield "data: [DONE]\n\n"
+
+    async def nxt(req):
+        return Response.sse(gen())
+
+    req = Request(method="POST", path="/v1/chat/completions",
+                  headers={}, query={}, body=b"{}")
+    resp = await mw(req, nxt)
+    # the middleware returned but the stream has not run: span open
+    assert [x.name for x in tracer.drain()] == []
+    chunks = [c async for c in resp.stream]
+    assert len(chunks) == 2
+    spans = tracer.drain()
+    assert [x.name for x in spans] == ["POST /v1/chat/completions"]
+    assert spans[0].kind == "server"
