@@ -113,6 +113,11 @@ class Telemetry:
         g_pfx = Gauge("igw_engine_prefix_hit_tokens_total",
                       "prompt tokens served from the prefix cache",
                       registry=r)
+        g_steps = Gauge("igw_engine_steps_total",
+                        "cumulative engine steps", registry=r)
+        g_mixed = Gauge("igw_engine_mixed_rounds_total",
+                        "cumulative mixed prefill+decode rounds",
+                        registry=r)
         sched = engine.scheduler
         bm = engine.block_manager
         g_run.set_function(lambda: len(sched.running))
@@ -124,6 +129,8 @@ class Telemetry:
         g_preempt.set_function(lambda: sched.num_preemptions)
         g_forks.set_function(lambda: engine.num_forks)
         g_pfx.set_function(lambda: bm.prefix_hit_tokens)
+        g_steps.set_function(lambda: engine.num_steps)
+        g_mixed.set_function(lambda: sched.num_mixed_rounds)
 
     def exposition(self) -> bytes:
         return generate_latest(self.registry)
