@@ -181,6 +181,8 @@ async def test_engine_gauges_in_metrics_exposition():
     assert "igw_engine_running_seqs 0.0" in text
     assert "igw_engine_generated_tokens_total 0.0" in text
     assert "igw_engine_preemptions_total 0.0" in text
+    assert "igw_engine_steps_total" in text
+    assert "igw_engine_mixed_rounds_total" in text
     provider.engine.shutdown()
 
 
