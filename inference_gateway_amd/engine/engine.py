@@ -156,6 +156,7 @@ class LLMEngine:
         self.num_prompt_tokens = 0
         self.num_forks = 0
         self.num_steps = 0        # scheduled engine steps (bench windows)
+        self._no_progress = 0     # consecutive unschedulable rounds
         # +1 block of headroom: the async decode pipeline can write KV for
         # one in-flight token past max_model_len before the stop check runs
         self.max_blocks_per_seq = (
@@ -269,7 +270,24 @@ class LLMEngine:
         async-scheduling trade."""
         batch = self.scheduler.schedule()
         if batch is None:
+            # nothing schedulable.  If requests are WAITING with nothing
+            # running, no future step can free blocks for them: a prompt
+            # (or a preempted sequence's folded prompt) larger than the
+            # whole pool would otherwise busy-spin here silently forever.
+            # Count consecutive no-progress rounds and fail loudly (the
+            # async engine surfaces this as EngineDeadError to clients).
+            if self.scheduler.waiting and not self.scheduler.running:
+                self._no_progress += 1
+                if self._no_progress > 4096:
+                    seq = self.scheduler.waiting[0]
+                    raise RuntimeError(
+                        f"scheduler made no progress for "
+                        f"{self._no_progress} rounds: seq "
+                        f"{seq.seq_id} needs {seq.prompt_len} prompt "
+                        "tokens but the KV pool cannot ever hold them "
+                        "— raise num_gpu_blocks or lower max_model_len")
             return self._drain_pending()
+        self._no_progress = 0
         self.num_steps += 1
         gr = self.graph_runner
         if not batch.is_prefill and gr is not None:

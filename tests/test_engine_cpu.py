@@ -1348,3 +1348,16 @@ def test_step_watchdog_fails_streams_on_wedged_step(monkeypatch):
     finally:
         release.set()
         eng.shutdown()
+
+
+def test_unschedulable_prompt_fails_loudly():
+    """A prompt that can NEVER fit the KV pool must raise (engine death
+    -> EngineDeadError at the gateway), not busy-spin silently."""
+    # pool: 8 blocks * 16 = 128 tokens usable minus reserved block;
+    # prompt of 120 tokens with blocks for it unavailable
+    eng = _engine(num_gpu_blocks=6, max_model_len=96)
+    eng.add_request(list(range(1, 90)),
+                    SamplingParams(max_new_tokens=2, temperature=0))
+    with pytest.raises(RuntimeError, match="no progress"):
+        for _ in range(10000):
+            eng.step()
