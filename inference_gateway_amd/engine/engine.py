@@ -207,6 +207,15 @@ class LLMEngine:
                     seq_id: Optional[int] = None) -> int:
         if seq_id is None:
             seq_id = next(self._seq_counter)
+        bs = self.cfg.block_size
+        need = (len(prompt_token_ids) + bs - 1) // bs + 1
+        if need > self.block_manager.num_blocks - 1:
+            # never-schedulable: reject per-request instead of letting
+            # the scheduler discover it as a no-progress spin
+            raise ValueError(
+                f"prompt needs {need} KV blocks but the pool has only "
+                f"{self.block_manager.num_blocks - 1} — raise "
+                "num_gpu_blocks / gpu_memory_utilization")
         seq = Sequence(
             seq_id=seq_id,
             prompt_token_ids=list(prompt_token_ids),

@@ -1350,14 +1350,29 @@ def test_step_watchdog_fails_streams_on_wedged_step(monkeypatch):
         eng.shutdown()
 
 
-def test_unschedulable_prompt_fails_loudly():
-    """A prompt that can NEVER fit the KV pool must raise (engine death
-    -> EngineDeadError at the gateway), not busy-spin silently."""
-    # pool: 8 blocks * 16 = 128 tokens usable minus reserved block;
-    # prompt of 120 tokens with blocks for it unavailable
+def test_unschedulable_prompt_rejected_at_admission():
+    """A prompt that can NEVER fit the KV pool is rejected per-request
+    at add time (the engine survives; other requests keep serving)."""
     eng = _engine(num_gpu_blocks=6, max_model_len=96)
-    eng.add_request(list(range(1, 90)),
-                    SamplingParams(max_new_tokens=2, temperature=0))
+    with pytest.raises(ValueError, match="KV blocks"):
+        eng.add_request(list(range(1, 90)),
+                        SamplingParams(max_new_tokens=2, temperature=0))
+    # the engine still serves a request that fits
+    rid = eng.add_request(list(range(1, 17)),
+                          SamplingParams(max_new_tokens=2, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(rid).finished
+
+
+def test_generation_outgrowing_pool_fails_loudly():
+    """The pathological fold case: a sequence whose PREEMPTION-FOLDED
+    prompt can no longer fit must raise (no-progress detection), not
+    busy-spin silently.  Requires pool < max_model_len — a
+    misconfiguration the startup path cannot always rule out."""
+    eng = _engine(num_gpu_blocks=6, max_model_len=96)
+    eng.add_request(list(range(1, 41)),
+                    SamplingParams(max_new_tokens=50, temperature=0))
     with pytest.raises(RuntimeError, match="no progress"):
         for _ in range(10000):
             eng.step()
