@@ -15,7 +15,7 @@ from hypothesis import strategies as st
 
 from inference_gateway_amd.gateway.validate import validate_chat_request
 
-_fast = settings(max_examples=120, deadline=None,
+_fast = settings(max_examples=120, deadline=None, derandomize=True,
                  suppress_health_check=[HealthCheck.too_slow])
 
 
