@@ -91,14 +91,17 @@ MODEL_PRESETS = {
         norm_eps=1e-6,
         qkv_bias=True,
     ),
-    "phi-3-mini": ModelConfig(
-        name="phi-3-mini",
-        hidden_size=3072,
-        num_layers=32,
-        num_heads=32,
-        num_kv_heads=32,
-        head_dim=96,
-        intermediate_size=8192,
+    # phi-3-MEDIUM, not mini: the MFMA attention kernels are tiled for
+    # head_dim=128 (mini's 96 would fail the kernels' TORCH_CHECK on
+    # GPU; engine init rejects non-128 head_dim on cuda devices)
+    "phi-3-medium": ModelConfig(
+        name="phi-3-medium",
+        hidden_size=5120,
+        num_layers=40,
+        num_heads=40,
+        num_kv_heads=10,
+        head_dim=128,
+        intermediate_size=17920,
         vocab_size=32064,
         rope_base=10000.0,
         max_position=4096,

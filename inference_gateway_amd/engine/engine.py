@@ -124,6 +124,11 @@ class LLMEngine:
             device if torch.cuda.is_available() else "cpu")
         _maybe_load_tuned_gemms()
         self.dtype = torch.bfloat16
+        if self.device.type == "cuda" and cfg.model.head_dim != 128:
+            raise ValueError(
+                f"model head_dim {cfg.model.head_dim} is not supported "
+                "on GPU: the MFMA attention kernels are tiled for "
+                "head_dim=128 (CPU reference path has no such limit)")
         if cfg.model.is_moe:
             from .models.mixtral import MixtralForCausalLM
 
