@@ -27,6 +27,9 @@ class ModelConfig:
     # Qwen2 family: bias on the fused qkv projection only (HF Qwen2
     # hardcodes q/k/v bias True, o/mlp bias False)
     qkv_bias: bool = False
+    # HF rope_scaling dict (Llama-3.1 "llama3" wavelength-banded
+    # rescale, or "linear"); None = plain RoPE
+    rope_scaling: Optional[dict] = None
     # MoE (Mixtral): 0 experts => dense MLP
     num_experts: int = 0
     num_experts_per_tok: int = 2
@@ -63,6 +66,21 @@ MODEL_PRESETS = {
         intermediate_size=28672,
         vocab_size=128256,
         rope_base=500000.0,
+    ),
+    "llama-3.1-8b": ModelConfig(
+        name="llama-3.1-8b",
+        hidden_size=4096,
+        num_layers=32,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=14336,
+        vocab_size=128256,
+        rope_base=500000.0,
+        max_position=131072,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
     ),
     "mixtral-8x7b": ModelConfig(
         name="mixtral-8x7b",

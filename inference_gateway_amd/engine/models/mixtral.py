@@ -375,7 +375,8 @@ class MixtralForCausalLM(nn.Module):
         self.lm_head = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size,
                                         device, dtype, _seed("lm_head"))
         self.cos_sin = ops.build_cos_sin_cache(
-            cfg.head_dim, cfg.max_position, cfg.rope_base, device=device)
+            cfg.head_dim, cfg.max_position, cfg.rope_base, device=device,
+            rope_scaling=cfg.rope_scaling)
 
     def forward(self, fb: ForwardBatch,
                 kv_caches: List[Tuple[torch.Tensor, torch.Tensor]]

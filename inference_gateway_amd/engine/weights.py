@@ -48,9 +48,14 @@ def model_config_from_hf(path: str, name: str = "hf-model"
     # field; Llama-family configs carry attention_bias (default False)
     qkv_bias = bool(cfg.get("attention_bias", mtype in ("qwen2",
                                                         "qwen2_moe")))
+    rs = cfg.get("rope_scaling")
+    if isinstance(rs, dict) and (rs.get("rope_type") or rs.get("type")) \
+            in ("default", None):
+        rs = None
     return ModelConfig(
         name=mtype,
         qkv_bias=qkv_bias,
+        rope_scaling=rs if isinstance(rs, dict) else None,
         hidden_size=int(cfg["hidden_size"]),
         num_layers=int(cfg["num_hidden_layers"]),
         num_heads=int(cfg["num_attention_heads"]),

@@ -50,13 +50,49 @@ def rope_ref(
 
 
 def build_cos_sin_cache(
-    head_dim: int, max_pos: int, base: float = 10000.0, device="cpu"
+    head_dim: int, max_pos: int, base: float = 10000.0, device="cpu",
+    rope_scaling: dict = None,
 ) -> torch.Tensor:
-    """[max_pos, HD] fp32 — cos in [0, HD/2), sin in [HD/2, HD)."""
+    """[max_pos, HD] fp32 — cos in [0, HD/2), sin in [HD/2, HD).
+
+    rope_scaling: HF `rope_scaling` dict.  Supported: the Llama-3.1
+    "llama3" scheme (wavelength-banded frequency rescale: short
+    wavelengths untouched, long wavelengths divided by `factor`, a
+    smooth ramp between — the published long-context extension) and
+    "linear" (all frequencies divided by `factor`).  None/unknown-less
+    configs keep plain RoPE; an unsupported rope_type raises (silent
+    wrong positions would corrupt every long-context generation)."""
     half = head_dim // 2
     inv_freq = 1.0 / (
         base ** (torch.arange(0, half, dtype=torch.float64, device=device) / half)
     )
+    if rope_scaling:
+        rtype = (rope_scaling.get("rope_type")
+                 or rope_scaling.get("type") or "llama3")
+        factor = float(rope_scaling.get("factor", 8.0))
+        if rtype == "llama3":
+            lo = float(rope_scaling.get("low_freq_factor", 1.0))
+            hi = float(rope_scaling.get("high_freq_factor", 4.0))
+            orig = float(rope_scaling.get(
+                "original_max_position_embeddings", 8192))
+            import math
+
+            wavelen = 2 * math.pi / inv_freq
+            low_wl = orig / lo            # longest untouched band edge
+            high_wl = orig / hi           # shortest scaled band edge
+            smooth = (orig / wavelen - lo) / (hi - lo)
+            scaled = torch.where(
+                wavelen > low_wl, inv_freq / factor, inv_freq)
+            ramp = ((1 - smooth) * inv_freq / factor
+                    + smooth * inv_freq)
+            inv_freq = torch.where(
+                (wavelen <= low_wl) & (wavelen >= high_wl), ramp, scaled)
+        elif rtype == "linear":
+            inv_freq = inv_freq / factor
+        else:
+            raise ValueError(
+                f"unsupported rope_scaling type {rtype!r} (supported: "
+                "llama3, linear)")
     t = torch.arange(max_pos, dtype=torch.float64, device=device)
     freqs = torch.outer(t, inv_freq)
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1).float()

@@ -483,3 +483,84 @@ def test_chat_template_from_checkpoint(tmp_path):
         assert ids == want
     finally:
         provider.engine.shutdown()
+
+
+def test_llama31_rope_scaling():
+    """Llama-3.1 "llama3" rope scaling: short wavelengths untouched,
+    wavelengths beyond original_max/low_freq_factor divided by
+    `factor`, smooth ramp between — verified against the published
+    piecewise formula computed independently here."""
+    import math
+
+    from inference_gateway_amd import ops
+
+    hd, base = 128, 500000.0
+    rs = {"rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+          "high_freq_factor": 4.0,
+          "original_max_position_embeddings": 8192}
+    plain = ops.build_cos_sin_cache(hd, 64, base)
+    scaled = ops.build_cos_sin_cache(hd, 64, base, rope_scaling=rs)
+    half = hd // 2
+
+    def freq_at(cache, i):
+        # angle(t=1) = f_i, read via atan2(sin, cos) — fp32-relative
+        # precision even for tiny frequencies (acos of cos~1 is not)
+        return math.atan2(cache[1, half + i].item(),
+                          cache[1, i].item())
+
+    for i in range(half):
+        f = 1.0 / (base ** (i / half))
+        wl = 2 * math.pi / f
+        want_f = None
+        if wl < 8192 / 4.0:                  # high-freq: untouched
+            want_f = f
+        elif wl > 8192 / 1.0:                # low-freq: / factor
+            want_f = f / 8.0
+        if want_f is not None:
+            got = freq_at(scaled, i)
+            assert abs(got - want_f) < 1e-6 + 1e-4 * want_f, (
+                i, got, want_f)
+    # the ramp band lies strictly between the two regimes
+    mid = [i for i in range(half)
+           if 8192 / 4.0 <= 2 * math.pi / (1.0 / (base ** (i / half)))
+           <= 8192]
+    assert mid, "test setup: ramp band must be non-empty"
+    for i in mid:
+        f = 1.0 / (base ** (i / half))
+        got = freq_at(scaled, i)
+        assert f / 8.0 - 1e-9 <= got <= f + 1e-9
+    # linear scaling divides everything
+    lin = ops.build_cos_sin_cache(hd, 64, base,
+                                  rope_scaling={"type": "linear",
+                                                "factor": 2.0})
+    assert abs(freq_at(lin, 0) - freq_at(plain, 0) / 2.0) < 1e-6
+    # unknown scheme fails loudly
+    with pytest.raises(ValueError, match="yarn"):
+        ops.build_cos_sin_cache(hd, 64, base,
+                                rope_scaling={"rope_type": "yarn",
+                                              "factor": 4.0})
+
+
+def test_rope_scaling_hf_ingestion_and_generation(tmp_path):
+    _tiny_llama_checkpoint(tmp_path)
+    cfgj = json.loads((tmp_path / "config.json").read_text())
+    cfgj["rope_scaling"] = {"rope_type": "llama3", "factor": 8.0,
+                            "low_freq_factor": 1.0,
+                            "high_freq_factor": 4.0,
+                            "original_max_position_embeddings": 1024}
+    (tmp_path / "config.json").write_text(json.dumps(cfgj))
+    cfg = W.model_config_from_hf(str(tmp_path))
+    assert cfg.rope_scaling and cfg.rope_scaling["factor"] == 8.0
+    eng = LLMEngine(EngineConfig(model=cfg, max_num_seqs=2,
+                                 max_model_len=128, num_gpu_blocks=32,
+                                 weights_path=str(tmp_path)),
+                    device="cpu")
+    rid = eng.add_request(list(range(1, 17)),
+                          SamplingParams(max_new_tokens=3, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    assert len(eng.get_output(rid).token_ids) == 3
+    # "default"/absent rope_type is treated as no scaling
+    cfgj["rope_scaling"] = {"rope_type": "default"}
+    (tmp_path / "config.json").write_text(json.dumps(cfgj))
+    assert W.model_config_from_hf(str(tmp_path)).rope_scaling is None
