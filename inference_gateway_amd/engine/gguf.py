@@ -233,6 +233,14 @@ def model_config_from_gguf(path: str) -> ModelConfig:
     n_heads = int(m("attention.head_count"))
     hidden = int(m("embedding_length"))
     tied = "output.weight" not in g.infos
+    # llama.cpp linear rope scaling metadata; llama-3.1-style GGUFs
+    # instead ship a precomputed rope_freqs tensor (not mapped here —
+    # use the HF safetensors form for 3.1 checkpoints)
+    rs = None
+    st = md.get(f"{a}.rope.scaling.type")
+    if st == "linear":
+        rs = {"type": "linear",
+              "factor": float(md.get(f"{a}.rope.scaling.factor", 1.0))}
     return ModelConfig(
         name=str(md.get("general.name") or a),
         hidden_size=hidden,
@@ -248,6 +256,7 @@ def model_config_from_gguf(path: str) -> ModelConfig:
         norm_eps=float(md.get(
             f"{a}.attention.layer_norm_rms_epsilon", 1e-5)),
         tie_embeddings=tied,
+        rope_scaling=rs,
         qkv_bias=("blk.0.attn_q.bias" in g.infos),
         num_experts=int(md.get(f"{a}.expert_count", 0) or 0),
         num_experts_per_tok=int(md.get(f"{a}.expert_used_count", 2) or 2),

@@ -230,8 +230,12 @@ def test_unsupported_quant_fails_loudly(tmp_path):
 def test_config_from_gguf(tmp_path):
     m = MODEL_PRESETS["tiny-llama"]
     t = _tiny_weights(m)
-    write_gguf(tmp_path / "m.gguf", _meta_for(m), _gguf_tensors(m, t))
+    meta = _meta_for(m) + [
+        _kv_str("llama.rope.scaling.type", "linear"),
+        _kv_f32("llama.rope.scaling.factor", 2.0)]
+    write_gguf(tmp_path / "m.gguf", meta, _gguf_tensors(m, t))
     cfg = G.model_config_from_gguf(str(tmp_path / "m.gguf"))
+    assert cfg.rope_scaling == {"type": "linear", "factor": 2.0}
     assert cfg.hidden_size == m.hidden_size
     assert cfg.num_layers == m.num_layers
     assert cfg.num_kv_heads == m.num_kv_heads
