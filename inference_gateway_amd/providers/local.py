@@ -290,6 +290,13 @@ class LocalProvider:
 
         self.chat_template = (load_chat_template(weights) if weights
                               else None)
+        # authoritative stop ids from the checkpoint: token-NAME
+        # heuristics miss chat-tuned EOS tokens (Qwen2's <|im_end|>,
+        # Llama-3-Instruct's <|eot_id|>) — generation_config.json /
+        # config.json carry the real eos_token_id (int or list)
+        self.eos_ids = self._checkpoint_eos_ids(weights)
+        if not self.eos_ids and self.tokenizer.eos_token_id is not None:
+            self.eos_ids = (self.tokenizer.eos_token_id,)
         # OpenAI system_fingerprint: identifies the backend configuration
         # (determinism domain for `seed`)
         self.fingerprint = f"fp_igw-amd_{local_cfg.model}"
@@ -297,6 +304,34 @@ class LocalProvider:
         # after the chunk's socket write completes) — bench.py's serving
         # mode measures the wire rate from this, not the engine rate
         self.streamed_tokens = 0
+
+    @staticmethod
+    def _checkpoint_eos_ids(weights) -> tuple:
+        """eos_token_id(s) from generation_config.json (preferred — the
+        chat EOS lives there) or config.json; () when absent."""
+        if not weights:
+            return ()
+        import json as _json
+        from pathlib import Path
+
+        d = Path(weights)
+        d = d if d.is_dir() else d.parent
+        ids: list = []
+        for fname in ("generation_config.json", "config.json"):
+            f = d / fname
+            if not f.is_file():
+                continue
+            try:
+                v = _json.loads(f.read_text()).get("eos_token_id")
+            except ValueError:
+                continue
+            if isinstance(v, int):
+                ids = [v]
+            elif isinstance(v, list):
+                ids = [int(x) for x in v if isinstance(x, int)]
+            if ids:
+                break
+        return tuple(ids)
 
     # ---- provider interface --------------------------------------------
     async def list_models(self) -> dict:
@@ -318,9 +353,7 @@ class LocalProvider:
     def _sampling_from_request(self, req: dict) -> SamplingParams:
         max_tokens = req.get("max_completion_tokens") or req.get(
             "max_tokens") or 256
-        stop_ids = ()
-        if self.tokenizer.eos_token_id is not None:
-            stop_ids = (self.tokenizer.eos_token_id,)
+        stop_ids = tuple(self.eos_ids)
         # OpenAI logprobs: bool `logprobs` + optional `top_logprobs` 0..20
         lp = -1
         if req.get("logprobs"):

@@ -564,3 +564,37 @@ def test_rope_scaling_hf_ingestion_and_generation(tmp_path):
     cfgj["rope_scaling"] = {"rope_type": "default"}
     (tmp_path / "config.json").write_text(json.dumps(cfgj))
     assert W.model_config_from_hf(str(tmp_path)).rope_scaling is None
+
+
+def test_checkpoint_eos_ids_from_generation_config(tmp_path):
+    """generation_config.json eos_token_id (list form — Llama-3-
+    Instruct/Qwen2 style) drives stop detection; forced via logit_bias
+    so the sampled token IS the configured eos."""
+    import asyncio
+
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.local import LocalProvider
+
+    _tiny_llama_checkpoint(tmp_path)
+    (tmp_path / "generation_config.json").write_text(
+        json.dumps({"eos_token_id": [5, 9]}))
+    lc = LocalConfig(enabled=True, model="hf", max_num_seqs=4,
+                     max_model_len=128)
+    lc.num_gpu_blocks = 64
+    lc.weights_path = str(tmp_path)
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        assert provider.eos_ids == (5, 9)
+
+        async def go():
+            return await provider.chat_completions({
+                "model": "local/hf",
+                "messages": [{"role": "user", "content": "hi"}],
+                "max_tokens": 30, "temperature": 0,
+                "logit_bias": {"9": 100}})
+
+        body = asyncio.run(go())
+        assert body["usage"]["completion_tokens"] == 1
+        assert body["choices"][0]["finish_reason"] == "stop"
+    finally:
+        provider.engine.shutdown()
