@@ -60,9 +60,9 @@ def accumulate_streaming_tool_calls(lines: List[str]) -> List[dict]:
             if fn.get("arguments"):
                 tc["function"]["arguments"] += fn["arguments"]
     out = []
-    for i in range(len(acc)):
-        tc = acc.get(i)
-        if tc and tc["function"]["name"]:
+    for i in sorted(acc):        # sparse indices must not drop calls
+        tc = acc[i]
+        if tc["function"]["name"]:
             out.append(tc)
     return out
 

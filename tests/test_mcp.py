@@ -344,3 +344,20 @@ async def test_reconnect_recovers_dead_server():
     names = [t["name"] for t in mcp.client.all_tools()]
     assert "late_tool" in names
     await mcp.stop()
+
+
+def test_accumulate_tolerates_sparse_indices():
+    """A provider that starts tool-call indices at 1 (or skips one)
+    must not lose calls."""
+    from inference_gateway_amd.mcp.agent import (
+        accumulate_streaming_tool_calls)
+
+    lines = [
+        'data: {"choices":[{"delta":{"tool_calls":[{"index":1,'
+        '"id":"b","function":{"name":"fs_read","arguments":"{}"}}]}}]}',
+        'data: {"choices":[{"delta":{"tool_calls":[{"index":3,'
+        '"id":"d","function":{"name":"time_now","arguments":""}}]}}]}',
+    ]
+    out = accumulate_streaming_tool_calls(lines)
+    assert [tc["function"]["name"] for tc in out] == ["fs_read",
+                                                      "time_now"]
