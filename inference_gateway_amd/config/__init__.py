@@ -157,6 +157,7 @@ class LocalConfig:
     chunked_prefill: bool = False
     prefill_first: bool = True        # strict admission priority
     mixed_batches: bool = False       # same-step prefill+decode rounds
+    max_queue: int = 0                # waiting-queue cap (0 = unbounded)
     weights_path: str = ""            # HF safetensors dir; "" = synthetic
     tokenizer_path: str = ""          # tokenizers json; "" = byte fallback
 
@@ -291,6 +292,7 @@ def load(env: Optional[Mapping[str, str]] = None) -> Config:
     lo.chunked_prefill = _bool(g("LOCAL_CHUNKED_PREFILL", "false"))
     lo.prefill_first = _bool(g("LOCAL_PREFILL_FIRST", "true"))
     lo.mixed_batches = _bool(g("LOCAL_MIXED_BATCHES", "false"))
+    lo.max_queue = int(g("LOCAL_MAX_QUEUE", "0"))
     lo.weights_path = g("LOCAL_WEIGHTS_PATH", "")
     lo.tokenizer_path = g("LOCAL_TOKENIZER_PATH", "")
 

@@ -304,6 +304,7 @@ class LocalProvider:
         # after the chunk's socket write completes) — bench.py's serving
         # mode measures the wire rate from this, not the engine rate
         self.streamed_tokens = 0
+        self.max_queue = int(getattr(local_cfg, "max_queue", 0) or 0)
 
     @staticmethod
     def _checkpoint_eos_ids(weights) -> tuple:
@@ -469,7 +470,19 @@ class LocalProvider:
         """Pre-stream validation hook (gateway/routes.py calls this
         before opening the SSE response)."""
         self._resolve_n(req)
+        self._check_queue()
         self._check_context(self._encode_request(req), req)
+
+    def _check_queue(self) -> None:
+        """Admission cap (LOCAL_MAX_QUEUE): shed load with 429 instead
+        of queueing unboundedly — waiting-queue depth is the TTFT a new
+        request will pay.  0 = unbounded (the bench's burst-hold mode
+        deliberately queues thousands)."""
+        cap = self.max_queue
+        if cap and len(self.engine.engine.scheduler.waiting) >= cap:
+            raise HTTPError(
+                429, f"server overloaded: {cap} requests already "
+                "queued for prefill — retry later")
 
     def _check_context(self, prompt_ids, req: Optional[dict] = None) -> None:
         limit = self.engine.engine.cfg.max_model_len
@@ -492,6 +505,7 @@ class LocalProvider:
     async def chat_completions(self, req: dict,
                                extra_headers=None) -> dict:
         prompt_ids = self._encode_request(req)
+        self._check_queue()
         self._check_context(prompt_ids, req)
         sampling = self._sampling_from_request(req)
         stops = self._stop_strings(req)

@@ -536,3 +536,46 @@ async def test_local_real_tokenizer_end_to_end(tmp_path):
                 r.json()["choices"][0]["message"]["content"], str)
     finally:
         provider.engine.shutdown()
+
+
+def test_max_queue_sheds_load_with_429():
+    """LOCAL_MAX_QUEUE: requests beyond the waiting-queue cap get 429
+    (load shedding), and the cap releases as the queue drains."""
+    import asyncio
+
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.core import HTTPError
+    from inference_gateway_amd.providers.local import LocalProvider
+
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=2,
+                     max_model_len=128)
+    lc.num_gpu_blocks = 64
+    lc.max_queue = 2
+    provider = LocalProvider(lc, device="cpu")
+    try:
+        eng = provider.engine
+        eng.hold()                     # requests queue, engine paused
+        async def go():
+            reqs = [asyncio.create_task(provider.chat_completions({
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": f"r{i}"}],
+                "max_tokens": 2, "temperature": 0})) for i in range(2)]
+            await asyncio.sleep(0.2)   # both now in waiting
+            with pytest.raises(HTTPError) as ei:
+                await provider.chat_completions({
+                    "model": "local/tiny-llama",
+                    "messages": [{"role": "user", "content": "over"}],
+                    "max_tokens": 2})
+            assert ei.value.status == 429
+            eng.release()              # drain; the cap must release too
+            out = await asyncio.gather(*reqs)
+            assert all(o["usage"]["completion_tokens"] == 2 for o in out)
+            ok = await provider.chat_completions({
+                "model": "local/tiny-llama",
+                "messages": [{"role": "user", "content": "after"}],
+                "max_tokens": 2, "temperature": 0})
+            assert ok["usage"]["completion_tokens"] == 2
+
+        asyncio.run(go())
+    finally:
+        provider.engine.shutdown()
