@@ -105,10 +105,17 @@ class MetricsIngestion:
                                   attrs.get("model", ""),
                                   attrs.get("source", ""),
                                   attrs.get("team", ""))
-                        counts = [int(c) for c in
-                                  dp.get("bucketCounts") or []]
-                        bounds = [float(b) for b in
-                                  dp.get("explicitBounds") or []]
+                        try:
+                            counts = [int(c) for c in
+                                      dp.get("bucketCounts") or []]
+                            bounds = [float(b) for b in
+                                      dp.get("explicitBounds") or []]
+                        except (TypeError, ValueError):
+                            rejected += 1     # malformed datapoint, not
+                            continue          # a 500
+                        if any(c < 0 for c in counts):
+                            rejected += 1
+                            continue
                         n = 0
                         for i, c in enumerate(counts):
                             if n >= MAX_OBSERVATIONS:

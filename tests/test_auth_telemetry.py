@@ -338,3 +338,21 @@ async def test_otlp_protobuf_ingestion():
     assert "gen_ai_client_operation_duration_seconds_count" in text
     # 1 obs in bucket 0 (mid 0.05) + 2 in bucket 1 (mid 0.55), twice
     assert 'provider="groq"' in text
+
+
+@pytest.mark.asyncio
+async def test_otlp_malformed_datapoints_rejected_not_500():
+    """Garbage bucketCounts/explicitBounds in a pushed histogram are
+    counted as rejected, never a 500."""
+    gw, _ = make_telemetry_gateway()
+    payload = {"resourceMetrics": [{"scopeMetrics": [{"metrics": [{
+        "name": "gen_ai.client.operation.duration",
+        "histogram": {"aggregationTemporality": 2, "dataPoints": [
+            {"bucketCounts": ["abc"], "explicitBounds": [0.1]},
+            {"bucketCounts": [-3], "explicitBounds": [0.1]},
+            {"bucketCounts": [1], "explicitBounds": ["x"]},
+        ]}}]}]}]}
+    async with gw_client(gw) as c:
+        r = await c.post("/v1/metrics", json=payload)
+        assert r.status_code == 200
+        assert r.json()["rejected"] >= 2
