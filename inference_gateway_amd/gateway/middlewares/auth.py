@@ -89,6 +89,10 @@ class OIDCAuthenticator:
                 return None
         elif alg == "HS256":
             secret = self.cfg.oidc_client_secret.encode()
+            if not secret:
+                # public client (no secret): HS256 would be verifiable
+                # with an EMPTY key — an attacker-forgeable token
+                return None
             expected = hmac.new(secret, signing_input,
                                 hashlib.sha256).digest()
             if not hmac.compare_digest(expected, sig):
@@ -96,13 +100,23 @@ class OIDCAuthenticator:
         else:
             return None
         now = time.time()
-        if "exp" in claims and now > float(claims["exp"]):
+        # exp is REQUIRED (go-oidc semantics — a token that never
+        # expires is rejected, not accepted forever)
+        if "exp" not in claims or now > float(claims["exp"]):
             return None
         if "nbf" in claims and now < float(claims["nbf"]):
             return None
         iss = claims.get("iss", "").rstrip("/")
         if iss and iss != self.issuer:
             return None
+        # audience check (go-oidc Config.ClientID): aud must name this
+        # client (string or array form)
+        cid = self.cfg.oidc_client_id
+        if cid:
+            aud = claims.get("aud")
+            auds = aud if isinstance(aud, list) else [aud]
+            if cid not in auds:
+                return None
         return claims
 
     def middleware(self):

@@ -132,21 +132,31 @@ async def test_auth_middleware_flow():
         r = await c.post("/v1/chat/completions", json=body,
                          headers={"Authorization": "Bearer garbage"})
         assert r.status_code == 401
-        good = make_hs256({"sub": "u1", "iss": ISSUER,
+        AUD = "inference-gateway-client"     # config default client id
+        good = make_hs256({"sub": "u1", "iss": ISSUER, "aud": AUD,
                            "exp": time.time() + 60}, "shh")
         r = await c.post("/v1/chat/completions", json=body,
                          headers={"Authorization": f"Bearer {good}"})
         assert r.status_code == 200
-        rs = make_rs256({"sub": "u2", "iss": ISSUER,
+        rs = make_rs256({"sub": "u2", "iss": ISSUER, "aud": [AUD, "x"],
                          "exp": time.time() + 60})
         r = await c.post("/v1/chat/completions", json=body,
                          headers={"Authorization": f"Bearer {rs}"})
         assert r.status_code == 200
-        expired = make_hs256({"sub": "u1", "iss": ISSUER,
+        expired = make_hs256({"sub": "u1", "iss": ISSUER, "aud": AUD,
                               "exp": time.time() - 10}, "shh")
         r = await c.post("/v1/chat/completions", json=body,
                          headers={"Authorization": f"Bearer {expired}"})
         assert r.status_code == 401
+        # go-oidc parity hardening: wrong audience, missing exp -> 401
+        for bad in (make_hs256({"sub": "u", "iss": ISSUER,
+                                "aud": "other-client",
+                                "exp": time.time() + 60}, "shh"),
+                    make_hs256({"sub": "u", "iss": ISSUER,
+                                "aud": AUD}, "shh")):
+            r = await c.post("/v1/chat/completions", json=body,
+                             headers={"Authorization": f"Bearer {bad}"})
+            assert r.status_code == 401
         # health stays open
         r = await c.get("/health")
         assert r.status_code == 200
@@ -356,3 +366,27 @@ async def test_otlp_malformed_datapoints_rejected_not_500():
         r = await c.post("/v1/metrics", json=payload)
         assert r.status_code == 200
         assert r.json()["rejected"] >= 2
+
+
+@pytest.mark.asyncio
+async def test_hs256_rejected_for_public_clients():
+    """A PUBLIC client (no client secret) must reject HS256 outright:
+    otherwise any attacker can forge tokens HMAC'd with the empty key."""
+    gw, _ = make_auth_gateway(secret="")
+    await gw.authenticator.initialize()
+    forged = make_hs256({"sub": "evil", "iss": ISSUER,
+                         "aud": "inference-gateway-client",
+                         "exp": time.time() + 3600}, "")
+    body = {"model": "groq/m", "messages": [{"role": "user",
+                                             "content": "x"}]}
+    async with gw_client(gw) as c:
+        r = await c.post("/v1/chat/completions", json=body,
+                         headers={"Authorization": f"Bearer {forged}"})
+        assert r.status_code == 401
+        # RS256 still works for the public client
+        rs = make_rs256({"sub": "u", "iss": ISSUER,
+                         "aud": "inference-gateway-client",
+                         "exp": time.time() + 60})
+        r = await c.post("/v1/chat/completions", json=body,
+                         headers={"Authorization": f"Bearer {rs}"})
+        assert r.status_code == 200
