@@ -371,6 +371,12 @@ class Router:
         if pid == LOCAL_PROVIDER_ID:
             return Response.error(400, "local provider has no proxy path")
         endpoint = "/" + rest if rest else "/"
+        if req.query:
+            # the caller's query string travels with the proxied path
+            # (reference ProxyHandler forwards the full RequestURI)
+            from urllib.parse import urlencode
+
+            endpoint += "?" + urlencode(req.query)
         headers = {}
         ct = req.headers.get("content-type")
         if ct:

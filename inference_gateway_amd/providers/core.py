@@ -76,7 +76,7 @@ class RemoteProvider:
     async def list_models(self) -> dict:
         url = self.url + self.meta.models_endpoint
         r = await self.client.get(url, headers=self.auth_headers(),
-                                  params=self.auth_params())
+                                  params=self.auth_params() or None)
         if r.status_code >= 400:
             raise HTTPError(r.status_code, r.text[:2048])
         return transform_list_models(self.id, r.json())
@@ -88,7 +88,7 @@ class RemoteProvider:
         if extra_headers:
             headers.update(extra_headers)
         r = await self.client.post(url, json=req, headers=headers,
-                                   params=self.auth_params())
+                                   params=self.auth_params() or None)
         if r.status_code >= 400:
             raise HTTPError(r.status_code, r.text[:2048])
         return r.json()
@@ -103,7 +103,7 @@ class RemoteProvider:
         if extra_headers:
             headers.update(extra_headers)
         async with self.client.stream("POST", url, json=req, headers=headers,
-                                      params=self.auth_params()) as r:
+                                      params=self.auth_params() or None) as r:
             if r.status_code >= 400:
                 body = await r.aread()
                 raise HTTPError(r.status_code,
@@ -121,7 +121,7 @@ class RemoteProvider:
             if k.lower() in ("content-type", "accept"):
                 h[k] = v
         return await self.client.request(method, url, content=body, headers=h,
-                                         params=self.auth_params())
+                                         params=self.auth_params() or None)
 
     def stream_raw(self, method: str, endpoint: str, body: bytes,
                    headers: Dict[str, str]):
@@ -131,7 +131,7 @@ class RemoteProvider:
             if k.lower() in ("content-type", "accept"):
                 h[k] = v
         return self.client.stream(method, url, content=body, headers=h,
-                                  params=self.auth_params())
+                                  params=self.auth_params() or None)
 
 
 class ProviderRegistry:

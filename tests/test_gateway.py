@@ -188,6 +188,18 @@ async def test_proxy_handler_strips_caller_auth():
 
 
 @pytest.mark.asyncio
+async def test_proxy_forwards_query_string():
+    gw, calls = make_gateway()
+    async with gw_client(gw) as c:
+        r = await c.get("/proxy/groq/models", params={"limit": "5",
+                                                      "after": "x"})
+        assert r.status_code == 200
+    hit = [x for x in calls if "models" in x.url.path][-1]
+    assert hit.url.params.get("limit") == "5"
+    assert hit.url.params.get("after") == "x"
+
+
+@pytest.mark.asyncio
 async def test_messages_passthrough_requires_anthropic():
     gw, calls = make_gateway()
     async with gw_client(gw) as c:
