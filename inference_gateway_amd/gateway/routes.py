@@ -27,6 +27,27 @@ from ..providers.core import HTTPError
 from .http import Request, Response
 
 
+
+async def _relay_sse_events(line_iter):
+    """Group upstream SSE lines into WHOLE events before re-framing.
+
+    The SSE writer appends the blank separator per yielded chunk, so
+    yielding line-by-line would split multi-line frames — Anthropic's
+    `event:` + `data:` pairs would become two frames and a
+    spec-compliant EventSource client would see the data under the
+    default event type instead of the named one."""
+    buf = []
+    async for line in line_iter:
+        if line == "":
+            if buf:
+                yield "\n".join(buf)
+                buf = []
+        else:
+            buf.append(line)
+    if buf:
+        yield "\n".join(buf)
+
+
 class Router:
     def __init__(self, cfg, registry, logger, selector=None, mcp=None,
                  telemetry=None):
@@ -278,9 +299,9 @@ class Router:
                             errors="replace")
                         yield "data: " + payload
                         return
-                    async for line in r.aiter_lines():
-                        if line:
-                            yield line
+                    async for ev in _relay_sse_events(
+                            r.aiter_lines()):
+                        yield ev
             return Response.sse(gen())
         try:
             r = await provider.forward_raw("POST", endpoint, raw, headers)
@@ -388,9 +409,9 @@ class Router:
                 async def gen():
                     async with provider.stream_raw(req.method, endpoint,
                                                    req.body, headers) as r:
-                        async for line in r.aiter_lines():
-                            if line:
-                                yield line
+                        async for ev in _relay_sse_events(
+                                r.aiter_lines()):
+                            yield ev
                 return Response.sse(gen())
             r = await provider.forward_raw(req.method, endpoint, req.body,
                                            headers)

@@ -438,16 +438,16 @@ async def test_messages_passthrough_sse():
         body = json.loads(request.content)
         assert body["model"] == "claude-3-5-sonnet"  # prefix stripped
         if body.get("stream"):
-            frames = ["event: message_start",
+            frames = ["event: message_start\n"
                       'data: {"type":"message_start"}',
-                      "event: content_block_delta",
+                      "event: content_block_delta\n"
                       'data: {"type":"content_block_delta",'
                       '"delta":{"type":"text_delta","text":"hi"}}',
-                      "event: message_stop",
+                      "event: message_stop\n"
                       'data: {"type":"message_stop"}']
             return httpx.Response(
                 200, headers={"content-type": "text/event-stream"},
-                content="\n".join(frames).encode())
+                content=("\n\n".join(frames) + "\n\n").encode())
         return httpx.Response(200, json={"type": "message"})
 
     cfg = config_mod.load({"ANTHROPIC_API_KEY": "k"})
@@ -464,6 +464,7 @@ async def test_messages_passthrough_sse():
                     lines.append(line)
     assert any("message_stop" in l for l in lines)
     assert any("text_delta" in l for l in lines)
+
 
 
 @pytest.mark.asyncio
@@ -516,3 +517,27 @@ async def test_chat_request_shape_validation():
             "messages": [{"role": "user", "content": [
                 {"type": "text", "text": "hi"}]}]})
         assert r.status_code == 200
+
+
+@pytest.mark.asyncio
+async def test_relay_sse_event_grouping():
+    """Multi-line SSE frames (Anthropic event:+data:) survive the relay
+    as ONE frame each — the writer adds one blank separator per yielded
+    chunk, so the relay must group lines by event boundary."""
+    from inference_gateway_amd.gateway.routes import _relay_sse_events
+
+    async def upstream_lines():
+        for line in ["event: message_start",
+                     'data: {"a":1}',
+                     "",
+                     "event: delta",
+                     'data: {"b":2}',
+                     "",
+                     "data: solo",
+                     ""]:
+            yield line
+
+    evs = [e async for e in _relay_sse_events(upstream_lines())]
+    assert evs == ['event: message_start\ndata: {"a":1}',
+                   'event: delta\ndata: {"b":2}',
+                   "data: solo"]

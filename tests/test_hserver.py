@@ -725,6 +725,29 @@ async def test_messages_passthrough_sse_over_socket():
         assert "event: message_start" in lines
         assert any("text_delta" in ln for ln in lines)
         assert "event: message_stop" in lines
+        # raw framing: read the same stream as BYTES on a second
+        # request — each event: line must share its frame with its
+        # data: line (no blank line between them)
+        host, port = "127.0.0.1", srv.port
+        reader, writer = await asyncio.open_connection(host, port)
+        body = json.dumps({"model": "anthropic/claude-x",
+                           "stream": True, "max_tokens": 8,
+                           "messages": [{"role": "user",
+                                         "content": "x"}]}).encode()
+        writer.write(b"POST /v1/messages HTTP/1.1\r\nhost: x\r\n"
+                     b"content-type: application/json\r\n"
+                     b"content-length: %d\r\n\r\n" % len(body) + body)
+        await writer.drain()
+        raw = b""
+        while b"message_stop" not in raw:
+            chunk = await asyncio.wait_for(reader.read(65536), 30)
+            if not chunk:
+                break
+            raw += chunk
+        writer.close()
+        text = raw.decode(errors="replace")
+        assert ("event: content_block_delta\ndata:" in text
+                or "event: content_block_delta\r\ndata:" in text), text
     finally:
         await srv.stop()
         await client.aclose()
