@@ -177,3 +177,52 @@ def test_q8_dequant_total(raw):
     n = (len(raw) // 34) * 32
     out = _dequant(np.frombuffer(raw, dtype=np.uint8), GGML_Q8_0, n)
     assert out.shape[0] == n
+
+
+# ---- config duration parser: total, non-negative ----------------------
+@_fast
+@given(st.text(max_size=16))
+def test_parse_duration_total(sraw):
+    from inference_gateway_amd.config import parse_duration
+
+    try:
+        v = parse_duration(sraw)
+    except ValueError:
+        return
+    assert isinstance(v, float)
+
+
+@_fast
+@given(st.integers(min_value=0, max_value=10**6),
+       st.sampled_from(["ms", "s", "m", "h"]))
+def test_parse_duration_units(n, unit):
+    from inference_gateway_amd.config import parse_duration
+
+    mult = {"ms": 0.001, "s": 1, "m": 60, "h": 3600}[unit]
+    assert abs(parse_duration(f"{n}{unit}") - n * mult) < 1e-6 * max(
+        1, n * mult)
+
+
+# ---- model routing: never raises, prefix split is consistent ----------
+@_fast
+@given(st.text(max_size=40))
+def test_model_routing_total(model):
+    from inference_gateway_amd.providers.routing import (
+        determine_provider_and_model)
+
+    pid, m = determine_provider_and_model(model)
+    if pid is None:
+        assert m == model
+    else:
+        assert model.lower().startswith(pid + "/")
+        assert model.endswith(m)
+
+
+# ---- MCP tool-name normalization idempotent ---------------------------
+@_fast
+@given(st.text(max_size=24))
+def test_tool_name_normalization_idempotent(name):
+    from inference_gateway_amd.mcp.client import normalize_tool_name
+
+    once = normalize_tool_name(name)
+    assert normalize_tool_name(once) == once
