@@ -1376,3 +1376,82 @@ def test_generation_outgrowing_pool_fails_loudly():
     with pytest.raises(RuntimeError, match="no progress"):
         for _ in range(10000):
             eng.step()
+
+
+def test_differential_plain_vs_full_stack_tie_aware():
+    """Randomized differential: staggered greedy request sets through
+    the PLAIN engine and the FULL feature stack (chunked + prefix +
+    mixed + prefill-first) must match token-for-token — except where
+    the two top logits sit within a couple of bf16 ULPs (mixed rounds
+    change reduction shapes; a one-quantum argmax tie may flip and is
+    NOT a scheduling/attention defect).  Any larger divergence fails."""
+    import random
+
+    def run(cfgkw, requests):
+        torch.manual_seed(0)
+        eng = LLMEngine(EngineConfig.tiny_llama(
+            num_gpu_blocks=128, max_model_len=192, max_num_seqs=6,
+            **cfgkw), device="cpu")
+        rids = []
+        added = 0
+        while True:
+            for _ in range(2):
+                if added < len(requests):
+                    pr, mnt = requests[added]
+                    rids.append(eng.add_request(pr, SamplingParams(
+                        max_new_tokens=mnt, temperature=0)))
+                    added += 1
+            if eng.is_finished() and added == len(requests):
+                break
+            eng.step()
+        while not eng.is_finished():
+            eng.step()
+        return [eng.get_output(r).token_ids for r in rids]
+
+    def logits_for(prompt):
+        torch.manual_seed(0)
+        eng = LLMEngine(EngineConfig.tiny_llama(
+            num_gpu_blocks=128, max_model_len=192, max_num_seqs=6),
+            device="cpu")
+        cap = {}
+        orig = eng.sampler.sample
+
+        def spy(lg, seqs):
+            cap["lg"] = lg.clone()
+            return orig(lg, seqs)
+
+        eng.sampler.sample = spy
+        eng.add_request(prompt, SamplingParams(max_new_tokens=1,
+                                               temperature=0))
+        while not eng.is_finished():
+            eng.step()
+        return cap["lg"][0]
+
+    rng = random.Random(99)
+    shared = list(range(300, 348))
+    for trial in range(8):
+        reqs = []
+        for _ in range(rng.randrange(3, 9)):
+            stem = shared[: rng.choice([0, 16, 32])]
+            pr = stem + [rng.randrange(1, 500)
+                         for _ in range(rng.randrange(4, 40))]
+            reqs.append((pr, rng.randrange(1, 12)))
+        plain = run({}, reqs)
+        full = run(dict(enable_chunked_prefill=True,
+                        enable_prefix_caching=True,
+                        enable_mixed_batches=True,
+                        scheduler_prefill_first=True,
+                        max_batched_tokens=48), reqs)
+        for i, (a, b) in enumerate(zip(plain, full)):
+            if a == b:
+                continue
+            # first divergence must be a near-tie
+            j = next(k for k in range(min(len(a), len(b)))
+                     if a[k] != b[k])
+            prompt = list(reqs[i][0]) + a[:j]
+            lg = logits_for(prompt)
+            gap = abs((lg[a[j]] - lg[b[j]]).item())
+            scale = max(abs(lg[a[j]].item()), 1.0)
+            assert gap <= 0.0078125 * 2 * scale, (
+                f"trial {trial} seq {i}: real divergence "
+                f"{a} vs {b} (logit gap {gap})")
