@@ -133,3 +133,30 @@ def test_nonpositive_body_cap_falls_back_to_default():
         assert cfg.server.max_request_body_size == 10 * 1024 * 1024
     cfg = config_mod.load({"SERVER_MAX_REQUEST_BODY_SIZE": "1234"})
     assert cfg.server.max_request_body_size == 1234
+
+
+def test_validate_checkpoint_cli(tmp_path, capsys):
+    """tools/validate_checkpoint.py dry-runs both formats."""
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path("tools").resolve()))
+    import validate_checkpoint as V
+
+    from test_weights import _tiny_llama_checkpoint
+    from test_gguf import (MODEL_PRESETS, _gguf_tensors, _meta_for,
+                           _tiny_weights, write_gguf)
+
+    hf = tmp_path / "hf"
+    hf.mkdir()
+    _tiny_llama_checkpoint(hf)
+    assert V.main([str(hf)]) == 0
+    out = capsys.readouterr().out
+    assert "VALID" in out and "hf safetensors" in out
+
+    m = MODEL_PRESETS["tiny-llama"]
+    gg = tmp_path / "m.gguf"
+    write_gguf(gg, _meta_for(m), _gguf_tensors(m, _tiny_weights(m)))
+    assert V.main([str(gg)]) == 0
+    out = capsys.readouterr().out
+    assert "VALID" in out and "gguf" in out
