@@ -1012,3 +1012,56 @@ async def test_chaos_soak_disconnects_and_garbage(provider):
                 break
             await asyncio.sleep(0.05)
         assert not eng.scheduler.running and not eng.scheduler.waiting
+
+
+def test_worker_main_process_serves():
+    """SERVER_WORKERS worker entry (_worker_main) in a real spawned
+    process: it loads config from env, binds with SO_REUSEPORT and
+    serves /health."""
+    import multiprocessing as mp
+    import os
+    import socket
+    import time
+    import urllib.request
+
+    # reserve a port: bind with SO_REUSEPORT so the worker can share it
+    probe = socket.socket()
+    probe.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEPORT, 1)
+    probe.bind(("127.0.0.1", 0))
+    port = probe.getsockname()[1]
+    probe.close()
+
+    env = dict(os.environ, SERVER_HOST="127.0.0.1",
+               SERVER_PORT=str(port), TELEMETRY_ENABLED="false")
+    ctx = mp.get_context("spawn")
+    proc = ctx.Process(target=_spawn_worker_entry, args=(env,),
+                       daemon=True)
+    proc.start()
+    try:
+        deadline = time.time() + 20
+        last = None
+        while time.time() < deadline:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/health",
+                        timeout=2) as r:
+                    assert r.status == 200
+                    assert b"healthy" in r.read()
+                    break
+            except Exception as e:
+                last = e
+                time.sleep(0.25)
+        else:
+            raise AssertionError(f"worker never served: {last}")
+    finally:
+        proc.terminate()
+        proc.join(timeout=10)
+
+
+def _spawn_worker_entry(env):
+    import os
+
+    os.environ.update(env)
+    from inference_gateway_amd.server import _worker_main
+
+    _worker_main(1)
