@@ -598,3 +598,36 @@ def test_checkpoint_eos_ids_from_generation_config(tmp_path):
         assert body["choices"][0]["finish_reason"] == "stop"
     finally:
         provider.engine.shutdown()
+
+
+def test_multi_shard_safetensors_checkpoint(tmp_path):
+    """Real 8B/70B checkpoints ship model-0000N-of-M.safetensors; the
+    shard index must resolve tensors across files and generate
+    identically to the single-file layout."""
+    t = _tiny_llama_checkpoint(tmp_path)
+    single = LLMEngine(EngineConfig.tiny_llama(weights_path=str(tmp_path)),
+                       device="cpu")
+    rid = single.add_request(list(range(1, 25)),
+                             SamplingParams(max_new_tokens=5,
+                                            temperature=0))
+    while not single.is_finished():
+        single.step()
+    want = single.get_output(rid).token_ids
+
+    sharded = tmp_path / "sharded"
+    sharded.mkdir()
+    names = sorted(t)
+    half = len(names) // 2
+    save_file({k: t[k] for k in names[:half]},
+              str(sharded / "model-00001-of-00002.safetensors"))
+    save_file({k: t[k] for k in names[half:]},
+              str(sharded / "model-00002-of-00002.safetensors"))
+    (sharded / "config.json").write_text(
+        (tmp_path / "config.json").read_text())
+    eng = LLMEngine(EngineConfig.tiny_llama(weights_path=str(sharded)),
+                    device="cpu")
+    rid = eng.add_request(list(range(1, 25)),
+                          SamplingParams(max_new_tokens=5, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    assert eng.get_output(rid).token_ids == want
