@@ -387,8 +387,13 @@ class HttpProtocol(asyncio.Protocol):
                 for k, v in hdrs:
                     head.append(k + b": " + v + b"\r\n")
                 if not has_len:
-                    self._chunked = True
-                    head.append(b"transfer-encoding: chunked\r\n")
+                    if scope["http_version"] == "1.0":
+                        # HTTP/1.0 has no chunked TE: stream raw bytes
+                        # and delimit the body by closing (RFC 1945)
+                        self._keep_alive = False
+                    else:
+                        self._chunked = True
+                        head.append(b"transfer-encoding: chunked\r\n")
                 if not self._keep_alive:
                     head.append(b"connection: close\r\n")
                 head.append(b"\r\n")

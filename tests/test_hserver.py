@@ -1065,3 +1065,34 @@ def _spawn_worker_entry(env):
     from inference_gateway_amd.server import _worker_main
 
     _worker_main(1)
+
+
+@pytest.mark.asyncio
+async def test_http10_sse_streams_raw_and_closes(provider):
+    """HTTP/1.0 has no chunked transfer-encoding: an SSE response to a
+    1.0 client streams raw bytes and ends by connection close."""
+    async with serve(provider) as base:
+        host, port = base[len("http://"):].split(":")
+        reader, writer = await asyncio.open_connection(host, int(port))
+        body = json.dumps({"model": "local/tiny-llama", "stream": True,
+                           "max_tokens": 4, "temperature": 0,
+                           "messages": [{"role": "user",
+                                         "content": "x"}]}).encode()
+        writer.write(b"POST /v1/chat/completions HTTP/1.0\r\nhost: x\r\n"
+                     b"content-type: application/json\r\n"
+                     b"content-length: %d\r\n\r\n" % len(body) + body)
+        await writer.drain()
+        raw = b""
+        while True:
+            chunk = await asyncio.wait_for(reader.read(65536), 30)
+            if not chunk:
+                break                     # server closed = end of body
+            raw += chunk
+        writer.close()
+        head, _, payload = raw.partition(b"\r\n\r\n")
+        assert b"200" in head.split(b"\r\n", 1)[0]
+        assert b"transfer-encoding" not in head.lower()
+        assert b"connection: close" in head.lower()
+        # raw SSE bytes, no chunk-size hex lines
+        assert payload.lstrip().startswith(b"data: ")
+        assert b"data: [DONE]" in payload
