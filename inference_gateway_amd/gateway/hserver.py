@@ -363,6 +363,7 @@ class HttpProtocol(asyncio.Protocol):
         self._pending_head = b""
         sent_body = False
         received = False
+        head_only = scope["method"] == "HEAD"   # RFC 9110: no body
 
         async def receive():
             nonlocal received
@@ -405,6 +406,8 @@ class HttpProtocol(asyncio.Protocol):
                 return
             chunk = message.get("body", b"")
             more = message.get("more_body", False)
+            if head_only:
+                chunk = b""
             if self._chunked:
                 parts = []
                 if chunk:

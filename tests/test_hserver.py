@@ -1096,3 +1096,26 @@ async def test_http10_sse_streams_raw_and_closes(provider):
         # raw SSE bytes, no chunk-size hex lines
         assert payload.lstrip().startswith(b"data: ")
         assert b"data: [DONE]" in payload
+
+
+@pytest.mark.asyncio
+async def test_head_request_returns_no_body(provider):
+    async with serve(provider) as base:
+        host, port = base[len("http://"):].split(":")
+        reader, writer = await asyncio.open_connection(host, int(port))
+        writer.write(b"HEAD /health HTTP/1.1\r\nhost: x\r\n"
+                     b"connection: close\r\n\r\n")
+        await writer.drain()
+        raw = b""
+        while True:
+            chunk = await asyncio.wait_for(reader.read(4096), 10)
+            if not chunk:
+                break
+            raw += chunk
+        writer.close()
+        head, _, payload = raw.partition(b"\r\n\r\n")
+        # parity: routes register GET only, so HEAD is 404 (gin does
+        # the same) — but the 404 BODY must still be suppressed
+        assert head.split(b"\r\n", 1)[0].endswith(b"404 Not Found")
+        assert b"healthy" not in payload
+        assert b"not found" not in payload.lower()
