@@ -274,6 +274,12 @@ class HttpProtocol(asyncio.Protocol):
                 need = n
             elif k == b"transfer-encoding":
                 chunked_req = b"chunked" in v.lower()
+            elif k == b"expect":
+                if v.lower() == b"100-continue":
+                    # RFC 9110 §10.1.1: tell the client to send the
+                    # body now instead of waiting out its timeout
+                    # (Go's net/http does this automatically)
+                    self.transport.write(b"HTTP/1.1 100 Continue\r\n\r\n")
             elif k == b"connection":
                 lv = v.lower()
                 if lv == b"close":

@@ -1119,3 +1119,34 @@ async def test_head_request_returns_no_body(provider):
         assert head.split(b"\r\n", 1)[0].endswith(b"404 Not Found")
         assert b"healthy" not in payload
         assert b"not found" not in payload.lower()
+
+
+@pytest.mark.asyncio
+async def test_expect_100_continue(provider):
+    """Clients sending Expect: 100-continue get the interim 100 before
+    the body is transmitted (curl does this for large bodies)."""
+    async with serve(provider) as base:
+        host, port = base[len("http://"):].split(":")
+        reader, writer = await asyncio.open_connection(host, int(port))
+        body = json.dumps({"model": "local/tiny-llama", "max_tokens": 2,
+                           "temperature": 0,
+                           "messages": [{"role": "user",
+                                         "content": "x"}]}).encode()
+        writer.write(b"POST /v1/chat/completions HTTP/1.1\r\nhost: x\r\n"
+                     b"content-type: application/json\r\n"
+                     b"expect: 100-continue\r\n"
+                     b"content-length: %d\r\n\r\n" % len(body))
+        await writer.drain()
+        interim = await asyncio.wait_for(reader.readline(), 10)
+        assert interim.startswith(b"HTTP/1.1 100")
+        await asyncio.wait_for(reader.readline(), 10)   # blank line
+        writer.write(body)
+        await writer.drain()
+        data = b""
+        while b"completion_tokens" not in data:
+            chunk = await asyncio.wait_for(reader.read(65536), 30)
+            if not chunk:
+                break
+            data += chunk
+        writer.close()
+        assert b"200" in data.split(b"\r\n", 1)[0]
