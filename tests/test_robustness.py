@@ -106,6 +106,9 @@ def test_head_parser_never_crashes(raw):
         def get_extra_info(self, k):
             return None
 
+        def write(self, data):      # 100-continue interim writes
+            pass
+
     p.transport = _T()
     try:
         p._parse_head(raw)
