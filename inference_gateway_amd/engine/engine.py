@@ -155,6 +155,9 @@ class LLMEngine:
         self.kv_caches = self._allocate_kv(num_blocks)
         self._seq_counter = itertools.count()
         self._seqs: Dict[int, Sequence] = {}
+        from collections import deque
+
+        self._finished_order: "deque[int]" = deque()
         self._fork_pending: Dict[int, List[Sequence]] = {}
         # cumulative serving counters (Prometheus gauges read these)
         self.num_generated_tokens = 0
@@ -260,6 +263,22 @@ class LLMEngine:
             for child in self._fork_pending.pop(seq_id, []):
                 if child.state == SeqState.WAITING:
                     self.scheduler.finish(child, "abort")
+        self._note_finished(seq_id)
+
+    _FINISHED_RETENTION = 4096
+
+    def _note_finished(self, seq_id: int) -> None:
+        """Bounded retention of finished Sequence bookkeeping: without
+        eviction a long-running server retains every prompt/output
+        token list ever served.  get_output stays valid for the last
+        _FINISHED_RETENTION finished requests (sync callers read right
+        after finishing; the async path never calls get_output)."""
+        self._finished_order.append(seq_id)
+        while len(self._finished_order) > self._FINISHED_RETENTION:
+            old = self._finished_order.popleft()
+            seq = self._seqs.get(old)
+            if seq is not None and seq.state == SeqState.FINISHED:
+                del self._seqs[old]
 
     def is_finished(self) -> bool:
         return not self.scheduler.has_work and self._pending is None
@@ -586,6 +605,7 @@ class LLMEngine:
             reason = "length"
         if reason:
             self.scheduler.finish(seq, reason)
+            self._note_finished(seq.seq_id)
         out = StepOutput(seq.seq_id, tok, reason is not None, reason)
         if lp is not None:
             out.logprob, out.top_logprobs = lp

@@ -1455,3 +1455,24 @@ def test_differential_plain_vs_full_stack_tie_aware():
             assert gap <= 0.0078125 * 2 * scale, (
                 f"trial {trial} seq {i}: real divergence "
                 f"{a} vs {b} (logit gap {gap})")
+
+
+def test_finished_requests_are_evicted_from_bookkeeping(monkeypatch):
+    """Serving-lifetime memory: finished Sequence bookkeeping is
+    retained for only the last _FINISHED_RETENTION requests — a
+    long-running server must not hold every prompt ever served.
+    get_output stays valid within the retention window (sync API)."""
+    eng = _engine()
+    monkeypatch.setattr(type(eng), "_FINISHED_RETENTION", 8)
+    rids = []
+    for i in range(20):
+        rid = eng.add_request(list(range(1, 9)),
+                              SamplingParams(max_new_tokens=2,
+                                             temperature=0))
+        while not eng.is_finished():
+            eng.step()
+        assert eng.get_output(rid).finished    # valid right after
+        rids.append(rid)
+    assert len(eng._seqs) <= 8
+    assert rids[0] not in eng._seqs            # oldest evicted
+    assert rids[-1] in eng._seqs               # recent retained
