@@ -215,6 +215,44 @@ def _gguf_tp_worker(rank: int, world: int):
     return eng.get_output(rid).token_ids
 
 
+def _qwen2_ckpt_tp_worker(rank: int, world: int):
+    """TP-sharded load + generate for the QKV-BIAS architecture
+    (tiny-qwen2, tied embeddings): bias slices must shard with the
+    head dims (engine/weights.py)."""
+    import tempfile
+    from pathlib import Path
+
+    from inference_gateway_amd.engine.config import (MODEL_PRESETS,
+                                                     EngineConfig,
+                                                     SamplingParams)
+    from inference_gateway_amd.engine.engine import LLMEngine
+    from inference_gateway_amd.parallel import comm
+    from test_weights import _tiny_qwen2_checkpoint
+
+    comm.init_tp(world)
+    d = Path(tempfile.gettempdir()) / "igw_qwen2_tp_test"
+    if world > 1:
+        import torch.distributed as dist
+
+        if rank == 0:
+            d.mkdir(exist_ok=True)
+            _tiny_qwen2_checkpoint(d, seed=13)
+        dist.barrier()
+    else:
+        d.mkdir(exist_ok=True)
+        _tiny_qwen2_checkpoint(d, seed=13)
+    eng = LLMEngine(
+        EngineConfig(model=MODEL_PRESETS["tiny-qwen2"], max_num_seqs=4,
+                     max_model_len=128, num_gpu_blocks=64,
+                     weights_path=str(d)),
+        device="cpu")
+    rid = eng.add_request(list(range(1, 25)),
+                          SamplingParams(max_new_tokens=6, temperature=0))
+    while not eng.is_finished():
+        eng.step()
+    return eng.get_output(rid).token_ids
+
+
 # ---- tests ---------------------------------------------------------------
 
 def _single_rank_reference(fn):
@@ -284,6 +322,12 @@ def test_gguf_load_tp2_matches_tp1(tmp_path):
     assert outs[0] == outs[1], "TP ranks must sample identical tokens"
     ref = _single_rank_reference(_gguf_tp_worker)
     assert outs[0] == ref, "TP=2 GGUF load must match TP=1"
+
+
+def test_qwen2_checkpoint_load_tp2_matches_tp1(tmp_path):
+    outs = _spawn("_qwen2_ckpt_tp_worker", tmp_path)
+    assert outs[0] == outs[1]
+    assert outs[0] == _single_rank_reference(_qwen2_ckpt_tp_worker)
 
 
 def test_bench_contract_torchrun_world2(tmp_path):
