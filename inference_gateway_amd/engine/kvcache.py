@@ -11,10 +11,15 @@ content hash of their token prefix.  A new prompt whose leading full
 blocks hash-match either a LIVE block (shared, ref++) or a RETAINED one
 (freed by its last owner but kept in an LRU pool instead of reclaimed)
 skips recomputing that prefix — the model extend-prefills only the
-suffix through `ops.flash_prefill_paged`.  Correctness hinges on two
+suffix through `ops.flash_prefill_paged`.  Correctness hinges on three
 invariants: decode only appends PAST a sequence's prompt, so shared full
-blocks are never written again; and the shared prefix must stay
-block-ALIGNED, so the suffix's KV lands in private blocks only.
+blocks are never written again; the shared prefix must stay
+block-ALIGNED, so the suffix's KV lands in private blocks only; and only
+the FRONT of the waiting queue may be mid-chunked-prefill (scheduler.py
+takes partial chunks from waiting[0] only), so a later prompt can never
+hash-match blocks whose KV a still-chunking sequence has not written
+yet.  (Same-round matches are safe: per layer, append_kv for the whole
+batch runs before attention reads the cache.)
 """
 from __future__ import annotations
 
