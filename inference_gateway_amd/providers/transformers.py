@@ -12,6 +12,12 @@ from typing import List
 
 def transform_list_models(provider_id: str, payload: dict) -> dict:
     data = payload.get("data") or []
+    if not data and isinstance(payload.get("models"), list):
+        # Cohere's native /v1/models shape ({"models": [{"name": ...}]})
+        # — the reference's generated transformer reads only data/id and
+        # returns an empty list there; tolerate the native shape instead
+        data = [{**m, "id": m.get("id") or m.get("name", "")}
+                for m in payload["models"] if isinstance(m, dict)]
     models: List[dict] = []
     for m in data:
         m = dict(m)
