@@ -82,6 +82,21 @@ MODEL_PRESETS = {
                       "low_freq_factor": 1.0, "high_freq_factor": 4.0,
                       "original_max_position_embeddings": 8192},
     ),
+    "llama-3.1-70b": ModelConfig(
+        name="llama-3.1-70b",
+        hidden_size=8192,
+        num_layers=80,
+        num_heads=64,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=28672,
+        vocab_size=128256,
+        rope_base=500000.0,
+        max_position=131072,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+    ),
     "mixtral-8x7b": ModelConfig(
         name="mixtral-8x7b",
         hidden_size=4096,
