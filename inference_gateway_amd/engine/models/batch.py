@@ -1,4 +1,5 @@
-"""Forward-batch metadata shared by all models."""
+"""Forward-batch metadata shared by all models (net-new engine scope,
+SURVEY.md §7 stage 3 — the reference gateway has no compute path)."""
 from __future__ import annotations
 
 from dataclasses import dataclass

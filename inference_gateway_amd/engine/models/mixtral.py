@@ -1,4 +1,5 @@
-"""Mixtral-8x7B MoE decoder for the MI355X engine (BASELINE config 5).
+"""Mixtral-8x7B MoE decoder for the MI355X engine (BASELINE config 5;
+net-new engine scope, SURVEY.md §2.3 expert parallelism).
 
 Same attention stack as Llama (hand-written gfx950 kernels); the MLP is a
 top-2 router + 8 SwiGLU experts. Expert parallelism strategy on the 8-GPU

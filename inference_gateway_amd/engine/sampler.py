@@ -1,4 +1,7 @@
-"""Batched sampling front-end over the fused gfx950 sampling kernel."""
+"""Batched sampling front-end over the fused gfx950 sampling kernel
+(net-new engine scope, SURVEY.md §7 stage 3e; OpenAI sampling-parameter
+semantics — seed/penalties/logit_bias — implemented for real here where
+the reference only forwards them upstream)."""
 from __future__ import annotations
 
 import random as _random

@@ -1,4 +1,5 @@
-"""Tokenizers for the local engine provider.
+"""Tokenizers for the local engine provider (net-new engine scope,
+SURVEY.md §7 — the reference proxies text and never tokenizes).
 
 Two implementations behind one interface:
 
