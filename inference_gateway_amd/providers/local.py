@@ -124,12 +124,13 @@ class DirectStream:
             text = self.tokenizer.decode(self.pending)
             if not text.endswith("�"):
                 nflush = len(self.pending)
+                if self.stops:
+                    i, buf = self.provider._scan_stop_progressive(
+                        self.emitted_tail, self.pending, self.stops)
                 self.pending = []
                 lp = self.pending_lp if self.want_lp else None
                 self.pending_lp = []
                 if self.stops:
-                    buf = self.emitted_tail + text
-                    i = self.provider._find_stop(buf, self.stops)
                     if i >= 0:
                         if buf[:i] and not self._emit_content(buf[:i], lp):
                             return
@@ -152,19 +153,16 @@ class DirectStream:
 
     def _end_stream(self) -> None:
         # final leftover: held-back stop tail + incomplete-UTF-8 pending
-        tail_text = (self.tokenizer.decode(self.pending) if self.pending
-                     else "")
         nflush = len(self.pending)
+        i, leftover = self.provider._scan_stop_progressive(
+            self.emitted_tail, self.pending, self.stops)
         self.pending = []
         lp = (self.pending_lp if self.want_lp and self.pending_lp
               else None)
         self.pending_lp = []
-        leftover = self.emitted_tail + tail_text
-        if self.stops and leftover:
-            i = self.provider._find_stop(leftover, self.stops)
-            if i >= 0:
-                leftover = leftover[:i]
-                self.finish_reason = "stop"
+        if i >= 0:
+            leftover = leftover[:i]
+            self.finish_reason = "stop"
         if leftover and not self._emit_content(leftover, lp):
             return
         self.provider.streamed_tokens += nflush
@@ -420,6 +418,29 @@ class LocalProvider:
             return [s for s in stop if isinstance(s, str)][:4]
         return []
 
+    def _scan_stop_progressive(self, tail: str, pending: list,
+                               stops) -> tuple:
+        """(cut_index, buffer) — stop scan with CANONICAL per-token
+        semantics for a BATCHED delivery: grow the text token by token
+        (exactly as the non-streaming _collect does) and return the
+        earliest-start match inside the FIRST prefix that contains any
+        stop.  A single whole-batch scan could cut at a globally
+        EARLIER index whose match only completes with later tokens —
+        tokens the per-token path never generates (it stops at the
+        first completed stop) — so batching must not change the text.
+        cut_index -1 => no stop; buffer is then tail + decode(all)."""
+        if stops:
+            i = self._find_stop(tail, stops)
+            if i >= 0:
+                return i, tail
+            for k in range(1, len(pending) + 1):
+                cand = tail + self.tokenizer.decode(pending[:k])
+                i = self._find_stop(cand, stops)
+                if i >= 0:
+                    return i, cand
+        return -1, tail + (self.tokenizer.decode(pending) if pending
+                           else "")
+
     @staticmethod
     def _find_stop(text: str, stops) -> int:
         """Earliest stop-string match index, or -1."""
@@ -641,14 +662,15 @@ class LocalProvider:
                                      else "length")
             text = self.tokenizer.decode(pending)
             # hold back partial UTF-8 sequences (byte tokenizer)
-            if text.endswith("\ufffd"):
+            if text.endswith("�"):
                 continue
             nflush = len(pending)
+            if stops:
+                i, buf = self._scan_stop_progressive(emitted_tail,
+                                                     pending, stops)
             pending = []
             lp, pending_lp = (pending_lp, []) if want_lp else (None, [])
             if stops:
-                buf = emitted_tail + text
-                i = self._find_stop(buf, stops)
                 if i >= 0:
                     # emit up to (excluding) the stop sequence, then end
                     if buf[:i]:
@@ -672,14 +694,12 @@ class LocalProvider:
             # any pending tokens that never completed a UTF-8 sequence
             # (emit them with replacement chars — dropping tokens would
             # leave a stream with content missing entirely)
-            tail_text = self.tokenizer.decode(pending) if pending else ""
-            leftover = emitted_tail + tail_text
+            i, leftover = self._scan_stop_progressive(emitted_tail,
+                                                       pending, stops)
             lp = pending_lp if want_lp and pending_lp else None
-            if stops and leftover:
-                i = self._find_stop(leftover, stops)
-                if i >= 0:
-                    leftover = leftover[:i]
-                    finish_reason = "stop"
+            if i >= 0:
+                leftover = leftover[:i]
+                finish_reason = "stop"
             if leftover:
                 yield chunk({"content": leftover}, lp=lp)
         usage = {
@@ -734,12 +754,13 @@ class LocalProvider:
                     text = self.tokenizer.decode(pending)
                     if text.endswith("�"):
                         continue
+                    if stops:
+                        i, buf = self._scan_stop_progressive(
+                            emitted_tail, pending, stops)
                     pending = []
                     lp = pending_lp if want_lp else None
                     pending_lp = []
                     if stops:
-                        buf = emitted_tail + text
-                        i = self._find_stop(buf, stops)
                         if i >= 0:
                             if buf[:i]:
                                 await queue.put(
@@ -755,14 +776,11 @@ class LocalProvider:
                         continue
                     await queue.put(chunk(idx, {"content": text}, lp=lp))
                 # flush the held-back tail + any incomplete-UTF-8 pending
-                tail_text = (self.tokenizer.decode(pending) if pending
-                             else "")
-                leftover = emitted_tail + tail_text
-                if stops and leftover:
-                    i = self._find_stop(leftover, stops)
-                    if i >= 0:
-                        leftover = leftover[:i]
-                        finish_reason = "stop"
+                i, leftover = self._scan_stop_progressive(
+                    emitted_tail, pending, stops)
+                if i >= 0:
+                    leftover = leftover[:i]
+                    finish_reason = "stop"
                 if leftover:
                     await queue.put(chunk(
                         idx, {"content": leftover},

@@ -915,13 +915,19 @@ def test_direct_stream_holdback_equivalence_fuzz(provider):
         return content
 
     def reference(tokens, stops):
-        full = tok.decode(list(tokens))
-        best = -1
-        for st_ in stops:
-            k = full.find(st_)
-            if k >= 0 and (best < 0 or k < best):
-                best = k
-        return full[:best] if best >= 0 else full
+        # CANONICAL per-token stop semantics (matches the non-streaming
+        # _collect): grow token by token, cut at the earliest-start
+        # match inside the FIRST prefix containing any stop
+        for k in range(1, len(tokens) + 1):
+            partial = tok.decode(list(tokens[:k]))
+            best = -1
+            for st_ in stops:
+                j = partial.find(st_)
+                if j >= 0 and (best < 0 or j < best):
+                    best = j
+            if best >= 0:
+                return partial[:best]
+        return tok.decode(list(tokens))
 
     rng = random.Random(7)
     for _ in range(300):
