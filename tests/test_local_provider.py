@@ -579,3 +579,52 @@ def test_max_queue_sheds_load_with_429():
         asyncio.run(go())
     finally:
         provider.engine.shutdown()
+
+
+def test_n_gt_1_stream_equals_nonstream_with_stops():
+    """Per-choice streamed content and finish_reason equal the
+    non-streaming response exactly — same seed, stops on, n in {2,3}
+    (cross-path consistency incl. the progressive stop scan)."""
+    import asyncio
+    import json
+    import random
+
+    from inference_gateway_amd.config import LocalConfig
+    from inference_gateway_amd.providers.local import LocalProvider
+
+    lc = LocalConfig(enabled=True, model="tiny-llama", max_num_seqs=8,
+                     max_model_len=128)
+    lc.num_gpu_blocks = 96
+    provider = LocalProvider(lc, device="cpu")
+
+    async def one(trial, rng):
+        req = {"model": "local/tiny-llama", "n": rng.choice([2, 3]),
+               "max_tokens": rng.randrange(3, 12), "temperature": 0.9,
+               "seed": trial,
+               "messages": [{"role": "user",
+                             "content": f"fuzz {trial}"}]}
+        if rng.random() < 0.7:
+            req["stop"] = [rng.choice(["<tok", ">", "ab", "<t"])
+                           for _ in range(rng.randrange(1, 3))]
+        ns = {c["index"]: (c["message"]["content"], c["finish_reason"])
+              for c in (await provider.chat_completions(
+                  dict(req)))["choices"]}
+        st = {}
+        async for line in provider.stream_chat_completions(dict(req)):
+            if not line.startswith("data: {"):
+                continue
+            ch = json.loads(line[6:])["choices"][0]
+            txt, fin = st.get(ch["index"], ("", None))
+            st[ch["index"]] = (txt + (ch["delta"].get("content") or ""),
+                               ch["finish_reason"] or fin)
+        assert st == ns, (trial, st, ns, req.get("stop"))
+
+    async def main():
+        rng = random.Random(77)
+        for trial in range(20):
+            await one(trial, rng)
+
+    try:
+        asyncio.run(main())
+    finally:
+        provider.engine.shutdown()
